@@ -1,0 +1,246 @@
+"""Client: the user entry point (parity: python/scannerpy/client.py).
+
+Local (in-process) execution runs the C++ engine directly; distributed
+execution (master/workers over gRPC) goes through scanner_amd.master /
+scanner_amd.worker, reusing the same graph assembly done here."""
+import os
+import tempfile
+
+import msgpack
+
+from . import _core
+from .common import (CacheMode, ColumnType, DeviceType, PerfParams,
+                     ScannerException)
+from .io import IOGenerator
+from .op import Op, OpColumn, OpGenerator
+from .profiler import Profile
+from .storage import NamedStream, NamedVideoStream
+from .streams import PartitionerGenerator, StreamsGenerator
+
+
+class Client:
+    def __init__(self, db_path=None, master=None, workers=None,
+                 start_cluster=True, recover=True):
+        self._db_path = db_path or os.path.join(
+            tempfile.gettempdir(), "scanner_amd_db")
+        os.makedirs(self._db_path, exist_ok=True)
+        self._db = _core.Database(self._db_path)
+        if recover:
+            self._db.recover()
+        self.ops = OpGenerator(self)
+        self.streams = StreamsGenerator(self)
+        self.partitioner = PartitionerGenerator()
+        self.io = IOGenerator(self)
+        self._op_info_cache = {}
+        self._last_profilers = None
+        # distributed mode
+        self._master_addr = master
+        self._worker_addrs = workers or []
+        self._cluster = None
+        if master is not None and start_cluster:
+            from .master import ClusterClient
+            self._cluster = ClusterClient(self, master, self._worker_addrs)
+
+    # ---- registry ----
+
+    def _op_info(self, name):
+        if name not in self._op_info_cache:
+            self._op_info_cache[name] = _core.op_info(name)
+        return self._op_info_cache[name]
+
+    def has_gpu(self):
+        return _core.have_gpu()
+
+    # ---- tables ----
+
+    def table_names(self):
+        return self._db.table_names()
+
+    def has_table(self, name):
+        return self._db.has_table(name)
+
+    def delete_table(self, name):
+        self._db.delete_table(name)
+
+    def table_info(self, name):
+        return self._db.table_info(name)
+
+    def new_table(self, name, columns, rows):
+        """rows: list (per row) of list (per column) of bytes."""
+        per_col = [[r[c] for r in rows] for c in range(len(columns))]
+        _core.write_bytes_table(self._db, name, columns, per_col, 128)
+        return NamedStream(self, name)
+
+    def sequence(self, name):
+        return NamedStream(self, name)
+
+    def summarize(self):
+        lines = []
+        for t in sorted(self.table_names()):
+            info = self.table_info(t)
+            cols = ", ".join(n for n, _ in info["columns"])
+            lines.append(f"{t}: {info['num_rows']} rows [{cols}]"
+                         f"{'' if info['committed'] else ' (uncommitted)'}")
+        return "\n".join(lines)
+
+    # ---- graph assembly ----
+
+    def _toposort(self, sink_op):
+        order = []
+        seen = {}
+
+        def visit(op):
+            if id(op) in seen:
+                if seen[id(op)] == 1:
+                    raise ScannerException("cycle in op graph")
+                return
+            seen[id(op)] = 1
+            for col in op._inputs:
+                visit(col.op)
+            seen[id(op)] = 2
+            order.append(op)
+
+        visit(sink_op)
+        return order
+
+    def _assemble(self, sink_op, n_jobs_hint=None):
+        """Returns (graph_bytes, jobs_bytes, out_streams, n_jobs)."""
+        ops = self._toposort(sink_op)
+        idx = {id(op): i for i, op in enumerate(ops)}
+
+        n_jobs = None
+        for op in ops:
+            if hasattr(op, "_streams"):
+                n = len(op._streams)
+                if n_jobs is None:
+                    n_jobs = n
+                elif n != n_jobs:
+                    raise ScannerException(
+                        f"stream count mismatch: {n} vs {n_jobs}")
+        if n_jobs is None:
+            raise ScannerException("graph has no Input op")
+
+        g_ops = []
+        for op in ops:
+            g_ops.append({
+                "name": op._name,
+                "inputs": [{"op": idx[id(c.op)], "column": c.name}
+                           for c in op._inputs],
+                "args": msgpack.packb(op._args) if op._args else b"",
+                "device": int(op._device),
+                "batch": int(op._batch),
+                "stencil": [int(s) for s in op._stencil],
+                "warmup": int(op._warmup),
+            })
+        graph_bytes = msgpack.packb({"ops": g_ops})
+
+        def stream_arg(lst, j):
+            if lst is None:
+                return None
+            if len(lst) == 1:
+                return lst[0]
+            if j >= len(lst):
+                raise ScannerException("fewer per-stream args than jobs")
+            return lst[j]
+
+        jobs = []
+        out_streams = sink_op._streams
+        for j in range(n_jobs):
+            sources = {}
+            sampling = {}
+            op_args = {}
+            for i, op in enumerate(ops):
+                if op._name == "Input":
+                    s = op._streams[j]
+                    sources[str(i)] = {"table": s.name,
+                                       "column": s.column_name()}
+                elif hasattr(op, "_sampling_kind"):
+                    if op._sampling_kind == "partition":
+                        args = dict(stream_arg(op._sampling_args, j))
+                    elif op._sampling_kind == "All":
+                        args = {"kind": "All"}
+                    else:
+                        args = dict(stream_arg(op._sampling_args, j))
+                        args.setdefault("kind", op._sampling_kind)
+
+                    def norm(a):
+                        full = {"kind": a.get("kind", "All"),
+                                "stride": a.get("stride", 1),
+                                "spacing": a.get("spacing", 1),
+                                "starts": a.get("starts", []),
+                                "ends": a.get("ends", []),
+                                "rows": a.get("rows", [])}
+                        if a.get("kind") == "PerGroup":
+                            full["groups"] = [norm(g) for g in a["groups"]]
+                        return full
+
+                    sampling[str(i)] = norm(args)
+                elif op._stream_args is not None:
+                    op_args[str(i)] = msgpack.packb(
+                        stream_arg(op._stream_args, j))
+            jobs.append({
+                "sources": sources,
+                "sampling": sampling,
+                "op_args": op_args,
+                "sink_table": out_streams[j].name,
+            })
+        return graph_bytes, jobs, out_streams, n_jobs
+
+    # ---- run ----
+
+    def run(self, outputs, perf_params=None, cache_mode=CacheMode.Error,
+            gpu_ids=None, pipeline_instances=None, show_progress=False):
+        """Execute the graph ending at `outputs` (an Output op) for every
+        bound stream (parity: Client.run client.py:1282)."""
+        if isinstance(outputs, (list, tuple)):
+            if len(outputs) != 1:
+                raise ScannerException(
+                    "multiple Output sinks per run() not yet supported; "
+                    "call run() per sink")
+            outputs = outputs[0]
+        perf = perf_params or PerfParams.estimate()
+        graph_bytes, jobs, out_streams, n_jobs = self._assemble(outputs)
+
+        # CacheMode handling (parity: client.py:1386-1432)
+        keep = []
+        for j, js in enumerate(jobs):
+            s = out_streams[j]
+            if s.exists() and self._db.table_committed(s.name):
+                if cache_mode == CacheMode.Error:
+                    raise ScannerException(
+                        f"output table '{s.name}' already exists "
+                        "(pass cache_mode=CacheMode.Ignore/Overwrite)")
+                if cache_mode == CacheMode.Ignore:
+                    continue
+            keep.append(js)
+        if not keep:
+            return None
+        jobs_bytes = msgpack.packb(keep)
+
+        if self._cluster is not None:
+            return self._cluster.run_job(graph_bytes, jobs_bytes, perf,
+                                         show_progress=show_progress)
+
+        # local execution
+        if gpu_ids is None:
+            needs_gpu = any(
+                o["device"] == int(DeviceType.GPU)
+                for o in msgpack.unpackb(graph_bytes)["ops"])
+            gpu_ids = list(range(_core.gpu_device_count())) if needs_gpu \
+                else []
+        if pipeline_instances is None:
+            pipeline_instances = (perf.pipeline_instances_per_node
+                                  or (len(gpu_ids) if gpu_ids else 1))
+        ex = _core.LocalExecutor(self._db, graph_bytes, jobs_bytes,
+                                 perf.to_dict(pipeline_instances), gpu_ids)
+        ex.run()
+        self._last_profilers = ex.profilers()
+        return Profile(self._last_profilers)
+
+    def profile(self):
+        return Profile(self._last_profilers or [])
+
+    def shutdown(self):
+        if self._cluster is not None:
+            self._cluster.shutdown()
+            self._cluster = None

@@ -1,0 +1,445 @@
+#include "executor.h"
+
+#include <algorithm>
+#include <set>
+
+#include "../memory.h"
+
+namespace sca {
+
+Element element_to_device(const Element& e, DeviceHandle target, bool& copied) {
+  copied = false;
+  if (e.is_null || e.device == target) return e;
+  Element out = e;
+  out.device = target;
+  out.buffer = new_buffer(target, e.size);
+  memcpy_buffer(out.buffer, target, e.buffer, e.device, e.size);
+  copied = true;
+  return out;
+}
+
+struct LocalExecutor::Instance {
+  i32 idx = 0;
+  DeviceHandle gpu{DeviceType::CPU, 0};  // GPU assigned to this instance
+  Profiler* profiler = nullptr;
+  // kernel instances per op (null for builtins); re-created per job when
+  // per-stream args change
+  std::vector<std::unique_ptr<BaseKernel>> kernels;
+  i32 kernels_job = -1;
+};
+
+LocalExecutor::LocalExecutor(std::shared_ptr<Database> db, JobGraph graph,
+                             std::vector<JobBinding> jobs, PerfParams pp,
+                             std::vector<i32> gpu_ids)
+    : db_(std::move(db)),
+      graph_(std::move(graph)),
+      jobs_(std::move(jobs)),
+      pp_(pp),
+      gpu_ids_(std::move(gpu_ids)) {
+  MemoryConfig mc;
+  mc.cpu_pool_size = pp_.cpu_pool_size;
+  mc.gpu_pool_size = pp_.gpu_pool_size;
+  mc.gpu_ids = gpu_ids_;
+  init_memory_allocators(mc);
+}
+
+LocalExecutor::~LocalExecutor() = default;
+
+void LocalExecutor::make_instance(i32 idx) {
+  auto inst = std::make_unique<Instance>();
+  inst->idx = idx;
+  if (!gpu_ids_.empty()) {
+    inst->gpu = DeviceHandle{DeviceType::GPU,
+                             gpu_ids_[idx % gpu_ids_.size()]};
+  }
+  profilers_.push_back(std::make_unique<Profiler>(
+      (ProfilerLevel)pp_.profiler_level));
+  inst->profiler = profilers_.back().get();
+  instances_.push_back(std::move(inst));
+}
+
+void LocalExecutor::prepare() {
+  if (prepared_) return;
+  validate_graph(graph_);
+  SourceRowsFn source_rows = [this](const SourceArgsC& s) -> i64 {
+    return db_->get_table(s.table).num_rows();
+  };
+  for (auto& job : jobs_) {
+    analyses_.push_back(analyze_job(graph_, job, source_rows));
+  }
+  // Create output tables (uncommitted), one item per task.
+  const OpNode& out_op = graph_.ops.back();
+  for (size_t j = 0; j < jobs_.size(); ++j) {
+    std::vector<std::string> col_names;
+    std::vector<ColumnType> col_types;
+    for (auto& e : out_op.inputs) {
+      col_names.push_back(e.column);
+      const auto& psi = analyses_[j].info[e.op];
+      size_t ci = 0;
+      for (; ci < psi.output_columns.size(); ++ci)
+        if (psi.output_columns[ci] == e.column) break;
+      SCA_CHECK(ci < psi.output_columns.size(),
+                "Output references unknown column '" + e.column + "'");
+      col_types.push_back(psi.output_types[ci]);
+    }
+    TableMetadata t = db_->new_table(jobs_[j].sink.table, col_names, col_types,
+                                     /*overwrite=*/true);
+    out_tables_.push_back(t);
+    // task boundaries
+    i64 rows = analyses_[j].output_rows;
+    std::vector<i64> ends;
+    for (i64 s = 0; s < rows; s += pp_.io_packet_size) {
+      ends.push_back(std::min(rows, s + pp_.io_packet_size));
+    }
+    if (ends.empty()) ends.push_back(0);
+    task_rows_.push_back(ends);
+  }
+  for (i32 i = 0; i < pp_.pipeline_instances; ++i) make_instance(i);
+  prepared_ = true;
+}
+
+std::vector<TaskDesc> LocalExecutor::all_tasks() const {
+  std::vector<TaskDesc> tasks;
+  for (size_t j = 0; j < task_rows_.size(); ++j) {
+    i64 prev = 0;
+    for (size_t t = 0; t < task_rows_[j].size(); ++t) {
+      tasks.push_back(TaskDesc{(i32)j, (i32)t, prev, task_rows_[j][t]});
+      prev = task_rows_[j][t];
+    }
+  }
+  return tasks;
+}
+
+i64 LocalExecutor::total_output_rows() const {
+  i64 total = 0;
+  for (auto& a : analyses_) total += a.output_rows;
+  return total;
+}
+
+void LocalExecutor::finalize_job(i32 job) {
+  TableMetadata t = out_tables_[job];
+  t.end_rows = task_rows_[job];
+  db_->update_table(t);
+  db_->commit_table(t.id);
+}
+
+void LocalExecutor::run() {
+  prepare();
+  auto tasks = all_tasks();
+  BoundedQueue<TaskDesc> q;
+  for (auto& t : tasks) q.push(t);
+  q.close();
+
+  std::atomic<bool> failed{false};
+  std::string fail_msg;
+  std::mutex fail_mu;
+
+  auto worker = [&](i32 idx) {
+    Instance& inst = *instances_[idx];
+    while (auto t = q.try_pop()) {
+      if (failed.load()) return;
+      try {
+        process_task(inst, *t);
+      } catch (const std::exception& ex) {
+        std::lock_guard<std::mutex> l(fail_mu);
+        failed.store(true);
+        fail_msg = ex.what();
+        return;
+      }
+    }
+  };
+
+  if (instances_.size() == 1) {
+    worker(0);
+  } else {
+    std::vector<std::thread> threads;
+    for (size_t i = 0; i < instances_.size(); ++i)
+      threads.emplace_back(worker, (i32)i);
+    for (auto& th : threads) th.join();
+  }
+  if (failed.load()) throw ScannerError("job failed: " + fail_msg);
+  for (size_t j = 0; j < jobs_.size(); ++j) finalize_job((i32)j);
+}
+
+void LocalExecutor::process_task_public(i32 instance, const TaskDesc& t) {
+  prepare();
+  while ((i32)instances_.size() <= instance) make_instance((i32)instances_.size());
+  process_task(*instances_[instance], t);
+}
+
+// Load Input-op rows into the per-op output maps.
+void LocalExecutor::load_inputs(
+    Instance& inst, const TaskDesc& t, const TaskPlan& plan,
+    std::vector<std::map<std::string, std::unordered_map<i64, Element>>>& outs) {
+  const JobBinding& job = jobs_[t.job];
+  for (auto& kv : plan.load_rows) {
+    i32 op_idx = kv.first;
+    const std::vector<i64>& rows = kv.second;
+    const SourceArgsC& src = job.sources.at(op_idx);
+    const OpStaticInfo& si = analyses_[t.job].info[op_idx];
+    TableMetadata table = db_->get_table(src.table);
+    const std::string& col_name = si.output_columns[0];
+    Profiler::Scope s(inst.profiler, "load:" + src.table);
+
+    if (si.output_types[0] == ColumnType::Video) {
+      // Raw-codec video: frames are stored as plain elements; geometry from
+      // the item's VideoMetadata. (svc-codec spans are decoded by the
+      // decode stage — wired in video/decoder.cpp.)
+      auto items = items_for_rows(table, rows);
+      std::unordered_map<i32, VideoMetadata> vm;
+      for (auto& ir : items)
+        vm[ir.item] = read_video_metadata(*db_, table, src.column, ir.item);
+      ElementVector elems =
+          read_column_rows(*db_, table, src.column, rows,
+                           pp_.sparsity_threshold);
+      size_t ei = 0;
+      for (auto& ir : items) {
+        const VideoMetadata& m = vm[ir.item];
+        SCA_CHECK(m.codec == "raw",
+                  "load_inputs only handles raw video here; decode stage "
+                  "handles codec '" + m.codec + "'");
+        while (ei < elems.size() && elems[ei].index < ir.row_end) {
+          Element& e = elems[ei];
+          e.is_frame = true;
+          e.frame_info.shape[0] = m.height;
+          e.frame_info.shape[1] = m.width;
+          e.frame_info.shape[2] = m.channels;
+          e.frame_info.type = m.frame_type;
+          outs[op_idx][col_name][e.index] = e;
+          ++ei;
+        }
+      }
+    } else {
+      ElementVector elems = read_column_rows(*db_, table, src.column, rows,
+                                             pp_.sparsity_threshold);
+      for (auto& e : elems) outs[op_idx][col_name][e.index] = e;
+    }
+  }
+}
+
+void LocalExecutor::process_task(Instance& inst, const TaskDesc& t) {
+  const JobBinding& job = jobs_[t.job];
+  const JobAnalysis& ja = analyses_[t.job];
+  size_t n = graph_.ops.size();
+
+  TaskPlan plan;
+  {
+    Profiler::Scope s(inst.profiler, "derive_task_plan");
+    plan = derive_task_plan(graph_, ja, job, t.start, t.end);
+  }
+
+  // (Re)build kernel instances for this job's per-stream args.
+  if (inst.kernels_job != t.job) {
+    inst.kernels.clear();
+    inst.kernels.resize(n);
+    for (size_t i = 0; i < n; ++i) {
+      const OpNode& op = graph_.ops[i];
+      if (is_builtin_op(op.name)) continue;
+      KernelConfig cfg;
+      cfg.device = op.device == DeviceType::GPU ? inst.gpu : CPU_DEVICE;
+      SCA_CHECK(op.device != DeviceType::GPU || cfg.device.is_gpu(),
+                "graph requests GPU but executor has no gpu_ids");
+      for (auto& e : op.inputs) cfg.input_columns.push_back(e.column);
+      cfg.output_columns = ja.info[i].output_columns;
+      cfg.args = op.args;
+      cfg.node_id = (i32)i;
+      cfg.max_batch = ja.info[i].batch;
+      cfg.profiler = inst.profiler;
+      const KernelFactory& kf = kernel_registry().get(op.name, op.device);
+      inst.kernels[i] = kf.make(cfg);
+      auto ait = job.op_args.find((i32)i);
+      inst.kernels[i]->new_stream(
+          ait == job.op_args.end() ? std::vector<u8>{} : ait->second);
+    }
+    inst.kernels_job = t.job;
+  }
+
+  // Per-op, per-column row->element maps.
+  std::vector<std::map<std::string, std::unordered_map<i64, Element>>> outs(n);
+  auto free_all = [&]() {
+    for (auto& per_op : outs) {
+      for (auto& per_col : per_op) {
+        for (auto& kv : per_col.second) {
+          Element& e = kv.second;
+          if (e.buffer) delete_buffer(e.device, e.buffer);
+        }
+      }
+    }
+  };
+
+  try {
+    load_inputs(inst, t, plan, outs);
+
+    for (size_t i = 0; i < n; ++i) {
+      const OpNode& op = graph_.ops[i];
+      const OpStaticInfo& si = ja.info[i];
+      OpTaskPlan& otp = plan.ops[i];
+      if (otp.required_rows.empty() && !is_output_op(op.name)) continue;
+
+      if (is_input_op(op.name)) continue;
+
+      if (is_sample_op(op.name) || is_slice_op(op.name) ||
+          is_unslice_op(op.name)) {
+        // remap: alias parent elements (zero-copy, extra ref per alias)
+        auto& parent_map = outs[op.inputs[0].op][op.inputs[0].column];
+        auto& out_map = outs[i][si.output_columns[0]];
+        for (size_t k = 0; k < otp.required_rows.size(); ++k) {
+          i64 row = otp.required_rows[k];
+          i64 up = otp.remap[k];
+          Element e;
+          if (up < 0) {
+            e.is_null = true;
+          } else {
+            auto pit = parent_map.find(up);
+            SCA_CHECK(pit != parent_map.end(),
+                      "remap source row missing (op " + op.name + ")");
+            e = pit->second;
+            if (e.buffer) add_buffer_ref(e.device, e.buffer);
+          }
+          e.index = row;
+          out_map[row] = e;
+        }
+        continue;
+      }
+
+      if (is_output_op(op.name)) {
+        // Sink: gather required rows from parents, move to CPU, write item.
+        Profiler::Scope s(inst.profiler, "save");
+        const TableMetadata& table = out_tables_[t.job];
+        std::vector<std::vector<Element>> cols(op.inputs.size());
+        std::vector<std::vector<u8>> owned(op.inputs.size());
+        for (size_t c = 0; c < op.inputs.size(); ++c) {
+          auto& parent_map = outs[op.inputs[c].op][op.inputs[c].column];
+          for (i64 row = t.start; row < t.end; ++row) {
+            auto pit = parent_map.find(row);
+            SCA_CHECK(pit != parent_map.end(), "sink missing row");
+            Element e = pit->second;
+            if (e.is_null) {
+              Element ne;
+              ne.is_null = true;
+              cols[c].push_back(ne);
+              owned[c].push_back(0);
+              continue;
+            }
+            bool copied;
+            Element ce = element_to_device(e, CPU_DEVICE, copied);
+            ce.index = row;
+            cols[c].push_back(ce);
+            if (copied) owned[c].push_back(1);
+            else owned[c].push_back(0);
+          }
+        }
+        for (size_t c = 0; c < op.inputs.size(); ++c) {
+          write_column_item(*db_, table, table.columns[c].name, t.task,
+                            cols[c]);
+          if (table.columns[c].type == ColumnType::Video) {
+            // Raw-stored frame column: record geometry so readers can
+            // reconstruct frames. (Codec-compressed sinks go through the
+            // encode stage instead.)
+            VideoMetadata vm;
+            vm.codec = "raw";
+            vm.num_frames = (i64)cols[c].size();
+            for (auto& e : cols[c]) {
+              if (e.is_null) continue;
+              vm.height = e.frame_info.shape[0];
+              vm.width = e.frame_info.shape[1];
+              vm.channels = e.frame_info.shape[2];
+              vm.frame_type = e.frame_info.type;
+              break;
+            }
+            for (i64 k = 0; k < vm.num_frames; ++k) {
+              vm.keyframe_indices.push_back(k);
+              vm.sample_sizes.push_back(cols[c][k].size);
+              vm.sample_offsets.push_back(
+                  k == 0 ? 0 : vm.sample_offsets[k - 1] +
+                                   vm.sample_sizes[k - 1]);
+            }
+            auto vbuf = vm.serialize();
+            db_->storage()->write_all(
+                db_->paths().video_metadata(
+                    table.id, table.column_id(table.columns[c].name), t.task),
+                vbuf.data(), vbuf.size());
+          }
+          for (size_t k = 0; k < cols[c].size(); ++k) {
+            if (owned[c][k] && cols[c][k].buffer)
+              delete_buffer(CPU_DEVICE, cols[c][k].buffer);
+          }
+        }
+        continue;
+      }
+
+      // ---- kernel op ----
+      BaseKernel* kernel = inst.kernels[i].get();
+      DeviceHandle kdev = kernel->config().device;
+      std::set<i64> required(otp.required_rows.begin(),
+                             otp.required_rows.end());
+      auto& out_maps = outs[i];
+
+      size_t bi = 0;
+      while (bi < otp.compute_rows.size()) {
+        // batch boundaries: batch size limit + reset markers
+        size_t be = bi + 1;
+        while (be < otp.compute_rows.size() &&
+               (i64)(be - bi) < (i64)si.batch && !otp.reset_before[be]) {
+          ++be;
+        }
+        if (otp.reset_before[bi]) kernel->reset();
+
+        // Gather stenciled inputs, moving to kernel device as needed.
+        StenciledElements input(op.inputs.size());
+        std::vector<Element> scratch;  // owned device copies to free
+        for (size_t c = 0; c < op.inputs.size(); ++c) {
+          auto& parent_map = outs[op.inputs[c].op][op.inputs[c].column];
+          input[c].resize(be - bi);
+          for (size_t r = bi; r < be; ++r) {
+            for (i64 wrow : otp.windows[r]) {
+              auto pit = parent_map.find(wrow);
+              SCA_CHECK(pit != parent_map.end(),
+                        "kernel input row missing for op '" + op.name + "'");
+              bool copied;
+              Element e = element_to_device(pit->second, kdev, copied);
+              if (copied) scratch.push_back(e);
+              input[c][r - bi].push_back(e);
+            }
+          }
+        }
+
+        BatchedElements output(si.output_columns.size());
+        {
+          Profiler::Scope s(inst.profiler, "op:" + op.name);
+          kernel->execute(input, output);
+        }
+        for (size_t c = 0; c < si.output_columns.size(); ++c) {
+          SCA_CHECK(output[c].size() == be - bi,
+                    "op '" + op.name + "' produced " +
+                        std::to_string(output[c].size()) + " rows, expected " +
+                        std::to_string(be - bi));
+          auto& out_map = out_maps[si.output_columns[c]];
+          for (size_t r = bi; r < be; ++r) {
+            Element& e = output[c][r - bi];
+            e.index = otp.compute_rows[r];
+            if (!e.is_null) e.device = kdev;
+            if (required.count(e.index)) {
+              out_map[e.index] = e;
+            } else if (e.buffer) {
+              delete_buffer(e.device, e.buffer);  // warmup-only row
+            }
+          }
+        }
+        // GPU kernels are async on this thread's stream; sync before
+        // releasing their input staging copies.
+        if (kdev.is_gpu()) sync_per_thread_stream();
+        for (auto& e : scratch) delete_buffer(e.device, e.buffer);
+        bi = be;
+      }
+    }
+  } catch (...) {
+    free_all();
+    throw;
+  }
+  free_all();
+  inst.profiler->increment("tasks", 1);
+  inst.profiler->increment("rows", t.end - t.start);
+}
+
+}  // namespace sca

@@ -1,0 +1,90 @@
+// Single-node executor: turns (graph, job bindings) into committed output
+// tables. Capability parity: the worker-side execution path of the
+// reference (worker.cpp process_job + load/pre-evaluate/evaluate/
+// post-evaluate/save workers). MI355X layout: one pipeline instance per GPU
+// (or N CPU instances), each instance a worker thread owning a HIP stream;
+// kernels and device copies are issued async on that stream and only the
+// sink synchronizes. Tasks are pulled from a shared queue (the in-process
+// analogue of the master's pull scheduler; the distributed version reuses
+// process_task via the worker service).
+#pragma once
+
+#include <atomic>
+#include <thread>
+
+#include "../dag/graph.h"
+#include "../ops/kernel.h"
+#include "../profiler.h"
+#include "../queue.h"
+#include "table_io.h"
+
+namespace sca {
+
+struct PerfParams {
+  i64 io_packet_size = 128;
+  i64 work_packet_size = 32;
+  i32 pipeline_instances = 1;
+  i64 queue_size = 4;
+  size_t cpu_pool_size = 0;
+  size_t gpu_pool_size = 0;
+  i32 sparsity_threshold = 8;
+  i32 profiler_level = 1;
+};
+
+struct TaskDesc {
+  i32 job = 0;
+  i32 task = 0;  // item index in the output table
+  i64 start = 0, end = 0;
+};
+
+class LocalExecutor {
+ public:
+  LocalExecutor(std::shared_ptr<Database> db, JobGraph graph,
+                std::vector<JobBinding> jobs, PerfParams pp,
+                std::vector<i32> gpu_ids);
+  ~LocalExecutor();
+
+  // Analyze + create output tables + run all tasks to completion. Throws on
+  // the first failure.
+  void run();
+
+  // Process one externally-scheduled task (distributed mode). Analysis and
+  // output tables must exist (call prepare() first).
+  void prepare();
+  void process_task_public(i32 instance, const TaskDesc& t);
+  std::vector<TaskDesc> all_tasks() const;
+  void finalize_job(i32 job);  // set end_rows + commit output table
+
+  const std::vector<std::unique_ptr<Profiler>>& profilers() const {
+    return profilers_;
+  }
+  i64 total_output_rows() const;
+
+ private:
+  struct Instance;
+  void process_task(Instance& inst, const TaskDesc& t);
+  void load_inputs(Instance& inst, const TaskDesc& t, const TaskPlan& plan,
+                   std::vector<std::map<std::string,
+                                        std::unordered_map<i64, Element>>>& outs);
+  void make_instance(i32 idx);
+
+  std::shared_ptr<Database> db_;
+  JobGraph graph_;
+  std::vector<JobBinding> jobs_;
+  PerfParams pp_;
+  std::vector<i32> gpu_ids_;
+
+  std::vector<JobAnalysis> analyses_;
+  std::vector<TableMetadata> out_tables_;   // per job
+  std::vector<std::vector<i64>> task_rows_; // per job: cumulative task ends
+  bool prepared_ = false;
+
+  std::vector<std::unique_ptr<Instance>> instances_;
+  std::vector<std::unique_ptr<Profiler>> profilers_;
+};
+
+// Move an element to the target device; returns the original element if
+// already resident (borrowed == true) else an owned copy.
+Element element_to_device(const Element& e, DeviceHandle target, bool& copied);
+
+}  // namespace sca

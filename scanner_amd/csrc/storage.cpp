@@ -1,0 +1,145 @@
+#include "storage.h"
+
+#include <dirent.h>
+#include <fcntl.h>
+#include <sys/stat.h>
+#include <sys/types.h>
+#include <unistd.h>
+
+#include <cerrno>
+#include <cstdio>
+#include <cstring>
+
+namespace sca {
+
+namespace {
+
+class PosixStorage : public StorageBackend {
+ public:
+  std::vector<u8> read_all(const std::string& path) override {
+    int fd = ::open(path.c_str(), O_RDONLY);
+    if (fd < 0) throw ScannerError("open failed: " + path + ": " + strerror(errno));
+    struct stat st;
+    if (fstat(fd, &st) != 0) {
+      ::close(fd);
+      throw ScannerError("fstat failed: " + path);
+    }
+    std::vector<u8> buf(st.st_size);
+    size_t off = 0;
+    while (off < buf.size()) {
+      ssize_t n = ::read(fd, buf.data() + off, buf.size() - off);
+      if (n <= 0) {
+        ::close(fd);
+        throw ScannerError("read failed: " + path);
+      }
+      off += n;
+    }
+    ::close(fd);
+    return buf;
+  }
+
+  void read_range(const std::string& path, u64 offset, u64 size,
+                  u8* out) override {
+    int fd = ::open(path.c_str(), O_RDONLY);
+    if (fd < 0) throw ScannerError("open failed: " + path + ": " + strerror(errno));
+    u64 off = 0;
+    while (off < size) {
+      ssize_t n = ::pread(fd, out + off, size - off, offset + off);
+      if (n <= 0) {
+        ::close(fd);
+        throw ScannerError("pread failed: " + path);
+      }
+      off += n;
+    }
+    ::close(fd);
+  }
+
+  u64 file_size(const std::string& path) override {
+    struct stat st;
+    if (stat(path.c_str(), &st) != 0)
+      throw ScannerError("stat failed: " + path);
+    return st.st_size;
+  }
+
+  void write_all(const std::string& path, const u8* data,
+                 size_t size) override {
+    // Write to temp + rename for atomicity (commit semantics depend on it).
+    std::string tmp = path + ".tmp";
+    int fd = ::open(tmp.c_str(), O_WRONLY | O_CREAT | O_TRUNC, 0644);
+    if (fd < 0) throw ScannerError("open for write failed: " + tmp + ": " + strerror(errno));
+    size_t off = 0;
+    while (off < size) {
+      ssize_t n = ::write(fd, data + off, size - off);
+      if (n < 0) {
+        ::close(fd);
+        throw ScannerError("write failed: " + tmp);
+      }
+      off += n;
+    }
+    if (fsync(fd) != 0 || ::close(fd) != 0)
+      throw ScannerError("fsync/close failed: " + tmp);
+    if (::rename(tmp.c_str(), path.c_str()) != 0)
+      throw ScannerError("rename failed: " + path);
+  }
+
+  bool exists(const std::string& path) override {
+    struct stat st;
+    return stat(path.c_str(), &st) == 0;
+  }
+
+  void remove(const std::string& path) override { ::unlink(path.c_str()); }
+
+  void remove_tree(const std::string& path) override {
+    DIR* d = opendir(path.c_str());
+    if (!d) {
+      ::unlink(path.c_str());
+      return;
+    }
+    struct dirent* e;
+    while ((e = readdir(d)) != nullptr) {
+      std::string name = e->d_name;
+      if (name == "." || name == "..") continue;
+      remove_tree(path + "/" + name);
+    }
+    closedir(d);
+    ::rmdir(path.c_str());
+  }
+
+  void make_dirs(const std::string& path) override {
+    std::string cur;
+    size_t i = 0;
+    while (i < path.size()) {
+      size_t j = path.find('/', i + 1);
+      if (j == std::string::npos) j = path.size();
+      cur = path.substr(0, j);
+      if (!cur.empty() && cur != "/") {
+        if (mkdir(cur.c_str(), 0755) != 0 && errno != EEXIST) {
+          throw ScannerError("mkdir failed: " + cur + ": " + strerror(errno));
+        }
+      }
+      i = j;
+    }
+  }
+
+  std::vector<std::string> list_dir(const std::string& path) override {
+    std::vector<std::string> out;
+    DIR* d = opendir(path.c_str());
+    if (!d) return out;
+    struct dirent* e;
+    while ((e = readdir(d)) != nullptr) {
+      std::string name = e->d_name;
+      if (name == "." || name == "..") continue;
+      out.push_back(name);
+    }
+    closedir(d);
+    return out;
+  }
+};
+
+}  // namespace
+
+std::unique_ptr<StorageBackend> StorageBackend::make_posix() {
+  return std::make_unique<PosixStorage>();
+}
+
+}  // namespace sca

@@ -1,0 +1,273 @@
+"""Op graph DSL (parity: python/scannerpy/op.py).
+
+`sc.ops.<Name>(input_col=..., arg=..., device=...)` materializes an Op node;
+`@scanner_amd.register_python_op()` registers a Python function/class as an
+op, deriving input/output columns from type annotations (FrameType ->
+frame column, bytes -> blob column, Sequence[...] -> batched,
+nested Sequence for stencil)."""
+import inspect
+import typing
+
+import msgpack
+
+from .common import ColumnType, DeviceType, FrameType, ScannerException
+
+
+class OpColumn:
+    def __init__(self, op, name, typ):
+        self.op = op
+        self.name = name
+        self.type = typ
+
+    def compress(self, **kw):
+        """Mark a frame column for codec compression at the sink (parity:
+        OpColumn.compress_video op.py:47-106)."""
+        self.op._compress[self.name] = dict(kw)
+        return self
+
+    def compress_video(self, **kw):
+        return self.compress(codec="svc", **kw)
+
+    def lossless(self):
+        return self.compress(codec="svc")
+
+
+class Op:
+    def __init__(self, client, name, inputs, args=None, device=DeviceType.CPU,
+                 batch=0, stencil=None, warmup=-1, stream_args=None,
+                 output_columns=None):
+        self._client = client
+        self._name = name
+        self._inputs = inputs              # list of OpColumn
+        self._args = args or {}
+        self._device = device
+        self._batch = batch
+        self._stencil = stencil or []
+        self._warmup = warmup
+        self._stream_args = stream_args    # per-stream args (list) or None
+        self._compress = {}
+        if output_columns is None:
+            info = client._op_info(name)
+            output_columns = [(n, ColumnType(t))
+                              for n, t in info["output_columns"]]
+        self._outputs = [OpColumn(self, n, t) for n, t in output_columns]
+
+    def outputs(self):
+        return list(self._outputs)
+
+    def __getitem__(self, name):
+        for c in self._outputs:
+            if c.name == name:
+                return c
+        raise ScannerException(f"op {self._name} has no output column {name}")
+
+    def _single(self):
+        if len(self._outputs) != 1:
+            raise ScannerException(
+                f"op {self._name} has {len(self._outputs)} outputs; "
+                "select one explicitly")
+        return self._outputs[0]
+
+
+def _as_column(x):
+    if isinstance(x, OpColumn):
+        return x
+    if isinstance(x, Op):
+        return x._single()
+    raise ScannerException(f"expected an op output column, got {type(x)}")
+
+
+class OpGenerator:
+    """`sc.ops.<OpName>(...)` (parity: OpGenerator op.py:121)."""
+
+    def __init__(self, client):
+        self._client = client
+
+    def __getattr__(self, name):
+        if name.startswith("_"):
+            raise AttributeError(name)
+        client = self._client
+
+        def make(*, device=DeviceType.CPU, batch=0, stencil=None,
+                 warmup=-1, args=None, stream_args=None, **kwargs):
+            info = client._op_info(name)
+            in_cols = [n for n, _ in info["input_columns"]]
+            inputs = []
+            op_args = dict(args or {})
+            if info["variadic"]:
+                cols = kwargs.pop("inputs")
+                inputs = [_as_column(c) for c in cols]
+                op_args.update(kwargs)
+            else:
+                for cn in in_cols:
+                    if cn not in kwargs:
+                        raise ScannerException(
+                            f"op {name} missing input column '{cn}'")
+                    inputs.append(_as_column(kwargs.pop(cn)))
+                # remaining kwargs are op args
+                op_args.update(kwargs)
+            return Op(client, name, inputs, args=op_args, device=device,
+                      batch=batch, stencil=stencil, warmup=warmup,
+                      stream_args=stream_args)
+
+        return make
+
+
+# ---------------- @register_python_op ----------------
+
+_FRAME_ANN = ("FrameType", FrameType)
+
+
+def _ann_to_column(ann):
+    """Returns (is_frame, batched, stencil) for an annotation."""
+    origin = typing.get_origin(ann)
+    if origin in (list, typing.Sequence) or str(origin).endswith("Sequence"):
+        inner = typing.get_args(ann)[0]
+        f, b, s = _ann_to_column(inner)
+        if not b:
+            return f, True, s
+        return f, True, True  # nested sequence => stencil within batch
+    if ann is bytes:
+        return False, False, False
+    if ann is FrameType or ann == "FrameType":
+        return True, False, False
+    if ann is typing.Any:
+        return False, False, False
+    return False, False, False
+
+
+def register_python_op(name=None, device_type=DeviceType.CPU, batch=0,
+                       stencil=None, bounded_state=False, warmup=0,
+                       unbounded_state=False):
+    """Decorator registering a Python function or Kernel class as an op
+    (parity: @scannerpy.register_python_op op.py:317). Input/output columns
+    derive from the annotations of `execute` (class) or the function."""
+
+    def deco(fn_or_cls):
+        from . import _core
+        opname = name or fn_or_cls.__name__
+        is_cls = inspect.isclass(fn_or_cls)
+        target = fn_or_cls.execute if is_cls else fn_or_cls
+        sig = inspect.signature(target)
+        params = [p for p in sig.parameters.values()
+                  if p.name not in ("self", "config")]
+        in_cols = []
+        batched = False
+        stenciled = stencil is not None
+        for p in params:
+            if p.annotation is inspect.Parameter.empty:
+                raise ScannerException(
+                    f"python op {opname}: parameter {p.name} needs a type "
+                    "annotation")
+            f, b, s = _ann_to_column(p.annotation)
+            batched = batched or b
+            in_cols.append((p.name, 1 if f else 0))
+        ret = sig.return_annotation
+        out_cols = []
+        if typing.get_origin(ret) is tuple:
+            for i, r in enumerate(typing.get_args(ret)):
+                f, _, _ = _ann_to_column(r)
+                out_cols.append((f"out{i}", 1 if f else 0))
+        else:
+            f, _, _ = _ann_to_column(ret)
+            out_cols.append(("out", 1 if f else 0))
+
+        the_stencil = list(stencil) if stencil else [0]
+        eff_batch = batch if batch > 0 else (1024 if batched else 1)
+
+        def factory(args_bytes):
+            return _PyKernelAdapter(fn_or_cls, is_cls, args_bytes, batched,
+                                    stenciled, len(in_cols), out_cols)
+
+        _core.register_python_op(
+            opname, factory, in_cols, out_cols, int(device_type),
+            eff_batch, the_stencil, bounded_state, warmup, unbounded_state)
+        fn_or_cls._scanner_op_name = opname
+        return fn_or_cls
+
+    return deco
+
+
+class Kernel:
+    """Base class for class-style Python ops (parity: scannerpy.Kernel)."""
+
+    def __init__(self, config):
+        self.config = config
+
+    def new_stream(self, args):
+        pass
+
+    def reset(self):
+        pass
+
+    def fetch_resources(self, args):
+        pass
+
+    def execute(self, *cols):
+        raise NotImplementedError
+
+
+class KernelConfigPy:
+    def __init__(self, args):
+        self.args = args
+
+
+class _PyKernelAdapter:
+    """Adapts user fn/class to the raw C++ interface
+    (cols[in][row][stencil] -> [out][row])."""
+
+    def __init__(self, fn_or_cls, is_cls, args_bytes, batched, stenciled,
+                 n_in, out_cols):
+        args = msgpack.unpackb(args_bytes) if args_bytes else {}
+        self._batched = batched
+        self._stenciled = stenciled
+        self._n_out = len(out_cols)
+        if is_cls:
+            self._obj = fn_or_cls(KernelConfigPy(args), **args)
+            self._fn = self._obj.execute
+        else:
+            self._obj = None
+            kw = args
+
+            def call(*cols):
+                return fn_or_cls(*cols, **kw)
+
+            self._fn = call
+
+    def new_stream(self, args_bytes):
+        if self._obj is not None and args_bytes:
+            args = msgpack.unpackb(args_bytes)
+            self._obj.new_stream(**args) if isinstance(args, dict) \
+                else self._obj.new_stream(args)
+
+    def reset(self):
+        if self._obj is not None:
+            self._obj.reset()
+
+    def execute(self, cols):
+        n_rows = len(cols[0]) if cols else 0
+        outs = [[] for _ in range(self._n_out)]
+        if self._batched:
+            ins = []
+            for col in cols:
+                if self._stenciled:
+                    ins.append([row for row in col])
+                else:
+                    ins.append([row[0] for row in col])
+            result = self._fn(*ins)
+            if self._n_out == 1:
+                result = (result,)
+            for c in range(self._n_out):
+                outs[c] = list(result[c])
+        else:
+            for r in range(n_rows):
+                ins = []
+                for col in cols:
+                    win = col[r]
+                    ins.append(win if self._stenciled else win[0])
+                result = self._fn(*ins)
+                if self._n_out == 1:
+                    result = (result,)
+                for c in range(self._n_out):
+                    outs[c].append(result[c])
+        return outs

@@ -1,0 +1,299 @@
+"""End-to-end CPU engine tests (parity role: tests/py_test.py in the
+reference — real in-process engine, temp storage, no mocks)."""
+import numpy as np
+import pytest
+
+import scanner_amd as sp
+from conftest import make_video
+
+
+def ref_histogram(frame):
+    h = np.zeros((3, 256), dtype=np.uint32)
+    for c in range(3):
+        h[c] = np.bincount(frame[:, :, c].ravel(), minlength=256)
+    return h
+
+
+def test_histogram_end_to_end(sc):
+    frames = make_video(n=30)
+    video = sp.NamedVideoStream(sc, "clip", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    hist = sc.ops.Histogram(frame=frame)
+    out = sp.NamedStream(sc, "clip_hist")
+    sink = sc.io.Output(hist, [out])
+    sc.run(sink, sp.PerfParams.manual(8, 16),
+           cache_mode=sp.CacheMode.Overwrite)
+
+    rows = list(out.load())
+    assert len(rows) == 30
+    for i, blob in enumerate(rows):
+        got = np.frombuffer(blob, dtype=np.uint32).reshape(3, 256)
+        np.testing.assert_array_equal(got, ref_histogram(frames[i]))
+
+
+def test_video_roundtrip_raw(sc):
+    frames = make_video(n=10)
+    video = sp.NamedVideoStream(sc, "rt", frames=frames, codec="raw")
+    got = np.stack(list(video.load()))
+    np.testing.assert_array_equal(got, frames)
+
+
+def test_resize(sc):
+    frames = make_video(n=6, h=32, w=40)
+    video = sp.NamedVideoStream(sc, "rs", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    small = sc.ops.Resize(frame=frame, width=20, height=16)
+    out = sp.NamedStream(sc, "rs_out")
+    sc.run(sc.io.Output(small, [out]), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite)
+    outs = list(sp.NamedVideoStream(sc, "rs_out").load())
+    assert len(outs) == 6
+    assert outs[0].shape == (16, 20, 3)
+
+
+def test_multi_job(sc):
+    vids = [make_video(n=8, seed=s) for s in range(3)]
+    streams = [sp.NamedVideoStream(sc, f"mj{i}", frames=v, codec="raw")
+               for i, v in enumerate(vids)]
+    frame = sc.io.Input(streams)
+    hist = sc.ops.Histogram(frame=frame)
+    outs = [sp.NamedStream(sc, f"mj{i}_hist") for i in range(3)]
+    sc.run(sc.io.Output(hist, outs), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite)
+    for i, o in enumerate(outs):
+        rows = list(o.load())
+        assert len(rows) == 8
+        got = np.frombuffer(rows[3], dtype=np.uint32).reshape(3, 256)
+        np.testing.assert_array_equal(got, ref_histogram(vids[i][3]))
+
+
+def test_multi_column_output(sc):
+    frames = make_video(n=6)
+    video = sp.NamedVideoStream(sc, "mc", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    hist = sc.ops.Histogram(frame=frame)
+    small = sc.ops.Resize(frame=frame, width=16, height=16)
+    out = sp.NamedStream(sc, "mc_out")
+    sc.run(sc.io.Output([hist, small], [out]), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite)
+    info = sc.table_info("mc_out")
+    names = [n for n, _ in info["columns"]]
+    assert names == ["histogram", "frame"]
+    assert info["num_rows"] == 6
+
+
+def test_stride_sampling(sc):
+    frames = make_video(n=20)
+    video = sp.NamedVideoStream(sc, "st", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    sampled = sc.streams.Stride(frame, [3])
+    hist = sc.ops.Histogram(frame=sampled)
+    out = sp.NamedStream(sc, "st_out")
+    sc.run(sc.io.Output(hist, [out]), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite)
+    rows = list(out.load())
+    assert len(rows) == 7  # ceil(20/3)
+    for k, blob in enumerate(rows):
+        got = np.frombuffer(blob, dtype=np.uint32).reshape(3, 256)
+        np.testing.assert_array_equal(got, ref_histogram(frames[k * 3]))
+
+
+def test_gather_range_samplers(sc):
+    frames = make_video(n=25)
+    video = sp.NamedVideoStream(sc, "sm", frames=frames, codec="raw")
+    # Gather
+    frame = sc.io.Input([video])
+    rows_wanted = [0, 3, 3, 17, 24]
+    g = sc.streams.Gather(frame, [rows_wanted])
+    out = sp.NamedStream(sc, "sm_g")
+    sc.run(sc.io.Output(sc.ops.Histogram(frame=g), [out]),
+           sp.PerfParams.manual(4, 8), cache_mode=sp.CacheMode.Overwrite)
+    rows = list(out.load())
+    assert len(rows) == len(rows_wanted)
+    for k, blob in enumerate(rows):
+        got = np.frombuffer(blob, dtype=np.uint32).reshape(3, 256)
+        np.testing.assert_array_equal(got,
+                                      ref_histogram(frames[rows_wanted[k]]))
+    # Range
+    frame = sc.io.Input([video])
+    r = sc.streams.Range(frame, [(5, 15)])
+    out2 = sp.NamedStream(sc, "sm_r")
+    sc.run(sc.io.Output(sc.ops.Histogram(frame=r), [out2]),
+           sp.PerfParams.manual(4, 8), cache_mode=sp.CacheMode.Overwrite)
+    assert len(list(out2.load())) == 10
+
+
+def test_space_repeat_null(sc):
+    n = 6
+    tab = sc.new_table("ints", ["col"],
+                       [[int(i).to_bytes(8, "little")] for i in range(n)])
+    col = sc.io.Input([tab])
+    spaced = sc.streams.RepeatNull(col, [3])
+    out = sp.NamedStream(sc, "spaced")
+    sc.run(sc.io.Output(spaced, [out]), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite)
+    rows = list(out.load())
+    assert len(rows) == n * 3
+    for i, rbytes in enumerate(rows):
+        if i % 3 == 0:
+            assert int.from_bytes(rbytes, "little") == i // 3
+        else:
+            assert rbytes is None
+
+
+def test_space_repeat(sc):
+    n = 4
+    tab = sc.new_table("ints2", ["col"],
+                       [[int(i).to_bytes(8, "little")] for i in range(n)])
+    col = sc.io.Input([tab])
+    spaced = sc.streams.Repeat(col, [2])
+    out = sp.NamedStream(sc, "rep")
+    sc.run(sc.io.Output(spaced, [out]), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite)
+    vals = [int.from_bytes(b, "little") for b in out.load()]
+    assert vals == [0, 0, 1, 1, 2, 2, 3, 3]
+
+
+def test_bounded_state(sc):
+    n = 20
+    tab = sc.new_table("bs", ["col"],
+                       [[int(i).to_bytes(8, "little")] for i in range(n)])
+    col = sc.io.Input([tab])
+    inc = sc.ops.TestIncrement(ignore=col)
+    out = sp.NamedStream(sc, "bs_out")
+    # tiny io packets force warmup recomputation across tasks
+    sc.run(sc.io.Output(inc, [out]), sp.PerfParams.manual(2, 4),
+           cache_mode=sp.CacheMode.Overwrite)
+    vals = [int.from_bytes(b, "little") for b in out.load()]
+    # warmup=0 bounded state: state resets at each task start, so each
+    # io-packet of 4 rows yields input + (1,2,3,4)
+    expect = [i + (i % 4) + 1 for i in range(n)]
+    assert vals == expect
+
+
+def test_unbounded_state(sc):
+    n = 12
+    tab = sc.new_table("us", ["col"],
+                       [[int(i).to_bytes(8, "little")] for i in range(n)])
+    col = sc.io.Input([tab])
+    inc = sc.ops.TestIncrementUnbounded(ignore=col)
+    out = sp.NamedStream(sc, "us_out")
+    sc.run(sc.io.Output(inc, [out]), sp.PerfParams.manual(2, 4),
+           cache_mode=sp.CacheMode.Overwrite)
+    vals = [int.from_bytes(b, "little") for b in out.load()]
+    # unbounded state recomputes from row 0 in every task: value = i + i+1
+    assert vals == [2 * i + 1 for i in range(n)]
+
+
+def test_stencil_wider_than_packet(sc):
+    frames = make_video(n=12)
+    video = sp.NamedVideoStream(sc, "stn", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    flow = sc.ops.OpticalFlow(frame=frame)
+    out = sp.NamedStream(sc, "stn_out")
+    # io_packet=2 < stencil reach across packets
+    sc.run(sc.io.Output(flow, [out]), sp.PerfParams.manual(1, 2),
+           cache_mode=sp.CacheMode.Overwrite)
+    rows = list(sp.NamedVideoStream(sc, "stn_out").load())
+    assert len(rows) == 12
+    assert rows[0].shape == (3, 4, 2)  # 48/16 x 64/16 x 2
+
+
+def test_slice_unslice(sc):
+    n = 16
+    tab = sc.new_table("sl", ["col"],
+                       [[int(i).to_bytes(8, "little")] for i in range(n)])
+    col = sc.io.Input([tab])
+    sliced = sc.streams.Slice(col, sc.partitioner.all(4))
+    inc = sc.ops.TestIncrement(ignore=sliced)
+    unsliced = sc.streams.Unslice(inc)
+    out = sp.NamedStream(sc, "sl_out")
+    sc.run(sc.io.Output(unsliced, [out]), sp.PerfParams.manual(8, 16),
+           cache_mode=sp.CacheMode.Overwrite)
+    vals = [int.from_bytes(b, "little") for b in out.load()]
+    # state resets every 4-row slice group: value = i + (i%4)+1
+    assert vals == [i + (i % 4) + 1 for i in range(n)]
+
+
+def test_overlapping_slices(sc):
+    n = 10
+    tab = sc.new_table("ov", ["col"],
+                       [[int(i).to_bytes(8, "little")] for i in range(n)])
+    col = sc.io.Input([tab])
+    # overlapping windows [0,6) and [4,10)
+    sliced = sc.streams.Slice(
+        col, sc.partitioner.ranges([(0, 6), (4, 10)]))
+    inc = sc.ops.TestIncrement(ignore=sliced)
+    unsliced = sc.streams.Unslice(inc)
+    out = sp.NamedStream(sc, "ov_out")
+    sc.run(sc.io.Output(unsliced, [out]), sp.PerfParams.manual(8, 16),
+           cache_mode=sp.CacheMode.Overwrite)
+    vals = [int.from_bytes(b, "little") for b in out.load()]
+    # group 0: rows 0..5 -> i + (i+1); group 1: rows 4..9 -> (4+i) + (i+1)
+    expect = [i + i + 1 for i in range(6)] + [4 + i + i + 1 for i in range(6)]
+    assert vals == expect
+
+
+def test_overlapping_slice_with_pergroup_sampling(sc):
+    """Mirror of reference test_overlapping_slice (py_test.py:361-375)."""
+    frames = make_video(n=35)
+    video = sp.NamedVideoStream(sc, "ovs", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    sliced = sc.streams.Slice(
+        frame, sc.partitioner.strided_ranges([(0, 15), (5, 25), (15, 35)], 1))
+    sampled = sc.streams.Range(frame=None, ranges=None) if False else \
+        sc.streams.Range(sliced, [sp.SliceList([
+            {"start": 0, "end": 10},
+            {"start": 5, "end": 15},
+            {"start": 5, "end": 15},
+        ])])
+    unsliced = sc.streams.Unslice(sampled)
+    out = sp.NamedStream(sc, "ovs_out")
+    sc.run(sc.io.Output(unsliced, [out]), sp.PerfParams.manual(8, 16),
+           cache_mode=sp.CacheMode.Overwrite)
+    rows = list(sp.NamedVideoStream(sc, "ovs_out").load())
+    assert len(rows) == 30
+    # group 1 covers source rows 5..25, its range picks local 5..15
+    # => source rows 10..20; group 2 covers 15..35 picking local 5..15
+    # => source rows 20..30
+    np.testing.assert_array_equal(rows[0], frames[0])
+    np.testing.assert_array_equal(rows[10], frames[10])
+    np.testing.assert_array_equal(rows[20], frames[20])
+    np.testing.assert_array_equal(rows[29], frames[29])
+    np.testing.assert_array_equal(rows[19], frames[19])
+
+
+def test_cache_mode(sc):
+    frames = make_video(n=4)
+    video = sp.NamedVideoStream(sc, "cm", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    hist = sc.ops.Histogram(frame=frame)
+    out = sp.NamedStream(sc, "cm_out")
+    sink = sc.io.Output(hist, [out])
+    sc.run(sink, sp.PerfParams.manual(4, 8))
+    with pytest.raises(sp.ScannerException):
+        frame = sc.io.Input([video])
+        sink2 = sc.io.Output(sc.ops.Histogram(frame=frame), [out])
+        sc.run(sink2, sp.PerfParams.manual(4, 8))
+    # Ignore: no-op
+    frame = sc.io.Input([video])
+    sink3 = sc.io.Output(sc.ops.Histogram(frame=frame), [out])
+    assert sc.run(sink3, sp.PerfParams.manual(4, 8),
+                  cache_mode=sp.CacheMode.Ignore) is None
+
+
+def test_profile(sc):
+    frames = make_video(n=8)
+    video = sp.NamedVideoStream(sc, "pf", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    hist = sc.ops.Histogram(frame=frame)
+    out = sp.NamedStream(sc, "pf_out")
+    prof = sc.run(sc.io.Output(hist, [out]), sp.PerfParams.manual(4, 8),
+                  cache_mode=sp.CacheMode.Overwrite)
+    stats = prof.statistics()
+    assert any(k.startswith("op:Histogram") for k in stats)
+    trace = prof.write_trace(str(sc._db_path) + "/trace.json")
+    import json
+    with open(trace) as f:
+        data = json.load(f)
+    assert len(data["traceEvents"]) > 0
