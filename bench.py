@@ -1,0 +1,170 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: frames/sec (whole node) on the BASELINE.json
+pipelines — 1080p histogram (SVC GPU decode + HIP histogram) and ResNet-50
+frame classification (bf16 MFMA) — on synthetic video with random-init
+weights (no network in this environment; BASELINE.md records that the
+reference repo publishes no in-repo numbers, so these runs establish the
+recorded baseline).
+
+Launched by the driver as:
+  python bench.py --gpus N --steps K --warmup W          (N=1, direct)
+  torchrun --nproc-per-node N bench.py --gpus N ...      (N>1, one rank/GPU)
+
+One step = processing FRAMES_PER_STEP 1080p frames per GPU through the
+engine pipeline (load -> SVC GPU decode -> op(s) -> save). Weak scaling:
+per-GPU work is fixed as N grows. Rank 0 prints one JSON line.
+"""
+import argparse
+import json
+import os
+import shutil
+import sys
+import tempfile
+import time
+
+import numpy as np
+
+FRAMES_PER_STEP = 512
+H, W = 1080, 1920
+
+
+def log(msg):
+    print(msg, file=sys.stderr, flush=True)
+
+
+def make_clip(n, h=H, w=W):
+    """Synthetic 1080p clip: smooth moving gradient + mild texture, the
+    shape/codec-behavior of real footage (temporally coherent)."""
+    yy, xx = np.mgrid[0:h, 0:w]
+    rng = np.random.RandomState(0)
+    texture = rng.randint(0, 32, size=(h, w, 3)).astype(np.int32)
+    frames = np.zeros((n, h, w, 3), np.uint8)
+    for i in range(n):
+        frames[i, :, :, 0] = (xx + i * 2 + texture[:, :, 0]) % 256
+        frames[i, :, :, 1] = (yy + i + texture[:, :, 1]) % 256
+        frames[i, :, :, 2] = (xx + yy + i * 3 + texture[:, :, 2]) % 256
+    return frames
+
+
+def build_pipeline(sc, sp, video, pipeline, device, out_name):
+    frame = sc.io.Input([video])
+    cols = []
+    if pipeline in ("hist", "full"):
+        cols.append(sc.ops.Histogram(frame=frame, device=device))
+    if pipeline in ("resnet", "full"):
+        cols.append(sc.ops.ResNet50(frame=frame, device=device))
+    out = sp.NamedStream(sc, out_name)
+    return sc.io.Output(cols, [out])
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--pipeline", default="full",
+                    choices=["hist", "resnet", "full"])
+    ap.add_argument("--frames-per-step", type=int, default=FRAMES_PER_STEP)
+    args = ap.parse_args()
+
+    import torch
+    import torch.distributed as dist
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    distributed = world > 1
+    if distributed:
+        dist.init_process_group(backend="nccl")
+        torch.cuda.set_device(local_rank)
+
+    import scanner_amd as sp
+    from scanner_amd import _core
+
+    have_gpu = _core.have_gpu()
+    device = sp.DeviceType.GPU if have_gpu else sp.DeviceType.CPU
+    if args.pipeline in ("resnet", "full") and "ResNet50" not in \
+            _core.registered_ops():
+        log("ResNet50 op not available; falling back to hist pipeline")
+        args.pipeline = "hist"
+
+    tmp = tempfile.mkdtemp(prefix=f"scanner_bench_r{rank}_")
+    sc = sp.Client(db_path=os.path.join(tmp, "db"))
+
+    n_frames = args.frames_per_step
+    log(f"[rank {rank}] ingesting {n_frames} synthetic 1080p frames (svc)")
+    clip = make_clip(n_frames)
+    video = sp.NamedVideoStream(sc, "bench_clip", frames=clip,
+                                codec="svc", io_packet_size=128)
+    del clip
+
+    gpu_ids = [local_rank] if have_gpu else []
+    perf = sp.PerfParams.manual(
+        work_packet_size=32, io_packet_size=128,
+        gpu_pool=0, cpu_pool=0)
+
+    def one_step(tag):
+        sink = build_pipeline(sc, sp, video, args.pipeline, device,
+                              f"bench_out_{tag}")
+        sc.run(sink, perf, cache_mode=sp.CacheMode.Overwrite,
+               gpu_ids=gpu_ids, pipeline_instances=1)
+
+    def sync():
+        if have_gpu:
+            torch.cuda.synchronize()
+        if distributed:
+            dist.barrier()
+
+    for i in range(args.warmup):
+        one_step(f"w{i}")
+        log(f"[rank {rank}] warmup {i} done")
+
+    sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        one_step(f"s{i}")
+    sync()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    if distributed:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device="cuda" if have_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = t.item()
+
+    total_frames = args.steps * n_frames * world
+    fps = total_frames / elapsed
+    if rank == 0:
+        result = {
+            "metric": "frames/sec (whole node), 1080p "
+                      + ("histogram+ResNet50" if args.pipeline == "full"
+                         else args.pipeline) + " pipeline",
+            "value": fps,
+            "unit": "frames/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if args.pipeline in ("resnet", "full") else "u8",
+            "data": "synthetic 1080p video (svc-encoded), random-init weights",
+            "config": {
+                "model": "1080p histogram + ResNet-50 classify"
+                         if args.pipeline == "full" else args.pipeline,
+                "global_batch": n_frames * world,
+                "seq_len": n_frames,
+                "parallelism": f"frame-shard dp{world}",
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if distributed:
+        dist.destroy_process_group()
+    shutil.rmtree(tmp, ignore_errors=True)
+
+
+if __name__ == "__main__":
+    main()

@@ -31,6 +31,7 @@
 // supergroup (128 groups = 4 KiB of payload).
 #pragma once
 
+#include "../element.h"
 #include "../metadata.h"
 
 namespace sca {
@@ -50,6 +51,15 @@ void svc_decode_cpu(const u8* stream, size_t size, const VideoMetadata& vm,
 // (keyframe-aligned spans; reference analogue: DecodeArgs GOP spans).
 std::vector<i64> svc_decode_span(const VideoMetadata& vm,
                                  const std::vector<i64>& want);
+
+// GPU decode (kernels/svc_codec.hip): uploads the needed GOP spans once,
+// then one kernel launch per frame on the calling thread's HIP stream;
+// decoded frames land in HBM as engine elements (device-resident), so a
+// downstream GPU op consumes them with zero copies.
+std::vector<Element> svc_decode_gpu(const u8* stream_host, size_t size,
+                                    const VideoMetadata& vm,
+                                    const std::vector<i64>& want,
+                                    DeviceHandle dev);
 
 struct SvcPacketView {
   bool is_key = false;

@@ -32,6 +32,17 @@ def sc(tmp_path):
     return sp.Client(db_path=str(tmp_path / "db"))
 
 
+def make_smooth_video(n=30, h=48, w=64):
+    """Moving-gradient clip: temporally smooth, realistic for codec tests."""
+    yy, xx = np.mgrid[0:h, 0:w]
+    frames = np.zeros((n, h, w, 3), np.uint8)
+    for i in range(n):
+        frames[i, :, :, 0] = (xx + i * 2) % 256
+        frames[i, :, :, 1] = (yy + i) % 256
+        frames[i, :, :, 2] = (xx + yy + i * 3) % 256
+    return frames
+
+
 def make_video(n=30, h=48, w=64, c=3, seed=0):
     """Synthetic moving-gradient clip (deterministic)."""
     rng = np.random.RandomState(seed)
