@@ -99,15 +99,19 @@ def main():
     del clip
 
     gpu_ids = [local_rank] if have_gpu else []
+    # Pools: steady-state allocation must never hit the driver (hipMalloc
+    # synchronizes the device; hipHostMalloc is ~ms per call).
     perf = sp.PerfParams.manual(
-        work_packet_size=32, io_packet_size=128,
-        gpu_pool=0, cpu_pool=0)
+        work_packet_size=32, io_packet_size=64,
+        gpu_pool=(24 << 30) if have_gpu else 0,
+        cpu_pool=(4 << 30) if have_gpu else 0)
+    instances = 4 if have_gpu else 1
 
     def one_step(tag):
         sink = build_pipeline(sc, sp, video, args.pipeline, device,
                               f"bench_out_{tag}")
         sc.run(sink, perf, cache_mode=sp.CacheMode.Overwrite,
-               gpu_ids=gpu_ids, pipeline_instances=1)
+               gpu_ids=gpu_ids, pipeline_instances=instances)
 
     def sync():
         if have_gpu:

@@ -358,6 +358,12 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t) {
         const TableMetadata& table = out_tables_[t.job];
         std::vector<std::vector<Element>> cols(op.inputs.size());
         std::vector<std::vector<u8>> owned(op.inputs.size());
+        // Batched D2H: gather all device-resident outputs into one
+        // memcpy_vec (async on the instance stream, one sync).
+        std::vector<u8*> d2h_dst;
+        std::vector<const u8*> d2h_src;
+        std::vector<size_t> d2h_sz;
+        DeviceHandle gpu_src = CPU_DEVICE;
         for (size_t c = 0; c < op.inputs.size(); ++c) {
           auto& parent_map = outs[op.inputs[c].op][op.inputs[c].column];
           for (i64 row = t.start; row < t.end; ++row) {
@@ -371,13 +377,26 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t) {
               owned[c].push_back(0);
               continue;
             }
-            bool copied;
-            Element ce = element_to_device(e, CPU_DEVICE, copied);
+            if (!e.device.is_gpu()) {
+              e.index = row;
+              cols[c].push_back(e);
+              owned[c].push_back(0);
+              continue;
+            }
+            Element ce = e;
+            ce.device = CPU_DEVICE;
+            ce.buffer = new_buffer(CPU_DEVICE, e.size);
             ce.index = row;
             cols[c].push_back(ce);
-            if (copied) owned[c].push_back(1);
-            else owned[c].push_back(0);
+            owned[c].push_back(1);
+            d2h_dst.push_back(ce.buffer);
+            d2h_src.push_back(e.buffer);
+            d2h_sz.push_back(e.size);
+            gpu_src = e.device;
           }
+        }
+        if (!d2h_dst.empty()) {
+          memcpy_vec(d2h_dst, CPU_DEVICE, d2h_src, gpu_src, d2h_sz);
         }
         for (size_t c = 0; c < op.inputs.size(); ++c) {
           write_column_item(*db_, table, table.columns[c].name, t.task,
@@ -476,9 +495,11 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t) {
             }
           }
         }
-        // GPU kernels are async on this thread's stream; sync before
-        // releasing their input staging copies.
-        if (kdev.is_gpu()) sync_per_thread_stream();
+        // GPU kernels are async on this thread's stream; staging copies
+        // can only be released after the kernels that read them complete.
+        // Everything else stays async — same-stream ordering covers all
+        // later consumers (kernels and D2H copies alike).
+        if (!scratch.empty() && kdev.is_gpu()) sync_per_thread_stream();
         for (auto& e : scratch) delete_buffer(e.device, e.buffer);
         bi = be;
       }

@@ -21,7 +21,10 @@
 //   u32 super_off[nsuper] — byte offset of each supergroup's packed data,
 //                           relative to the packed region start
 //   u8  width[ngroups]    — bits per value in the group (0..8)
-//   packed region: per group, 32 values * width bits = 4*width bytes
+//   (pad to 4-byte alignment)
+//   packed region: per group, 32 values * width bits = 4*width bytes —
+//   every group payload is 4-byte aligned, so the GPU decoder unpacks with
+//   u32 loads
 //
 // Residuals: delta frames predict from the previous frame byte; key frames
 // predict from the previous byte within the group (first byte from 128).

@@ -67,7 +67,8 @@ void encode_frame(const u8* cur, const u8* prev, u32 nbytes, bool key,
   size_t header = 4 + 4 + 4 + 4 + 4;
   size_t packed_bytes = 0;
   for (u32 g = 0; g < ngroups; ++g) packed_bytes += 4u * widths[g];
-  size_t total = header + nsuper * 4 + ngroups + packed_bytes;
+  size_t widths_padded = (ngroups + 3) / 4 * 4;  // 4B-align packed region
+  size_t total = header + nsuper * 4 + widths_padded + packed_bytes;
   size_t base = out.size();
   out.resize(base + total, 0);
   u8* p = out.data() + base;
@@ -84,7 +85,7 @@ void encode_frame(const u8* cur, const u8* prev, u32 nbytes, bool key,
   std::memcpy(p, super_off.data(), nsuper * 4);
   p += nsuper * 4;
   std::memcpy(p, widths.data(), ngroups);
-  p += ngroups;
+  p += widths_padded;
   // pack
   for (u32 g = 0; g < ngroups; ++g) {
     u32 w = widths[g];
@@ -158,10 +159,12 @@ SvcPacketView svc_parse_packet(const u8* pkt, size_t size) {
   std::memcpy(&v.ngroups, pkt + 12, 4);
   std::memcpy(&v.nsuper, pkt + 16, 4);
   size_t header = 20;
-  SCA_CHECK(size >= header + v.nsuper * 4 + v.ngroups, "svc packet truncated");
+  size_t widths_padded = (v.ngroups + 3) / 4 * 4;
+  SCA_CHECK(size >= header + v.nsuper * 4 + widths_padded,
+            "svc packet truncated");
   v.super_off = reinterpret_cast<const u32*>(pkt + header);
   v.widths = pkt + header + v.nsuper * 4;
-  v.packed = v.widths + v.ngroups;
+  v.packed = v.widths + widths_padded;
   return v;
 }
 

@@ -31,7 +31,10 @@ __device__ inline u8 dev_unzigzag(u8 z) {
   return (u8)v;
 }
 
-// One block per supergroup, 128 threads.
+// One block per supergroup, 128 threads (one lane per group). Group
+// payloads are 4*width bytes and 4-byte aligned, so the unpack runs on u32
+// loads; the delta path reads prev and writes cur as u32s (the frame base
+// of a group is 32-byte aligned).
 __global__ void __launch_bounds__(128)
     svc_decode_frame_kernel(const u8* __restrict__ pkt_widths,
                             const u32* __restrict__ pkt_super_off,
@@ -39,30 +42,30 @@ __global__ void __launch_bounds__(128)
                             const u8* __restrict__ prev, bool is_key,
                             u32 nbytes, u32 ngroups,
                             u8* __restrict__ cur) {
-  __shared__ u32 offs[129];
+  __shared__ u32 offs[128];
   u32 s = blockIdx.x;
   u32 g0 = s * 128;
   u32 local_n = min(128u, ngroups - g0);
   u32 tid = threadIdx.x;
 
-  u32 w = tid < local_n ? pkt_widths[g0 + tid] : 0;
-  // simple LDS scan over 128 widths (exclusive)
-  offs[0] = 0;
+  u32 w = tid < local_n ? (u32)pkt_widths[g0 + tid] : 0;
+  // Hillis-Steele exclusive scan of 4*w over the 128 lanes.
+  u32 val = 4u * w;
+  offs[tid] = val;
   __syncthreads();
-  // serial scan by lane 0 (128 adds — negligible vs unpack work)
-  if (tid == 0) {
-    u32 acc = 0;
-    for (u32 i = 0; i < local_n; ++i) {
-      offs[i] = acc;
-      acc += 4u * pkt_widths[g0 + i];
-    }
-    offs[local_n] = acc;
+#pragma unroll
+  for (u32 d = 1; d < 128; d <<= 1) {
+    u32 x = tid >= d ? offs[tid - d] : 0;
+    __syncthreads();
+    offs[tid] += x;
+    __syncthreads();
   }
-  __syncthreads();
+  u32 my_off = offs[tid] - val;  // exclusive
   if (tid >= local_n) return;
 
   u32 g = g0 + tid;
-  const u8* q = pkt_packed + pkt_super_off[s] + offs[tid];
+  const u32* q =
+      reinterpret_cast<const u32*>(pkt_packed + pkt_super_off[s] + my_off);
   u8 res[32];
   if (w == 0) {
 #pragma unroll
@@ -72,10 +75,11 @@ __global__ void __launch_bounds__(128)
     u32 nacc = 0;
     u32 qi = 0;
     u32 mask = (1u << w) - 1;
+#pragma unroll 4
     for (int k = 0; k < 32; ++k) {
-      while (nacc < w) {
+      if (nacc < w) {
         acc |= ((u64)q[qi++]) << nacc;
-        nacc += 8;
+        nacc += 32;
       }
       res[k] = (u8)(acc & mask);
       acc >>= w;
@@ -83,20 +87,46 @@ __global__ void __launch_bounds__(128)
     }
   }
   u32 base = g * 32;
+  if (base + 32 > nbytes) {  // scalar tail group
+    u32 n = nbytes - base;
+    if (is_key) {
+      u8 p = 128;
+      for (u32 k = 0; k < n; ++k) {
+        p = (u8)(p + dev_unzigzag(res[k]));
+        cur[base + k] = p;
+      }
+    } else {
+      for (u32 k = 0; k < n; ++k)
+        cur[base + k] = (u8)(prev[base + k] + dev_unzigzag(res[k]));
+    }
+    return;
+  }
+  u32 out[8];
   if (is_key) {
     u8 p = 128;
+#pragma unroll
     for (int k = 0; k < 32; ++k) {
-      u32 i = base + k;
-      if (i >= nbytes) break;
       p = (u8)(p + dev_unzigzag(res[k]));
-      cur[i] = p;
+      out[k / 4] = (k % 4 == 0) ? p : (out[k / 4] | ((u32)p << (8 * (k % 4))));
     }
   } else {
-    u32 n = min(32u, nbytes - base);
-    for (u32 k = 0; k < n; ++k) {
-      cur[base + k] = (u8)(prev[base + k] + dev_unzigzag(res[k]));
+    const u32* prev32 = reinterpret_cast<const u32*>(prev + base);
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      u32 pv = prev32[k];
+      u32 o = 0;
+#pragma unroll
+      for (int b = 0; b < 4; ++b) {
+        u8 byte = (u8)((pv >> (8 * b)) & 0xff);
+        byte = (u8)(byte + dev_unzigzag(res[k * 4 + b]));
+        o |= (u32)byte << (8 * b);
+      }
+      out[k] = o;
     }
   }
+  uint4* dst = reinterpret_cast<uint4*>(cur + base);
+  dst[0] = make_uint4(out[0], out[1], out[2], out[3]);
+  dst[1] = make_uint4(out[4], out[5], out[6], out[7]);
 }
 
 }  // namespace
@@ -138,7 +168,7 @@ std::vector<Element> svc_decode_gpu(const u8* stream_host, size_t size,
     const u8* pkt_d = d_stream + pkt_off;
     const u32* super_off_d = reinterpret_cast<const u32*>(pkt_d + 20);
     const u8* widths_d = pkt_d + 20 + v.nsuper * 4;
-    const u8* packed_d = widths_d + v.ngroups;
+    const u8* packed_d = widths_d + (v.ngroups + 3) / 4 * 4;
 
     bool wanted = wi < want.size() && want[wi] == f;
     u8* cur;
