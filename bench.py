@@ -131,6 +131,11 @@ def main():
     t1 = time.perf_counter()
 
     elapsed = t1 - t0
+    if os.environ.get("SCANNER_PROFILE"):
+        stats = sc.profile().statistics()
+        for k in sorted(stats, key=lambda k: -stats[k]["total_ms"]):
+            log(f"[prof] {k}: {stats[k]['total_ms']:.1f} ms "
+                f"x{stats[k]['count']}")
     if distributed:
         t = torch.tensor([elapsed], dtype=torch.float64,
                          device="cuda" if have_gpu else "cpu")

@@ -206,23 +206,36 @@ void LocalExecutor::load_inputs(
             local.push_back(rows[ri] - ir.row_start);
             ++ri;
           }
-          auto stream = db_->storage()->read_all(
-              db_->paths().item(table.id, table.column_id(src.column),
-                                ir.item));
+          // Read only the byte range covering the keyframe-aligned decode
+          // span, into pinned memory (fast DMA H2D).
+          std::vector<i64> span = svc_decode_span(m, local);
+          if (span.empty()) continue;
+          u64 lo = m.sample_offsets[span.front()];
+          u64 hi = m.sample_offsets[span.back()] + m.sample_sizes[span.back()];
+          u8* stream_buf = new_buffer(CPU_DEVICE, hi - lo);
+          {
+            Profiler::Scope sl(inst.profiler, "load:video");
+            db_->storage()->read_range(
+                db_->paths().item(table.id, table.column_id(src.column),
+                                  ir.item),
+                lo, hi - lo, stream_buf);
+          }
           if (inst.gpu.is_gpu()) {
             Profiler::Scope sd(inst.profiler, "decode:gpu");
-            auto elems = svc_decode_gpu(stream.data(), stream.size(), m,
-                                        local, inst.gpu);
+            auto elems =
+                svc_decode_gpu(stream_buf, hi - lo, m, local, inst.gpu, lo);
             for (auto& e : elems) {
               i64 grow = e.index + ir.row_start;
               Element ge = e;
               ge.index = grow;
               outs[op_idx][col_name][grow] = ge;
             }
+            delete_buffer(CPU_DEVICE, stream_buf);
           } else {
             Profiler::Scope sd(inst.profiler, "decode:cpu");
             std::vector<std::vector<u8>> frames;
-            svc_decode_cpu(stream.data(), stream.size(), m, local, frames);
+            svc_decode_cpu(stream_buf, hi - lo, m, local, frames, lo);
+            delete_buffer(CPU_DEVICE, stream_buf);
             for (size_t k = 0; k < local.size(); ++k) {
               Element e;
               e.is_frame = true;

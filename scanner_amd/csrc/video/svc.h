@@ -46,9 +46,12 @@ void svc_encode_cpu(const u8* frames, i64 n, i32 h, i32 w, i32 c, i32 gop,
                     std::vector<u8>& stream, VideoMetadata& vm);
 
 // Decode the given (sorted, item-local) frame indices. out[i] = raw frame.
+// `stream` holds bytes [stream_offset, stream_offset+size) of the item —
+// callers read only the GOP-span byte range (reference analogue:
+// keyframe-aligned encoded-range reads, column_source.cpp:209).
 void svc_decode_cpu(const u8* stream, size_t size, const VideoMetadata& vm,
                     const std::vector<i64>& want,
-                    std::vector<std::vector<u8>>& out);
+                    std::vector<std::vector<u8>>& out, u64 stream_offset = 0);
 
 // Given wanted frames, the minimal list of frames that must be decoded
 // (keyframe-aligned spans; reference analogue: DecodeArgs GOP spans).
@@ -62,7 +65,7 @@ std::vector<i64> svc_decode_span(const VideoMetadata& vm,
 std::vector<Element> svc_decode_gpu(const u8* stream_host, size_t size,
                                     const VideoMetadata& vm,
                                     const std::vector<i64>& want,
-                                    DeviceHandle dev);
+                                    DeviceHandle dev, u64 stream_offset = 0);
 
 struct SvcPacketView {
   bool is_key = false;

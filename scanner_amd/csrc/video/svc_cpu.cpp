@@ -216,7 +216,7 @@ std::vector<i64> svc_decode_span(const VideoMetadata& vm,
 
 void svc_decode_cpu(const u8* stream, size_t size, const VideoMetadata& vm,
                     const std::vector<i64>& want,
-                    std::vector<std::vector<u8>>& out) {
+                    std::vector<std::vector<u8>>& out, u64 stream_offset) {
   u32 nbytes = (u32)((i64)vm.height * vm.width * vm.channels);
   std::vector<i64> span = svc_decode_span(vm, want);
   std::vector<u8> prev(nbytes), cur(nbytes);
@@ -224,7 +224,11 @@ void svc_decode_cpu(const u8* stream, size_t size, const VideoMetadata& vm,
   i64 last = -2;
   for (i64 f : span) {
     SCA_CHECK(f < (i64)vm.sample_offsets.size(), "frame beyond stream");
-    const u8* pkt = stream + vm.sample_offsets[f];
+    SCA_CHECK(vm.sample_offsets[f] >= stream_offset &&
+                  vm.sample_offsets[f] + vm.sample_sizes[f] <=
+                      stream_offset + size,
+              "svc stream range does not cover decode span");
+    const u8* pkt = stream + (vm.sample_offsets[f] - stream_offset);
     SvcPacketView v = svc_parse_packet(pkt, vm.sample_sizes[f]);
     SCA_CHECK(v.is_key || f == last + 1,
               "svc decode: non-contiguous delta frame");

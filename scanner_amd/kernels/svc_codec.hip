@@ -134,7 +134,7 @@ __global__ void __launch_bounds__(128)
 std::vector<Element> svc_decode_gpu(const u8* stream_host, size_t size,
                                     const VideoMetadata& vm,
                                     const std::vector<i64>& want,
-                                    DeviceHandle dev) {
+                                    DeviceHandle dev, u64 stream_offset) {
   hipStream_t s = (hipStream_t)per_thread_hip_stream();
   u32 nbytes = (u32)((i64)vm.height * vm.width * vm.channels);
   std::vector<i64> span = svc_decode_span(vm, want);
@@ -146,9 +146,11 @@ std::vector<Element> svc_decode_gpu(const u8* stream_host, size_t size,
   // adjacent so over-read is bounded by skipped GOPs).
   u64 lo = vm.sample_offsets[span.front()];
   u64 hi = vm.sample_offsets[span.back()] + vm.sample_sizes[span.back()];
-  SCA_CHECK(hi <= size, "svc stream truncated");
+  SCA_CHECK(lo >= stream_offset && hi <= stream_offset + size,
+            "svc stream range does not cover decode span");
   u8* d_stream = new_buffer(dev, hi - lo);
-  memcpy_buffer(d_stream, dev, stream_host + lo, CPU_DEVICE, hi - lo);
+  memcpy_buffer(d_stream, dev, stream_host + (lo - stream_offset),
+                CPU_DEVICE, hi - lo);
 
   u8* scratch[2] = {nullptr, nullptr};
   auto get_scratch = [&](int i) {
@@ -160,7 +162,7 @@ std::vector<Element> svc_decode_gpu(const u8* stream_host, size_t size,
   u8* prev = nullptr;
   int flip = 0;
   for (i64 f : span) {
-    const u8* pkt_h = stream_host + vm.sample_offsets[f];
+    const u8* pkt_h = stream_host + (vm.sample_offsets[f] - stream_offset);
     SvcPacketView v = svc_parse_packet(pkt_h, vm.sample_sizes[f]);
     SCA_CHECK(v.nbytes == nbytes, "svc frame size mismatch");
     // device pointers into d_stream at the same relative offsets
