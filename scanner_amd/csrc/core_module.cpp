@@ -6,6 +6,7 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include "../kernels/dnn.h"
 #include "dag/graph.h"
 #include "engine/executor.h"
 #include "hip_util.h"
@@ -86,9 +87,62 @@ PYBIND11_MODULE(_core, m) {
 
   register_stdlib_ops();
   register_gpu_ops();
+  register_resnet50_op();
 
   m.def("have_gpu", &have_gpu);
   m.def("gpu_device_count", &gpu_device_count);
+
+  // Raw MFMA GEMM entry for numerics tests: f32 inputs are rounded to
+  // bf16 on the host, C = A @ B_nk^T computed on GPU 0.
+  m.def("gemm_bf16_test",
+        [](py::array_t<float, py::array::c_style | py::array::forcecast> A,
+           py::array_t<float, py::array::c_style | py::array::forcecast> Bnk,
+           bool relu) {
+          SCA_CHECK(have_gpu(), "gemm_bf16_test needs a GPU");
+          int M = (int)A.shape(0), K = (int)A.shape(1);
+          int N = (int)Bnk.shape(0);
+          SCA_CHECK((int)Bnk.shape(1) == K, "K mismatch");
+          DeviceHandle dev{DeviceType::GPU, 0};
+          auto to_bf16 = [](const float* p, size_t n) {
+            std::vector<u16> v(n);
+            for (size_t i = 0; i < n; ++i) {
+              u32 bits;
+              std::memcpy(&bits, &p[i], 4);
+              bits += 0x7fff + ((bits >> 16) & 1);
+              v[i] = (u16)(bits >> 16);
+            }
+            return v;
+          };
+          auto ah = to_bf16(A.data(), (size_t)M * K);
+          auto bh = to_bf16(Bnk.data(), (size_t)N * K);
+          u8* dA = new_buffer(dev, ah.size() * 2);
+          u8* dB = new_buffer(dev, bh.size() * 2);
+          u8* dC = new_buffer(dev, (size_t)M * N * 2);
+          memcpy_buffer(dA, dev, (u8*)ah.data(), CPU_DEVICE, ah.size() * 2);
+          memcpy_buffer(dB, dev, (u8*)bh.data(), CPU_DEVICE, bh.size() * 2);
+          GemmArgs g;
+          g.A = dA;
+          g.B = dB;
+          g.C = dC;
+          g.M = M;
+          g.N = N;
+          g.K = K;
+          g.relu = relu;
+          gemm_bf16(g, per_thread_hip_stream());
+          sync_per_thread_stream();
+          std::vector<u16> ch((size_t)M * N);
+          memcpy_buffer((u8*)ch.data(), CPU_DEVICE, dC, dev, ch.size() * 2);
+          delete_buffer(dev, dA);
+          delete_buffer(dev, dB);
+          delete_buffer(dev, dC);
+          py::array_t<float> out({M, N});
+          float* op = out.mutable_data();
+          for (size_t i = 0; i < ch.size(); ++i) {
+            u32 bits = (u32)ch[i] << 16;
+            std::memcpy(&op[i], &bits, 4);
+          }
+          return out;
+        });
 
   m.def("init_memory", [](size_t cpu_pool, size_t gpu_pool,
                           std::vector<i32> gpu_ids) {

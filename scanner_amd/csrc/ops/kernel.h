@@ -153,5 +153,6 @@ struct KernelRegistrar {
 // Register all built-in C++ ops/kernels (called once from module init; we
 // avoid static-initializer ordering issues by explicit registration).
 void register_stdlib_ops();
+void register_resnet50_op();
 
 }  // namespace sca

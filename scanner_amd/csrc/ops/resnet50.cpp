@@ -1,0 +1,396 @@
+// ResNet-50 bf16 inference op — the DNN-inference capability of the
+// framework (parity role: the reference's scannertools DNN ops, e.g.
+// caffe-based frame classification). All conv/GEMM work runs on the
+// hand-written MFMA GEMM (kernels/gemm_mfma.hip) via im2col; BN is folded
+// into the GEMM epilogue (scale/bias), ReLU and residual adds are fused.
+// Weights are random-init (He) by default — there is no network in this
+// environment — or loaded from a tensor file for numerics tests
+// (tests/test_resnet_gpu.py compares against a PyTorch fp32 reference).
+#include <random>
+
+#include "../../kernels/dnn.h"
+#include "../memory.h"
+#include "../msgpack.h"
+#include "../serialize.h"
+#include "../storage.h"
+#include "kernel.h"
+
+namespace sca {
+
+namespace {
+
+struct ConvSpec {
+  std::string name;
+  int in_c, out_c, r, s, stride, pad;
+  bool relu;
+  int residual_from = -1;  // index into block-local buffers (see forward)
+  int kp() const { return (r * s * in_c + 63) / 64 * 64; }
+  int np() const { return (out_c + 63) / 64 * 64; }
+};
+
+// The full layer list, grouped into bottleneck blocks.
+struct Bottleneck {
+  int in_c, mid_c, out_c, stride;
+  bool downsample;
+};
+
+struct ResNet50Config {
+  std::vector<Bottleneck> blocks;
+  ResNet50Config() {
+    auto stage = [&](int n, int in_c, int mid, int out, int stride) {
+      for (int i = 0; i < n; ++i) {
+        blocks.push_back(Bottleneck{i == 0 ? in_c : out, mid, out,
+                                    i == 0 ? stride : 1, i == 0});
+      }
+    };
+    stage(3, 64, 64, 256, 1);
+    stage(4, 256, 128, 512, 2);
+    stage(6, 512, 256, 1024, 2);
+    stage(3, 1024, 512, 2048, 2);
+  }
+};
+
+struct Tensors {
+  // name -> f32 host data
+  std::map<std::string, std::vector<f32>> t;
+  bool has(const std::string& n) const { return t.count(n); }
+  std::vector<f32>& operator[](const std::string& n) { return t[n]; }
+};
+
+// Deterministic random weights: He-normal conv weights, BN folded to
+// scale ~ U(0.7, 1.3), bias ~ N(0, 0.05).
+void random_init(Tensors& ts, const std::string& name, int out_c, int in_c,
+                 int r, int s, std::mt19937& rng) {
+  std::normal_distribution<f32> nd(0.f, std::sqrt(2.f / (in_c * r * s)));
+  auto& w = ts[name + ".weight"];
+  w.resize((size_t)out_c * r * s * in_c);
+  for (auto& v : w) v = nd(rng);
+  std::uniform_real_distribution<f32> su(0.7f, 1.3f);
+  std::normal_distribution<f32> bn(0.f, 0.05f);
+  auto& sc = ts[name + ".scale"];
+  auto& bi = ts[name + ".bias"];
+  sc.resize(out_c);
+  bi.resize(out_c);
+  for (auto& v : sc) v = su(rng);
+  for (auto& v : bi) v = bn(rng);
+}
+
+Tensors load_tensor_file(const std::string& path) {
+  auto storage = StorageBackend::make_posix();
+  auto buf = storage->read_all(path);
+  BinReader r(buf);
+  u32 magic = r.u32v();
+  SCA_CHECK(magic == 0x52534E54, "bad tensor file magic");  // 'TNSR'
+  u32 n = r.u32v();
+  Tensors ts;
+  for (u32 i = 0; i < n; ++i) {
+    std::string name = r.str();
+    ts[name] = r.vec_pod<f32>();
+  }
+  return ts;
+}
+
+// Device-side model: one bf16 weight blob ([out][kp] per conv,
+// concatenated), one f32 scale+bias blob.
+struct DeviceModel {
+  DeviceHandle dev;
+  u8* weights = nullptr;   // bf16
+  u8* scalebias = nullptr; // f32: per conv scale then bias (padded np)
+  struct Entry {
+    ConvSpec spec;
+    size_t w_off;   // bf16 elements
+    size_t sb_off;  // f32 elements (scale at sb_off, bias at sb_off+np)
+  };
+  std::vector<Entry> convs;  // ordered; name->idx
+  std::map<std::string, int> by_name;
+  float* mean = nullptr;  // 3 floats + 3 std
+  ~DeviceModel() {
+    if (weights) delete_buffer(dev, weights);
+    if (scalebias) delete_buffer(dev, scalebias);
+    if (mean) delete_buffer(dev, (u8*)mean);
+  }
+};
+
+u16 f32_to_bf16_host(f32 v) {
+  u32 bits;
+  std::memcpy(&bits, &v, 4);
+  // round-to-nearest-even
+  u32 lsb = (bits >> 16) & 1;
+  bits += 0x7fff + lsb;
+  return (u16)(bits >> 16);
+}
+
+std::shared_ptr<DeviceModel> build_model(DeviceHandle dev,
+                                         const std::string& weights_file,
+                                         u64 seed) {
+  ResNet50Config cfg;
+  std::vector<ConvSpec> specs;
+  specs.push_back({"conv1", 3, 64, 7, 7, 2, 3, true});
+  for (size_t b = 0; b < cfg.blocks.size(); ++b) {
+    const Bottleneck& bk = cfg.blocks[b];
+    std::string p = "block" + std::to_string(b);
+    specs.push_back({p + ".conv1", bk.in_c, bk.mid_c, 1, 1, 1, 0, true});
+    specs.push_back(
+        {p + ".conv2", bk.mid_c, bk.mid_c, 3, 3, bk.stride, 1, true});
+    specs.push_back({p + ".conv3", bk.mid_c, bk.out_c, 1, 1, 1, 0, true});
+    if (bk.downsample) {
+      specs.push_back(
+          {p + ".downsample", bk.in_c, bk.out_c, 1, 1, bk.stride, 0, false});
+    }
+  }
+  specs.push_back({"fc", 2048, 1000, 1, 1, 1, 0, false});
+
+  Tensors ts;
+  if (!weights_file.empty()) {
+    ts = load_tensor_file(weights_file);
+  } else {
+    std::mt19937 rng((u32)seed);
+    for (auto& sp : specs) {
+      random_init(ts, sp.name, sp.out_c, sp.in_c, sp.r, sp.s, rng);
+    }
+    // fc: plain linear (scale 1)
+    auto& sc = ts["fc.scale"];
+    std::fill(sc.begin(), sc.end(), 1.f);
+  }
+
+  auto model = std::make_shared<DeviceModel>();
+  model->dev = dev;
+  size_t w_elems = 0, sb_elems = 0;
+  for (auto& sp : specs) {
+    model->by_name[sp.name] = (int)model->convs.size();
+    model->convs.push_back({sp, w_elems, sb_elems});
+    w_elems += (size_t)sp.np() * sp.kp();
+    sb_elems += 2 * (size_t)sp.np();
+  }
+  std::vector<u16> wh(w_elems, 0);
+  std::vector<f32> sbh(sb_elems, 0.f);
+  for (auto& e : model->convs) {
+    const ConvSpec& sp = e.spec;
+    auto& w = ts[sp.name + ".weight"];
+    SCA_CHECK((i64)w.size() == (i64)sp.out_c * sp.r * sp.s * sp.in_c,
+              "weight size mismatch for " + sp.name);
+    int krs = sp.r * sp.s * sp.in_c;
+    // tensor file layout: [out][r][s][in] — same k ordering as im2col
+    // (cell-major, channel-minor)
+    for (int o = 0; o < sp.out_c; ++o) {
+      for (int k = 0; k < krs; ++k) {
+        wh[e.w_off + (size_t)o * sp.kp() + k] =
+            f32_to_bf16_host(w[(size_t)o * krs + k]);
+      }
+    }
+    auto& sc = ts[sp.name + ".scale"];
+    auto& bi = ts[sp.name + ".bias"];
+    SCA_CHECK((i64)sc.size() == sp.out_c && (i64)bi.size() == sp.out_c,
+              "scale/bias size mismatch for " + sp.name);
+    for (int o = 0; o < sp.out_c; ++o) {
+      sbh[e.sb_off + o] = sc[o];
+      sbh[e.sb_off + sp.np() + o] = bi[o];
+    }
+  }
+  model->weights = new_buffer(dev, w_elems * 2);
+  memcpy_buffer(model->weights, dev, (const u8*)wh.data(), CPU_DEVICE,
+                w_elems * 2);
+  model->scalebias = new_buffer(dev, sb_elems * 4);
+  memcpy_buffer(model->scalebias, dev, (const u8*)sbh.data(), CPU_DEVICE,
+                sb_elems * 4);
+  f32 ms[6] = {0.485f, 0.456f, 0.406f, 0.229f, 0.224f, 0.225f};
+  model->mean = (float*)new_buffer(dev, 6 * 4);
+  memcpy_buffer((u8*)model->mean, dev, (const u8*)ms, CPU_DEVICE, 24);
+  return model;
+}
+
+// Model cache: one DeviceModel per (device, weights_file, seed), shared
+// across pipeline instances (reference analogue: fetch_resources +
+// setup_with_resources barrier, evaluate_worker.cpp:493-550).
+std::mutex g_model_mu;
+std::map<std::string, std::shared_ptr<DeviceModel>> g_models;
+
+std::shared_ptr<DeviceModel> get_model(DeviceHandle dev,
+                                       const std::string& weights_file,
+                                       u64 seed) {
+  std::string key = dev.to_string() + "|" + weights_file + "|" +
+                    std::to_string(seed);
+  std::lock_guard<std::mutex> l(g_model_mu);
+  auto it = g_models.find(key);
+  if (it != g_models.end()) return it->second;
+  auto m = build_model(dev, weights_file, seed);
+  g_models[key] = m;
+  return m;
+}
+
+class ResNet50KernelGPU : public BatchedKernel {
+ public:
+  explicit ResNet50KernelGPU(const KernelConfig& cfg) : BatchedKernel(cfg) {
+    auto a = mp::decode(cfg.args);
+    weights_file_ = a.get_str("weights_file", "");
+    seed_ = (u64)a.get_int("seed", 1234);
+    batch_ = cfg.max_batch;
+    model_ = get_model(cfg.device, weights_file_, seed_);
+  }
+
+  void execute_batch(const BatchedElements& in, BatchedElements& out) override;
+
+ private:
+  std::string weights_file_;
+  u64 seed_;
+  i32 batch_;
+  std::shared_ptr<DeviceModel> model_;
+};
+
+void ResNet50KernelGPU::execute_batch(const BatchedElements& in,
+                                      BatchedElements& out) {
+  void* s = per_thread_hip_stream();
+  DeviceHandle dev = config_.device;
+  int n = (int)in[0].size();
+  if (n == 0) return;
+  const Element& f0 = in[0][0];
+  SCA_CHECK(f0.is_frame && f0.device.is_gpu(),
+            "ResNet50 needs GPU frame input");
+  int ih = f0.frame_info.shape[0], iw = f0.frame_info.shape[1],
+      ic = f0.frame_info.shape[2];
+
+  // device pointer array for the input frames
+  std::vector<const u8*> ptrs(n);
+  for (int i = 0; i < n; ++i) ptrs[i] = in[0][i].buffer;
+  u8* d_ptrs = new_buffer(dev, n * sizeof(u8*));
+  memcpy_buffer(d_ptrs, dev, (const u8*)ptrs.data(), CPU_DEVICE,
+                n * sizeof(u8*));
+
+  // workspace (act elements peak: n*112*112*64 == n*802816)
+  size_t act_elems = (size_t)n * 802816;
+  size_t im2col_elems = (size_t)n * 112 * 112 * 192;  // conv1 is the max
+  u8* act0 = new_buffer(dev, act_elems * 2);
+  u8* act1 = new_buffer(dev, act_elems * 2);
+  u8* act2 = new_buffer(dev, act_elems * 2);
+  u8* resid = new_buffer(dev, act_elems * 2);
+  u8* colbuf = new_buffer(dev, im2col_elems * 2);
+  u8* pre = new_buffer(dev, (size_t)n * 224 * 224 * 3 * 2);
+
+  auto conv = [&](const char* name, const u8* x, int h, int w, u8* y,
+                  const u8* residual, int& oh, int& ow) -> int {
+    const auto& e = model_->convs[model_->by_name.at(name)];
+    const ConvSpec& sp = e.spec;
+    oh = (h + 2 * sp.pad - sp.r) / sp.stride + 1;
+    ow = (w + 2 * sp.pad - sp.s) / sp.stride + 1;
+    int M = n * oh * ow;
+    const u8* A = x;
+    if (!(sp.r == 1 && sp.s == 1 && sp.stride == 1 && sp.pad == 0)) {
+      im2col_bf16(x, n, h, w, sp.in_c, sp.r, sp.s, sp.stride, sp.pad,
+                  colbuf, oh, ow, sp.kp(), s);
+      A = colbuf;
+    }
+    GemmArgs g;
+    g.A = A;
+    g.B = model_->weights + e.w_off * 2;
+    g.C = y;
+    g.M = M;
+    g.N = sp.np();
+    g.K = sp.kp();
+    g.scale = (const float*)model_->scalebias + e.sb_off;
+    g.bias = (const float*)model_->scalebias + e.sb_off + sp.np();
+    g.residual = residual;
+    g.relu = sp.relu;
+    gemm_bf16(g, s);
+    return sp.np();
+  };
+
+  // ---- forward ----
+  f32* mean = model_->mean;
+  preprocess_frames_bf16(d_ptrs, n, ih, iw, ic, 224, pre, mean, mean + 3, s);
+  int h = 224, w = 224, oh, ow;
+  conv("conv1", pre, h, w, act0, nullptr, oh, ow);  // -> n,112,112,64
+  h = oh;
+  w = ow;
+  maxpool3x3s2_bf16(act0, n, h, w, 64, act1, 56, 56, s);
+  h = w = 56;
+  // Three rotating activation buffers: x holds the block input; a/b are
+  // the two others. conv3 writes into a (its conv1 temp is dead by then),
+  // never into the residual source.
+  u8* bufs[3] = {act1, act0, act2};
+  u8* x = bufs[0];
+  ResNet50Config cfg;
+  for (size_t b = 0; b < cfg.blocks.size(); ++b) {
+    const Bottleneck& bk = cfg.blocks[b];
+    std::string p = "block" + std::to_string(b);
+    u8* others[2];
+    int oi = 0;
+    for (int i = 0; i < 3; ++i)
+      if (bufs[i] != x) others[oi++] = bufs[i];
+    u8* a = others[0];
+    u8* bbuf = others[1];
+    const u8* identity = x;
+    if (bk.downsample) {
+      conv((p + ".downsample").c_str(), x, h, w, resid, nullptr, oh, ow);
+      identity = resid;
+    }
+    conv((p + ".conv1").c_str(), x, h, w, a, nullptr, oh, ow);
+    conv((p + ".conv2").c_str(), a, h, w, bbuf, nullptr, oh, ow);
+    h = oh;
+    w = ow;
+    conv((p + ".conv3").c_str(), bbuf, h, w, a, identity, oh, ow);
+    x = a;
+  }
+  // x: [n,7,7,2048]
+  u8* y = bufs[0] == x ? bufs[1] : bufs[0];
+  global_avgpool_bf16(x, n, 7, 7, 2048, y, s);
+  // fc: M=n, K=2048, N=1024 (padded from 1000)
+  {
+    const auto& e = model_->convs[model_->by_name.at("fc")];
+    GemmArgs g;
+    g.A = y;
+    g.B = model_->weights + e.w_off * 2;
+    g.C = x;
+    g.M = n;
+    g.N = e.spec.np();
+    g.K = e.spec.kp();
+    g.scale = (const float*)model_->scalebias + e.sb_off;
+    g.bias = (const float*)model_->scalebias + e.sb_off + e.spec.np();
+    g.relu = false;
+    gemm_bf16(g, s);
+  }
+
+  // logits: first 1000 of each padded row, bf16 -> f32, one block buffer
+  size_t logit_bytes = (size_t)n * 1000 * 4;
+  u8* out_block = new_block_buffer(dev, logit_bytes, n);
+  bf16_rows_to_f32(x, n, 1024, 1000, out_block, s);
+
+  sync_per_thread_stream();
+  delete_buffer(dev, d_ptrs);
+  delete_buffer(dev, act0);
+  delete_buffer(dev, act1);
+  delete_buffer(dev, act2);
+  delete_buffer(dev, resid);
+  delete_buffer(dev, colbuf);
+  delete_buffer(dev, pre);
+
+  for (int i = 0; i < n; ++i) {
+    Element e;
+    e.buffer = out_block + (size_t)i * 1000 * 4;
+    e.size = 1000 * 4;
+    e.device = dev;
+    out[0].push_back(e);
+  }
+}
+
+}  // namespace
+
+void register_resnet50_op() {
+  static bool done = false;
+  if (done) return;
+  done = true;
+  OpInfo o;
+  o.name = "ResNet50";
+  o.input_columns = {{"frame", ColumnType::Video}};
+  o.output_columns = {{"logits", ColumnType::Bytes}};
+  op_registry().add(o);
+  KernelFactory f;
+  f.op_name = "ResNet50";
+  f.device_type = DeviceType::GPU;
+  f.preferred_batch = 16;
+  f.make = [](const KernelConfig& c) -> std::unique_ptr<BaseKernel> {
+    return std::make_unique<ResNet50KernelGPU>(c);
+  };
+  kernel_registry().add(f);
+}
+
+}  // namespace sca

@@ -1,0 +1,48 @@
+// Launch APIs for the DNN kernels (implemented in gemm_mfma.hip and
+// dnn_ops.hip; host orchestration in csrc/ops/resnet50.cpp). All pointers
+// are device memory; all launches go onto the caller's HIP stream.
+#pragma once
+
+#include "../csrc/common.h"
+
+namespace sca {
+
+struct GemmArgs {
+  const void* A = nullptr;  // bf16 [M][K] row-major
+  const void* B = nullptr;  // bf16 [N][K] row-major (transposed weights)
+  void* C = nullptr;        // bf16 [M][N]
+  int M = 0, N = 0, K = 0;  // K, N multiples of 64
+  const float* scale = nullptr;     // per-col (folded BN gamma/sqrt(var))
+  const float* bias = nullptr;      // per-col
+  const void* residual = nullptr;   // bf16 [M][N] added before activation
+  bool relu = false;
+};
+
+void gemm_bf16(const GemmArgs& g, void* stream);
+
+// u8 HWC frames (device pointer array) -> normalized bf16 NHWC out_hw x
+// out_hw x 3, bilinear resize. mean/std per channel (device, 3 floats).
+void preprocess_frames_bf16(const void* frames_ptr_array, int n, int in_h,
+                            int in_w, int in_c, int out_hw, void* out,
+                            const float* mean, const float* std_,
+                            void* stream);
+
+// NHWC bf16 -> im2col rows [n*out_h*out_w][k_padded] (zero-padded past
+// r*s*c). stride/pad symmetric.
+void im2col_bf16(const void* in, int n, int h, int w, int c, int r, int s,
+                 int stride, int pad, void* out, int out_h, int out_w,
+                 int k_padded, void* stream);
+
+// 3x3 stride-2 pad-1 max pool, NHWC bf16.
+void maxpool3x3s2_bf16(const void* in, int n, int h, int w, int c, void* out,
+                       int out_h, int out_w, void* stream);
+
+// global average pool NHWC -> [n][c] bf16
+void global_avgpool_bf16(const void* in, int n, int h, int w, int c,
+                         void* out, void* stream);
+
+// [n][stride_cols] bf16 -> [n][ncols] f32 (row-sliced cast)
+void bf16_rows_to_f32(const void* in, int n, int stride_cols, int ncols,
+                      void* out, void* stream);
+
+}  // namespace sca

@@ -1,0 +1,198 @@
+// Hand-written bf16 GEMM on MFMA for gfx950 — the conv/GEMM engine of the
+// DNN-inference op (ResNet-50). Structure follows the CDNA4 canonical GEMM
+// anatomy (cdna_hip_programming.md §5): 128-wide tiles, global_load_lds
+// 16-byte staging into double-buffered LDS, v_mfma_f32_16x16x32_bf16 inner
+// loop accumulating fp32, fused epilogue (per-channel scale+bias = folded
+// BN, ReLU, residual add) writing bf16.
+//
+// C[M,N] = A[M,K] @ B[N,K]^T   (A row-major [M][K]; B stored [N][K] so
+// both operands are contiguous in K — weights are laid out at init time).
+// K and N must be multiples of 64 (callers pad); M is arbitrary.
+#include <hip/hip_runtime.h>
+
+#include "../csrc/memory.h"
+#include "dnn.h"
+
+namespace sca {
+
+namespace {
+
+using bf16 = __bf16;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+__device__ inline float bf16_to_f32(bf16 v) { return (float)v; }
+
+__device__ inline bf16 f32_to_bf16(float v) { return (bf16)v; }
+
+// One workgroup = 256 threads = 4 waves in a WM x WN grid; each wave owns
+// a (BM/WM) x (BN/WN) output sub-tile as FM x FN fragments of 16x16.
+template <int BM, int BN, int WM, int WN, bool RELU, bool RESIDUAL>
+__global__ void __launch_bounds__(256, 2)
+    gemm_bf16_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
+                     bf16* __restrict__ C, int M, int N, int K,
+                     const float* __restrict__ scale,
+                     const float* __restrict__ bias,
+                     const bf16* __restrict__ residual) {
+  constexpr int BK = 64;
+  constexpr int FM = BM / WM / 16;
+  constexpr int FN = BN / WN / 16;
+  // LDS: [2 buffers][A BM rows + B BN rows][BK]
+  __shared__ bf16 lds[2 * (BM + BN) * BK];
+
+  int tid = threadIdx.x;
+  int lane = tid & 63;
+  int wave = tid >> 6;
+  int wrow = wave / WN;
+  int wcol = wave % WN;
+
+  // XCD-aware swizzle (T1): consecutive blockIdx.x stay on one XCD's share
+  // of the N dimension when the grid allows.
+  int nwg = gridDim.x;
+  int wgid = blockIdx.x;
+  if (nwg % 8 == 0) {
+    int q = nwg / 8;
+    wgid = (wgid % q) * 8 + wgid / q;
+  }
+  int ntiles_n = (N + BN - 1) / BN;
+  int m0 = (wgid / ntiles_n) * BM;
+  int n0 = (wgid % ntiles_n) * BN;
+
+  f32x4 acc[FM][FN];
+#pragma unroll
+  for (int i = 0; i < FM; ++i)
+#pragma unroll
+    for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // Staging: each wave covers 8 rows x 64 k per glds instruction
+  // (64 lanes x 16 B = 8 bf16 each). A tile: BM rows -> BM/8 instrs
+  // over 4 waves; same for B.
+  constexpr int A_INSTRS = BM / 8 / 4;  // per wave
+  constexpr int B_INSTRS = BN / 8 / 4;
+  auto stage = [&](int buf, int k0) {
+    const int lrow = lane >> 3;          // 0..7 row within instr block
+    const int lk = (lane & 7) * 8;       // k offset (8 bf16 = 16B)
+    bf16* lds_a = lds + buf * (BM + BN) * BK;
+    bf16* lds_b = lds_a + BM * BK;
+#pragma unroll
+    for (int i = 0; i < A_INSTRS; ++i) {
+      int row = (wave * A_INSTRS + i) * 8 + lrow;
+      int grow = m0 + row;
+      if (grow >= M) grow = M - 1;  // clamp: garbage rows masked at store
+      const bf16* src = A + (size_t)grow * K + k0 + lk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3)))
+               uint32_t*)(lds_a + (size_t)(wave * A_INSTRS + i) * 8 * BK),
+          16, 0, 0);
+    }
+#pragma unroll
+    for (int i = 0; i < B_INSTRS; ++i) {
+      int row = (wave * B_INSTRS + i) * 8 + lrow;
+      const bf16* src = B + (size_t)(n0 + row) * K + k0 + lk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3)))
+               uint32_t*)(lds_b + (size_t)(wave * B_INSTRS + i) * 8 * BK),
+          16, 0, 0);
+    }
+  };
+
+  int ksteps = K / BK;
+  stage(0, 0);
+  for (int ks = 0; ks < ksteps; ++ks) {
+    int buf = ks & 1;
+    // prefetch next while computing current
+    __builtin_amdgcn_s_waitcnt(/*vmcnt(0) lgkmcnt(0)*/ 0);
+    __syncthreads();
+    if (ks + 1 < ksteps) stage(buf ^ 1, (ks + 1) * BK);
+
+    const bf16* lds_a = lds + buf * (BM + BN) * BK;
+    const bf16* lds_b = lds_a + BM * BK;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {  // two K=32 chunks per BK
+      int kbase = kk * 32 + (lane >> 4) * 8;
+      bf16x8 afrag[FM], bfrag[FN];
+#pragma unroll
+      for (int i = 0; i < FM; ++i) {
+        int row = wrow * (BM / WM) + i * 16 + (lane & 15);
+        afrag[i] = *reinterpret_cast<const bf16x8*>(
+            lds_a + (size_t)row * BK + kbase);
+      }
+#pragma unroll
+      for (int j = 0; j < FN; ++j) {
+        int col = wcol * (BN / WN) + j * 16 + (lane & 15);
+        bfrag[j] = *reinterpret_cast<const bf16x8*>(
+            lds_b + (size_t)col * BK + kbase);
+      }
+#pragma unroll
+      for (int i = 0; i < FM; ++i)
+#pragma unroll
+        for (int j = 0; j < FN; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: D = act(acc * scale[col] + bias[col] [+ residual])
+#pragma unroll
+  for (int i = 0; i < FM; ++i) {
+#pragma unroll
+    for (int j = 0; j < FN; ++j) {
+      int col = n0 + wcol * (BN / WN) + j * 16 + (lane & 15);
+      float sc = scale ? scale[col] : 1.f;
+      float bi = bias ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wrow * (BM / WM) + i * 16 + (lane >> 4) * 4 + r;
+        if (row >= M) continue;
+        float v = acc[i][j][r] * sc + bi;
+        if constexpr (RESIDUAL) {
+          v += bf16_to_f32(residual[(size_t)row * N + col]);
+        }
+        if constexpr (RELU) v = v > 0.f ? v : 0.f;
+        C[(size_t)row * N + col] = f32_to_bf16(v);
+      }
+    }
+  }
+}
+
+template <int BM, int BN, int WM, int WN>
+void launch_variant(const GemmArgs& g, hipStream_t s) {
+  int grid = ((g.M + BM - 1) / BM) * (g.N / BN);
+  auto disp = [&](auto relu, auto res) {
+    gemm_bf16_kernel<BM, BN, WM, WN, decltype(relu)::value,
+                     decltype(res)::value><<<grid, 256, 0, s>>>(
+        (const bf16*)g.A, (const bf16*)g.B, (bf16*)g.C, g.M, g.N, g.K,
+        g.scale, g.bias, (const bf16*)g.residual);
+  };
+  if (g.relu && g.residual)
+    disp(std::true_type{}, std::true_type{});
+  else if (g.relu)
+    disp(std::true_type{}, std::false_type{});
+  else if (g.residual)
+    disp(std::false_type{}, std::true_type{});
+  else
+    disp(std::false_type{}, std::false_type{});
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess) {
+    throw ScannerError(std::string("gemm launch failed: ") +
+                       hipGetErrorString(e));
+  }
+}
+
+}  // namespace
+
+void gemm_bf16(const GemmArgs& g, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  SCA_CHECK(g.K % 64 == 0, "gemm K must be a multiple of 64");
+  SCA_CHECK(g.N % 64 == 0, "gemm N must be a multiple of 64");
+  if (g.N % 128 == 0 && g.M > 64) {
+    launch_variant<128, 128, 2, 2>(g, s);
+  } else {
+    launch_variant<64, 64, 2, 2>(g, s);
+  }
+}
+
+}  // namespace sca

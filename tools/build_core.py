@@ -23,6 +23,7 @@ CPP_SOURCES = [
     "csrc/ops/registry.cpp",
     "csrc/ops/stdlib_cpu.cpp",
     "csrc/ops/python_kernel.cpp",
+    "csrc/ops/resnet50.cpp",
     "csrc/engine/table_io.cpp",
     "csrc/engine/executor.cpp",
     "csrc/video/svc_cpu.cpp",
@@ -34,6 +35,7 @@ HIP_SOURCES = [
     "kernels/color.hip",
     "kernels/optflow.hip",
     "kernels/gemm_mfma.hip",
+    "kernels/dnn_ops.hip",
 ]
 
 
