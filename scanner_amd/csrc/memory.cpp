@@ -218,8 +218,19 @@ DeviceAllocators& allocators_for(DeviceHandle device) {
   }
   if (!device.is_gpu()) return g_mem.cpu;
   auto it = g_mem.gpus.find(device.id);
-  SCA_CHECK(it != g_mem.gpus.end(),
-            "memory not initialized for GPU " + std::to_string(device.id));
+  if (it == g_mem.gpus.end()) {
+    // GPU not covered by the explicit init (or lazy default init): give it
+    // system+block allocators (no pool) on demand.
+    std::lock_guard<std::mutex> l(g_mem.mu);
+    it = g_mem.gpus.find(device.id);
+    if (it == g_mem.gpus.end()) {
+      SCA_CHECK(device.id < gpu_device_count(),
+                "no such GPU " + std::to_string(device.id));
+      DeviceAllocators da;
+      setup_device(da, device, 0, false);
+      it = g_mem.gpus.emplace(device.id, std::move(da)).first;
+    }
+  }
   return it->second;
 }
 
