@@ -209,6 +209,9 @@ class BorrowedAllocator : public Allocator {
   Allocator* a_;
 };
 
+void setup_device(DeviceAllocators& da, DeviceHandle dev, size_t pool_size,
+                  bool pinned);
+
 DeviceAllocators& allocators_for(DeviceHandle device) {
   if (!g_mem.initialized) {
     // Lazy default init (CPU only, no pools) so unit tests and simple tools
