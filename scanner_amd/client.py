@@ -218,8 +218,21 @@ class Client:
         jobs_bytes = msgpack.packb(keep)
 
         if self._cluster is not None:
-            return self._cluster.run_job(graph_bytes, jobs_bytes, perf,
-                                         show_progress=show_progress)
+            # ship python ops used by this graph to the master (workers
+            # sync them on NewJob)
+            from .op import _PY_OP_PICKLES
+            for o in msgpack.unpackb(graph_bytes)["ops"]:
+                name = o["name"]
+                if (name in _PY_OP_PICKLES
+                        and name not in self._cluster._py_ops_sent):
+                    self._cluster.register_python_op(
+                        name, {}, _PY_OP_PICKLES[name])
+            result = self._cluster.run_job(graph_bytes, jobs_bytes, perf,
+                                           show_progress=show_progress)
+            # reload metadata: output tables were created/committed by the
+            # master process (shared storage)
+            self._db = _core.Database(self._db_path)
+            return result
 
         # local execution
         if gpu_ids is None:

@@ -436,7 +436,8 @@ PYBIND11_MODULE(_core, m) {
              py::gil_scoped_release rel;
              ex.run();
            })
-      .def("prepare", &LocalExecutor::prepare)
+      .def("prepare", &LocalExecutor::prepare,
+           py::arg("create_outputs") = true)
       .def("all_tasks",
            [](LocalExecutor& ex) {
              py::list out;

@@ -59,7 +59,7 @@ void LocalExecutor::make_instance(i32 idx) {
   instances_.push_back(std::move(inst));
 }
 
-void LocalExecutor::prepare() {
+void LocalExecutor::prepare(bool create_outputs) {
   if (prepared_) return;
   validate_graph(graph_);
   SourceRowsFn source_rows = [this](const SourceArgsC& s) -> i64 {
@@ -83,8 +83,11 @@ void LocalExecutor::prepare() {
                 "Output references unknown column '" + e.column + "'");
       col_types.push_back(psi.output_types[ci]);
     }
-    TableMetadata t = db_->new_table(jobs_[j].sink.table, col_names, col_types,
-                                     /*overwrite=*/true);
+    TableMetadata t =
+        create_outputs
+            ? db_->new_table(jobs_[j].sink.table, col_names, col_types,
+                             /*overwrite=*/true)
+            : db_->get_table(jobs_[j].sink.table);
     out_tables_.push_back(t);
     // task boundaries
     i64 rows = analyses_[j].output_rows;

@@ -49,8 +49,10 @@ class LocalExecutor {
   void run();
 
   // Process one externally-scheduled task (distributed mode). Analysis and
-  // output tables must exist (call prepare() first).
-  void prepare();
+  // output tables must exist (call prepare() first). Workers pass
+  // create_outputs=false: the master already created the output tables and
+  // they attach to the existing descriptors (shared storage).
+  void prepare(bool create_outputs = true);
   void process_task_public(i32 instance, const TaskDesc& t);
   std::vector<TaskDesc> all_tasks() const;
   void finalize_job(i32 job);  // set end_rows + commit output table

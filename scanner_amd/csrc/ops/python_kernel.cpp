@@ -88,6 +88,10 @@ class PythonKernel : public BaseKernel {
     obj_ = factory(args);
   }
   ~PythonKernel() override {
+    if (!Py_IsInitialized()) {
+      obj_.release();  // interpreter gone; leak instead of crashing
+      return;
+    }
     py::gil_scoped_acquire gil;
     obj_ = py::object();
   }
@@ -166,8 +170,18 @@ void register_python_op_binding(
   op_registry().add(o);
 
   size_t n_out = output_columns.size();
-  // Keep the factory alive for process lifetime (registry is global).
-  auto holder = std::make_shared<py::object>(std::move(factory));
+  // Keep the factory alive for process lifetime. The registry is a static;
+  // its destruction runs after Py_Finalize, so the py::object must never
+  // be destroyed — leak it deliberately.
+  std::shared_ptr<py::object> holder(new py::object(std::move(factory)),
+                                     [](py::object* p) {
+                                       if (Py_IsInitialized()) {
+                                         py::gil_scoped_acquire g;
+                                         delete p;
+                                       } else {
+                                         p->release();
+                                       }
+                                     });
   KernelFactory f;
   f.op_name = name;
   f.device_type = (DeviceType)device_type;

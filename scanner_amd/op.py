@@ -136,6 +136,12 @@ def _ann_to_column(ann):
     return False, False, False
 
 
+# name -> cloudpickled re-registration thunk, for shipping python ops to
+# worker processes (parity: RegisterOp/RegisterPythonKernel broadcast,
+# master.cpp:751-813)
+_PY_OP_PICKLES = {}
+
+
 def register_python_op(name=None, device_type=DeviceType.CPU, batch=0,
                        stencil=None, bounded_state=False, warmup=0,
                        unbounded_state=False):
@@ -183,6 +189,19 @@ def register_python_op(name=None, device_type=DeviceType.CPU, batch=0,
             opname, factory, in_cols, out_cols, int(device_type),
             eff_batch, the_stencil, bounded_state, warmup, unbounded_state)
         fn_or_cls._scanner_op_name = opname
+        try:
+            import cloudpickle
+
+            def _remote_register(fn_or_cls=fn_or_cls, kw=dict(
+                    name=opname, device_type=device_type, batch=batch,
+                    stencil=stencil, bounded_state=bounded_state,
+                    warmup=warmup, unbounded_state=unbounded_state)):
+                from scanner_amd.op import register_python_op as rpo
+                rpo(**kw)(fn_or_cls)
+
+            _PY_OP_PICKLES[opname] = cloudpickle.dumps(_remote_register)
+        except Exception:
+            pass  # op still usable locally
         return fn_or_cls
 
     return deco

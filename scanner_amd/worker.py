@@ -1,0 +1,184 @@
+"""Worker server: pulls tasks from the master, runs them on the C++
+engine's pipeline instances, acks completions (parity:
+scanner/engine/worker.cpp — registration, op sync, pull loop with backoff,
+least-loaded dispatch, watchdog self-termination)."""
+import argparse
+import os
+import queue
+import threading
+import time
+
+import msgpack
+
+from . import _core
+from .rpc import RpcClient, RpcError, RpcServer
+
+WATCHDOG_TIMEOUT = 60.0
+
+
+class WorkerServer:
+    def __init__(self, master_addr, db_path, addr="127.0.0.1:0",
+                 pipeline_instances=1, gpu_ids=None, watchdog=True):
+        self._db_path = db_path
+        self._db = _core.Database(db_path)
+        self._master = RpcClient(master_addr)
+        self._instances = pipeline_instances
+        self._gpu_ids = gpu_ids or []
+        self._last_poke = time.time()
+        self._shutdown = threading.Event()
+        self._job_lock = threading.Lock()
+        self._job_id = -1
+        self._executor = None
+        self._registered_py_ops = set()
+
+        self._server = RpcServer(addr, {
+            "Ping": self._ping,
+            "Shutdown": self._shutdown_rpc,
+            "PokeWatchdog": self._poke,
+        })
+        self.port = self._server.port
+        self.addr = f"127.0.0.1:{self.port}"
+        resp = self._master.call("RegisterWorker", {"addr": self.addr})
+        self.worker_id = resp["worker_id"]
+
+        self._task_q = queue.Queue()
+        self._threads = []
+        for i in range(pipeline_instances):
+            t = threading.Thread(target=self._instance_loop, args=(i,),
+                                 daemon=True)
+            t.start()
+            self._threads.append(t)
+        self._pull = threading.Thread(target=self._pull_loop, daemon=True)
+        self._pull.start()
+        if watchdog:
+            self._wd = threading.Thread(target=self._watchdog, daemon=True)
+            self._wd.start()
+
+    # ---- rpc ----
+
+    def _ping(self, req):
+        self._last_poke = time.time()
+        return {"ok": True}
+
+    def _poke(self, req):
+        self._last_poke = time.time()
+        return {"ok": True}
+
+    def _shutdown_rpc(self, req):
+        threading.Thread(target=self.shutdown, daemon=True).start()
+        return {"ok": True}
+
+    # ---- job setup ----
+
+    def _ensure_job(self, job_id):
+        with self._job_lock:
+            if self._job_id == job_id and self._executor is not None:
+                return
+            params = self._master.call("GetJob", {"job_id": job_id},
+                                       timeout=60)
+            # sync python ops from the master (reference: worker op sync
+            # worker.cpp:868-938)
+            for op in params.get("py_ops", []):
+                if op["name"] in self._registered_py_ops:
+                    continue
+                import cloudpickle
+                reg = cloudpickle.loads(op["pickled"])
+                reg()  # re-runs the registration in this process
+                self._registered_py_ops.add(op["name"])
+            perf = dict(params["perf"])
+            perf["pipeline_instances"] = self._instances
+            # reload metadata (ingests + master-created output tables)
+            self._db = _core.Database(self._db_path)
+            ex = _core.LocalExecutor(
+                self._db, params["graph"], msgpack.packb(params["jobs"]),
+                perf, self._gpu_ids)
+            ex.prepare(False)
+            self._executor = ex
+            self._job_id = job_id
+
+    # ---- loops ----
+
+    def _pull_loop(self):
+        backoff = 0.05
+        while not self._shutdown.is_set():
+            if self._task_q.qsize() >= self._instances:
+                time.sleep(0.01)
+                continue
+            try:
+                resp = self._master.call("NextWork", {
+                    "worker_id": self.worker_id,
+                    "max_tasks": self._instances}, timeout=30)
+            except RpcError:
+                time.sleep(min(backoff, 2.0))
+                backoff *= 1.5
+                continue
+            backoff = 0.05
+            if resp["job_id"] < 0 or not resp["tasks"]:
+                time.sleep(resp.get("wait", 0.2) or 0.2)
+                continue
+            for t in resp["tasks"]:
+                self._task_q.put((resp["job_id"], t))
+
+    def _instance_loop(self, idx):
+        while not self._shutdown.is_set():
+            try:
+                job_id, t = self._task_q.get(timeout=0.2)
+            except queue.Empty:
+                continue
+            stream, task, start, end = t
+            ok, err = True, None
+            try:
+                self._ensure_job(job_id)
+                self._executor.process_task(idx, stream, task, start, end)
+            except Exception as e:
+                ok, err = False, f"{type(e).__name__}: {e}"
+            try:
+                self._master.call("FinishedWork", {
+                    "worker_id": self.worker_id, "job_id": job_id,
+                    "stream": stream, "task": task, "success": ok,
+                    "error": err}, timeout=30)
+            except RpcError:
+                pass
+
+    def _watchdog(self):
+        # self-terminate if the master goes silent (reference:
+        # worker.cpp:647-673)
+        while not self._shutdown.wait(5.0):
+            if time.time() - self._last_poke > WATCHDOG_TIMEOUT:
+                os._exit(3)
+
+    def shutdown(self):
+        self._shutdown.set()
+        self._master.try_call("UnregisterWorker",
+                              {"worker_id": self.worker_id}, timeout=5)
+        self._server.stop()
+
+    def wait(self):
+        self._server.wait()
+
+
+def start_worker(master_addr, db_path, addr="127.0.0.1:0", block=False,
+                 **kw):
+    w = WorkerServer(master_addr, db_path, addr, **kw)
+    if block:
+        w.wait()
+    return w
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--master", required=True)
+    ap.add_argument("--db-path", required=True)
+    ap.add_argument("--addr", default="127.0.0.1:0")
+    ap.add_argument("--instances", type=int, default=1)
+    ap.add_argument("--gpu-ids", default="")
+    ap.add_argument("--no-watchdog", action="store_true")
+    args = ap.parse_args()
+    gpu_ids = [int(x) for x in args.gpu_ids.split(",") if x != ""]
+    start_worker(args.master, args.db_path, args.addr, block=True,
+                 pipeline_instances=args.instances, gpu_ids=gpu_ids,
+                 watchdog=not args.no_watchdog)
+
+
+if __name__ == "__main__":
+    main()
