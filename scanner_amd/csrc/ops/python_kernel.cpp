@@ -188,6 +188,9 @@ void register_python_op_binding(
   f.preferred_batch = batch > 0 ? batch : 1;
   f.make = [holder, n_out](const KernelConfig& cfg)
       -> std::unique_ptr<BaseKernel> {
+    // executor threads call this without the GIL; copying the factory
+    // py::object inc_refs it
+    py::gil_scoped_acquire gil;
     return std::make_unique<PythonKernel>(cfg, *holder, n_out);
   };
   kernel_registry().add(f);

@@ -119,21 +119,18 @@ _FRAME_ANN = ("FrameType", FrameType)
 
 
 def _ann_to_column(ann):
-    """Returns (is_frame, batched, stencil) for an annotation."""
+    """Returns (is_frame, seq_depth) for an annotation. Depth 1 means a
+    batch (or, when the op declares stencil=, the stencil window); depth 2
+    means stencil-within-batch (parity: reference op.py:389-535)."""
+    import collections.abc
     origin = typing.get_origin(ann)
-    if origin in (list, typing.Sequence) or str(origin).endswith("Sequence"):
+    if origin in (list, collections.abc.Sequence):
         inner = typing.get_args(ann)[0]
-        f, b, s = _ann_to_column(inner)
-        if not b:
-            return f, True, s
-        return f, True, True  # nested sequence => stencil within batch
-    if ann is bytes:
-        return False, False, False
+        f, d = _ann_to_column(inner)
+        return f, d + 1
     if ann is FrameType or ann == "FrameType":
-        return True, False, False
-    if ann is typing.Any:
-        return False, False, False
-    return False, False, False
+        return True, 0
+    return False, 0
 
 
 # name -> cloudpickled re-registration thunk, for shipping python ops to
@@ -158,24 +155,27 @@ def register_python_op(name=None, device_type=DeviceType.CPU, batch=0,
         params = [p for p in sig.parameters.values()
                   if p.name not in ("self", "config")]
         in_cols = []
-        batched = False
         stenciled = stencil is not None
+        max_depth = 0
         for p in params:
             if p.annotation is inspect.Parameter.empty:
                 raise ScannerException(
                     f"python op {opname}: parameter {p.name} needs a type "
                     "annotation")
-            f, b, s = _ann_to_column(p.annotation)
-            batched = batched or b
+            f, d = _ann_to_column(p.annotation)
+            max_depth = max(max_depth, d)
             in_cols.append((p.name, 1 if f else 0))
+        # depth 1 = batch, unless a stencil is declared (then it is the
+        # stencil window); depth 2 = stencil within batch
+        batched = max_depth >= 2 or (max_depth == 1 and not stenciled)
         ret = sig.return_annotation
         out_cols = []
         if typing.get_origin(ret) is tuple:
             for i, r in enumerate(typing.get_args(ret)):
-                f, _, _ = _ann_to_column(r)
+                f, _ = _ann_to_column(r)
                 out_cols.append((f"out{i}", 1 if f else 0))
         else:
-            f, _, _ = _ann_to_column(ret)
+            f, _ = _ann_to_column(ret)
             out_cols.append(("out", 1 if f else 0))
 
         the_stencil = list(stencil) if stencil else [0]

@@ -1,0 +1,118 @@
+"""Python op tests (parity: reference py_test.py:558-729 — plain, batch,
+stencil, stencil+batch, per-stream args, stateful python kernels)."""
+import numpy as np
+import pytest
+from typing import Any, Sequence
+
+import scanner_amd as sp
+from scanner_amd import FrameType, register_python_op
+from conftest import make_video
+
+
+@register_python_op()
+def DoubleIt(col: bytes) -> bytes:
+    v = int.from_bytes(col, "little")
+    return (2 * v).to_bytes(8, "little")
+
+
+@register_python_op(name="FrameMean")
+def frame_mean(frame: FrameType) -> bytes:
+    return float(np.mean(frame)).hex().encode()
+
+
+@register_python_op(batch=4)
+def BatchedDouble(cols: Sequence[bytes]) -> Sequence[bytes]:
+    return [(2 * int.from_bytes(c, "little")).to_bytes(8, "little")
+            for c in cols]
+
+
+@register_python_op(stencil=[-1, 0, 1])
+def StencilSum(cols: Sequence[bytes]) -> bytes:
+    s = sum(int.from_bytes(c, "little") for c in cols)
+    return s.to_bytes(8, "little")
+
+
+@register_python_op()
+class StatefulCounter(sp.Kernel):
+    def __init__(self, config, **kwargs):
+        self.count = 0
+        self.base = 0
+
+    def new_stream(self, base=0):
+        self.base = base
+
+    def reset(self):
+        self.count = 0
+
+    def execute(self, col: bytes) -> bytes:
+        self.count += 1
+        return (self.base + self.count).to_bytes(8, "little")
+
+
+def int_table(sc, name, n):
+    return sc.new_table(name, ["col"],
+                        [[int(i).to_bytes(8, "little")] for i in range(n)])
+
+
+def test_python_op_plain(sc):
+    tab = int_table(sc, "p1", 10)
+    col = sc.io.Input([tab])
+    doubled = sc.ops.DoubleIt(col=col)
+    out = sp.NamedStream(sc, "p1_out")
+    sc.run(sc.io.Output(doubled, [out]), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite)
+    vals = [int.from_bytes(b, "little") for b in out.load()]
+    assert vals == [2 * i for i in range(10)]
+
+
+def test_python_op_frame(sc):
+    frames = make_video(n=5)
+    video = sp.NamedVideoStream(sc, "p2", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    means = sc.ops.FrameMean(frame=frame)
+    out = sp.NamedStream(sc, "p2_out")
+    sc.run(sc.io.Output(means, [out]), sp.PerfParams.manual(2, 4),
+           cache_mode=sp.CacheMode.Overwrite)
+    got = [float.fromhex(b.decode()) for b in out.load()]
+    for i in range(5):
+        assert abs(got[i] - frames[i].mean()) < 1e-6
+
+
+def test_python_op_batched(sc):
+    tab = int_table(sc, "p3", 11)
+    col = sc.io.Input([tab])
+    doubled = sc.ops.BatchedDouble(cols=col)
+    out = sp.NamedStream(sc, "p3_out")
+    sc.run(sc.io.Output(doubled, [out]), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite)
+    vals = [int.from_bytes(b, "little") for b in out.load()]
+    assert vals == [2 * i for i in range(11)]
+
+
+def test_python_op_stencil(sc):
+    n = 9
+    tab = int_table(sc, "p4", n)
+    col = sc.io.Input([tab])
+    summed = sc.ops.StencilSum(cols=col)
+    out = sp.NamedStream(sc, "p4_out")
+    sc.run(sc.io.Output(summed, [out]), sp.PerfParams.manual(2, 3),
+           cache_mode=sp.CacheMode.Overwrite)
+    vals = [int.from_bytes(b, "little") for b in out.load()]
+    # REPEAT_EDGE clamping at both ends
+    expect = [min(max(i - 1, 0), n - 1) + i + min(i + 1, n - 1)
+              for i in range(n)]
+    assert vals == expect
+
+
+def test_python_op_stateful_with_stream_args(sc):
+    n = 6
+    tab = int_table(sc, "p5", n)
+    col = sc.io.Input([tab])
+    counted = sc.ops.StatefulCounter(col=col,
+                                     stream_args=[{"base": 100}])
+    out = sp.NamedStream(sc, "p5_out")
+    # one task: state runs through the whole stream
+    sc.run(sc.io.Output(counted, [out]), sp.PerfParams.manual(8, 16),
+           cache_mode=sp.CacheMode.Overwrite)
+    vals = [int.from_bytes(b, "little") for b in out.load()]
+    assert vals == [100 + i + 1 for i in range(n)]
