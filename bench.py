@@ -53,6 +53,11 @@ def build_pipeline(sc, sp, video, pipeline, device, out_name):
         cols.append(sc.ops.Histogram(frame=frame, device=device))
     if pipeline in ("resnet", "full"):
         cols.append(sc.ops.ResNet50(frame=frame, device=device))
+    if pipeline == "flow":
+        # BASELINE config 4: dense optical flow; per-frame flow summary is
+        # the saved column (the 16 MB/frame flow field stays on-GPU).
+        flow = sc.ops.OpticalFlow(frame=frame, device=device)
+        cols.append(sc.ops.FlowStats(flow=flow, device=device))
     out = sp.NamedStream(sc, out_name)
     return sc.io.Output(cols, [out])
 
@@ -63,7 +68,7 @@ def main():
     ap.add_argument("--steps", type=int, default=5)
     ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--pipeline", default="full",
-                    choices=["hist", "resnet", "full"])
+                    choices=["hist", "resnet", "full", "flow", "pose"])
     ap.add_argument("--frames-per-step", type=int, default=FRAMES_PER_STEP)
     args = ap.parse_args()
 
@@ -158,7 +163,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if args.pipeline in ("resnet", "full") else "u8",
+            "dtype": {"resnet": "bf16", "full": "bf16", "pose": "bf16",
+                      "flow": "f32", "hist": "u8"}[args.pipeline],
             "data": "synthetic 1080p video (svc-encoded), random-init weights",
             "config": {
                 "model": "1080p histogram + ResNet-50 classify"

@@ -88,6 +88,7 @@ PYBIND11_MODULE(_core, m) {
   register_stdlib_ops();
   register_gpu_ops();
   register_resnet50_op();
+  register_optflow_gpu();
 
   m.def("have_gpu", &have_gpu);
   m.def("gpu_device_count", &gpu_device_count);

@@ -154,5 +154,6 @@ struct KernelRegistrar {
 // avoid static-initializer ordering issues by explicit registration).
 void register_stdlib_ops();
 void register_resnet50_op();
+void register_optflow_gpu();
 
 }  // namespace sca

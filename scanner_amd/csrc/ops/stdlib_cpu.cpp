@@ -4,12 +4,16 @@
 // OpticalFlow, Resize, TestIncrement bounded/unbounded, Blur, Sleep);
 // here they are first-party, with GPU (HIP) counterparts in
 // kernels/image_ops.hip registered by ops/stdlib_gpu.cpp.
+#include <algorithm>
 #include <cmath>
+#include <cstring>
 #include <thread>
+#include <vector>
 
 #include "../memory.h"
 #include "../msgpack.h"
 #include "kernel.h"
+#include "optflow_common.h"
 
 namespace sca {
 
@@ -186,57 +190,180 @@ class SleepKernel : public Kernel {
   i64 ms_;
 };
 
-// ---- OpticalFlowCPU: coarse diamond-search block matching, stencil [0,1].
-// (Parity role of the reference's Farneback test op; the production dense
-// flow lives in kernels/optflow.hip.) Output: H/16 x W/16 x 2 f32 motion
-// vectors per 16x16 block.
+// ---- OpticalFlowCPU: dense pyramidal Lucas-Kanade, stencil [0,1].
+// Same algorithm and constants (optflow_common.h) as the CDNA4 kernel in
+// kernels/optflow.hip, so GPU numerics tests compare against this directly.
+// Parity: the reference's OpticalFlow op wraps OpenCV Farneback
+// (tests/test_ops.cpp:63-113) with the same dense H x W x 2 f32 contract.
 class OpticalFlowKernelCPU : public StenciledKernel {
  public:
-  using StenciledKernel::StenciledKernel;
+  explicit OpticalFlowKernelCPU(const KernelConfig& cfg)
+      : StenciledKernel(cfg) {
+    auto a = mp::decode(cfg.args);
+    radius_ = (int)a.get_int("radius", optflow::kDefaultRadius);
+    iters_ = (int)a.get_int("iters", optflow::kDefaultIters);
+    max_levels_ = (int)a.get_int("levels", optflow::kDefaultMaxLevels);
+  }
   void execute_stencil(const BatchedElements& in, ElementVector& out) override {
     const Element& f0 = in[0][0];
     const Element& f1 = in[0][1];
     i32 h = f0.frame_info.shape[0], w = f0.frame_info.shape[1],
         c = f0.frame_info.shape[2];
-    i32 bh = h / 16, bw = w / 16;
-    Element e = alloc_frame(config_.device, bh, bw, 2, FrameType::F32);
-    f32* flow = reinterpret_cast<f32*>(e.buffer);
-    auto lum = [&](const u8* p, i32 y, i32 x) -> i32 {
-      i64 off = ((i64)y * w + x) * c;
-      i32 s = 0;
-      for (i32 ch = 0; ch < c; ++ch) s += p[off + ch];
-      return s;
-    };
-    for (i32 by = 0; by < bh; ++by) {
-      for (i32 bx = 0; bx < bw; ++bx) {
-        i32 y0 = by * 16, x0 = bx * 16;
-        i32 best_dy = 0, best_dx = 0;
-        i64 best = INT64_MAX;
-        for (i32 dy = -8; dy <= 8; dy += 4) {
-          for (i32 dx = -8; dx <= 8; dx += 4) {
-            i64 sad = 0;
-            for (i32 y = 0; y < 16; y += 4) {
-              for (i32 x = 0; x < 16; x += 4) {
-                i32 sy = y0 + y + dy, sx = x0 + x + dx;
-                if (sy < 0 || sy >= h || sx < 0 || sx >= w) {
-                  sad += 255;
-                  continue;
-                }
-                sad += std::abs(lum(f1.buffer, sy, sx) -
-                                lum(f0.buffer, y0 + y, x0 + x));
-              }
-            }
-            if (sad < best) {
-              best = sad;
-              best_dy = dy;
-              best_dx = dx;
-            }
-          }
-        }
-        flow[((i64)by * bw + bx) * 2 + 0] = (f32)best_dx;
-        flow[((i64)by * bw + bx) * 2 + 1] = (f32)best_dy;
+    int levels = optflow::num_levels(h, w, max_levels_);
+    // Gray pyramids.
+    std::vector<std::vector<f32>> p0(levels), p1(levels);
+    std::vector<int> lh(levels), lw(levels);
+    lh[0] = h;
+    lw[0] = w;
+    to_gray(f0.buffer, h, w, c, p0[0]);
+    to_gray(f1.buffer, h, w, c, p1[0]);
+    for (int l = 1; l < levels; ++l) {
+      lh[l] = lh[l - 1] / 2;
+      lw[l] = lw[l - 1] / 2;
+      down2x(p0[l - 1], lh[l - 1], lw[l - 1], p0[l], lh[l], lw[l]);
+      down2x(p1[l - 1], lh[l - 1], lw[l - 1], p1[l], lh[l], lw[l]);
+    }
+    std::vector<f32> flow(2 * (size_t)lh[levels - 1] * lw[levels - 1], 0.f);
+    std::vector<f32> next;
+    for (int l = levels - 1; l >= 0; --l) {
+      for (int it = 0; it < iters_; ++it) {
+        lk_iter(p0[l], p1[l], lh[l], lw[l], flow, next);
+        std::swap(flow, next);
+      }
+      if (l > 0) {
+        upsample(flow, lh[l], lw[l], next, lh[l - 1], lw[l - 1]);
+        std::swap(flow, next);
       }
     }
+    Element e = alloc_frame(config_.device, h, w, 2, FrameType::F32);
+    std::memcpy(e.buffer, flow.data(), flow.size() * sizeof(f32));
+    out[0] = e;
+  }
+
+ private:
+  static void to_gray(const u8* p, int h, int w, int c, std::vector<f32>& g) {
+    g.resize((size_t)h * w);
+    for (i64 i = 0; i < (i64)h * w; ++i) {
+      const u8* px = p + i * c;
+      g[i] = c == 3 ? 0.299f * px[0] + 0.587f * px[1] + 0.114f * px[2]
+                    : (f32)px[0];
+    }
+  }
+  static void down2x(const std::vector<f32>& s, int sh, int sw,
+                     std::vector<f32>& d, int dh, int dw) {
+    d.resize((size_t)dh * dw);
+    for (int y = 0; y < dh; ++y)
+      for (int x = 0; x < dw; ++x) {
+        int y0 = 2 * y, x0 = 2 * x;
+        int y1 = std::min(y0 + 1, sh - 1), x1 = std::min(x0 + 1, sw - 1);
+        d[(size_t)y * dw + x] =
+            0.25f * (s[(size_t)y0 * sw + x0] + s[(size_t)y0 * sw + x1] +
+                     s[(size_t)y1 * sw + x0] + s[(size_t)y1 * sw + x1]);
+      }
+  }
+  static f32 bilerp(const std::vector<f32>& im, int h, int w, f32 fx, f32 fy) {
+    int x0 = (int)std::floor(fx), y0 = (int)std::floor(fy);
+    f32 ax = fx - x0, ay = fy - y0;
+    int x0c = std::min(std::max(x0, 0), w - 1);
+    int x1c = std::min(std::max(x0 + 1, 0), w - 1);
+    int y0c = std::min(std::max(y0, 0), h - 1);
+    int y1c = std::min(std::max(y0 + 1, 0), h - 1);
+    f32 v00 = im[(size_t)y0c * w + x0c], v01 = im[(size_t)y0c * w + x1c];
+    f32 v10 = im[(size_t)y1c * w + x0c], v11 = im[(size_t)y1c * w + x1c];
+    return v00 * (1 - ay) * (1 - ax) + v01 * (1 - ay) * ax +
+           v10 * ay * (1 - ax) + v11 * ay * ax;
+  }
+  void lk_iter(const std::vector<f32>& I0, const std::vector<f32>& I1, int h,
+               int w, const std::vector<f32>& fin, std::vector<f32>& fout) {
+    fout.resize((size_t)h * w * 2);
+    int R = radius_;
+    auto at = [&](const std::vector<f32>& im, int y, int x) {
+      return im[(size_t)std::min(std::max(y, 0), h - 1) * w +
+                std::min(std::max(x, 0), w - 1)];
+    };
+    for (int y = 0; y < h; ++y) {
+      for (int x = 0; x < w; ++x) {
+        f32 u = fin[((size_t)y * w + x) * 2 + 0];
+        f32 v = fin[((size_t)y * w + x) * 2 + 1];
+        f32 a11 = 0, a12 = 0, a22 = 0, b1 = 0, b2 = 0;
+        for (int dy = -R; dy <= R; ++dy)
+          for (int dx = -R; dx <= R; ++dx) {
+            int yy = y + dy, xx = x + dx;
+            f32 ix = 0.5f * (at(I0, yy, xx + 1) - at(I0, yy, xx - 1));
+            f32 iy = 0.5f * (at(I0, yy + 1, xx) - at(I0, yy - 1, xx));
+            f32 it = bilerp(I1, h, w, xx + u, yy + v) - at(I0, yy, xx);
+            a11 += ix * ix;
+            a12 += ix * iy;
+            a22 += iy * iy;
+            b1 += ix * it;
+            b2 += iy * it;
+          }
+        f32 det = a11 * a22 - a12 * a12;
+        if (det > optflow::kDetEps) {
+          u -= (a22 * b1 - a12 * b2) / det;
+          v -= (a11 * b2 - a12 * b1) / det;
+        }
+        fout[((size_t)y * w + x) * 2 + 0] = u;
+        fout[((size_t)y * w + x) * 2 + 1] = v;
+      }
+    }
+  }
+  void upsample(const std::vector<f32>& s, int sh, int sw,
+                std::vector<f32>& d, int dh, int dw) {
+    d.resize((size_t)dh * dw * 2);
+    f32 fx_scale = (f32)dw / sw, fy_scale = (f32)dh / sh;
+    for (int y = 0; y < dh; ++y)
+      for (int x = 0; x < dw; ++x) {
+        f32 sxf = (x + 0.5f) / fx_scale - 0.5f;
+        f32 syf = (y + 0.5f) / fy_scale - 0.5f;
+        int x0 = (int)std::floor(sxf), y0 = (int)std::floor(syf);
+        f32 ax = sxf - x0, ay = syf - y0;
+        int x0c = std::min(std::max(x0, 0), sw - 1);
+        int x1c = std::min(std::max(x0 + 1, 0), sw - 1);
+        int y0c = std::min(std::max(y0, 0), sh - 1);
+        int y1c = std::min(std::max(y0 + 1, 0), sh - 1);
+        for (int ch = 0; ch < 2; ++ch) {
+          f32 v00 = s[((size_t)y0c * sw + x0c) * 2 + ch];
+          f32 v01 = s[((size_t)y0c * sw + x1c) * 2 + ch];
+          f32 v10 = s[((size_t)y1c * sw + x0c) * 2 + ch];
+          f32 v11 = s[((size_t)y1c * sw + x1c) * 2 + ch];
+          f32 vv = v00 * (1 - ay) * (1 - ax) + v01 * (1 - ay) * ax +
+                   v10 * ay * (1 - ax) + v11 * ay * ax;
+          d[((size_t)y * dw + x) * 2 + ch] =
+              vv * (ch == 0 ? fx_scale : fy_scale);
+        }
+      }
+  }
+
+  int radius_, iters_, max_levels_;
+};
+
+// ---- FlowStats: dense flow frame -> {mean|u|, mean|v|, max|u|, max|v|} ----
+class FlowStatsKernelCPU : public Kernel {
+ public:
+  using Kernel::Kernel;
+  void execute_row(const ElementVector& in, ElementVector& out) override {
+    const Element& f = in[0];
+    SCA_CHECK(f.is_frame && f.frame_info.type == FrameType::F32 &&
+                  f.frame_info.shape[2] == 2,
+              "FlowStats expects H x W x 2 f32 flow frames");
+    i64 npix = (i64)f.frame_info.shape[0] * f.frame_info.shape[1];
+    const f32* p = reinterpret_cast<const f32*>(f.buffer);
+    // match the GPU reduction: sums accumulate in f32
+    f32 su = 0, sv = 0, mu = 0, mv = 0;
+    for (i64 i = 0; i < npix; ++i) {
+      f32 u = std::fabs(p[i * 2]), v = std::fabs(p[i * 2 + 1]);
+      su += u;
+      sv += v;
+      mu = std::max(mu, u);
+      mv = std::max(mv, v);
+    }
+    Element e = alloc_bytes(config_.device, 4 * sizeof(f32));
+    f32* o = reinterpret_cast<f32*>(e.buffer);
+    o[0] = su / npix;
+    o[1] = sv / npix;
+    o[2] = mu;
+    o[3] = mv;
     out[0] = e;
   }
 };
@@ -338,6 +465,20 @@ void register_stdlib_ops() {
     f.device_type = DeviceType::CPU;
     f.make = [](const KernelConfig& c) -> std::unique_ptr<BaseKernel> {
       return std::make_unique<SleepKernel>(c);
+    };
+    kernel_registry().add(f);
+  }
+  {
+    OpInfo o;
+    o.name = "FlowStats";
+    o.input_columns = {{"flow", ColumnType::Video}};
+    o.output_columns = {{"stats", ColumnType::Bytes}};
+    op_registry().add(o);
+    KernelFactory f;
+    f.op_name = "FlowStats";
+    f.device_type = DeviceType::CPU;
+    f.make = [](const KernelConfig& c) -> std::unique_ptr<BaseKernel> {
+      return std::make_unique<FlowStatsKernelCPU>(c);
     };
     kernel_registry().add(f);
   }

@@ -1,0 +1,507 @@
+// Dense optical flow for gfx950 (MI355X): pyramidal Lucas-Kanade.
+//
+// Capability parity: the reference's OpticalFlow op (tests/test_ops.cpp:
+// 63-113) wraps OpenCV's Farneback — stencil [0,1], dense H x W x 2 f32
+// flow. This is a from-scratch CDNA4 implementation of the same contract:
+//   * all frames of a work packet are processed in ONE set of batched
+//     kernel launches per pyramid level (blocks carry a frame/pair index),
+//     so coarse levels still put >> 256 workgroups on the 8 XCDs;
+//   * consecutive stencil windows share frames — gray pyramids are built
+//     once per unique frame, not once per pair;
+//   * the LK iteration stages the reference image tile (incl. gradient +
+//     window halo) in LDS; warped-image taps go through L2 (they are
+//     flow-dependent gather reads, uncacheable in LDS by construction).
+// All launches go on the pipeline instance's per-thread HIP stream; the
+// kernel never synchronizes the device.
+#include <hip/hip_runtime.h>
+
+#include <map>
+#include <vector>
+
+#include "../csrc/memory.h"
+#include "../csrc/msgpack.h"
+#include "../csrc/ops/kernel.h"
+#include "../csrc/ops/optflow_common.h"
+
+namespace sca {
+
+namespace {
+
+namespace of = ::sca::optflow;
+
+#define OF_CHECK(expr)                                                    \
+  do {                                                                    \
+    hipError_t _e = (expr);                                               \
+    if (_e != hipSuccess) {                                               \
+      throw ScannerError(std::string("HIP error in OpticalFlow: ") +      \
+                         hipGetErrorString(_e));                          \
+    }                                                                     \
+  } while (0)
+
+inline hipStream_t cur_stream() {
+  return (hipStream_t)per_thread_hip_stream();
+}
+
+// u8 HWC (c=1 or 3) -> f32 luma planes, one plane per unique frame.
+// frames: device array of device pointers. out: [frame][h*w].
+__global__ void __launch_bounds__(256)
+    of_gray_kernel(const u8* const* __restrict__ frames, int nframes, int h,
+                   int w, int c, float* __restrict__ out) {
+  i64 total = (i64)nframes * h * w;
+  i64 stride = (i64)gridDim.x * blockDim.x;
+  for (i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    int f = (int)(i / ((i64)h * w));
+    i64 pix = i % ((i64)h * w);
+    const u8* p = frames[f] + pix * c;
+    float v;
+    if (c == 3)
+      v = 0.299f * p[0] + 0.587f * p[1] + 0.114f * p[2];
+    else
+      v = p[0];
+    out[i] = v;
+  }
+}
+
+// 2x2 box downsample of nframes planes. src stride sh*sw, dst stride dh*dw.
+__global__ void __launch_bounds__(256)
+    of_down2x_kernel(const float* __restrict__ src, int sh, int sw,
+                     float* __restrict__ dst, int dh, int dw, int nframes) {
+  i64 total = (i64)nframes * dh * dw;
+  i64 stride = (i64)gridDim.x * blockDim.x;
+  for (i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    int f = (int)(i / ((i64)dh * dw));
+    i64 pix = i % ((i64)dh * dw);
+    int y = (int)(pix / dw), x = (int)(pix % dw);
+    const float* s = src + (i64)f * sh * sw;
+    int y0 = 2 * y, x0 = 2 * x;
+    int y1 = min(y0 + 1, sh - 1), x1 = min(x0 + 1, sw - 1);
+    dst[i] = 0.25f * (s[(i64)y0 * sw + x0] + s[(i64)y0 * sw + x1] +
+                      s[(i64)y1 * sw + x0] + s[(i64)y1 * sw + x1]);
+  }
+}
+
+// Bilinear-upsample flow from (sh,sw) to (dh,dw) for npairs pairs, scaling
+// vector components by the dimension ratio. Layout: [pair][h*w*2].
+__global__ void __launch_bounds__(256)
+    of_upsample_kernel(const float* __restrict__ src, int sh, int sw,
+                       float* __restrict__ dst, int dh, int dw, int npairs) {
+  i64 total = (i64)npairs * dh * dw;
+  i64 stride = (i64)gridDim.x * blockDim.x;
+  float fx_scale = (float)dw / sw, fy_scale = (float)dh / sh;
+  for (i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    int p = (int)(i / ((i64)dh * dw));
+    i64 pix = i % ((i64)dh * dw);
+    int y = (int)(pix / dw), x = (int)(pix % dw);
+    float sxf = (x + 0.5f) / fx_scale - 0.5f;
+    float syf = (y + 0.5f) / fy_scale - 0.5f;
+    int x0 = (int)floorf(sxf), y0 = (int)floorf(syf);
+    float ax = sxf - x0, ay = syf - y0;
+    int x0c = min(max(x0, 0), sw - 1), x1c = min(max(x0 + 1, 0), sw - 1);
+    int y0c = min(max(y0, 0), sh - 1), y1c = min(max(y0 + 1, 0), sh - 1);
+    const float* s = src + (i64)p * sh * sw * 2;
+    float* d = dst + ((i64)p * dh * dw + (i64)y * dw + x) * 2;
+#pragma unroll
+    for (int ch = 0; ch < 2; ++ch) {
+      float v00 = s[((i64)y0c * sw + x0c) * 2 + ch];
+      float v01 = s[((i64)y0c * sw + x1c) * 2 + ch];
+      float v10 = s[((i64)y1c * sw + x0c) * 2 + ch];
+      float v11 = s[((i64)y1c * sw + x1c) * 2 + ch];
+      float v = v00 * (1 - ay) * (1 - ax) + v01 * (1 - ay) * ax +
+                v10 * ay * (1 - ax) + v11 * ay * ax;
+      d[ch] = v * (ch == 0 ? fx_scale : fy_scale);
+    }
+  }
+}
+
+// One LK iteration at one pyramid level for all pairs. Block = 16x16 pixel
+// tile of one pair's plane (blockIdx.z = pair); I0 tile + halo staged in
+// LDS. RADIUS is a compile-time template so the window loops fully unroll.
+template <int RADIUS>
+__global__ void __launch_bounds__(256)
+    of_lk_kernel(const float* __restrict__ gray, i64 level_off, int h, int w,
+                 const int* __restrict__ pair_f0,
+                 const int* __restrict__ pair_f1,
+                 const float* __restrict__ flow_in,
+                 float* __restrict__ flow_out) {
+  constexpr int TILE = 16;
+  constexpr int HALO = RADIUS + 1;           // gradient needs +-1 past window
+  constexpr int LW = TILE + 2 * HALO;        // staged tile width
+  __shared__ float lds_i0[LW * LW];
+
+  int pair = blockIdx.z;
+  const float* I0 = gray + level_off + (i64)pair_f0[pair] * h * w;
+  const float* I1 = gray + level_off + (i64)pair_f1[pair] * h * w;
+
+  int tx0 = blockIdx.x * TILE, ty0 = blockIdx.y * TILE;
+  // Cooperative LDS fill with clamped loads.
+  for (int i = threadIdx.y * TILE + threadIdx.x; i < LW * LW;
+       i += TILE * TILE) {
+    int ly = i / LW, lx = i % LW;
+    int gy = min(max(ty0 + ly - HALO, 0), h - 1);
+    int gx = min(max(tx0 + lx - HALO, 0), w - 1);
+    lds_i0[i] = I0[(i64)gy * w + gx];
+  }
+  __syncthreads();
+
+  int x = tx0 + threadIdx.x, y = ty0 + threadIdx.y;
+  if (x >= w || y >= h) return;
+
+  const float* fin = flow_in + (i64)pair * h * w * 2;
+  float* fout = flow_out + (i64)pair * h * w * 2;
+  float u = fin[((i64)y * w + x) * 2 + 0];
+  float v = fin[((i64)y * w + x) * 2 + 1];
+
+  float a11 = 0, a12 = 0, a22 = 0, b1 = 0, b2 = 0;
+#pragma unroll
+  for (int dy = -RADIUS; dy <= RADIUS; ++dy) {
+#pragma unroll
+    for (int dx = -RADIUS; dx <= RADIUS; ++dx) {
+      int lx = (int)threadIdx.x + HALO + dx;
+      int ly = (int)threadIdx.y + HALO + dy;
+      float ix = 0.5f * (lds_i0[ly * LW + lx + 1] - lds_i0[ly * LW + lx - 1]);
+      float iy = 0.5f * (lds_i0[(ly + 1) * LW + lx] - lds_i0[(ly - 1) * LW + lx]);
+      float i0v = lds_i0[ly * LW + lx];
+      // Bilinear tap of I1 at the flow-warped position (clamped).
+      float wx = x + dx + u, wy = y + dy + v;
+      int wx0 = (int)floorf(wx), wy0 = (int)floorf(wy);
+      float ax = wx - wx0, ay = wy - wy0;
+      int x0c = min(max(wx0, 0), w - 1), x1c = min(max(wx0 + 1, 0), w - 1);
+      int y0c = min(max(wy0, 0), h - 1), y1c = min(max(wy0 + 1, 0), h - 1);
+      float w00 = I1[(i64)y0c * w + x0c], w01 = I1[(i64)y0c * w + x1c];
+      float w10 = I1[(i64)y1c * w + x0c], w11 = I1[(i64)y1c * w + x1c];
+      float i1v = w00 * (1 - ay) * (1 - ax) + w01 * (1 - ay) * ax +
+                  w10 * ay * (1 - ax) + w11 * ay * ax;
+      float it = i1v - i0v;
+      a11 += ix * ix;
+      a12 += ix * iy;
+      a22 += iy * iy;
+      b1 += ix * it;
+      b2 += iy * it;
+    }
+  }
+  float det = a11 * a22 - a12 * a12;
+  if (det > of::kDetEps) {
+    u -= (a22 * b1 - a12 * b2) / det;
+    v -= (a11 * b2 - a12 * b1) / det;
+  }
+  fout[((i64)y * w + x) * 2 + 0] = u;
+  fout[((i64)y * w + x) * 2 + 1] = v;
+}
+
+inline int grid_1d(i64 total, int block = 256, int cap = 4096) {
+  return (int)std::min<i64>(cap, (total + block - 1) / block);
+}
+
+// ---------------- FlowStats: dense flow -> {mean|u|,mean|v|,max|u|,max|v|}
+// Two-stage reduction (partials per chunk, then per-frame merge) — one
+// workgroup can't fill an XCD, and f32 global atomics would serialize.
+__global__ void __launch_bounds__(256)
+    flowstats_partial_kernel(const float* const* __restrict__ flows, i64 npix,
+                             float* __restrict__ partials, int chunks) {
+  __shared__ float lsum[2][64], lmax[2][64];
+  int frame = blockIdx.y, chunk = blockIdx.x;
+  const float* f = flows[frame];
+  float su = 0, sv = 0, mu = 0, mv = 0;
+  for (i64 i = (i64)chunk * 256 + threadIdx.x; i < npix;
+       i += (i64)chunks * 256) {
+    float u = fabsf(f[i * 2]), v = fabsf(f[i * 2 + 1]);
+    su += u;
+    sv += v;
+    mu = fmaxf(mu, u);
+    mv = fmaxf(mv, v);
+  }
+  // wave64 shuffle reduce, then LDS across the 4 waves
+  for (int off = 32; off; off >>= 1) {
+    su += __shfl_down(su, off, 64);
+    sv += __shfl_down(sv, off, 64);
+    mu = fmaxf(mu, __shfl_down(mu, off, 64));
+    mv = fmaxf(mv, __shfl_down(mv, off, 64));
+  }
+  int lane = threadIdx.x % 64, wave = threadIdx.x / 64;
+  if (lane == 0) {
+    lsum[0][wave] = su;
+    lsum[1][wave] = sv;
+    lmax[0][wave] = mu;
+    lmax[1][wave] = mv;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float tsu = 0, tsv = 0, tmu = 0, tmv = 0;
+    for (int wv = 0; wv < (int)blockDim.x / 64; ++wv) {
+      tsu += lsum[0][wv];
+      tsv += lsum[1][wv];
+      tmu = fmaxf(tmu, lmax[0][wv]);
+      tmv = fmaxf(tmv, lmax[1][wv]);
+    }
+    float* p = partials + ((i64)frame * chunks + chunk) * 4;
+    p[0] = tsu;
+    p[1] = tsv;
+    p[2] = tmu;
+    p[3] = tmv;
+  }
+}
+
+__global__ void flowstats_reduce_kernel(const float* __restrict__ partials,
+                                        int chunks, i64 npix,
+                                        float* __restrict__ out) {
+  int frame = blockIdx.x;
+  if (threadIdx.x != 0) return;
+  const float* p = partials + (i64)frame * chunks * 4;
+  float su = 0, sv = 0, mu = 0, mv = 0;
+  for (int cc = 0; cc < chunks; ++cc) {
+    su += p[cc * 4 + 0];
+    sv += p[cc * 4 + 1];
+    mu = fmaxf(mu, p[cc * 4 + 2]);
+    mv = fmaxf(mv, p[cc * 4 + 3]);
+  }
+  float* o = out + (i64)frame * 4;
+  o[0] = su / npix;
+  o[1] = sv / npix;
+  o[2] = mu;
+  o[3] = mv;
+}
+
+class FlowStatsKernelGPU : public BatchedKernel {
+ public:
+  using BatchedKernel::BatchedKernel;
+  void execute_batch(const BatchedElements& in, BatchedElements& out) override {
+    hipStream_t s = cur_stream();
+    size_t n = in[0].size();
+    if (n == 0) return;
+    const Element& f0 = in[0][0];
+    SCA_CHECK(f0.is_frame && f0.frame_info.type == FrameType::F32 &&
+                  f0.frame_info.shape[2] == 2,
+              "FlowStats expects H x W x 2 f32 flow frames");
+    i64 npix = (i64)f0.frame_info.shape[0] * f0.frame_info.shape[1];
+    int chunks = 32;
+    auto dev = config_.device;
+    std::vector<const u8*> ptrs(n);
+    for (size_t i = 0; i < n; ++i) ptrs[i] = in[0][i].buffer;
+    u8* d_ptrs = new_buffer(dev, n * sizeof(u8*));
+    OF_CHECK(hipMemcpyAsync(d_ptrs, ptrs.data(), n * sizeof(u8*),
+                            hipMemcpyHostToDevice, s));
+    u8* scratch = new_buffer(dev, n * chunks * 4 * sizeof(float));
+    u8* out_block = new_block_buffer(dev, n * 4 * sizeof(float), (i32)n);
+    dim3 grid(chunks, (u32)n);
+    flowstats_partial_kernel<<<grid, 256, 0, s>>>(
+        (const float* const*)d_ptrs, npix, (float*)scratch, chunks);
+    OF_CHECK(hipGetLastError());
+    flowstats_reduce_kernel<<<(u32)n, 64, 0, s>>>((const float*)scratch,
+                                                  chunks, npix,
+                                                  (float*)out_block);
+    OF_CHECK(hipGetLastError());
+    OF_CHECK(hipStreamSynchronize(s));
+    delete_buffer(dev, scratch);
+    delete_buffer(dev, d_ptrs);
+    for (size_t i = 0; i < n; ++i) {
+      Element e;
+      e.buffer = out_block + i * 4 * sizeof(float);
+      e.size = 4 * sizeof(float);
+      e.device = dev;
+      out[0].push_back(e);
+    }
+  }
+};
+
+// ---------------- host kernel ----------------
+
+class OpticalFlowKernelGPU : public BaseKernel {
+ public:
+  explicit OpticalFlowKernelGPU(const KernelConfig& cfg) : BaseKernel(cfg) {
+    auto a = mp::decode(cfg.args);
+    radius_ = (int)a.get_int("radius", of::kDefaultRadius);
+    iters_ = (int)a.get_int("iters", of::kDefaultIters);
+    max_levels_ = (int)a.get_int("levels", of::kDefaultMaxLevels);
+    SCA_CHECK(radius_ == 2 || radius_ == 3 || radius_ == 4,
+              "OpticalFlow radius must be 2, 3 or 4");
+  }
+
+  void execute(const StenciledElements& in, BatchedElements& out) override {
+    hipStream_t s = cur_stream();
+    size_t n = in[0].size();  // pairs
+    if (n == 0) return;
+    const Element& f00 = in[0][0][0];
+    SCA_CHECK(f00.is_frame && f00.device.is_gpu(),
+              "GPU OpticalFlow needs GPU frame input");
+    int h = f00.frame_info.shape[0], w = f00.frame_info.shape[1],
+        c = f00.frame_info.shape[2];
+    SCA_CHECK(f00.frame_info.type == FrameType::U8,
+              "OpticalFlow expects u8 frames");
+
+    // Dedup unique frames across stencil windows (frame i closes pair i-1
+    // and opens pair i).
+    std::map<const u8*, int> uniq;
+    std::vector<const u8*> frame_ptrs;
+    std::vector<int> pf0(n), pf1(n);
+    for (size_t i = 0; i < n; ++i) {
+      SCA_CHECK(in[0][i].size() == 2, "OpticalFlow stencil must be [0,1]");
+      for (int j = 0; j < 2; ++j) {
+        const Element& e = in[0][i][j];
+        SCA_CHECK(e.frame_info.shape[0] == h && e.frame_info.shape[1] == w,
+                  "OpticalFlow batch with mixed frame sizes");
+        auto it = uniq.find(e.buffer);
+        int idx;
+        if (it == uniq.end()) {
+          idx = (int)frame_ptrs.size();
+          uniq.emplace(e.buffer, idx);
+          frame_ptrs.push_back(e.buffer);
+        } else {
+          idx = it->second;
+        }
+        (j == 0 ? pf0 : pf1)[i] = idx;
+      }
+    }
+    int nf = (int)frame_ptrs.size();
+
+    int levels = num_levels_clamped(h, w);
+    // Pyramid geometry + one gray buffer holding all levels of all frames:
+    // [level][frame][h_l*w_l], levels concatenated.
+    std::vector<int> lh(levels), lw(levels);
+    std::vector<i64> loff(levels);
+    i64 gray_elems = 0;
+    for (int l = 0; l < levels; ++l) {
+      lh[l] = l ? lh[l - 1] / 2 : h;
+      lw[l] = l ? lw[l - 1] / 2 : w;
+      loff[l] = gray_elems;
+      gray_elems += (i64)nf * lh[l] * lw[l];
+    }
+
+    auto dev = config_.device;
+    u8* gray = new_buffer(dev, gray_elems * sizeof(float));
+    u8* d_ptrs = new_buffer(dev, nf * sizeof(u8*));
+    u8* d_pf = new_buffer(dev, 2 * n * sizeof(int));
+    OF_CHECK(hipMemcpyAsync(d_ptrs, frame_ptrs.data(), nf * sizeof(u8*),
+                            hipMemcpyHostToDevice, s));
+    OF_CHECK(hipMemcpyAsync(d_pf, pf0.data(), n * sizeof(int),
+                            hipMemcpyHostToDevice, s));
+    OF_CHECK(hipMemcpyAsync(d_pf + n * sizeof(int), pf1.data(),
+                            n * sizeof(int), hipMemcpyHostToDevice, s));
+
+    float* gbase = (float*)gray;
+    of_gray_kernel<<<grid_1d((i64)nf * h * w), 256, 0, s>>>(
+        (const u8* const*)d_ptrs, nf, h, w, c, gbase);
+    OF_CHECK(hipGetLastError());
+    for (int l = 1; l < levels; ++l) {
+      of_down2x_kernel<<<grid_1d((i64)nf * lh[l] * lw[l]), 256, 0, s>>>(
+          gbase + loff[l - 1], lh[l - 1], lw[l - 1], gbase + loff[l], lh[l],
+          lw[l], nf);
+      OF_CHECK(hipGetLastError());
+    }
+
+    // Flow ping-pong buffers sized for level 0; output is a block buffer so
+    // the whole packet's flow is one allocation.
+    i64 flow_bytes = (i64)n * h * w * 2 * sizeof(float);
+    u8* ping = new_buffer(dev, flow_bytes);
+    u8* pong = new_buffer(dev, flow_bytes);
+    u8* out_block = new_block_buffer(dev, flow_bytes, (i32)n);
+    float* cur = (float*)ping;
+    float* alt = (float*)pong;
+
+    int top = levels - 1;
+    OF_CHECK(hipMemsetAsync(cur, 0,
+                            (i64)n * lh[top] * lw[top] * 2 * sizeof(float),
+                            s));
+    const int* d_pf0 = (const int*)d_pf;
+    const int* d_pf1 = (const int*)(d_pf + n * sizeof(int));
+    for (int l = top; l >= 0; --l) {
+      dim3 grid((lw[l] + 15) / 16, (lh[l] + 15) / 16, (u32)n);
+      dim3 block(16, 16);
+      for (int it = 0; it < iters_; ++it) {
+        float* dst = (l == 0 && it == iters_ - 1) ? (float*)out_block : alt;
+        launch_lk(grid, block, s, gbase, loff[l], lh[l], lw[l], d_pf0, d_pf1,
+                  cur, dst);
+        std::swap(cur, alt);
+        if (dst == (float*)out_block) cur = (float*)out_block;
+      }
+      if (l > 0) {
+        of_upsample_kernel<<<grid_1d((i64)n * lh[l - 1] * lw[l - 1]), 256, 0,
+                             s>>>(cur, lh[l], lw[l], alt, lh[l - 1],
+                                  lw[l - 1], (int)n);
+        OF_CHECK(hipGetLastError());
+        std::swap(cur, alt);
+      }
+    }
+
+    // Scratch feeds kernels on this stream; sync before returning it to the
+    // shared pool.
+    OF_CHECK(hipStreamSynchronize(s));
+    delete_buffer(dev, gray);
+    delete_buffer(dev, d_ptrs);
+    delete_buffer(dev, d_pf);
+    delete_buffer(dev, ping);
+    delete_buffer(dev, pong);
+
+    i64 per_pair = (i64)h * w * 2 * sizeof(float);
+    for (size_t i = 0; i < n; ++i) {
+      Element e;
+      e.is_frame = true;
+      e.frame_info.shape[0] = h;
+      e.frame_info.shape[1] = w;
+      e.frame_info.shape[2] = 2;
+      e.frame_info.type = FrameType::F32;
+      e.size = per_pair;
+      e.buffer = out_block + i * per_pair;
+      e.device = dev;
+      out[0].push_back(e);
+    }
+  }
+
+ private:
+  int num_levels_clamped(int h, int w) const {
+    return of::num_levels(h, w, max_levels_);
+  }
+
+  void launch_lk(dim3 grid, dim3 block, hipStream_t s, const float* gray,
+                 i64 off, int h, int w, const int* pf0, const int* pf1,
+                 const float* fin, float* fout) {
+    switch (radius_) {
+      case 2:
+        of_lk_kernel<2><<<grid, block, 0, s>>>(gray, off, h, w, pf0, pf1,
+                                               fin, fout);
+        break;
+      case 3:
+        of_lk_kernel<3><<<grid, block, 0, s>>>(gray, off, h, w, pf0, pf1,
+                                               fin, fout);
+        break;
+      default:
+        of_lk_kernel<4><<<grid, block, 0, s>>>(gray, off, h, w, pf0, pf1,
+                                               fin, fout);
+        break;
+    }
+    OF_CHECK(hipGetLastError());
+  }
+
+  int radius_, iters_, max_levels_;
+};
+
+}  // namespace
+
+void register_optflow_gpu() {
+  static bool done = false;
+  if (done) return;
+  done = true;
+  KernelFactory f;
+  f.op_name = "OpticalFlow";
+  f.device_type = DeviceType::GPU;
+  f.preferred_batch = 16;
+  f.make = [](const KernelConfig& c) -> std::unique_ptr<BaseKernel> {
+    return std::make_unique<OpticalFlowKernelGPU>(c);
+  };
+  kernel_registry().add(f);
+  {
+    // OpInfo for FlowStats is registered by register_stdlib_ops().
+    KernelFactory fs;
+    fs.op_name = "FlowStats";
+    fs.device_type = DeviceType::GPU;
+    fs.preferred_batch = 32;
+    fs.make = [](const KernelConfig& c) -> std::unique_ptr<BaseKernel> {
+      return std::make_unique<FlowStatsKernelGPU>(c);
+    };
+    kernel_registry().add(fs);
+  }
+}
+
+}  // namespace sca

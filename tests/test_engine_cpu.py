@@ -196,7 +196,7 @@ def test_stencil_wider_than_packet(sc):
            cache_mode=sp.CacheMode.Overwrite)
     rows = list(sp.NamedVideoStream(sc, "stn_out").load())
     assert len(rows) == 12
-    assert rows[0].shape == (3, 4, 2)  # 48/16 x 64/16 x 2
+    assert rows[0].shape == (48, 64, 2)  # dense per-pixel flow
 
 
 def test_slice_unslice(sc):
@@ -355,3 +355,54 @@ def test_svc_strided_decode(sc):
     for k, blob in enumerate(rows):
         got = np.frombuffer(blob, dtype=np.uint32).reshape(3, 256)
         np.testing.assert_array_equal(got, ref_histogram(frames[k * 7]))
+
+
+def make_textured_pair(h=96, w=128, dx=2, dy=1, seed=3):
+    """Frame pair with known integer translation: band-limited random
+    texture (LK needs gradients at every scale)."""
+    from scipy.ndimage import gaussian_filter
+    rng = np.random.RandomState(seed)
+    base = gaussian_filter(rng.rand(h, w) * 255, sigma=2.0)
+    base = ((base - base.min()) / (np.ptp(base) + 1e-9) * 255).astype(np.uint8)
+    f0 = np.stack([base] * 3, axis=-1)
+    f1 = np.roll(f0, (dy, dx), axis=(0, 1))
+    return np.stack([f0, f1])
+
+
+def test_optical_flow_translation(sc):
+    dx, dy = 2, 1
+    frames = make_textured_pair(dx=dx, dy=dy)
+    video = sp.NamedVideoStream(sc, "of", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    flow = sc.ops.OpticalFlow(frame=frame)
+    out = sp.NamedStream(sc, "of_out")
+    sc.run(sc.io.Output(flow, [out]), sp.PerfParams.manual(2, 2),
+           cache_mode=sp.CacheMode.Overwrite)
+    rows = list(sp.NamedVideoStream(sc, "of_out").load())
+    assert len(rows) == 2
+    f = rows[0]
+    assert f.shape == (96, 128, 2) and f.dtype == np.float32
+    # interior flow recovers the translation (frame1 = frame0 shifted by
+    # (dy,dx), so I1(x+dx,y+dy)=I0(x,y) => LK converges to u=(dx,dy))
+    interior = f[16:-16, 16:-16]
+    err = np.abs(interior - np.array([dx, dy], np.float32))
+    assert np.median(err[..., 0]) < 0.25, np.median(err[..., 0])
+    assert np.median(err[..., 1]) < 0.25, np.median(err[..., 1])
+
+
+def test_flow_stats(sc):
+    frames = make_textured_pair(dx=2, dy=1)
+    video = sp.NamedVideoStream(sc, "fs", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    flow = sc.ops.OpticalFlow(frame=frame)
+    stats = sc.ops.FlowStats(flow=flow)
+    out = sp.NamedStream(sc, "fs_out")
+    sc.run(sc.io.Output(stats, [out]), sp.PerfParams.manual(2, 2),
+           cache_mode=sp.CacheMode.Overwrite)
+    rows = list(out.load())
+    assert len(rows) == 2
+    s = np.frombuffer(rows[0], np.float32)
+    assert s.shape == (4,)
+    # mean |u| ~ 2, mean |v| ~ 1 for the (2,1) translation pair
+    assert 1.0 < s[0] < 3.0 and 0.4 < s[1] < 2.0
+    assert s[2] >= s[0] and s[3] >= s[1]  # max >= mean

@@ -100,3 +100,26 @@ def test_gpu_pool_allocator():
            cache_mode=sp.CacheMode.Overwrite, gpu_ids=[0])
     assert len(list(out.load())) == 4
     _core.destroy_memory()
+
+
+def test_gpu_optical_flow_matches_cpu(sc):
+    from test_engine_cpu import make_textured_pair
+    pair = make_textured_pair(h=96, w=128, dx=2, dy=1)
+    # 5 frames: repeat the pair pattern so several stencil windows run
+    frames = np.concatenate([pair, pair[::-1], pair[:1]])
+    video = sp.NamedVideoStream(sc, "g_of", frames=frames, codec="raw")
+    for dev, name in ((sp.DeviceType.CPU, "g_of_cpu"),
+                      (sp.DeviceType.GPU, "g_of_gpu")):
+        frame = sc.io.Input([video])
+        flow = sc.ops.OpticalFlow(frame=frame, device=dev)
+        out = sp.NamedStream(sc, name)
+        sc.run(sc.io.Output(flow, [out]), sp.PerfParams.manual(4, 8),
+               cache_mode=sp.CacheMode.Overwrite,
+               gpu_ids=[0] if dev == sp.DeviceType.GPU else [])
+    cpu = np.stack(list(sp.NamedVideoStream(sc, "g_of_cpu").load()))
+    gpu = np.stack(list(sp.NamedVideoStream(sc, "g_of_gpu").load()))
+    assert cpu.shape == gpu.shape == (5, 96, 128, 2)
+    # same algorithm in f32; differences only from fma contraction
+    diff = np.abs(cpu - gpu)
+    assert np.median(diff) < 1e-3, np.median(diff)
+    assert np.percentile(diff, 99) < 0.1, np.percentile(diff, 99)
