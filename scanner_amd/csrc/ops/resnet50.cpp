@@ -6,27 +6,19 @@
 // Weights are random-init (He) by default — there is no network in this
 // environment — or loaded from a tensor file for numerics tests
 // (tests/test_resnet_gpu.py compares against a PyTorch fp32 reference).
-#include <random>
-
 #include "../../kernels/dnn.h"
 #include "../memory.h"
 #include "../msgpack.h"
-#include "../serialize.h"
-#include "../storage.h"
+#include "dnn_model.h"
 #include "kernel.h"
 
 namespace sca {
 
 namespace {
 
-struct ConvSpec {
-  std::string name;
-  int in_c, out_c, r, s, stride, pad;
-  bool relu;
-  int residual_from = -1;  // index into block-local buffers (see forward)
-  int kp() const { return (r * s * in_c + 63) / 64 * 64; }
-  int np() const { return (out_c + 63) / 64 * 64; }
-};
+using dnn::ConvSpec;
+using dnn::DeviceModel;
+using dnn::Tensors;
 
 // The full layer list, grouped into bottleneck blocks.
 struct Bottleneck {
@@ -50,79 +42,7 @@ struct ResNet50Config {
   }
 };
 
-struct Tensors {
-  // name -> f32 host data
-  std::map<std::string, std::vector<f32>> t;
-  bool has(const std::string& n) const { return t.count(n); }
-  std::vector<f32>& operator[](const std::string& n) { return t[n]; }
-};
-
-// Deterministic random weights: He-normal conv weights, BN folded to
-// scale ~ U(0.7, 1.3), bias ~ N(0, 0.05).
-void random_init(Tensors& ts, const std::string& name, int out_c, int in_c,
-                 int r, int s, std::mt19937& rng) {
-  std::normal_distribution<f32> nd(0.f, std::sqrt(2.f / (in_c * r * s)));
-  auto& w = ts[name + ".weight"];
-  w.resize((size_t)out_c * r * s * in_c);
-  for (auto& v : w) v = nd(rng);
-  std::uniform_real_distribution<f32> su(0.7f, 1.3f);
-  std::normal_distribution<f32> bn(0.f, 0.05f);
-  auto& sc = ts[name + ".scale"];
-  auto& bi = ts[name + ".bias"];
-  sc.resize(out_c);
-  bi.resize(out_c);
-  for (auto& v : sc) v = su(rng);
-  for (auto& v : bi) v = bn(rng);
-}
-
-Tensors load_tensor_file(const std::string& path) {
-  auto storage = StorageBackend::make_posix();
-  auto buf = storage->read_all(path);
-  BinReader r(buf);
-  u32 magic = r.u32v();
-  SCA_CHECK(magic == 0x52534E54, "bad tensor file magic");  // 'TNSR'
-  u32 n = r.u32v();
-  Tensors ts;
-  for (u32 i = 0; i < n; ++i) {
-    std::string name = r.str();
-    ts[name] = r.vec_pod<f32>();
-  }
-  return ts;
-}
-
-// Device-side model: one bf16 weight blob ([out][kp] per conv,
-// concatenated), one f32 scale+bias blob.
-struct DeviceModel {
-  DeviceHandle dev;
-  u8* weights = nullptr;   // bf16
-  u8* scalebias = nullptr; // f32: per conv scale then bias (padded np)
-  struct Entry {
-    ConvSpec spec;
-    size_t w_off;   // bf16 elements
-    size_t sb_off;  // f32 elements (scale at sb_off, bias at sb_off+np)
-  };
-  std::vector<Entry> convs;  // ordered; name->idx
-  std::map<std::string, int> by_name;
-  float* mean = nullptr;  // 3 floats + 3 std
-  ~DeviceModel() {
-    if (weights) delete_buffer(dev, weights);
-    if (scalebias) delete_buffer(dev, scalebias);
-    if (mean) delete_buffer(dev, (u8*)mean);
-  }
-};
-
-u16 f32_to_bf16_host(f32 v) {
-  u32 bits;
-  std::memcpy(&bits, &v, 4);
-  // round-to-nearest-even
-  u32 lsb = (bits >> 16) & 1;
-  bits += 0x7fff + lsb;
-  return (u16)(bits >> 16);
-}
-
-std::shared_ptr<DeviceModel> build_model(DeviceHandle dev,
-                                         const std::string& weights_file,
-                                         u64 seed) {
+std::vector<ConvSpec> resnet50_specs() {
   ResNet50Config cfg;
   std::vector<ConvSpec> specs;
   specs.push_back({"conv1", 3, 64, 7, 7, 2, 3, true});
@@ -139,83 +59,19 @@ std::shared_ptr<DeviceModel> build_model(DeviceHandle dev,
     }
   }
   specs.push_back({"fc", 2048, 1000, 1, 1, 1, 0, false});
-
-  Tensors ts;
-  if (!weights_file.empty()) {
-    ts = load_tensor_file(weights_file);
-  } else {
-    std::mt19937 rng((u32)seed);
-    for (auto& sp : specs) {
-      random_init(ts, sp.name, sp.out_c, sp.in_c, sp.r, sp.s, rng);
-    }
-    // fc: plain linear (scale 1)
-    auto& sc = ts["fc.scale"];
-    std::fill(sc.begin(), sc.end(), 1.f);
-  }
-
-  auto model = std::make_shared<DeviceModel>();
-  model->dev = dev;
-  size_t w_elems = 0, sb_elems = 0;
-  for (auto& sp : specs) {
-    model->by_name[sp.name] = (int)model->convs.size();
-    model->convs.push_back({sp, w_elems, sb_elems});
-    w_elems += (size_t)sp.np() * sp.kp();
-    sb_elems += 2 * (size_t)sp.np();
-  }
-  std::vector<u16> wh(w_elems, 0);
-  std::vector<f32> sbh(sb_elems, 0.f);
-  for (auto& e : model->convs) {
-    const ConvSpec& sp = e.spec;
-    auto& w = ts[sp.name + ".weight"];
-    SCA_CHECK((i64)w.size() == (i64)sp.out_c * sp.r * sp.s * sp.in_c,
-              "weight size mismatch for " + sp.name);
-    int krs = sp.r * sp.s * sp.in_c;
-    // tensor file layout: [out][r][s][in] — same k ordering as im2col
-    // (cell-major, channel-minor)
-    for (int o = 0; o < sp.out_c; ++o) {
-      for (int k = 0; k < krs; ++k) {
-        wh[e.w_off + (size_t)o * sp.kp() + k] =
-            f32_to_bf16_host(w[(size_t)o * krs + k]);
-      }
-    }
-    auto& sc = ts[sp.name + ".scale"];
-    auto& bi = ts[sp.name + ".bias"];
-    SCA_CHECK((i64)sc.size() == sp.out_c && (i64)bi.size() == sp.out_c,
-              "scale/bias size mismatch for " + sp.name);
-    for (int o = 0; o < sp.out_c; ++o) {
-      sbh[e.sb_off + o] = sc[o];
-      sbh[e.sb_off + sp.np() + o] = bi[o];
-    }
-  }
-  model->weights = new_buffer(dev, w_elems * 2);
-  memcpy_buffer(model->weights, dev, (const u8*)wh.data(), CPU_DEVICE,
-                w_elems * 2);
-  model->scalebias = new_buffer(dev, sb_elems * 4);
-  memcpy_buffer(model->scalebias, dev, (const u8*)sbh.data(), CPU_DEVICE,
-                sb_elems * 4);
-  f32 ms[6] = {0.485f, 0.456f, 0.406f, 0.229f, 0.224f, 0.225f};
-  model->mean = (float*)new_buffer(dev, 6 * 4);
-  memcpy_buffer((u8*)model->mean, dev, (const u8*)ms, CPU_DEVICE, 24);
-  return model;
+  return specs;
 }
-
-// Model cache: one DeviceModel per (device, weights_file, seed), shared
-// across pipeline instances (reference analogue: fetch_resources +
-// setup_with_resources barrier, evaluate_worker.cpp:493-550).
-std::mutex g_model_mu;
-std::map<std::string, std::shared_ptr<DeviceModel>> g_models;
 
 std::shared_ptr<DeviceModel> get_model(DeviceHandle dev,
                                        const std::string& weights_file,
                                        u64 seed) {
-  std::string key = dev.to_string() + "|" + weights_file + "|" +
-                    std::to_string(seed);
-  std::lock_guard<std::mutex> l(g_model_mu);
-  auto it = g_models.find(key);
-  if (it != g_models.end()) return it->second;
-  auto m = build_model(dev, weights_file, seed);
-  g_models[key] = m;
-  return m;
+  return dnn::get_model("resnet50", dev, weights_file, seed, [&]() {
+    Tensors ts;
+    if (!weights_file.empty()) ts = dnn::load_tensor_file(weights_file);
+    auto m = dnn::build_device_model(dev, resnet50_specs(), std::move(ts),
+                                     seed);
+    return m;
+  });
 }
 
 class ResNet50KernelGPU : public BatchedKernel {

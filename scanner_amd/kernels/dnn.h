@@ -45,4 +45,15 @@ void global_avgpool_bf16(const void* in, int n, int h, int w, int c,
 void bf16_rows_to_f32(const void* in, int n, int stride_cols, int ncols,
                       void* out, void* stream);
 
+// NHWC channel concat of three bf16 tensors sharing npix pixels; per-input
+// channel strides let branches read from GEMM-padded buffers.
+void concat3_bf16(const void* a, int ca, int stride_a, const void* b, int cb,
+                  int stride_b, const void* c, int cc, int stride_c, i64 npix,
+                  void* out, void* stream);
+
+// Per-(frame,channel) spatial argmax of NHWC bf16 maps (padded channel
+// stride c_stride): out[n][nch][3] f32 = {x, y, peak}.
+void heatmap_argmax(const void* maps, int n, int h, int w, int c_stride,
+                    int nch, void* out, void* stream);
+
 }  // namespace sca

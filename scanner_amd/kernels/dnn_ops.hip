@@ -220,4 +220,99 @@ void bf16_rows_to_f32(const void* in, int n, int stride_cols, int ncols,
   DNN_CHECK();
 }
 
+
+// ---- pose-net helpers (csrc/ops/pose.cpp) ----
+
+namespace {
+// NHWC channel concat of three tensors (b and c channel counts may read
+// from GEMM-padded buffers via stride args).
+__global__ void __launch_bounds__(256)
+    concat3_kernel(const bf16* __restrict__ a, int ca, int sa,
+                   const bf16* __restrict__ b, int cb, int sb,
+                   const bf16* __restrict__ c, int cc, int scc, i64 npix,
+                   bf16* __restrict__ out) {
+  int ct = ca + cb + cc;
+  i64 total = npix * ct;
+  i64 gs = (i64)gridDim.x * blockDim.x;
+  for (i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gs) {
+    i64 pix = i / ct;
+    int ch = (int)(i % ct);
+    bf16 v;
+    if (ch < ca)
+      v = a[pix * sa + ch];
+    else if (ch < ca + cb)
+      v = b[pix * sb + (ch - ca)];
+    else
+      v = c[pix * scc + (ch - ca - cb)];
+    out[i] = v;
+  }
+}
+
+// Per-(frame, channel) spatial argmax over an NHWC bf16 map with padded
+// channel stride: out[frame][ch] = {x, y, score} f32. One workgroup per
+// (frame, channel); wave64 shuffle reduction.
+__global__ void __launch_bounds__(256)
+    heatmap_argmax_kernel(const bf16* __restrict__ maps, int h, int w,
+                          int c_stride, int nch, float* __restrict__ out) {
+  int frame = blockIdx.y, ch = blockIdx.x;
+  const bf16* m = maps + (i64)frame * h * w * c_stride + ch;
+  float best = -1e30f;
+  int besti = 0;
+  for (int i = threadIdx.x; i < h * w; i += blockDim.x) {
+    float v = (float)m[(i64)i * c_stride];
+    if (v > best) {
+      best = v;
+      besti = i;
+    }
+  }
+  __shared__ float lbest[4];
+  __shared__ int lidx[4];
+  for (int off = 32; off; off >>= 1) {
+    float ov = __shfl_down(best, off, 64);
+    int oi = __shfl_down(besti, off, 64);
+    if (ov > best) {
+      best = ov;
+      besti = oi;
+    }
+  }
+  int lane = threadIdx.x % 64, wave = threadIdx.x / 64;
+  if (lane == 0) {
+    lbest[wave] = best;
+    lidx[wave] = besti;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int wv = 1; wv < (int)blockDim.x / 64; ++wv) {
+      if (lbest[wv] > best) {
+        best = lbest[wv];
+        besti = lidx[wv];
+      }
+    }
+    float* o = out + ((i64)frame * nch + ch) * 3;
+    o[0] = (float)(besti % w);
+    o[1] = (float)(besti / w);
+    o[2] = best;
+  }
+}
+}  // namespace
+
+void concat3_bf16(const void* a, int ca, int stride_a, const void* b, int cb,
+                  int stride_b, const void* c, int cc, int stride_c, i64 npix,
+                  void* out, void* stream) {
+  i64 total = npix * (ca + cb + cc);
+  concat3_kernel<<<grid_for(total), 256, 0, (hipStream_t)stream>>>(
+      (const bf16*)a, ca, stride_a, (const bf16*)b, cb, stride_b,
+      (const bf16*)c, cc, stride_c, npix, (bf16*)out);
+  DNN_CHECK();
+}
+
+void heatmap_argmax(const void* maps, int n, int h, int w, int c_stride,
+                    int nch, void* out, void* stream) {
+  dim3 grid(nch, n);
+  heatmap_argmax_kernel<<<grid, 256, 0, (hipStream_t)stream>>>(
+      (const bf16*)maps, h, w, c_stride, nch, (float*)out);
+  DNN_CHECK();
+}
+
 }  // namespace sca
