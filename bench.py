@@ -58,6 +58,11 @@ def build_pipeline(sc, sp, video, pipeline, device, out_name):
         # the saved column (the 16 MB/frame flow field stays on-GPU).
         flow = sc.ops.OpticalFlow(frame=frame, device=device)
         cols.append(sc.ops.FlowStats(flow=flow, device=device))
+    if pipeline == "pose":
+        # BASELINE config 5: multi-DNN graph on 4K — pose keypoints and
+        # ResNet-50 classification of every frame, decode+DNN overlapped.
+        cols.append(sc.ops.Pose(frame=frame, device=device))
+        cols.append(sc.ops.ResNet50(frame=frame, device=device))
     out = sp.NamedStream(sc, out_name)
     return sc.io.Output(cols, [out])
 
@@ -69,7 +74,7 @@ def main():
     ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--pipeline", default="full",
                     choices=["hist", "resnet", "full", "flow", "pose"])
-    ap.add_argument("--frames-per-step", type=int, default=FRAMES_PER_STEP)
+    ap.add_argument("--frames-per-step", type=int, default=None)
     args = ap.parse_args()
 
     import torch
@@ -96,9 +101,12 @@ def main():
     tmp = tempfile.mkdtemp(prefix=f"scanner_bench_r{rank}_")
     sc = sp.Client(db_path=os.path.join(tmp, "db"))
 
-    n_frames = args.frames_per_step
-    log(f"[rank {rank}] ingesting {n_frames} synthetic 1080p frames (svc)")
-    clip = make_clip(n_frames)
+    # pose runs on 4K per BASELINE config 5; everything else on 1080p
+    h, w = (2160, 3840) if args.pipeline == "pose" else (H, W)
+    n_frames = args.frames_per_step or \
+        (128 if args.pipeline == "pose" else FRAMES_PER_STEP)
+    log(f"[rank {rank}] ingesting {n_frames} synthetic {h}x{w} frames (svc)")
+    clip = make_clip(n_frames, h=h, w=w)
     video = sp.NamedVideoStream(sc, "bench_clip", frames=clip,
                                 codec="svc", io_packet_size=128)
     del clip
@@ -150,10 +158,12 @@ def main():
     total_frames = args.steps * n_frames * world
     fps = total_frames / elapsed
     if rank == 0:
+        pipe_name = {"full": "histogram+ResNet50", "pose": "Pose+ResNet50",
+                     }.get(args.pipeline, args.pipeline)
+        res_name = "4K" if args.pipeline == "pose" else "1080p"
         result = {
-            "metric": "frames/sec (whole node), 1080p "
-                      + ("histogram+ResNet50" if args.pipeline == "full"
-                         else args.pipeline) + " pipeline",
+            "metric": f"frames/sec (whole node), {res_name} "
+                      f"{pipe_name} pipeline",
             "value": fps,
             "unit": "frames/s",
             "n_gpus": world,
@@ -165,10 +175,12 @@ def main():
             "vs_baseline": None,
             "dtype": {"resnet": "bf16", "full": "bf16", "pose": "bf16",
                       "flow": "f32", "hist": "u8"}[args.pipeline],
-            "data": "synthetic 1080p video (svc-encoded), random-init weights",
+            "data": f"synthetic {res_name} video (svc-encoded), "
+                    "random-init weights",
             "config": {
-                "model": "1080p histogram + ResNet-50 classify"
-                         if args.pipeline == "full" else args.pipeline,
+                "model": {"full": "1080p histogram + ResNet-50 classify",
+                          "pose": "4K 3-stage 2-branch pose CNN + ResNet-50",
+                          }.get(args.pipeline, args.pipeline),
                 "global_batch": n_frames * world,
                 "seq_len": n_frames,
                 "parallelism": f"frame-shard dp{world}",

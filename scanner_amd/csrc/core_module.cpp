@@ -89,6 +89,7 @@ PYBIND11_MODULE(_core, m) {
   register_gpu_ops();
   register_resnet50_op();
   register_optflow_gpu();
+  register_pose_op();
 
   m.def("have_gpu", &have_gpu);
   m.def("gpu_device_count", &gpu_device_count);

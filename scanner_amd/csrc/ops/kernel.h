@@ -155,5 +155,6 @@ struct KernelRegistrar {
 void register_stdlib_ops();
 void register_resnet50_op();
 void register_optflow_gpu();
+void register_pose_op();
 
 }  // namespace sca

@@ -74,3 +74,29 @@ def test_resnet50_vs_torch(sc, tmp_path):
             f"frame {i}: top1 {np.argmax(got[i])} vs {np.argmax(ref[i])}"
         corr = np.corrcoef(got[i], ref[i])[0, 1]
         assert corr > 0.98, f"frame {i}: corr {corr}"
+
+
+def test_pose_op(sc):
+    from conftest import make_video
+    frames = make_video(n=6, h=480, w=640)
+    video = sp.NamedVideoStream(sc, "pose_v", frames=frames, codec="raw")
+
+    def run(tag):
+        frame = sc.io.Input([video])
+        pose = sc.ops.Pose(frame=frame, device=sp.DeviceType.GPU)
+        out = sp.NamedStream(sc, f"pose_out_{tag}")
+        sc.run(sc.io.Output(pose, [out]), sp.PerfParams.manual(4, 8),
+               cache_mode=sp.CacheMode.Overwrite, gpu_ids=[0])
+        return [np.frombuffer(b, np.float32).reshape(19, 3)
+                for b in out.load()]
+
+    a = run("a")
+    assert len(a) == 6
+    for kp in a:
+        assert np.isfinite(kp).all()
+        assert (kp[:, 0] >= 0).all() and (kp[:, 0] < 46).all()
+        assert (kp[:, 1] >= 0).all() and (kp[:, 1] < 46).all()
+    # deterministic: same seed => identical keypoints across jobs
+    b = run("b")
+    for x, y in zip(a, b):
+        np.testing.assert_array_equal(x, y)

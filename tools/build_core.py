@@ -24,6 +24,7 @@ CPP_SOURCES = [
     "csrc/ops/stdlib_cpu.cpp",
     "csrc/ops/python_kernel.cpp",
     "csrc/ops/resnet50.cpp",
+    "csrc/ops/pose.cpp",
     "csrc/engine/table_io.cpp",
     "csrc/engine/executor.cpp",
     "csrc/video/svc_cpu.cpp",
