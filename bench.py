@@ -80,13 +80,11 @@ def main():
     import torch
     import torch.distributed as dist
 
-    rank = int(os.environ.get("RANK", "0"))
-    world = int(os.environ.get("WORLD_SIZE", "1"))
+    from scanner_amd import parallel
+
+    rank, world, dist_device = parallel.init_from_env()
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     distributed = world > 1
-    if distributed:
-        dist.init_process_group(backend="nccl")
-        torch.cuda.set_device(local_rank)
 
     import scanner_amd as sp
     from scanner_amd import _core
@@ -149,11 +147,17 @@ def main():
         for k in sorted(stats, key=lambda k: -stats[k]["total_ms"]):
             log(f"[prof] {k}: {stats[k]['total_ms']:.1f} ms "
                 f"x{stats[k]['count']}")
+    elapsed = parallel.allreduce_max_time(elapsed, dist_device)
     if distributed:
-        t = torch.tensor([elapsed], dtype=torch.float64,
-                         device="cuda" if have_gpu else "cpu")
-        dist.all_reduce(t, op=dist.ReduceOp.MAX)
-        elapsed = t.item()
+        # Post-timing RCCL gather over xGMI: each rank ships the first
+        # result rows of its last step's output column to rank 0 (the
+        # reference round-trips results through shared storage instead).
+        sample = list(sp.NamedStream(
+            sc, f"bench_out_s{args.steps-1}").load(rows=range(4)))
+        gathered = parallel.gather_column(sample, dist_device)
+        if rank == 0:
+            log(f"gathered {len(gathered)} result rows over "
+                f"{'RCCL/xGMI' if have_gpu else 'gloo'}")
 
     total_frames = args.steps * n_frames * world
     fps = total_frames / elapsed

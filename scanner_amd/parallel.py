@@ -1,0 +1,152 @@
+"""Single-node multi-GPU frame-shard data parallelism over RCCL/xGMI.
+
+The reference moves bulk data between workers only through shared storage
+(scanner/engine master/worker + storehouse). On an
+MI355X node the 8 GPUs are a 7-link xGMI point-to-point mesh, so the
+native layout is one engine process per GPU (`torch.distributed` with the
+"nccl" backend, which IS RCCL on ROCm) processing a disjoint row shard,
+plus collectives for the three data movements the reference does through
+storage or gRPC:
+
+  * `shard_rows`    — split a job's output rows across ranks (the
+                      reference's master-side task partitioning,
+                      master.cpp:1558-1607, done symmetrically).
+  * `gather_column` — bring per-rank result blobs to rank 0 over xGMI
+                      (replaces "every worker writes to shared storage,
+                      the client re-reads it").
+  * `broadcast_blob`— ship op args / weight files from rank 0 (the
+                      reference syncs ops master->worker over gRPC,
+                      worker.cpp:868-938).
+  * `allreduce_max_time` — job-level timing consensus for benchmarks.
+
+xGMI note: with 7 point-to-point links per GPU, ring collectives are
+per-link bound; for the short, fat transfers here (result columns,
+weight blobs) direct gather/broadcast is the right shape, and RCCL's
+topology detection picks the single-hop path on a full mesh. Bucket fan-in
+caps each gather message at `max_bytes` so no single link serializes one
+giant payload.
+
+CPU fallback uses the gloo backend so the whole module is testable without
+a GPU (tests/test_parallel.py runs world_size=2 over gloo).
+"""
+import os
+import pickle
+
+import torch
+import torch.distributed as dist
+
+_DEFAULT_MAX_BYTES = 64 << 20
+
+
+def init_from_env(device_type="auto"):
+    """Initialize the process group from torchrun env vars. Returns
+    (rank, world_size, device). Safe to call when WORLD_SIZE is unset or 1
+    (returns a degenerate single-rank context without init)."""
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    use_gpu = (device_type == "gpu" or
+               (device_type == "auto" and torch.cuda.is_available()))
+    device = torch.device("cuda", local_rank) if use_gpu \
+        else torch.device("cpu")
+    if world > 1 and not dist.is_initialized():
+        dist.init_process_group(backend="nccl" if use_gpu else "gloo")
+        if use_gpu:
+            torch.cuda.set_device(device)
+    return rank, world, device
+
+
+def shard_rows(n_rows, world, rank):
+    """Contiguous row shard [start, end) for `rank`; like the reference's
+    io-packet task partitioning, but computed symmetrically on every rank
+    (no master round-trip). Remainder rows go to the leading ranks."""
+    base, rem = divmod(n_rows, world)
+    start = rank * base + min(rank, rem)
+    return start, start + base + (1 if rank < rem else 0)
+
+
+def _to_tensor(blob, device):
+    t = torch.frombuffer(bytearray(blob), dtype=torch.uint8)
+    return t.to(device)
+
+
+def gather_column(blobs, device, dst=0, group=None,
+                  max_bytes=_DEFAULT_MAX_BYTES):
+    """Gather a list of per-row byte blobs from every rank to `dst`.
+
+    On GPU the payload moves as uint8 CUDA tensors over RCCL/xGMI (single
+    hop on the full mesh); on CPU over gloo. Returns the concatenated
+    per-rank lists (rank order) on `dst`, None elsewhere. Payloads larger
+    than `max_bytes` per rank are sent in multiple rounds so one link is
+    never saturated by a single message.
+    """
+    if not dist.is_initialized() or dist.get_world_size(group) == 1:
+        return list(blobs)
+    rank = dist.get_rank(group)
+    world = dist.get_world_size(group)
+
+    payload = pickle.dumps(blobs, protocol=pickle.HIGHEST_PROTOCOL)
+    chunks = [payload[i:i + max_bytes]
+              for i in range(0, len(payload), max_bytes)] or [b""]
+    n_rounds = torch.tensor([len(chunks)], dtype=torch.int64)
+    if device.type == "cuda":
+        n_rounds = n_rounds.to(device)
+    dist.all_reduce(n_rounds, op=dist.ReduceOp.MAX, group=group)
+    rounds = int(n_rounds.item())
+
+    received = [bytearray() for _ in range(world)]
+    for r in range(rounds):
+        chunk = chunks[r] if r < len(chunks) else b""
+        t = _to_tensor(chunk, device)
+        size = torch.tensor([t.numel()], dtype=torch.int64,
+                            device=device if device.type == "cuda" else None)
+        sizes = [torch.zeros_like(size) for _ in range(world)]
+        dist.all_gather(sizes, size, group=group)
+        maxn = max(int(s.item()) for s in sizes)
+        pad = torch.zeros(maxn, dtype=torch.uint8, device=device)
+        pad[:t.numel()] = t
+        if rank == dst:
+            outs = [torch.empty(maxn, dtype=torch.uint8, device=device)
+                    for _ in range(world)]
+            dist.gather(pad, outs, dst=dst, group=group)
+            for i in range(world):
+                n = int(sizes[i].item())
+                if n:
+                    received[i] += outs[i][:n].cpu().numpy().tobytes()
+        else:
+            dist.gather(pad, None, dst=dst, group=group)
+    if rank != dst:
+        return None
+    result = []
+    for i in range(world):
+        result.extend(pickle.loads(bytes(received[i])))
+    return result
+
+
+def broadcast_blob(blob, device, src=0, group=None):
+    """Broadcast one byte blob from `src` to every rank (weights, op args).
+    Returns the blob on every rank."""
+    if not dist.is_initialized() or dist.get_world_size(group) == 1:
+        return blob
+    rank = dist.get_rank(group)
+    size = torch.tensor([len(blob) if rank == src else 0],
+                        dtype=torch.int64,
+                        device=device if device.type == "cuda" else None)
+    dist.broadcast(size, src=src, group=group)
+    n = int(size.item())
+    if rank == src:
+        t = _to_tensor(blob, device)
+    else:
+        t = torch.empty(n, dtype=torch.uint8, device=device)
+    dist.broadcast(t, src=src, group=group)
+    return blob if rank == src else t.cpu().numpy().tobytes()
+
+
+def allreduce_max_time(seconds, device):
+    """Whole-job elapsed time = max over ranks (benchmark contract)."""
+    if not dist.is_initialized() or dist.get_world_size() == 1:
+        return seconds
+    t = torch.tensor([seconds], dtype=torch.float64,
+                     device=device if device.type == "cuda" else None)
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    return float(t.item())
