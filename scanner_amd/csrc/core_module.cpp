@@ -90,6 +90,8 @@ PYBIND11_MODULE(_core, m) {
   register_resnet50_op();
   register_optflow_gpu();
   register_pose_op();
+  register_color_gpu();
+  register_image_encoder_op();
 
   m.def("have_gpu", &have_gpu);
   m.def("gpu_device_count", &gpu_device_count);

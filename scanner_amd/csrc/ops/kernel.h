@@ -156,5 +156,7 @@ void register_stdlib_ops();
 void register_resnet50_op();
 void register_optflow_gpu();
 void register_pose_op();
+void register_color_gpu();
+void register_image_encoder_op();
 
 }  // namespace sca

@@ -338,6 +338,91 @@ class OpticalFlowKernelCPU : public StenciledKernel {
   int radius_, iters_, max_levels_;
 };
 
+// ---- Crop: HWC u8 frame -> window (x, y, width, height) ----
+class CropKernelCPU : public BatchedKernel {
+ public:
+  explicit CropKernelCPU(const KernelConfig& cfg) : BatchedKernel(cfg) {
+    auto a = mp::decode(cfg.args);
+    x_ = (i32)a.get_int("x", 0);
+    y_ = (i32)a.get_int("y", 0);
+    w_ = (i32)a.get_int("width", 0);
+    h_ = (i32)a.get_int("height", 0);
+    SCA_CHECK(w_ > 0 && h_ > 0, "Crop needs width/height args");
+  }
+  void execute_batch(const BatchedElements& in, BatchedElements& out) override {
+    for (const Element& f : in[0]) {
+      i32 h = f.frame_info.shape[0], w = f.frame_info.shape[1],
+          c = f.frame_info.shape[2];
+      SCA_CHECK(x_ + w_ <= w && y_ + h_ <= h, "Crop outside frame bounds");
+      Element e = alloc_frame(config_.device, h_, w_, c, FrameType::U8);
+      for (i32 y = 0; y < h_; ++y) {
+        std::memcpy(e.buffer + (i64)y * w_ * c,
+                    f.buffer + (((i64)(y_ + y)) * w + x_) * c,
+                    (size_t)w_ * c);
+      }
+      out[0].push_back(e);
+    }
+  }
+
+ private:
+  i32 x_, y_, w_, h_;
+};
+
+// ---- ColorConvert: gray (BT.601 luma) | yuv (BT.601 full) | planar ----
+class ColorConvertKernelCPU : public BatchedKernel {
+ public:
+  explicit ColorConvertKernelCPU(const KernelConfig& cfg)
+      : BatchedKernel(cfg) {
+    auto a = mp::decode(cfg.args);
+    mode_ = a.get_str("format", "gray");
+    SCA_CHECK(mode_ == "gray" || mode_ == "yuv" || mode_ == "planar",
+              "ColorConvert format must be gray|yuv|planar");
+  }
+  void execute_batch(const BatchedElements& in, BatchedElements& out) override {
+    auto cl = [](f32 v) -> u8 {
+      return (u8)std::min(std::max(v, 0.f), 255.f);
+    };
+    for (const Element& f : in[0]) {
+      i32 h = f.frame_info.shape[0], w = f.frame_info.shape[1],
+          c = f.frame_info.shape[2];
+      i64 npix = (i64)h * w;
+      const u8* p = f.buffer;
+      Element e;
+      if (mode_ == "gray") {
+        SCA_CHECK(c == 3, "gray conversion needs RGB input");
+        e = alloc_frame(config_.device, h, w, 1, FrameType::U8);
+        for (i64 i = 0; i < npix; ++i) {
+          e.buffer[i] = cl(0.299f * p[i * 3] + 0.587f * p[i * 3 + 1] +
+                           0.114f * p[i * 3 + 2] + 0.5f);
+        }
+      } else if (mode_ == "yuv") {
+        SCA_CHECK(c == 3, "yuv conversion needs RGB input");
+        e = alloc_frame(config_.device, h, w, 3, FrameType::U8);
+        for (i64 i = 0; i < npix; ++i) {
+          f32 r = p[i * 3], g = p[i * 3 + 1], b = p[i * 3 + 2];
+          e.buffer[i * 3 + 0] =
+              cl(0.299f * r + 0.587f * g + 0.114f * b + 0.5f);
+          e.buffer[i * 3 + 1] =
+              cl(-0.169f * r - 0.331f * g + 0.5f * b + 128.5f);
+          e.buffer[i * 3 + 2] =
+              cl(0.5f * r - 0.419f * g - 0.081f * b + 128.5f);
+        }
+      } else {  // planar HWC -> CHW (reference parity: image.cu:202-220)
+        e = alloc_frame(config_.device, c, h, w, FrameType::U8);
+        for (i32 ch = 0; ch < c; ++ch) {
+          for (i64 i = 0; i < npix; ++i) {
+            e.buffer[(i64)ch * npix + i] = p[i * c + ch];
+          }
+        }
+      }
+      out[0].push_back(e);
+    }
+  }
+
+ private:
+  std::string mode_;
+};
+
 // ---- FlowStats: dense flow frame -> {mean|u|, mean|v|, max|u|, max|v|} ----
 class FlowStatsKernelCPU : public Kernel {
  public:
@@ -465,6 +550,36 @@ void register_stdlib_ops() {
     f.device_type = DeviceType::CPU;
     f.make = [](const KernelConfig& c) -> std::unique_ptr<BaseKernel> {
       return std::make_unique<SleepKernel>(c);
+    };
+    kernel_registry().add(f);
+  }
+  {
+    OpInfo o;
+    o.name = "Crop";
+    o.input_columns = {frame_in};
+    o.output_columns = {frame_out};
+    op_registry().add(o);
+    KernelFactory f;
+    f.op_name = "Crop";
+    f.device_type = DeviceType::CPU;
+    f.preferred_batch = 8;
+    f.make = [](const KernelConfig& c) -> std::unique_ptr<BaseKernel> {
+      return std::make_unique<CropKernelCPU>(c);
+    };
+    kernel_registry().add(f);
+  }
+  {
+    OpInfo o;
+    o.name = "ColorConvert";
+    o.input_columns = {frame_in};
+    o.output_columns = {frame_out};
+    op_registry().add(o);
+    KernelFactory f;
+    f.op_name = "ColorConvert";
+    f.device_type = DeviceType::CPU;
+    f.preferred_batch = 8;
+    f.make = [](const KernelConfig& c) -> std::unique_ptr<BaseKernel> {
+      return std::make_unique<ColorConvertKernelCPU>(c);
     };
     kernel_registry().add(f);
   }

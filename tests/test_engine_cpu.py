@@ -406,3 +406,85 @@ def test_flow_stats(sc):
     # mean |u| ~ 2, mean |v| ~ 1 for the (2,1) translation pair
     assert 1.0 < s[0] < 3.0 and 0.4 < s[1] < 2.0
     assert s[2] >= s[0] and s[3] >= s[1]  # max >= mean
+
+
+def test_crop(sc):
+    frames = make_video(n=4, h=40, w=60)
+    video = sp.NamedVideoStream(sc, "cr", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    crop = sc.ops.Crop(frame=frame, x=10, y=5, width=32, height=24)
+    out = sp.NamedStream(sc, "cr_out")
+    sc.run(sc.io.Output(crop, [out]), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite)
+    outs = list(sp.NamedVideoStream(sc, "cr_out").load())
+    assert len(outs) == 4
+    for i, o in enumerate(outs):
+        np.testing.assert_array_equal(o, frames[i][5:29, 10:42])
+
+
+def test_color_convert(sc):
+    frames = make_video(n=3, h=24, w=32)
+    video = sp.NamedVideoStream(sc, "ccv", frames=frames, codec="raw")
+    for mode in ("gray", "yuv", "planar"):
+        frame = sc.io.Input([video])
+        cc = sc.ops.ColorConvert(frame=frame, format=mode)
+        out = sp.NamedStream(sc, f"ccv_{mode}")
+        sc.run(sc.io.Output(cc, [out]), sp.PerfParams.manual(4, 8),
+               cache_mode=sp.CacheMode.Overwrite)
+        outs = list(sp.NamedVideoStream(sc, f"ccv_{mode}").load())
+        f = frames[0].astype(np.float32)
+        if mode == "gray":
+            assert outs[0].shape == (24, 32, 1)
+            ref = (0.299 * f[..., 0] + 0.587 * f[..., 1]
+                   + 0.114 * f[..., 2] + 0.5).astype(np.uint8)
+            np.testing.assert_allclose(outs[0][..., 0].astype(int),
+                                       ref.astype(int), atol=1)
+        elif mode == "yuv":
+            assert outs[0].shape == (24, 32, 3)
+            ref_y = (0.299 * f[..., 0] + 0.587 * f[..., 1]
+                     + 0.114 * f[..., 2] + 0.5).astype(np.uint8)
+            np.testing.assert_allclose(outs[0][..., 0].astype(int),
+                                       ref_y.astype(int), atol=1)
+        else:
+            assert outs[0].shape == (3, 24, 32)
+            np.testing.assert_array_equal(
+                outs[0], np.transpose(frames[0], (2, 0, 1)))
+
+
+def decode_png_filter0(blob):
+    """Minimal PNG reader for the encoder's output (filter-0 scanlines)."""
+    import struct
+    import zlib
+    assert blob[:8] == b"\x89PNG\r\n\x1a\x0a"
+    pos = 8
+    idat = b""
+    w = h = ct = None
+    while pos < len(blob):
+        ln, typ = struct.unpack(">I4s", blob[pos:pos + 8])
+        data = blob[pos + 8:pos + 8 + ln]
+        if typ == b"IHDR":
+            w, h, depth, ct = struct.unpack(">IIBB", data[:10])
+            assert depth == 8
+        elif typ == b"IDAT":
+            idat += data
+        pos += 12 + ln
+    c = 3 if ct == 2 else 1
+    raw = zlib.decompress(idat)
+    stride = 1 + w * c
+    rows = [raw[i * stride + 1:(i + 1) * stride] for i in range(h)]
+    assert all(raw[i * stride] == 0 for i in range(h))
+    return np.frombuffer(b"".join(rows), np.uint8).reshape(h, w, c)
+
+
+def test_image_encoder_png(sc):
+    frames = make_video(n=3, h=24, w=32)
+    video = sp.NamedVideoStream(sc, "png", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    enc = sc.ops.ImageEncoder(frame=frame, format="png")
+    out = sp.NamedStream(sc, "png_out")
+    sc.run(sc.io.Output(enc, [out]), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite)
+    rows = list(out.load())
+    assert len(rows) == 3
+    for i, blob in enumerate(rows):
+        np.testing.assert_array_equal(decode_png_filter0(blob), frames[i])

@@ -123,3 +123,35 @@ def test_gpu_optical_flow_matches_cpu(sc):
     diff = np.abs(cpu - gpu)
     assert np.median(diff) < 1e-3, np.median(diff)
     assert np.percentile(diff, 99) < 0.1, np.percentile(diff, 99)
+
+
+def test_gpu_color_ops_match_cpu(sc):
+    frames = make_video(n=4, h=48, w=64)
+    video = sp.NamedVideoStream(sc, "g_c", frames=frames, codec="raw")
+
+    def run(dev, tag, make_op):
+        frame = sc.io.Input([video])
+        col = make_op(frame, dev)
+        out = sp.NamedStream(sc, tag)
+        sc.run(sc.io.Output(col, [out]), sp.PerfParams.manual(4, 8),
+               cache_mode=sp.CacheMode.Overwrite,
+               gpu_ids=[0] if dev == sp.DeviceType.GPU else [])
+        return np.stack(list(sp.NamedVideoStream(sc, tag).load()))
+
+    cases = {
+        "crop": lambda f, d: sc.ops.Crop(frame=f, x=8, y=4, width=32,
+                                         height=24, device=d),
+        "gray": lambda f, d: sc.ops.ColorConvert(frame=f, format="gray",
+                                                 device=d),
+        "yuv": lambda f, d: sc.ops.ColorConvert(frame=f, format="yuv",
+                                                device=d),
+        "planar": lambda f, d: sc.ops.ColorConvert(frame=f, format="planar",
+                                                   device=d),
+        "blur": lambda f, d: sc.ops.Blur(frame=f, kernel_size=5, device=d),
+    }
+    for name, mk in cases.items():
+        cpu = run(sp.DeviceType.CPU, f"g_c_{name}_cpu", mk)
+        gpu = run(sp.DeviceType.GPU, f"g_c_{name}_gpu", mk)
+        assert cpu.shape == gpu.shape, name
+        diff = np.abs(cpu.astype(int) - gpu.astype(int))
+        assert diff.max() <= 1, (name, diff.max())

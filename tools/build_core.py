@@ -25,6 +25,7 @@ CPP_SOURCES = [
     "csrc/ops/python_kernel.cpp",
     "csrc/ops/resnet50.cpp",
     "csrc/ops/pose.cpp",
+    "csrc/ops/image_encoder.cpp",
     "csrc/engine/table_io.cpp",
     "csrc/engine/executor.cpp",
     "csrc/video/svc_cpu.cpp",
@@ -69,7 +70,7 @@ def main():
         "  deps = gcc",
         "rule link",
         "  command = $hipcc -shared -fPIC $in -o $out -L/opt/rocm/lib "
-        "-lamdhip64",
+        "-lamdhip64 -lz",
     ]
     objs = []
     for s in CPP_SOURCES:
