@@ -96,6 +96,46 @@ PYBIND11_MODULE(_core, m) {
   m.def("have_gpu", &have_gpu);
   m.def("gpu_device_count", &gpu_device_count);
 
+  // GEMM micro-benchmark: device-resident buffers, hipEvent timing.
+  // Returns ms per iteration (tools/gemm_tune.py computes TFLOP/s).
+  m.def("gemm_bench", [](int M, int N, int K, int iters, bool relu) {
+    SCA_CHECK(have_gpu(), "gemm_bench needs a GPU");
+    DeviceHandle dev{DeviceType::GPU, 0};
+    u8* dA = new_buffer(dev, (size_t)M * K * 2);
+    u8* dB = new_buffer(dev, (size_t)N * K * 2);
+    u8* dC = new_buffer(dev, (size_t)M * N * 2);
+    u8* dS = new_buffer(dev, (size_t)N * 8);
+    GemmArgs g;
+    g.A = dA;
+    g.B = dB;
+    g.C = dC;
+    g.M = M;
+    g.N = N;
+    g.K = K;
+    g.scale = (const float*)dS;
+    g.bias = (const float*)(dS + (size_t)N * 4);
+    g.relu = relu;
+    void* s = per_thread_hip_stream();
+    for (int i = 0; i < 3; ++i) gemm_bf16(g, s);
+    sync_per_thread_stream();
+    hipEvent_t e0, e1;
+    (void)hipEventCreate(&e0);
+    (void)hipEventCreate(&e1);
+    (void)hipEventRecord(e0, (hipStream_t)s);
+    for (int i = 0; i < iters; ++i) gemm_bf16(g, s);
+    (void)hipEventRecord(e1, (hipStream_t)s);
+    sync_per_thread_stream();
+    float ms = 0.f;
+    (void)hipEventElapsedTime(&ms, e0, e1);
+    (void)hipEventDestroy(e0);
+    (void)hipEventDestroy(e1);
+    delete_buffer(dev, dA);
+    delete_buffer(dev, dB);
+    delete_buffer(dev, dC);
+    delete_buffer(dev, dS);
+    return ms / iters;
+  });
+
   // Raw MFMA GEMM entry for numerics tests: f32 inputs are rounded to
   // bf16 on the host, C = A @ B_nk^T computed on GPU 0.
   m.def("gemm_bf16_test",

@@ -154,32 +154,76 @@ __global__ void __launch_bounds__(256)
   float u = fin[((i64)y * w + x) * 2 + 0];
   float v = fin[((i64)y * w + x) * 2 + 1];
 
+  // The warp offset (u,v) is constant across the window, so the bilinear
+  // fractions are too: the 49 taps read a contiguous (2R+2)^2 region of I1
+  // at one fractional offset. Interior fast path: slide two unrolled row
+  // register arrays down that region — 64 loads/pixel instead of 196, no
+  // per-tap floor/clamp. Border threads take the clamped slow path.
   float a11 = 0, a12 = 0, a22 = 0, b1 = 0, b2 = 0;
+  float wxf = x + u, wyf = y + v;
+  int xi = (int)floorf(wxf), yi = (int)floorf(wyf);
+  float ax = wxf - xi, ay = wyf - yi;
+  constexpr int RW = 2 * RADIUS + 2;
+  bool interior = xi - RADIUS >= 0 && xi + RADIUS + 1 < w &&
+                  yi - RADIUS >= 0 && yi + RADIUS + 1 < h;
+  if (interior) {
+    float r0[RW], r1[RW];
+    const float* row = I1 + (i64)(yi - RADIUS) * w + (xi - RADIUS);
 #pragma unroll
-  for (int dy = -RADIUS; dy <= RADIUS; ++dy) {
+    for (int j = 0; j < RW; ++j) r0[j] = row[j];
 #pragma unroll
-    for (int dx = -RADIUS; dx <= RADIUS; ++dx) {
-      int lx = (int)threadIdx.x + HALO + dx;
+    for (int dy = -RADIUS; dy <= RADIUS; ++dy) {
+      const float* nrow = I1 + (i64)(yi + dy + 1) * w + (xi - RADIUS);
+#pragma unroll
+      for (int j = 0; j < RW; ++j) r1[j] = nrow[j];
       int ly = (int)threadIdx.y + HALO + dy;
-      float ix = 0.5f * (lds_i0[ly * LW + lx + 1] - lds_i0[ly * LW + lx - 1]);
-      float iy = 0.5f * (lds_i0[(ly + 1) * LW + lx] - lds_i0[(ly - 1) * LW + lx]);
-      float i0v = lds_i0[ly * LW + lx];
-      // Bilinear tap of I1 at the flow-warped position (clamped).
-      float wx = x + dx + u, wy = y + dy + v;
-      int wx0 = (int)floorf(wx), wy0 = (int)floorf(wy);
-      float ax = wx - wx0, ay = wy - wy0;
-      int x0c = min(max(wx0, 0), w - 1), x1c = min(max(wx0 + 1, 0), w - 1);
-      int y0c = min(max(wy0, 0), h - 1), y1c = min(max(wy0 + 1, 0), h - 1);
-      float w00 = I1[(i64)y0c * w + x0c], w01 = I1[(i64)y0c * w + x1c];
-      float w10 = I1[(i64)y1c * w + x0c], w11 = I1[(i64)y1c * w + x1c];
-      float i1v = w00 * (1 - ay) * (1 - ax) + w01 * (1 - ay) * ax +
-                  w10 * ay * (1 - ax) + w11 * ay * ax;
-      float it = i1v - i0v;
-      a11 += ix * ix;
-      a12 += ix * iy;
-      a22 += iy * iy;
-      b1 += ix * it;
-      b2 += iy * it;
+#pragma unroll
+      for (int dx = -RADIUS; dx <= RADIUS; ++dx) {
+        int lx = (int)threadIdx.x + HALO + dx;
+        float ix =
+            0.5f * (lds_i0[ly * LW + lx + 1] - lds_i0[ly * LW + lx - 1]);
+        float iy = 0.5f *
+                   (lds_i0[(ly + 1) * LW + lx] - lds_i0[(ly - 1) * LW + lx]);
+        int j = dx + RADIUS;
+        float top = r0[j] + ax * (r0[j + 1] - r0[j]);
+        float bot = r1[j] + ax * (r1[j + 1] - r1[j]);
+        float it = top + ay * (bot - top) - lds_i0[ly * LW + lx];
+        a11 += ix * ix;
+        a12 += ix * iy;
+        a22 += iy * iy;
+        b1 += ix * it;
+        b2 += iy * it;
+      }
+#pragma unroll
+      for (int j = 0; j < RW; ++j) r0[j] = r1[j];
+    }
+  } else {
+#pragma unroll
+    for (int dy = -RADIUS; dy <= RADIUS; ++dy) {
+#pragma unroll
+      for (int dx = -RADIUS; dx <= RADIUS; ++dx) {
+        int lx = (int)threadIdx.x + HALO + dx;
+        int ly = (int)threadIdx.y + HALO + dy;
+        float ix =
+            0.5f * (lds_i0[ly * LW + lx + 1] - lds_i0[ly * LW + lx - 1]);
+        float iy = 0.5f *
+                   (lds_i0[(ly + 1) * LW + lx] - lds_i0[(ly - 1) * LW + lx]);
+        float i0v = lds_i0[ly * LW + lx];
+        int x0c = min(max(xi + dx, 0), w - 1);
+        int x1c = min(max(xi + dx + 1, 0), w - 1);
+        int y0c = min(max(yi + dy, 0), h - 1);
+        int y1c = min(max(yi + dy + 1, 0), h - 1);
+        float w00 = I1[(i64)y0c * w + x0c], w01 = I1[(i64)y0c * w + x1c];
+        float w10 = I1[(i64)y1c * w + x0c], w11 = I1[(i64)y1c * w + x1c];
+        float i1v = w00 * (1 - ay) * (1 - ax) + w01 * (1 - ay) * ax +
+                    w10 * ay * (1 - ax) + w11 * ay * ax;
+        float it = i1v - i0v;
+        a11 += ix * ix;
+        a12 += ix * iy;
+        a22 += iy * iy;
+        b1 += ix * it;
+        b2 += iy * it;
+      }
     }
   }
   float det = a11 * a22 - a12 * a12;

@@ -1,0 +1,27 @@
+#!/usr/bin/env python3
+"""GEMM micro-benchmark at the DNN ops' real shapes (run on GPU via
+gpurun). Prints ms and TFLOP/s per shape."""
+import sys
+sys.path.insert(0, ".")
+from scanner_amd import _core
+
+# (label, M, N, K): ResNet-50 @ batch 16 and Pose @ batch 8 hot shapes
+SHAPES = [
+    ("rn.conv1   ", 16 * 112 * 112, 64, 192),
+    ("rn.l1.c2   ", 16 * 56 * 56, 64, 576),
+    ("rn.l1.c3   ", 16 * 56 * 56, 256, 64),
+    ("rn.l2.c2   ", 16 * 28 * 28, 128, 1152),
+    ("rn.l3.c2   ", 16 * 14 * 14, 256, 2304),
+    ("rn.l4.c2   ", 16 * 7 * 7, 512, 4608),
+    ("pose.b2    ", 8 * 184 * 184, 64, 576),
+    ("pose.b4    ", 8 * 92 * 92, 128, 1152),
+    ("pose.stage ", 8 * 46 * 46, 128, 1728),
+    ("square4k   ", 4096, 4096, 4096),
+    ("square8k   ", 8192, 8192, 8192),
+]
+
+for label, M, N, K in SHAPES:
+    ms = _core.gemm_bench(M, N, K, 20, True)
+    tf = 2.0 * M * N * K / (ms * 1e-3) / 1e12
+    print(f"{label} M={M:<8} N={N:<5} K={K:<5} {ms:8.3f} ms  {tf:7.1f} TF/s",
+          flush=True)
