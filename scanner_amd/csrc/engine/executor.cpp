@@ -414,13 +414,68 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t) {
         if (!d2h_dst.empty()) {
           memcpy_vec(d2h_dst, CPU_DEVICE, d2h_src, gpu_src, d2h_sz);
         }
+        // Sink-side codec annotations (OpColumn.compress_video; parity:
+        // reference compressed-output columns + PostEvaluateWorker encode,
+        // evaluate_worker.cpp:1329-1560).
+        std::map<std::string, std::string> compress;
+        if (!op.args.empty()) {
+          auto a = mp::decode(op.args);
+          auto& am = a.as_map();
+          auto cit = am.find("compress");
+          if (cit != am.end()) {
+            for (auto& kv : cit->second.as_map())
+              compress[kv.first] = kv.second.as_str();
+          }
+        }
         for (size_t c = 0; c < op.inputs.size(); ++c) {
+          const std::string& cname = table.columns[c].name;
+          auto cmp = compress.find(cname);
+          if (cmp != compress.end() &&
+              table.columns[c].type == ColumnType::Video) {
+            SCA_CHECK(cmp->second == "svc",
+                      "unknown sink codec '" + cmp->second + "'");
+            i32 h = 0, w = 0, ch = 0;
+            size_t fsize = 0;
+            std::vector<u8> contig;
+            i64 nf = 0;
+            for (auto& e : cols[c]) {
+              SCA_CHECK(!e.is_null,
+                        "cannot codec-compress a column with null rows");
+              SCA_CHECK(e.is_frame && e.frame_info.type == FrameType::U8,
+                        "svc compression needs u8 frames");
+              if (nf == 0) {
+                h = e.frame_info.shape[0];
+                w = e.frame_info.shape[1];
+                ch = e.frame_info.shape[2];
+                fsize = e.size;
+                contig.reserve(fsize * cols[c].size());
+              }
+              SCA_CHECK(e.size == fsize,
+                        "svc compression needs uniform frame sizes");
+              contig.insert(contig.end(), e.buffer, e.buffer + e.size);
+              ++nf;
+            }
+            VideoMetadata vm;
+            vm.width = w;
+            vm.height = h;
+            vm.channels = ch;
+            vm.frame_type = FrameType::U8;
+            vm.num_frames = nf;
+            std::vector<u8> stream;
+            svc_encode_cpu(contig.data(), nf, h, w, ch, /*gop=*/16, stream,
+                           vm);
+            write_video_item(*db_, table, cname, t.task, stream, vm);
+            for (size_t k = 0; k < cols[c].size(); ++k) {
+              if (owned[c][k] && cols[c][k].buffer)
+                delete_buffer(CPU_DEVICE, cols[c][k].buffer);
+            }
+            continue;
+          }
           write_column_item(*db_, table, table.columns[c].name, t.task,
                             cols[c]);
           if (table.columns[c].type == ColumnType::Video) {
             // Raw-stored frame column: record geometry so readers can
-            // reconstruct frames. (Codec-compressed sinks go through the
-            // encode stage instead.)
+            // reconstruct frames.
             VideoMetadata vm;
             vm.codec = "raw";
             vm.num_frames = (i64)cols[c].size();

@@ -37,6 +37,11 @@ class IOGenerator:
                 raise ScannerException(
                     f"duplicate output column name '{c.name}'")
             seen.add(c.name)
-        op = Op(self._client, "Output", cols, output_columns=[])
+        # sink-side codec annotations (OpColumn.compress_video / lossless)
+        compress = {c.name: c.op._compress[c.name].get("codec", "svc")
+                    for c in cols if c.name in c.op._compress}
+        op = Op(self._client, "Output", cols,
+                args={"compress": compress} if compress else None,
+                output_columns=[])
         op._streams = list(streams)
         return op
