@@ -9,6 +9,7 @@
 #include <cerrno>
 #include <cstdio>
 #include <cstring>
+#include <thread>
 
 namespace sca {
 
@@ -42,16 +43,40 @@ class PosixStorage : public StorageBackend {
                   u8* out) override {
     int fd = ::open(path.c_str(), O_RDONLY);
     if (fd < 0) throw ScannerError("open failed: " + path + ": " + strerror(errno));
-    u64 off = 0;
-    while (off < size) {
-      ssize_t n = ::pread(fd, out + off, size - off, offset + off);
-      if (n <= 0) {
-        ::close(fd);
-        throw ScannerError("pread failed: " + path);
+    // One pread from page cache is a single-thread memcpy (~10 GB/s); the
+    // decode stage eats multi-GB video spans per step, so fan large reads
+    // out over a few threads (the reference reads items on a thread pool
+    // too — load_worker + storehouse).
+    constexpr u64 kParallelMin = 16 << 20;
+    constexpr u64 kChunk = 8 << 20;
+    bool fail = false;
+    auto read_span = [&](u64 lo, u64 hi) {
+      u64 off = lo;
+      while (off < hi) {
+        ssize_t n = ::pread(fd, out + off, hi - off, offset + off);
+        if (n <= 0) {
+          fail = true;
+          return;
+        }
+        off += n;
       }
-      off += n;
+    };
+    if (size < kParallelMin) {
+      read_span(0, size);
+    } else {
+      int nthreads = (int)std::min<u64>(8, (size + kChunk - 1) / kChunk);
+      u64 per = (size + nthreads - 1) / nthreads;
+      std::vector<std::thread> ts;
+      for (int i = 1; i < nthreads; ++i) {
+        u64 lo = i * per;
+        if (lo >= size) break;
+        ts.emplace_back(read_span, lo, std::min(size, lo + per));
+      }
+      read_span(0, std::min(size, per));
+      for (auto& th : ts) th.join();
     }
     ::close(fd);
+    if (fail) throw ScannerError("pread failed: " + path);
   }
 
   u64 file_size(const std::string& path) override {

@@ -158,6 +158,171 @@ __global__ void __launch_bounds__(256, 2)
   }
 }
 
+// Small-K variant (K = 64: the DNN ops' 1x1 convolutions out of 64-channel
+// activations). A single K-step leaves the main kernel with no prefetch
+// overlap — every A tile is a cold HBM read the MFMAs must wait on
+// (measured 69 TF/s at M=50176,N=256,K=64). Here a workgroup walks
+// MULTIPLE M-tiles of one N-strip, double-buffering the A staging across
+// tiles: tile t+1's global_load_lds runs while tile t's MFMAs and stores
+// execute. B ([BN][64], 16 KB) is staged once and reused for the whole
+// walk. K=64 keeps the 128-byte LDS row layout the lane-contiguous
+// LDS-DMA scatter requires.
+template <int BM, int BN, int WM, int WN, bool RELU, bool RESIDUAL>
+__global__ void __launch_bounds__(256, 2)
+    gemm_bf16_smallk_kernel(const bf16* __restrict__ A,
+                            const bf16* __restrict__ B, bf16* __restrict__ C,
+                            int M, int N, const float* __restrict__ scale,
+                            const float* __restrict__ bias,
+                            const bf16* __restrict__ residual,
+                            int tiles_per_wg) {
+  constexpr int K = 64;
+  constexpr int FM = BM / WM / 16;
+  constexpr int FN = BN / WN / 16;
+  __shared__ bf16 lds_b[BN * K];
+  __shared__ bf16 lds_a[2 * BM * K];
+
+  int tid = threadIdx.x;
+  int lane = tid & 63;
+  int wave = tid >> 6;
+  int wrow = wave / WN;
+  int wcol = wave % WN;
+
+  int ntiles_n = (N + BN - 1) / BN;
+  int ntiles_m = (M + BM - 1) / BM;
+  int strip = blockIdx.x / ntiles_n;
+  int n0 = (blockIdx.x % ntiles_n) * BN;
+  int mt0 = strip * tiles_per_wg;
+  int mt_end = min(mt0 + tiles_per_wg, ntiles_m);
+  if (mt0 >= ntiles_m) return;
+
+  const int lrow = lane >> 3;
+  const int lk = (lane & 7) * 8;
+  constexpr int B_INSTRS = BN / 8 / 4;
+  constexpr int A_INSTRS = BM / 8 / 4;
+
+#pragma unroll
+  for (int i = 0; i < B_INSTRS; ++i) {
+    int row = (wave * B_INSTRS + i) * 8 + lrow;
+    const bf16* src = B + (size_t)(n0 + row) * K + lk;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) uint32_t*)src,
+        (__attribute__((address_space(3)))
+             uint32_t*)(lds_b + (size_t)(wave * B_INSTRS + i) * 8 * K),
+        16, 0, 0);
+  }
+
+  auto stage_a = [&](int buf, int mt) {
+    int m0 = mt * BM;
+    bf16* dst = lds_a + buf * BM * K;
+#pragma unroll
+    for (int i = 0; i < A_INSTRS; ++i) {
+      int row = (wave * A_INSTRS + i) * 8 + lrow;
+      int grow = m0 + row;
+      if (grow >= M) grow = M - 1;
+      const bf16* src = A + (size_t)grow * K + lk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3)))
+               uint32_t*)(dst + (size_t)(wave * A_INSTRS + i) * 8 * K),
+          16, 0, 0);
+    }
+  };
+
+  stage_a(0, mt0);
+  for (int mt = mt0; mt < mt_end; ++mt) {
+    int buf = (mt - mt0) & 1;
+    __builtin_amdgcn_s_waitcnt(0);
+    __syncthreads();
+    if (mt + 1 < mt_end) stage_a(buf ^ 1, mt + 1);
+
+    f32x4 acc[FM][FN];
+#pragma unroll
+    for (int i = 0; i < FM; ++i)
+#pragma unroll
+      for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    const bf16* la = lds_a + buf * BM * K;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      int kbase = kk * 32 + (lane >> 4) * 8;
+      bf16x8 afrag[FM], bfrag[FN];
+#pragma unroll
+      for (int i = 0; i < FM; ++i) {
+        int row = wrow * (BM / WM) + i * 16 + (lane & 15);
+        afrag[i] =
+            *reinterpret_cast<const bf16x8*>(la + (size_t)row * K + kbase);
+      }
+#pragma unroll
+      for (int j = 0; j < FN; ++j) {
+        int col = wcol * (BN / WN) + j * 16 + (lane & 15);
+        bfrag[j] = *reinterpret_cast<const bf16x8*>(lds_b +
+                                                    (size_t)col * K + kbase);
+      }
+#pragma unroll
+      for (int i = 0; i < FM; ++i)
+#pragma unroll
+        for (int j = 0; j < FN; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    }
+
+    int m0 = mt * BM;
+#pragma unroll
+    for (int i = 0; i < FM; ++i) {
+#pragma unroll
+      for (int j = 0; j < FN; ++j) {
+        int col = n0 + wcol * (BN / WN) + j * 16 + (lane & 15);
+        float sc = scale ? scale[col] : 1.f;
+        float bi = bias ? bias[col] : 0.f;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = m0 + wrow * (BM / WM) + i * 16 + (lane >> 4) * 4 + r;
+          if (row >= M) continue;
+          float v = acc[i][j][r] * sc + bi;
+          if constexpr (RESIDUAL) {
+            v += bf16_to_f32(residual[(size_t)row * N + col]);
+          }
+          if constexpr (RELU) v = v > 0.f ? v : 0.f;
+          C[(size_t)row * N + col] = f32_to_bf16(v);
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
+void launch_smallk(const GemmArgs& g, hipStream_t s) {
+  constexpr int BM = 128, BN = 128;
+  int ntiles_m = (g.M + BM - 1) / BM;
+  int ntiles_n = (g.N + BN - 1) / BN;
+  // Enough strips to fill the chip (>=2048 workgroups when the shape
+  // allows), each walking a run of consecutive M-tiles.
+  int target_wgs = 2048;
+  int strips = std::max(1, std::min(ntiles_m, target_wgs / ntiles_n));
+  int tiles_per_wg = (ntiles_m + strips - 1) / strips;
+  strips = (ntiles_m + tiles_per_wg - 1) / tiles_per_wg;
+  int grid = strips * ntiles_n;
+  auto disp = [&](auto relu, auto res) {
+    gemm_bf16_smallk_kernel<BM, BN, 2, 2, decltype(relu)::value,
+                            decltype(res)::value><<<grid, 256, 0, s>>>(
+        (const bf16*)g.A, (const bf16*)g.B, (bf16*)g.C, g.M, g.N, g.scale,
+        g.bias, (const bf16*)g.residual, tiles_per_wg);
+  };
+  if (g.relu && g.residual)
+    disp(std::true_type{}, std::true_type{});
+  else if (g.relu)
+    disp(std::true_type{}, std::false_type{});
+  else if (g.residual)
+    disp(std::false_type{}, std::true_type{});
+  else
+    disp(std::false_type{}, std::false_type{});
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess) {
+    throw ScannerError(std::string("gemm smallk launch failed: ") +
+                       hipGetErrorString(e));
+  }
+}
+
 template <int BM, int BN, int WM, int WN>
 void launch_variant(const GemmArgs& g, hipStream_t s) {
   int grid = ((g.M + BM - 1) / BM) * (g.N / BN);
@@ -188,7 +353,9 @@ void gemm_bf16(const GemmArgs& g, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   SCA_CHECK(g.K % 64 == 0, "gemm K must be a multiple of 64");
   SCA_CHECK(g.N % 64 == 0, "gemm N must be a multiple of 64");
-  if (g.N % 128 == 0 && g.M > 64) {
+  if (g.K == 64 && g.N % 128 == 0 && g.M >= 1024) {
+    launch_smallk(g, s);
+  } else if (g.N % 128 == 0 && g.M > 64) {
     launch_variant<128, 128, 2, 2>(g, s);
   } else {
     launch_variant<64, 64, 2, 2>(g, s);

@@ -24,8 +24,10 @@ def test_gemm_bf16_identity():
 def test_gemm_bf16_random():
     from scanner_amd import _core
     rng = np.random.RandomState(0)
+    # 2000x256x64 and 5000x128x64 take the small-K M-walking kernel
     for (M, N, K) in [(64, 64, 64), (128, 128, 128), (300, 256, 192),
-                      (1000, 512, 576), (16, 1024, 2048)]:
+                      (1000, 512, 576), (16, 1024, 2048), (2000, 256, 64),
+                      (5000, 128, 64)]:
         A = rng.randn(M, K).astype(np.float32)
         B = rng.randn(N, K).astype(np.float32)
         C = _core.gemm_bf16_test(A, B, False)

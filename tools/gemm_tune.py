@@ -16,6 +16,8 @@ SHAPES = [
     ("pose.b2    ", 8 * 184 * 184, 64, 576),
     ("pose.b4    ", 8 * 92 * 92, 128, 1152),
     ("pose.stage ", 8 * 46 * 46, 128, 1728),
+    ("smallk.c3  ", 16 * 56 * 56, 256, 64),   # small-K M-walk path
+    ("smallk.wide", 16 * 56 * 56, 128, 64),
     ("square4k   ", 4096, 4096, 4096),
     ("square8k   ", 8192, 8192, 8192),
 ]
