@@ -20,7 +20,16 @@ from .streams import PartitionerGenerator, StreamsGenerator
 
 class Client:
     def __init__(self, db_path=None, master=None, workers=None,
-                 start_cluster=True, recover=True):
+                 start_cluster=True, recover=True, config=None,
+                 config_path=None):
+        if config is None and (config_path is not None
+                               or db_path is None or master is None):
+            from .config import Config
+            config = Config(config_path=config_path, db_path=db_path,
+                            master=master)
+        if config is not None:
+            db_path = db_path or config.db_path
+            master = master or config.master_address
         self._db_path = db_path or os.path.join(
             tempfile.gettempdir(), "scanner_amd_db")
         os.makedirs(self._db_path, exist_ok=True)

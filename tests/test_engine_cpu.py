@@ -522,3 +522,15 @@ def test_compressed_output_column(sc):
         total = sum(os.path.getsize(os.path.join(tdir, f))
                     for f in os.listdir(tdir))
         assert total < frames.nbytes * limit, (total, frames.nbytes)
+
+
+def test_config_file(tmp_path, monkeypatch):
+    cfg = tmp_path / "scanner_amd.toml"
+    cfg.write_text(
+        '[storage]\ntype = "posix"\ndb_path = "%s"\n' % (tmp_path / "cdb"))
+    monkeypatch.setenv("SCANNER_AMD_CONFIG", str(cfg))
+    sc2 = sp.Client(start_cluster=False)
+    assert sc2._db_path == str(tmp_path / "cdb")
+    # ops still work against the configured db
+    tab = sc2.new_table("cfg_t", ["col"], [[b"x"]])
+    assert sc2.has_table("cfg_t")
