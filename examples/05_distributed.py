@@ -31,8 +31,8 @@ def main():
         print("distributed job OK across", len(workers), "workers")
     finally:
         for w in workers:
-            w.stop()
-        master.stop()
+            w.shutdown()
+        master.shutdown()
 
 
 if __name__ == "__main__":
