@@ -1,11 +1,17 @@
 #include "memory.h"
 
+#include <atomic>
+
 #include <algorithm>
 #include <cstdlib>
 
 #include "hip_util.h"
 
 namespace sca {
+
+namespace {
+std::atomic<u64> g_mem_generation{0};
+}  // namespace
 
 // ---------------- SystemAllocator ----------------
 
@@ -271,9 +277,29 @@ void init_memory_allocators(const MemoryConfig& cfg) {
     g_mem.gpus.emplace(id, std::move(da));
   }
   g_mem.initialized = true;
+  g_mem_generation.fetch_add(1);
+}
+
+namespace {
+std::mutex g_teardown_mu;
+std::vector<std::function<void()>>& teardown_callbacks() {
+  static std::vector<std::function<void()>> cbs;
+  return cbs;
+}
+}  // namespace
+
+void register_memory_teardown_callback(std::function<void()> cb) {
+  std::lock_guard<std::mutex> l(g_teardown_mu);
+  teardown_callbacks().push_back(std::move(cb));
 }
 
 void destroy_memory_allocators() {
+  // Let caches drop allocator-backed buffers while allocators still exist
+  // (outside g_mem.mu: the callbacks call delete_buffer).
+  {
+    std::lock_guard<std::mutex> l(g_teardown_mu);
+    for (auto& cb : teardown_callbacks()) cb();
+  }
   std::lock_guard<std::mutex> l(g_mem.mu);
   g_mem.gpus.clear();
   g_mem.cpu = DeviceAllocators{};
@@ -281,6 +307,8 @@ void destroy_memory_allocators() {
 }
 
 bool memory_initialized() { return g_mem.initialized; }
+
+u64 memory_generation() { return g_mem_generation.load(); }
 
 u8* new_buffer(DeviceHandle device, size_t size) {
   // Plain buffers are block buffers with one ref — uniform delete path.

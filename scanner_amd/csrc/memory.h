@@ -15,6 +15,7 @@
 //     reference's memcpy_vec.
 #pragma once
 
+#include <functional>
 #include <map>
 #include <mutex>
 #include <unordered_map>
@@ -116,6 +117,14 @@ struct MemoryConfig {
 void init_memory_allocators(const MemoryConfig& cfg);
 void destroy_memory_allocators();
 bool memory_initialized();
+// Increments on every init_memory_allocators; buffers allocated under an
+// older generation must not be freed into the current allocators.
+u64 memory_generation();
+
+// Callbacks run at the START of destroy_memory_allocators() so caches
+// holding allocator-backed buffers (e.g. the DNN model cache) can release
+// them while the allocators still exist. Survives re-init.
+void register_memory_teardown_callback(std::function<void()> cb);
 
 u8* new_buffer(DeviceHandle device, size_t size);
 u8* new_block_buffer(DeviceHandle device, size_t size, i32 refs);

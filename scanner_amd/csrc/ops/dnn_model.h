@@ -92,7 +92,12 @@ struct DeviceModel {
   std::vector<Entry> convs;
   std::map<std::string, int> by_name;
   float* mean = nullptr;  // 3 floats mean + 3 std (imagenet normalization)
+  u64 generation = 0;     // memory_generation() at build time
   ~DeviceModel() {
+    // Free only into the allocator generation that produced these buffers
+    // (skip at interpreter exit after teardown, and after a
+    // destroy+re-init cycle where the pointers would be stale).
+    if (!memory_initialized() || generation != memory_generation()) return;
     if (weights) delete_buffer(dev, weights);
     if (scalebias) delete_buffer(dev, scalebias);
     if (mean) delete_buffer(dev, (u8*)mean);
@@ -110,6 +115,7 @@ inline std::shared_ptr<DeviceModel> build_device_model(
   }
   auto model = std::make_shared<DeviceModel>();
   model->dev = dev;
+  model->generation = memory_generation();
   size_t w_elems = 0, sb_elems = 0;
   for (const auto& sp : specs) {
     model->by_name[sp.name] = (int)model->convs.size();
@@ -153,13 +159,24 @@ inline std::shared_ptr<DeviceModel> build_device_model(
   return model;
 }
 
-// Global cache keyed by (model name, device, weights file, seed).
+// Global cache keyed by (model name, device, weights file, seed). The
+// cache registers a memory-teardown callback so destroy_memory_allocators
+// frees the cached device models while the allocators still exist (and a
+// later re-init rebuilds models instead of using stale slab pointers).
 inline std::shared_ptr<DeviceModel> get_model(
     const std::string& model_name, DeviceHandle dev,
     const std::string& weights_file, u64 seed,
     const std::function<std::shared_ptr<DeviceModel>()>& build) {
   static std::mutex mu;
   static std::map<std::string, std::shared_ptr<DeviceModel>> cache;
+  static bool registered = []() {
+    register_memory_teardown_callback([]() {
+      std::lock_guard<std::mutex> l(mu);
+      cache.clear();
+    });
+    return true;
+  }();
+  (void)registered;
   std::string key = model_name + "|" + dev.to_string() + "|" + weights_file +
                     "|" + std::to_string(seed);
   std::lock_guard<std::mutex> l(mu);
