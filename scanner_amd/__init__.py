@@ -12,11 +12,13 @@ from .client import Client
 from .job import Job
 from .op import Kernel, register_python_op
 from .storage import NamedStream, NamedVideoStream
+from . import parallel, types
 
 __version__ = "0.1.0"
 
 __all__ = [
     "CacheMode", "Client", "ColumnType", "DeviceType", "FrameType", "Job",
     "Kernel", "NamedStream", "NamedVideoStream", "PerfParams",
-    "ScannerException", "default_machine_params", "register_python_op",
+    "ScannerException", "default_machine_params", "parallel",
+    "register_python_op", "types",
 ]

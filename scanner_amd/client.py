@@ -259,6 +259,18 @@ class Client:
         self._last_profilers = ex.profilers()
         return Profile(self._last_profilers)
 
+    def batch_load(self, streams, fn=None, workers=8, rows=None):
+        """Load several streams' rows in parallel (parity: Client.batch_load
+        client.py:1270-1281 — thread-pooled column fetch). Returns a list of
+        per-stream row lists, in input order."""
+        from concurrent.futures import ThreadPoolExecutor
+
+        def one(s):
+            return list(s.load(fn=fn, rows=rows))
+
+        with ThreadPoolExecutor(max_workers=workers) as pool:
+            return list(pool.map(one, streams))
+
     def profile(self):
         return Profile(self._last_profilers or [])
 

@@ -534,3 +534,13 @@ def test_config_file(tmp_path, monkeypatch):
     # ops still work against the configured db
     tab = sc2.new_table("cfg_t", ["col"], [[b"x"]])
     assert sc2.has_table("cfg_t")
+
+
+def test_batch_load(sc):
+    vids = [make_video(n=6, seed=s) for s in range(3)]
+    streams = [sp.NamedVideoStream(sc, f"bl{i}", frames=v, codec="raw")
+               for i, v in enumerate(vids)]
+    loaded = sc.batch_load(streams)
+    assert len(loaded) == 3
+    for i, rows in enumerate(loaded):
+        np.testing.assert_array_equal(np.stack(rows), vids[i])
