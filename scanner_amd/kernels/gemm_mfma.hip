@@ -59,15 +59,17 @@ __device__ inline void epilogue_store(f32x4 (&acc)[FM][FN], float* scratch,
       }
     }
     asm volatile("s_waitcnt lgkmcnt(0)");  // writes visible to own reads
+    constexpr int CHUNK = COLS / 4;        // contiguous cols per lane (>=8)
+    static_assert(CHUNK >= 8 && CHUNK % 8 == 0, "wave tile too narrow");
     int r2 = lane >> 2;                    // 16 rows
-    int q = lane & 3;                      // 4 x 16-col chunks
+    int q = lane & 3;                      // 4 chunks per row
     int row = m0 + wrow * (FM * 16) + i * 16 + r2;
     if (row < M) {
-      int colbase = n0 + wcol * COLS + q * 16;
-      const float* src = scratch + r2 * STRIDE + q * 16;
-      bf16 out[16];
+      int colbase = n0 + wcol * COLS + q * CHUNK;
+      const float* src = scratch + r2 * STRIDE + q * CHUNK;
+      bf16 out[CHUNK];
 #pragma unroll
-      for (int k = 0; k < 16; ++k) {
+      for (int k = 0; k < CHUNK; ++k) {
         float v = src[k];
         if constexpr (RESIDUAL) {
           v += bf16_to_f32(residual[(size_t)row * N + colbase + k]);
@@ -75,11 +77,12 @@ __device__ inline void epilogue_store(f32x4 (&acc)[FM][FN], float* scratch,
         if constexpr (RELU) v = v > 0.f ? v : 0.f;
         out[k] = f32_to_bf16(v);
       }
-      // two 16-byte stores per lane, contiguous across the 4 q-lanes
-      *reinterpret_cast<bf16x8*>(C + (size_t)row * N + colbase) =
-          *reinterpret_cast<const bf16x8*>(out);
-      *reinterpret_cast<bf16x8*>(C + (size_t)row * N + colbase + 8) =
-          *reinterpret_cast<const bf16x8*>(out + 8);
+      // 16-byte stores, contiguous across the 4 q-lanes of each row
+#pragma unroll
+      for (int c8 = 0; c8 < CHUNK; c8 += 8) {
+        *reinterpret_cast<bf16x8*>(C + (size_t)row * N + colbase + c8) =
+            *reinterpret_cast<const bf16x8*>(out + c8);
+      }
     }
   }
 }
