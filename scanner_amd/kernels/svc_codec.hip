@@ -148,9 +148,13 @@ std::vector<Element> svc_decode_gpu(const u8* stream_host, size_t size,
   u64 hi = vm.sample_offsets[span.back()] + vm.sample_sizes[span.back()];
   SCA_CHECK(lo >= stream_offset && hi <= stream_offset + size,
             "svc stream range does not cover decode span");
+  // Async upload on the decode stream: the decode kernels queue right
+  // behind it (same stream), so the CPU thread never blocks here and the
+  // DMA overlaps other pipeline instances' compute. stream_host is pinned
+  // (CPU pool uses hipHostMalloc) and outlives the end-of-function sync.
   u8* d_stream = new_buffer(dev, hi - lo);
-  memcpy_buffer(d_stream, dev, stream_host + (lo - stream_offset),
-                CPU_DEVICE, hi - lo);
+  SVC_CHECK(hipMemcpyAsync(d_stream, stream_host + (lo - stream_offset),
+                           hi - lo, hipMemcpyHostToDevice, s));
 
   u8* scratch[2] = {nullptr, nullptr};
   auto get_scratch = [&](int i) {
