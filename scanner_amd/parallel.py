@@ -105,16 +105,17 @@ def gather_column(blobs, device, dst=0, group=None,
         maxn = max(int(s.item()) for s in sizes)
         pad = torch.zeros(maxn, dtype=torch.uint8, device=device)
         pad[:t.numel()] = t
+        # all_gather rather than gather: supported uniformly on RCCL and
+        # gloo (plain gather is not implemented by the nccl backend), and
+        # on the xGMI full mesh the extra fan-out is one hop per link.
+        outs = [torch.empty(maxn, dtype=torch.uint8, device=device)
+                for _ in range(world)]
+        dist.all_gather(outs, pad, group=group)
         if rank == dst:
-            outs = [torch.empty(maxn, dtype=torch.uint8, device=device)
-                    for _ in range(world)]
-            dist.gather(pad, outs, dst=dst, group=group)
             for i in range(world):
                 n = int(sizes[i].item())
                 if n:
                     received[i] += outs[i][:n].cpu().numpy().tobytes()
-        else:
-            dist.gather(pad, None, dst=dst, group=group)
     if rank != dst:
         return None
     result = []

@@ -69,3 +69,38 @@ def test_parallel_gloo_world2(tmp_path):
     for rank, (p, out) in enumerate(zip(procs, outs)):
         assert p.returncode == 0, f"rank {rank} failed:\n{out}"
         assert f"rank {rank} OK" in out
+
+
+def test_bench_cpu_world2(tmp_path):
+    """bench.py under the driver's multi-rank contract, CPU/gloo: two
+    ranks, tiny step counts; rank 0 must print one JSON line with the
+    whole-job aggregate."""
+    import json
+    procs = []
+    outs = []
+    for rank in range(2):
+        env = dict(os.environ)
+        env.update({
+            "PYTHONPATH": REPO + os.pathsep + env.get("PYTHONPATH", ""),
+            "RANK": str(rank),
+            "LOCAL_RANK": str(rank),
+            "WORLD_SIZE": "2",
+            "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": "29617",
+        })
+        procs.append(subprocess.Popen(
+            [sys.executable, os.path.join(REPO, "bench.py"), "--pipeline",
+             "hist", "--steps", "1", "--warmup", "0", "--frames-per-step",
+             "8"],
+            env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE))
+    for p in procs:
+        out, err = p.communicate(timeout=300)
+        outs.append((p.returncode, out.decode(), err.decode()))
+    for rank, (rc, out, err) in enumerate(outs):
+        assert rc == 0, f"rank {rank}: {err[-2000:]}"
+    result = json.loads(outs[0][1].strip().splitlines()[-1])
+    assert result["n_gpus"] == 2
+    assert result["config"]["parallelism"] == "frame-shard dp2"
+    assert result["value"] > 0
+    # only rank 0 prints the result line (gloo chatter may appear)
+    assert not any(ln.startswith("{") for ln in outs[1][1].splitlines())
