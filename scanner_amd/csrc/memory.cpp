@@ -371,6 +371,18 @@ void memcpy_vec(const std::vector<u8*>& dests, DeviceHandle dest_device,
 
 void* per_thread_hip_stream() { return (void*)thread_copy_stream(); }
 
+// Small per-thread pool of auxiliary streams for intra-op parallelism
+// (e.g. GOP-parallel video decode chains). Lazily created, never freed.
+void* per_thread_aux_stream(int i) {
+  constexpr int kAux = 4;
+  thread_local hipStream_t aux[kAux] = {nullptr, nullptr, nullptr, nullptr};
+  int k = ((i % kAux) + kAux) % kAux;
+  if (aux[k] == nullptr) {
+    HIP_CHECK(hipStreamCreateWithFlags(&aux[k], hipStreamNonBlocking));
+  }
+  return (void*)aux[k];
+}
+
 void sync_per_thread_stream() {
   HIP_CHECK(hipStreamSynchronize(thread_copy_stream()));
 }

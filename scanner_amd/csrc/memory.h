@@ -149,5 +149,8 @@ size_t gpu_pool_bytes_in_use(i32 gpu_id);
 // ordering within an instance is by construction.
 void* per_thread_hip_stream();
 void sync_per_thread_stream();
+// Auxiliary per-thread streams (i mod 4) for intra-op parallel chains
+// (GOP-parallel decode); fork/join with events against the main stream.
+void* per_thread_aux_stream(int i);
 
 }  // namespace sca

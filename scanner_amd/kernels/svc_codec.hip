@@ -141,35 +141,53 @@ std::vector<Element> svc_decode_gpu(const u8* stream_host, size_t size,
   std::vector<Element> out;
   if (span.empty()) return out;
 
-  // Upload the byte range covering the span (span frames are contiguous
-  // per GOP but may have gaps; upload the covering range — packets are
-  // adjacent so over-read is bounded by skipped GOPs).
+  // Upload the byte range covering the span asynchronously on the main
+  // stream; decode chains fork off it via an event. stream_host is pinned
+  // (CPU pool = hipHostMalloc) and outlives the end-of-function sync.
   u64 lo = vm.sample_offsets[span.front()];
   u64 hi = vm.sample_offsets[span.back()] + vm.sample_sizes[span.back()];
   SCA_CHECK(lo >= stream_offset && hi <= stream_offset + size,
             "svc stream range does not cover decode span");
-  // Async upload on the decode stream: the decode kernels queue right
-  // behind it (same stream), so the CPU thread never blocks here and the
-  // DMA overlaps other pipeline instances' compute. stream_host is pinned
-  // (CPU pool uses hipHostMalloc) and outlives the end-of-function sync.
   u8* d_stream = new_buffer(dev, hi - lo);
   SVC_CHECK(hipMemcpyAsync(d_stream, stream_host + (lo - stream_offset),
                            hi - lo, hipMemcpyHostToDevice, s));
+  hipEvent_t up_ev;
+  SVC_CHECK(hipEventCreateWithFlags(&up_ev, hipEventDisableTiming));
+  SVC_CHECK(hipEventRecord(up_ev, s));
 
-  u8* scratch[2] = {nullptr, nullptr};
-  auto get_scratch = [&](int i) {
-    if (!scratch[i]) scratch[i] = new_buffer(dev, nbytes);
-    return scratch[i];
-  };
-
+  // Frames inside a GOP are a serial prediction chain, but GOPs are
+  // independent: each keyframe starts a new chain on one of 4 auxiliary
+  // streams (fork/join with events), so a work packet's GOPs decode
+  // concurrently. (The reference's decoder automaton is serial per item —
+  // decoder_automata.cpp; GOP concurrency is the MI355X-native upgrade.)
+  constexpr int kChains = 4;
+  hipStream_t chain_stream[kChains];
+  bool chain_used[kChains] = {false, false, false, false};
+  std::vector<u8*> scratches;
   size_t wi = 0;
+  int chain = -1;
+  hipStream_t cs = s;
   u8* prev = nullptr;
+  u8* chain_scratch[2] = {nullptr, nullptr};
   int flip = 0;
   for (i64 f : span) {
     const u8* pkt_h = stream_host + (vm.sample_offsets[f] - stream_offset);
     SvcPacketView v = svc_parse_packet(pkt_h, vm.sample_sizes[f]);
     SCA_CHECK(v.nbytes == nbytes, "svc frame size mismatch");
-    // device pointers into d_stream at the same relative offsets
+    if (v.is_key) {
+      // new chain
+      chain = (chain + 1);
+      int ci = chain % kChains;
+      cs = (hipStream_t)per_thread_aux_stream(ci);
+      if (!chain_used[ci]) {
+        chain_used[ci] = true;
+        chain_stream[ci] = cs;
+        SVC_CHECK(hipStreamWaitEvent(cs, up_ev, 0));
+      }
+      prev = nullptr;
+      chain_scratch[0] = chain_scratch[1] = nullptr;
+      flip = 0;
+    }
     u64 pkt_off = vm.sample_offsets[f] - lo;
     const u8* pkt_d = d_stream + pkt_off;
     const u32* super_off_d = reinterpret_cast<const u32*>(pkt_d + 20);
@@ -191,10 +209,14 @@ std::vector<Element> svc_decode_gpu(const u8* stream_host, size_t size,
       e.index = f;
       cur = e.buffer;
     } else {
-      cur = get_scratch(flip);
+      if (!chain_scratch[flip]) {
+        chain_scratch[flip] = new_buffer(dev, nbytes);
+        scratches.push_back(chain_scratch[flip]);
+      }
+      cur = chain_scratch[flip];
     }
     u32 blocks = v.nsuper;
-    svc_decode_frame_kernel<<<blocks, 128, 0, s>>>(
+    svc_decode_frame_kernel<<<blocks, 128, 0, cs>>>(
         widths_d, super_off_d, packed_d, prev, v.is_key, nbytes, v.ngroups,
         cur);
     SVC_CHECK(hipGetLastError());
@@ -205,12 +227,20 @@ std::vector<Element> svc_decode_gpu(const u8* stream_host, size_t size,
     prev = cur;
     flip ^= 1;
   }
-  // Frames decode async on the stream; the stream buffer and scratch may
-  // not be released until the launches complete.
+  // Join: main stream waits every used chain, then sync before returning
+  // scratch + stream buffers to the shared pool.
+  for (int ci = 0; ci < kChains; ++ci) {
+    if (!chain_used[ci]) continue;
+    hipEvent_t ev;
+    SVC_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+    SVC_CHECK(hipEventRecord(ev, chain_stream[ci]));
+    SVC_CHECK(hipStreamWaitEvent(s, ev, 0));
+    SVC_CHECK(hipEventDestroy(ev));
+  }
+  SVC_CHECK(hipEventDestroy(up_ev));
   SVC_CHECK(hipStreamSynchronize(s));
   delete_buffer(dev, d_stream);
-  if (scratch[0]) delete_buffer(dev, scratch[0]);
-  if (scratch[1]) delete_buffer(dev, scratch[1]);
+  for (u8* sc : scratches) delete_buffer(dev, sc);
   SCA_CHECK(wi == want.size(), "svc gpu decode: not all frames produced");
   return out;
 }
