@@ -72,10 +72,15 @@ class PerfParams:
     @classmethod
     def estimate(cls, total_rows=None, element_size=None, n_gpus=None,
                  **kw):
-        """Heuristic auto-tuner (parity: PerfParams.estimate common.py:149).
+        """Heuristic auto-tuner (parity: PerfParams.estimate common.py:149,
+        which probes GPUtil/psutil).
 
         Sizes packets so a task's working set stays within a fraction of
-        memory, with sane defaults for video frames.
+        memory, and — when a GPU is visible and the caller didn't choose —
+        sizes the GPU pool from the device's actual free HBM so steady-state
+        allocation never touches the driver (288 GB per MI355X: default to
+        an 1/8 slab per pipeline-instance set, leaving room for 8 ranks per
+        node).
         """
         element_size = element_size or (1920 * 1080 * 3)
         # keep one task's decoded frames under ~2 GB
@@ -84,6 +89,15 @@ class PerfParams:
         if total_rows is not None:
             io = min(io, max(1, int(math.ceil(total_rows / 4))))
             work = min(work, io)
+        if "gpu_pool" not in kw:
+            try:
+                from . import _core
+                if _core.have_gpu():
+                    free = _core.gpu_free_memory(0)
+                    kw["gpu_pool"] = min(free // 2, 32 << 30)
+                    kw.setdefault("cpu_pool", 4 << 30)
+            except Exception:
+                pass
         return cls(work_packet_size=work, io_packet_size=io, **kw)
 
     def to_dict(self, n_instances=1):

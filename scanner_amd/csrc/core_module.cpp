@@ -95,6 +95,17 @@ PYBIND11_MODULE(_core, m) {
 
   m.def("have_gpu", &have_gpu);
   m.def("gpu_device_count", &gpu_device_count);
+  m.def("gpu_free_memory", [](int device) -> size_t {
+    SCA_CHECK(have_gpu(), "gpu_free_memory needs a GPU");
+    int prev = 0;
+    (void)hipGetDevice(&prev);
+    (void)hipSetDevice(device);
+    size_t free_b = 0, total_b = 0;
+    hipError_t e = hipMemGetInfo(&free_b, &total_b);
+    (void)hipSetDevice(prev);
+    SCA_CHECK(e == hipSuccess, "hipMemGetInfo failed");
+    return free_b;
+  });
 
   // GEMM micro-benchmark: device-resident buffers, hipEvent timing.
   // Returns ms per iteration (tools/gemm_tune.py computes TFLOP/s).

@@ -544,3 +544,36 @@ def test_batch_load(sc):
     assert len(loaded) == 3
     for i, rows in enumerate(loaded):
         np.testing.assert_array_equal(np.stack(rows), vids[i])
+
+
+def test_stencil_clamps_at_slice_boundary(sc):
+    """Stencil windows must clamp at slice-group bounds (REPEAT_EDGE,
+    reference dag_analysis.cpp:1450-1469): the last row of a group sees
+    (last, last), not the first frame of the next group."""
+    frames = np.concatenate([make_textured_pair(dx=3, dy=0, seed=7)[0:1]] * 4
+                            + [make_textured_pair(dx=3, dy=0, seed=8)[1:2]]
+                            * 4)
+    # frames 0-3 identical, frames 4-7 identical, 0->4 is a translation
+    video = sp.NamedVideoStream(sc, "slb", frames=frames, codec="raw")
+
+    def flow_mags(sliced):
+        frame = sc.io.Input([video])
+        col = sc.streams.Slice(frame, sc.partitioner.all(4)) if sliced \
+            else frame
+        flow = sc.ops.OpticalFlow(frame=col)
+        stats = sc.ops.FlowStats(flow=flow)
+        col_out = sc.streams.Unslice(stats) if sliced else stats
+        out = sp.NamedStream(sc, f"slb_out_{sliced}")
+        sc.run(sc.io.Output(col_out, [out]), sp.PerfParams.manual(4, 8),
+               cache_mode=sp.CacheMode.Overwrite)
+        return [np.frombuffer(b, np.float32)[0] for b in out.load()]
+
+    unsliced = flow_mags(False)
+    sliced = flow_mags(True)
+    assert len(unsliced) == len(sliced) == 8
+    # row 3: unsliced pairs (3,4) = translation -> mean|u| near 3;
+    # sliced clamps to (3,3) within group -> ~0 flow
+    assert unsliced[3] > 0.5, unsliced
+    assert sliced[3] < 0.2, sliced
+    # interior rows agree
+    assert abs(unsliced[1] - sliced[1]) < 1e-4
