@@ -241,6 +241,8 @@ class Client:
             # reload metadata: output tables were created/committed by the
             # master process (shared storage)
             self._db = _core.Database(self._db_path)
+            if result is not None:
+                self._last_profilers = result._profilers
             return result
 
         # local execution

@@ -78,6 +78,7 @@ class MasterServer:
             "RegisterOp": self._register_op,
             "PokeWatchdog": self._poke,
             "Shutdown": self._shutdown_rpc,
+            "CollectProfiles": self._collect_profiles,
         })
         self.port = self._server.port
         self.addr = f"127.0.0.1:{self.port}"
@@ -91,6 +92,20 @@ class MasterServer:
 
     def _ping(self, req):
         return {"ok": True}
+
+    def _collect_profiles(self, req):
+        """Ask every live worker to write its profiler file for a job
+        (reference: per-node profile_<node>.bin after a job)."""
+        job_id = req["job_id"]
+        with self._lock:
+            workers = [w for w in self._workers.values() if w.alive]
+        written = 0
+        for w in workers:
+            r = w.client.try_call("WriteProfile", {"job_id": job_id},
+                                  timeout=15)
+            if r and r.get("written"):
+                written += 1
+        return {"written": written}
 
     def _poke(self, req):
         return {"ok": True}
@@ -348,8 +363,24 @@ class ClusterClient:
                     raise ScannerException(
                         "streams failed (blacklisted after repeated task "
                         f"failures): {st['blacklisted_streams']}")
-                return None
+                return self._load_profiles(jid)
             time.sleep(0.1)
+
+    def _load_profiles(self, jid):
+        """Gather per-worker profiler files written via CollectProfiles
+        (parity: Profile parsing jobs/<id>/profile_<node>.bin)."""
+        import glob
+        import os
+
+        from .profiler import Profile
+        self._rpc.try_call("CollectProfiles", {"job_id": jid}, timeout=30)
+        profs = []
+        pat = os.path.join(self._client._db_path, "jobs", str(jid),
+                           "profile_*.bin")
+        for path in sorted(glob.glob(pat)):
+            with open(path, "rb") as f:
+                profs.extend(msgpack.unpackb(f.read(), raw=False))
+        return Profile(profs)
 
     def shutdown(self):
         self._heartbeat_stop.set()

@@ -35,6 +35,7 @@ class WorkerServer:
             "Ping": self._ping,
             "Shutdown": self._shutdown_rpc,
             "PokeWatchdog": self._poke,
+            "WriteProfile": self._write_profile,
         })
         self.port = self._server.port
         self.addr = f"127.0.0.1:{self.port}"
@@ -67,6 +68,22 @@ class WorkerServer:
     def _shutdown_rpc(self, req):
         threading.Thread(target=self.shutdown, daemon=True).start()
         return {"ok": True}
+
+    def _write_profile(self, req):
+        """Serialize this worker's pipeline profilers for the given job to
+        shared storage (parity: write_profiler_to_file profiler.h:82 —
+        jobs/<id>/profile_<node>.bin, parsed by scannerpy Profile)."""
+        job_id = req["job_id"]
+        with self._job_lock:
+            if self._job_id != job_id or self._executor is None:
+                return {"written": False}
+            profs = self._executor.profilers()
+        d = os.path.join(self._db_path, "jobs", str(job_id))
+        os.makedirs(d, exist_ok=True)
+        path = os.path.join(d, f"profile_{self.worker_id}.bin")
+        with open(path, "wb") as f:
+            f.write(msgpack.packb(profs))
+        return {"written": True}
 
     # ---- job setup ----
 

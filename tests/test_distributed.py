@@ -183,3 +183,31 @@ def test_late_worker_join(tmp_path):
         if "w" in worker_holder:
             worker_holder["w"].shutdown()
         master.shutdown()
+
+
+def test_distributed_profile_collection(tmp_path):
+    """Workers write per-node profiler files; client gets a merged Profile
+    (parity: jobs/<id>/profile_<node>.bin + scannerpy Profile)."""
+    db = _mk_db(tmp_path)
+    master = MasterServer(db)
+    workers = [start_worker(master.addr, db) for _ in range(2)]
+    try:
+        sc = sp.Client(db_path=db, master=master.addr)
+        frames = make_video(n=12)
+        video = sp.NamedVideoStream(sc, "dp", frames=frames, codec="raw")
+        frame = sc.io.Input([video])
+        hist = sc.ops.Histogram(frame=frame)
+        out = sp.NamedStream(sc, "dp_hist")
+        prof = sc.run(sc.io.Output(hist, [out]), sp.PerfParams.manual(2, 4),
+                      cache_mode=sp.CacheMode.Overwrite)
+        assert prof is not None
+        stats = prof.statistics()
+        assert any(k.startswith("op:Histogram") for k in stats), stats
+        trace = prof.write_trace(str(tmp_path / "trace.json"))
+        import json
+        with open(trace) as f:
+            assert len(json.load(f)["traceEvents"]) > 0
+    finally:
+        for w in workers:
+            w.shutdown()
+        master.shutdown()
