@@ -314,8 +314,22 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t) {
       const KernelFactory& kf = kernel_registry().get(op.name, op.device);
       inst.kernels[i] = kf.make(cfg);
       auto ait = job.op_args.find((i32)i);
-      inst.kernels[i]->new_stream(
-          ait == job.op_args.end() ? std::vector<u8>{} : ait->second);
+      const std::vector<u8>& sargs =
+          ait == job.op_args.end() ? std::vector<u8>{} : ait->second;
+      // fetch-once resources (weights etc.) — first fetch per op name
+      // process-wide, then every instance sets up (reference:
+      // evaluate_worker.cpp:493-550 worker-0 fetch + setup barrier).
+      {
+        static std::mutex fetch_mu;
+        static std::set<std::string> fetched;
+        std::lock_guard<std::mutex> fl(fetch_mu);
+        if (!fetched.count(op.name)) {
+          inst.kernels[i]->fetch_resources(op.args);
+          fetched.insert(op.name);
+        }
+      }
+      inst.kernels[i]->setup_with_resources(op.args);
+      inst.kernels[i]->new_stream(sargs);
     }
     inst.kernels_job = t.job;
   }

@@ -117,6 +117,14 @@ class PythonKernel : public BaseKernel {
     }
   }
 
+  void setup_with_resources(const std::vector<u8>& args) override {
+    py::gil_scoped_acquire gil;
+    if (py::hasattr(obj_, "setup_with_resources")) {
+      obj_.attr("setup_with_resources")(
+          py::bytes(reinterpret_cast<const char*>(args.data()), args.size()));
+    }
+  }
+
   void execute(const StenciledElements& input,
                BatchedElements& output) override {
     py::gil_scoped_acquire gil;

@@ -240,6 +240,9 @@ class Kernel:
     def fetch_resources(self, args):
         pass
 
+    def setup_with_resources(self, args):
+        pass
+
     def execute(self, *cols):
         raise NotImplementedError
 
@@ -280,6 +283,17 @@ class _PyKernelAdapter:
     def reset(self):
         if self._obj is not None:
             self._obj.reset()
+
+    def fetch_resources(self, args_bytes):
+        if self._obj is not None:
+            args = msgpack.unpackb(args_bytes) if args_bytes else {}
+            self._obj.fetch_resources(args)
+
+    def setup_with_resources(self, args_bytes):
+        if self._obj is not None and hasattr(self._obj,
+                                             "setup_with_resources"):
+            args = msgpack.unpackb(args_bytes) if args_bytes else {}
+            self._obj.setup_with_resources(args)
 
     def execute(self, cols):
         n_rows = len(cols[0]) if cols else 0

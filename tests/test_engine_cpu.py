@@ -577,3 +577,25 @@ def test_stencil_clamps_at_slice_boundary(sc):
     assert sliced[3] < 0.2, sliced
     # interior rows agree
     assert abs(unsliced[1] - sliced[1]) < 1e-4
+
+
+def test_overlapping_slices(sc):
+    """Slice with overlapping range partitions: each group processes its
+    full range independently (parity: reference overlapping-slice test
+    py_test.py:350-406)."""
+    n = 12
+    tab = sc.new_table("ovl", ["col"],
+                       [[int(i).to_bytes(8, "little")] for i in range(n)])
+    col = sc.io.Input([tab])
+    sliced = sc.streams.Slice(col, sc.partitioner.ranges([(0, 8), (4, 12)]))
+    inc = sc.ops.TestIncrement(ignore=sliced)
+    unsliced = sc.streams.Unslice(inc)
+    out = sp.NamedStream(sc, "ovl_out")
+    sc.run(sc.io.Output(unsliced, [out]), sp.PerfParams.manual(4, 16),
+           cache_mode=sp.CacheMode.Overwrite)
+    vals = [int.from_bytes(b, "little") for b in out.load()]
+    # group 0: rows 0-7 -> value row + k+1 (k = offset in group);
+    # group 1: rows 4-11 -> same pattern; unslice concatenates both groups
+    expect = [i + (i % 8) + 1 for i in range(8)] + \
+             [4 + i + (i % 8) + 1 for i in range(8)]
+    assert vals == expect, vals

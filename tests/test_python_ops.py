@@ -116,3 +116,36 @@ def test_python_op_stateful_with_stream_args(sc):
            cache_mode=sp.CacheMode.Overwrite)
     vals = [int.from_bytes(b, "little") for b in out.load()]
     assert vals == [100 + i + 1 for i in range(n)]
+
+
+@register_python_op()
+class ResourceKernel(sp.Kernel):
+    """fetch_resources runs once per op process-wide; setup runs per
+    instance (parity: reference fetch_resources test py_test.py:587-622)."""
+    FETCHES = []
+    SETUPS = []
+
+    def fetch_resources(self, args):
+        ResourceKernel.FETCHES.append(1)
+
+    def setup_with_resources(self, args):
+        ResourceKernel.SETUPS.append(1)
+
+    def execute(self, col: bytes) -> bytes:
+        assert len(ResourceKernel.FETCHES) >= 1
+        assert len(ResourceKernel.SETUPS) >= 1
+        return col
+
+
+def test_python_op_fetch_resources(sc):
+    n = 6
+    tab = sc.new_table("fr", ["col"],
+                       [[int(i).to_bytes(4, "little")] for i in range(n)])
+    col = sc.io.Input([tab])
+    rk = sc.ops.ResourceKernel(col=col)
+    out = sp.NamedStream(sc, "fr_out")
+    sc.run(sc.io.Output(rk, [out]), sp.PerfParams.manual(2, 4),
+           cache_mode=sp.CacheMode.Overwrite, pipeline_instances=2)
+    assert len(list(out.load())) == n
+    assert len(ResourceKernel.FETCHES) == 1      # fetched exactly once
+    assert len(ResourceKernel.SETUPS) >= 1       # per instance
