@@ -121,7 +121,11 @@ class PoseKernelGPU : public BatchedKernel {
       oh = (h + 2 * sp.pad - sp.r) / sp.stride + 1;
       ow = (w + 2 * sp.pad - sp.s) / sp.stride + 1;
       const u8* A = x;
-      if (!(sp.r == 1 && sp.s == 1 && sp.stride == 1 && sp.pad == 0)) {
+      bool direct = sp.r == 1 && sp.s == 1 && sp.stride == 1 && sp.pad == 0;
+      // Implicit GEMM when c % 8 == 0 (everything except b1's c=3): the
+      // GEMM stages im2col rows straight from the activations.
+      bool implicit = !direct && sp.in_c % 8 == 0;
+      if (!direct && !implicit) {
         im2col_bf16(x, n, h, w, sp.in_c, sp.r, sp.s, sp.stride, sp.pad,
                     colbuf, oh, ow, sp.kp(), s);
         A = colbuf;
@@ -136,7 +140,12 @@ class PoseKernelGPU : public BatchedKernel {
       g.scale = (const float*)model_->scalebias + e.sb_off;
       g.bias = (const float*)model_->scalebias + e.sb_off + sp.np();
       g.relu = sp.relu;
-      gemm_bf16(g, s);
+      if (implicit) {
+        ConvDesc d{n, h, w, sp.in_c, sp.r, sp.s, sp.stride, sp.pad, oh, ow};
+        conv_gemm_bf16(g, d, s);
+      } else {
+        gemm_bf16(g, s);
+      }
     };
 
     // ---- backbone ----

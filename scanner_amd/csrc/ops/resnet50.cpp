@@ -130,7 +130,11 @@ void ResNet50KernelGPU::execute_batch(const BatchedElements& in,
     ow = (w + 2 * sp.pad - sp.s) / sp.stride + 1;
     int M = n * oh * ow;
     const u8* A = x;
-    if (!(sp.r == 1 && sp.s == 1 && sp.stride == 1 && sp.pad == 0)) {
+    bool direct = sp.r == 1 && sp.s == 1 && sp.stride == 1 && sp.pad == 0;
+    // Implicit GEMM for spatial convs with c % 8 == 0: the GEMM stages
+    // im2col rows straight from the activation tensor (no HBM round trip).
+    bool implicit = !direct && sp.in_c % 8 == 0;
+    if (!direct && !implicit) {
       im2col_bf16(x, n, h, w, sp.in_c, sp.r, sp.s, sp.stride, sp.pad,
                   colbuf, oh, ow, sp.kp(), s);
       A = colbuf;
@@ -146,7 +150,12 @@ void ResNet50KernelGPU::execute_batch(const BatchedElements& in,
     g.bias = (const float*)model_->scalebias + e.sb_off + sp.np();
     g.residual = residual;
     g.relu = sp.relu;
-    gemm_bf16(g, s);
+    if (implicit) {
+      ConvDesc d{n, h, w, sp.in_c, sp.r, sp.s, sp.stride, sp.pad, oh, ow};
+      conv_gemm_bf16(g, d, s);
+    } else {
+      gemm_bf16(g, s);
+    }
     return sp.np();
   };
 

@@ -20,6 +20,19 @@ struct GemmArgs {
 
 void gemm_bf16(const GemmArgs& g, void* stream);
 
+// Implicit-GEMM convolution: the GEMM's A operand is gathered straight
+// from the NHWC activation tensor during LDS staging (no im2col buffer in
+// HBM). g.A = activations (NHWC bf16); g.M = n*oh*ow; g.K = kp (padded
+// r*s*c). Requires c % 8 == 0 (each 16-byte staged chunk must stay within
+// one (r,s) cell's channel run).
+struct ConvDesc {
+  int n, h, w, c;       // input NHWC
+  int r, s, stride, pad;
+  int oh, ow;
+};
+
+void conv_gemm_bf16(const GemmArgs& g, const ConvDesc& d, void* stream);
+
 // u8 HWC frames (device pointer array) -> normalized bf16 NHWC out_hw x
 // out_hw x 3, bilinear resize. mean/std per channel (device, 3 floats).
 void preprocess_frames_bf16(const void* frames_ptr_array, int n, int in_h,

@@ -10,6 +10,9 @@
 // K and N must be multiples of 64 (callers pad); M is arbitrary.
 #include <hip/hip_runtime.h>
 
+#include <map>
+#include <mutex>
+
 #include "../csrc/memory.h"
 #include "dnn.h"
 
@@ -89,13 +92,19 @@ __device__ inline void epilogue_store(f32x4 (&acc)[FM][FN], float* scratch,
 
 // One workgroup = 256 threads = 4 waves in a WM x WN grid; each wave owns
 // a (BM/WM) x (BN/WN) output sub-tile as FM x FN fragments of 16x16.
-template <int BM, int BN, int WM, int WN, bool RELU, bool RESIDUAL>
+// When IMPLICIT, A is the NHWC activation tensor and the A staging
+// gathers im2col rows on the fly (ConvDesc d gives the geometry; zero
+// points at 16+ zero bytes for padding / k-pad lanes). Lane-addressed
+// global_load_lds makes the gather free of any HBM im2col round trip.
+template <int BM, int BN, int WM, int WN, bool RELU, bool RESIDUAL,
+          bool IMPLICIT>
 __global__ void __launch_bounds__(256, 2)
     gemm_bf16_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                      bf16* __restrict__ C, int M, int N, int K,
                      const float* __restrict__ scale,
                      const float* __restrict__ bias,
-                     const bf16* __restrict__ residual) {
+                     const bf16* __restrict__ residual, ConvDesc d,
+                     const bf16* __restrict__ zero) {
   constexpr int BK = 64;
   constexpr int FM = BM / WM / 16;
   constexpr int FN = BN / WN / 16;
@@ -141,7 +150,28 @@ __global__ void __launch_bounds__(256, 2)
       int row = (wave * A_INSTRS + i) * 8 + lrow;
       int grow = m0 + row;
       if (grow >= M) grow = M - 1;  // clamp: garbage rows masked at store
-      const bf16* src = A + (size_t)grow * K + k0 + lk;
+      const bf16* src;
+      if constexpr (IMPLICIT) {
+        int k = k0 + lk;
+        int cell = k / d.c;
+        int cj = k - cell * d.c;
+        if (cell >= d.r * d.s) {
+          src = zero;
+        } else {
+          int dr = cell / d.s, ds = cell - dr * d.s;
+          int q = grow % d.ow;
+          int t = grow / d.ow;
+          int p = t % d.oh;
+          int nn = t / d.oh;
+          int hh = p * d.stride - d.pad + dr;
+          int ww = q * d.stride - d.pad + ds;
+          src = (hh >= 0 && hh < d.h && ww >= 0 && ww < d.w)
+                    ? A + ((((i64)nn * d.h + hh) * d.w + ww) * d.c + cj)
+                    : zero;
+        }
+      } else {
+        src = A + (size_t)grow * K + k0 + lk;
+      }
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) uint32_t*)src,
           (__attribute__((address_space(3)))
@@ -372,9 +402,9 @@ void launch_variant(const GemmArgs& g, hipStream_t s) {
   int grid = ((g.M + BM - 1) / BM) * (g.N / BN);
   auto disp = [&](auto relu, auto res) {
     gemm_bf16_kernel<BM, BN, WM, WN, decltype(relu)::value,
-                     decltype(res)::value><<<grid, 256, 0, s>>>(
+                     decltype(res)::value, false><<<grid, 256, 0, s>>>(
         (const bf16*)g.A, (const bf16*)g.B, (bf16*)g.C, g.M, g.N, g.K,
-        g.scale, g.bias, (const bf16*)g.residual);
+        g.scale, g.bias, (const bf16*)g.residual, ConvDesc{}, nullptr);
   };
   if (g.relu && g.residual)
     disp(std::true_type{}, std::true_type{});
@@ -391,7 +421,65 @@ void launch_variant(const GemmArgs& g, hipStream_t s) {
   }
 }
 
+// 16+ zero bytes in device memory for implicit-conv padding lanes; one
+// per device, allocated outside the pool so memory teardown cycles never
+// invalidate it.
+const bf16* device_zero_chunk() {
+  static std::mutex mu;
+  static std::map<int, bf16*> per_dev;
+  int dev = 0;
+  (void)hipGetDevice(&dev);
+  std::lock_guard<std::mutex> l(mu);
+  auto it = per_dev.find(dev);
+  if (it != per_dev.end()) return it->second;
+  void* p = nullptr;
+  hipError_t e = hipMalloc(&p, 256);
+  if (e != hipSuccess || hipMemset(p, 0, 256) != hipSuccess) {
+    throw ScannerError("failed to allocate implicit-conv zero chunk");
+  }
+  per_dev[dev] = (bf16*)p;
+  return (bf16*)p;
+}
+
+template <int BM, int BN, int WM, int WN>
+void launch_conv_variant(const GemmArgs& g, const ConvDesc& d,
+                         hipStream_t s) {
+  int grid = ((g.M + BM - 1) / BM) * (g.N / BN);
+  const bf16* zero = device_zero_chunk();
+  auto disp = [&](auto relu, auto res) {
+    gemm_bf16_kernel<BM, BN, WM, WN, decltype(relu)::value,
+                     decltype(res)::value, true><<<grid, 256, 0, s>>>(
+        (const bf16*)g.A, (const bf16*)g.B, (bf16*)g.C, g.M, g.N, g.K,
+        g.scale, g.bias, (const bf16*)g.residual, d, zero);
+  };
+  if (g.relu && g.residual)
+    disp(std::true_type{}, std::true_type{});
+  else if (g.relu)
+    disp(std::true_type{}, std::false_type{});
+  else if (g.residual)
+    disp(std::false_type{}, std::true_type{});
+  else
+    disp(std::false_type{}, std::false_type{});
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess) {
+    throw ScannerError(std::string("conv gemm launch failed: ") +
+                       hipGetErrorString(e));
+  }
+}
+
 }  // namespace
+
+void conv_gemm_bf16(const GemmArgs& g, const ConvDesc& d, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  SCA_CHECK(g.K % 64 == 0 && g.N % 64 == 0, "conv gemm K/N must be x64");
+  SCA_CHECK(d.c % 8 == 0, "implicit conv needs c % 8 == 0");
+  SCA_CHECK(g.M == d.n * d.oh * d.ow, "conv gemm M mismatch");
+  if (g.N % 128 == 0) {
+    launch_conv_variant<128, 128, 2, 2>(g, d, s);
+  } else {
+    launch_conv_variant<64, 64, 2, 2>(g, d, s);
+  }
+}
 
 void gemm_bf16(const GemmArgs& g, void* stream) {
   hipStream_t s = (hipStream_t)stream;
