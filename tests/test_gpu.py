@@ -155,3 +155,48 @@ def test_gpu_color_ops_match_cpu(sc):
         assert cpu.shape == gpu.shape, name
         diff = np.abs(cpu.astype(int) - gpu.astype(int))
         assert diff.max() <= 1, (name, diff.max())
+
+
+def test_gpu_to_cpu_device_crossing(sc):
+    """GPU Resize feeding the CPU PNG encoder exercises the automatic
+    kernel-group D2H transfer (reference: copy_or_ref_elements,
+    evaluate_worker.cpp:798-806)."""
+    from test_engine_cpu import decode_png_filter0
+    frames = make_video(n=4, h=64, w=96)
+    video = sp.NamedVideoStream(sc, "g_x", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    small = sc.ops.Resize(frame=frame, width=48, height=32,
+                          device=sp.DeviceType.GPU)
+    png = sc.ops.ImageEncoder(frame=small, format="png")  # CPU op
+    out = sp.NamedStream(sc, "g_x_out")
+    sc.run(sc.io.Output(png, [out]), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite, gpu_ids=[0])
+    rows = list(out.load())
+    assert len(rows) == 4
+    img = decode_png_filter0(rows[0])
+    assert img.shape == (32, 48, 3)
+
+
+def test_gpu_compressed_output(sc):
+    """GPU pipeline with an SVC-compressed output column (sink-side
+    encode of device-produced frames)."""
+    from conftest import make_smooth_video
+    frames = make_smooth_video(n=24, h=48, w=64)
+    video = sp.NamedVideoStream(sc, "g_cmp", frames=frames, codec="svc")
+    frame = sc.io.Input([video])
+    blur = sc.ops.Blur(frame=frame, kernel_size=3, device=sp.DeviceType.GPU)
+    blur.compress_video()
+    out = sp.NamedStream(sc, "g_cmp_out")
+    sc.run(sc.io.Output(blur, [out]), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite, gpu_ids=[0])
+    got = np.stack(list(sp.NamedVideoStream(sc, "g_cmp_out").load()))
+    assert got.shape == (24, 48, 64, 3)
+
+    # CPU blur reference: identical box blur math
+    frame = sc.io.Input([video])
+    blur2 = sc.ops.Blur(frame=frame, kernel_size=3)
+    out2 = sp.NamedStream(sc, "g_cmp_ref")
+    sc.run(sc.io.Output(blur2, [out2]), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite)
+    ref = np.stack(list(sp.NamedVideoStream(sc, "g_cmp_ref").load()))
+    assert np.abs(got.astype(int) - ref.astype(int)).max() <= 1
