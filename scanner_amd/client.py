@@ -261,6 +261,13 @@ class Client:
         self._last_profilers = ex.profilers()
         return Profile(self._last_profilers)
 
+    def load_op(self, path):
+        """Load a user C++/HIP op plugin .so built with tools/build_op.py
+        (parity: Client.load_op client.py:514 — REGISTER_OP static
+        registrars run at load time)."""
+        _core.load_op_library(path)
+        self._op_info_cache.clear()
+
     def batch_load(self, streams, fn=None, workers=8, rows=None):
         """Load several streams' rows in parallel (parity: Client.batch_load
         client.py:1270-1281 — thread-pooled column fetch). Returns a list of

@@ -2,6 +2,8 @@
 // module (parity role: scanner/engine/python.cpp). Job specs travel as
 // msgpack bytes (packed with the msgpack wheel on the Python side, decoded
 // by csrc/msgpack.h here).
+#include <dlfcn.h>
+
 #include <pybind11/numpy.h>
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
@@ -92,6 +94,18 @@ PYBIND11_MODULE(_core, m) {
   register_pose_op();
   register_color_gpu();
   register_image_encoder_op();
+
+  // Load a user op plugin .so built with tools/build_op.py (parity:
+  // Client.load_op / REGISTER_OP static registrars in user libraries,
+  // scannerpy client.py:514). The dlopen runs the plugin's SCA_REGISTER_*
+  // constructors against this module's registries.
+  m.def("load_op_library", [](const std::string& path) {
+    void* h = dlopen(path.c_str(), RTLD_NOW | RTLD_LOCAL);
+    if (!h) {
+      throw ScannerError(std::string("load_op_library failed: ") +
+                         dlerror());
+    }
+  });
 
   m.def("have_gpu", &have_gpu);
   m.def("gpu_device_count", &gpu_device_count);

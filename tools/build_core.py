@@ -50,7 +50,7 @@ def main():
     arch = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
     common = (f"-O3 -std=c++17 -fPIC -I{py_inc} {pybind_inc} "
               f"-I{SRC} -I/opt/rocm/include -D__HIP_PLATFORM_AMD__ "
-              f"-Wno-unused-result -fvisibility=hidden")
+              f"-Wno-unused-result")
     hipflags = f"--offload-arch={arch}"
 
     hip_srcs = [s for s in HIP_SOURCES
