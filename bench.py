@@ -117,7 +117,7 @@ def main():
         gpu_pool=(24 << 30) if have_gpu else 0,
         cpu_pool=(4 << 30) if have_gpu else 0)
     instances = int(os.environ.get("SCANNER_BENCH_INSTANCES",
-                                   "4" if have_gpu else "1"))
+                                   "6" if have_gpu else "1"))
 
     def one_step(tag):
         sink = build_pipeline(sc, sp, video, args.pipeline, device,

@@ -120,7 +120,7 @@ __global__ void __launch_bounds__(256)
 // tile of one pair's plane (blockIdx.z = pair); I0 tile + halo staged in
 // LDS. RADIUS is a compile-time template so the window loops fully unroll.
 template <int RADIUS>
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(256, 5)
     of_lk_kernel(const float* __restrict__ gray, i64 level_off, int h, int w,
                  const int* __restrict__ pair_f0,
                  const int* __restrict__ pair_f1,
@@ -189,11 +189,6 @@ __global__ void __launch_bounds__(256)
       const float* nrow = I1 + (i64)(yi + dy + 1) * w + (xi - RADIUS);
 #pragma unroll
       for (int j = 0; j < RW; ++j) r1[j] = nrow[j];
-      // separable bilinear: vertical lerp hoisted per row pair (8 fma)
-      // instead of per tap (the fractions are constant per center pixel)
-      float vert[RW];
-#pragma unroll
-      for (int j = 0; j < RW; ++j) vert[j] = r0[j] + ay * (r1[j] - r0[j]);
       int ly = (int)threadIdx.y + HALO + dy;
       int gy_ = (int)threadIdx.y + RADIUS + dy;
 #pragma unroll
@@ -203,8 +198,12 @@ __global__ void __launch_bounds__(256)
         float ix = lds_gx[gy_ * GW + gx_];
         float iy = lds_gy[gy_ * GW + gx_];
         int j = dx + RADIUS;
-        float it =
-            vert[j] + ax * (vert[j + 1] - vert[j]) - lds_i0[ly * LW + lx];
+        // per-tap 2D lerp: a hoisted vertical-lerp row array costs 8
+        // VGPRs and one wave of occupancy — latency, not VALU, is the
+        // bound here (PMC: ACTIVE 26%, WAIT 49%)
+        float top = r0[j] + ax * (r0[j + 1] - r0[j]);
+        float bot = r1[j] + ax * (r1[j + 1] - r1[j]);
+        float it = top + ay * (bot - top) - lds_i0[ly * LW + lx];
         a11 += ix * ix;
         a12 += ix * iy;
         a22 += iy * iy;
