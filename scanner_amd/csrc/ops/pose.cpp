@@ -230,7 +230,7 @@ void register_pose_op() {
   KernelFactory f;
   f.op_name = "Pose";
   f.device_type = DeviceType::GPU;
-  f.preferred_batch = 8;
+  f.preferred_batch = 16;
   f.make = [](const KernelConfig& c) -> std::unique_ptr<BaseKernel> {
     return std::make_unique<PoseKernelGPU>(c);
   };

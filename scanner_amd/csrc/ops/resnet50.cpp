@@ -251,7 +251,7 @@ void register_resnet50_op() {
   KernelFactory f;
   f.op_name = "ResNet50";
   f.device_type = DeviceType::GPU;
-  f.preferred_batch = 16;
+  f.preferred_batch = 32;
   f.make = [](const KernelConfig& c) -> std::unique_ptr<BaseKernel> {
     return std::make_unique<ResNet50KernelGPU>(c);
   };
