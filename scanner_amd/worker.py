@@ -188,10 +188,15 @@ def main():
     ap.add_argument("--db-path", required=True)
     ap.add_argument("--addr", default="127.0.0.1:0")
     ap.add_argument("--instances", type=int, default=1)
-    ap.add_argument("--gpu-ids", default="")
+    ap.add_argument("--gpu-ids", default=None,
+                    help="comma-separated; default = all visible GPUs "
+                         "(reference: default_machine_params)")
     ap.add_argument("--no-watchdog", action="store_true")
     args = ap.parse_args()
-    gpu_ids = [int(x) for x in args.gpu_ids.split(",") if x != ""]
+    if args.gpu_ids is None:
+        gpu_ids = list(range(_core.gpu_device_count()))
+    else:
+        gpu_ids = [int(x) for x in args.gpu_ids.split(",") if x != ""]
     start_worker(args.master, args.db_path, args.addr, block=True,
                  pipeline_instances=args.instances, gpu_ids=gpu_ids,
                  watchdog=not args.no_watchdog)
