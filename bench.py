@@ -49,10 +49,12 @@ def make_clip(n, h=H, w=W):
 def build_pipeline(sc, sp, video, pipeline, device, out_name):
     frame = sc.io.Input([video])
     cols = []
+    dnn_batch = int(os.environ.get("SCANNER_BENCH_BATCH", "0"))
     if pipeline in ("hist", "full"):
         cols.append(sc.ops.Histogram(frame=frame, device=device))
     if pipeline in ("resnet", "full"):
-        cols.append(sc.ops.ResNet50(frame=frame, device=device))
+        cols.append(sc.ops.ResNet50(frame=frame, device=device,
+                                    batch=dnn_batch))
     if pipeline == "flow":
         # BASELINE config 4: dense optical flow; per-frame flow summary is
         # the saved column (the 16 MB/frame flow field stays on-GPU).
@@ -113,7 +115,8 @@ def main():
     # Pools: steady-state allocation must never hit the driver (hipMalloc
     # synchronizes the device; hipHostMalloc is ~ms per call).
     perf = sp.PerfParams.manual(
-        work_packet_size=32, io_packet_size=64,
+        work_packet_size=int(os.environ.get("SCANNER_BENCH_WORK", "32")),
+        io_packet_size=int(os.environ.get("SCANNER_BENCH_IO", "64")),
         gpu_pool=(24 << 30) if have_gpu else 0,
         cpu_pool=(4 << 30) if have_gpu else 0)
     instances = int(os.environ.get("SCANNER_BENCH_INSTANCES",
