@@ -121,6 +121,9 @@ void ResNet50KernelGPU::execute_batch(const BatchedElements& in,
   u8* resid = new_buffer(dev, act_elems * 2);
   u8* colbuf = new_buffer(dev, im2col_elems * 2);
   u8* pre = new_buffer(dev, (size_t)n * 224 * 224 * 3 * 2);
+  // split-K f32 partials for the launch-bound stage-3/4 convs + fc
+  constexpr size_t kSplitkBytes = 64u << 20;
+  u8* skbuf = new_buffer(dev, kSplitkBytes);
 
   auto conv = [&](const char* name, const u8* x, int h, int w, u8* y,
                   const u8* residual, int& oh, int& ow) -> int {
@@ -150,6 +153,8 @@ void ResNet50KernelGPU::execute_batch(const BatchedElements& in,
     g.bias = (const float*)model_->scalebias + e.sb_off + sp.np();
     g.residual = residual;
     g.relu = sp.relu;
+    g.splitk_scratch = skbuf;
+    g.splitk_scratch_bytes = kSplitkBytes;
     if (implicit) {
       ConvDesc d{n, h, w, sp.in_c, sp.r, sp.s, sp.stride, sp.pad, oh, ow};
       conv_gemm_bf16(g, d, s);
@@ -211,6 +216,8 @@ void ResNet50KernelGPU::execute_batch(const BatchedElements& in,
     g.scale = (const float*)model_->scalebias + e.sb_off;
     g.bias = (const float*)model_->scalebias + e.sb_off + e.spec.np();
     g.relu = false;
+    g.splitk_scratch = skbuf;
+    g.splitk_scratch_bytes = kSplitkBytes;
     gemm_bf16(g, s);
   }
 
@@ -227,6 +234,7 @@ void ResNet50KernelGPU::execute_batch(const BatchedElements& in,
   delete_buffer(dev, resid);
   delete_buffer(dev, colbuf);
   delete_buffer(dev, pre);
+  delete_buffer(dev, skbuf);
 
   for (int i = 0; i < n; ++i) {
     Element e;

@@ -16,6 +16,13 @@ struct GemmArgs {
   const float* bias = nullptr;      // per-col
   const void* residual = nullptr;   // bf16 [M][N] added before activation
   bool relu = false;
+  // Optional split-K workspace (f32 partials). When set and the shape is
+  // launch-bound (few output tiles, deep K), gemm_bf16 splits K across
+  // workgroups and reduces. Caller guarantees the buffer lives until the
+  // stream syncs (the DNN ops pass a slice of their per-execute
+  // workspace).
+  void* splitk_scratch = nullptr;
+  size_t splitk_scratch_bytes = 0;
 };
 
 void gemm_bf16(const GemmArgs& g, void* stream);

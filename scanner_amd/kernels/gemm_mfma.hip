@@ -10,6 +10,7 @@
 // K and N must be multiples of 64 (callers pad); M is arbitrary.
 #include <hip/hip_runtime.h>
 
+#include <algorithm>
 #include <map>
 #include <mutex>
 
@@ -365,6 +366,229 @@ __global__ void __launch_bounds__(256, 2)
   }
 }
 
+// ---- split-K path ----
+// For launch-bound shapes (few output tiles, deep K — e.g. ResNet stage-4
+// conv2: M-tiles x N-tiles = 52 workgroups on a 256-CU chip, measured
+// ~59 TF/s), S workgroup groups each compute a K-slice partial into f32
+// scratch and a reduce kernel applies the epilogue. No inter-workgroup
+// hand-off inside a launch (guide correctness boundary G16) — the split
+// and reduce are separate kernels on the same stream.
+template <int BM, int BN, int WM, int WN, bool IMPLICIT>
+__global__ void __launch_bounds__(256, 2)
+    gemm_bf16_splitk_kernel(const bf16* __restrict__ A,
+                            const bf16* __restrict__ B, int M, int N, int K,
+                            int ksteps_per_split,
+                            float* __restrict__ partials, ConvDesc d,
+                            const bf16* __restrict__ zero) {
+  constexpr int BK = 64;
+  constexpr int FM = BM / WM / 16;
+  constexpr int FN = BN / WN / 16;
+  __shared__ bf16 lds[2 * (BM + BN) * BK];
+
+  int tid = threadIdx.x;
+  int lane = tid & 63;
+  int wave = tid >> 6;
+  int wrow = wave / WN;
+  int wcol = wave % WN;
+
+  int ntiles_n = (N + BN - 1) / BN;
+  int ntiles_m = (M + BM - 1) / BM;
+  int tiles = ntiles_m * ntiles_n;
+  int split = blockIdx.x / tiles;
+  int tile = blockIdx.x % tiles;
+  int m0 = (tile / ntiles_n) * BM;
+  int n0 = (tile % ntiles_n) * BN;
+  int ks_begin = split * ksteps_per_split;
+  int ks_end = min(ks_begin + ksteps_per_split, K / BK);
+  if (ks_begin >= ks_end) return;
+
+  f32x4 acc[FM][FN];
+#pragma unroll
+  for (int i = 0; i < FM; ++i)
+#pragma unroll
+    for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  constexpr int A_INSTRS = BM / 8 / 4;
+  constexpr int B_INSTRS = BN / 8 / 4;
+  auto stage = [&](int buf, int k0) {
+    const int lrow = lane >> 3;
+    const int lk = (lane & 7) * 8;
+    bf16* lds_a = lds + buf * (BM + BN) * BK;
+    bf16* lds_b = lds_a + BM * BK;
+#pragma unroll
+    for (int i = 0; i < A_INSTRS; ++i) {
+      int row = (wave * A_INSTRS + i) * 8 + lrow;
+      int grow = m0 + row;
+      if (grow >= M) grow = M - 1;
+      const bf16* src;
+      if constexpr (IMPLICIT) {
+        int k = k0 + lk;
+        int cell = k / d.c;
+        int cj = k - cell * d.c;
+        if (cell >= d.r * d.s) {
+          src = zero;
+        } else {
+          int dr = cell / d.s, ds = cell - dr * d.s;
+          int q = grow % d.ow;
+          int t = grow / d.ow;
+          int p = t % d.oh;
+          int nn = t / d.oh;
+          int hh = p * d.stride - d.pad + dr;
+          int ww = q * d.stride - d.pad + ds;
+          src = (hh >= 0 && hh < d.h && ww >= 0 && ww < d.w)
+                    ? A + ((((i64)nn * d.h + hh) * d.w + ww) * d.c + cj)
+                    : zero;
+        }
+      } else {
+        src = A + (size_t)grow * K + k0 + lk;
+      }
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3)))
+               uint32_t*)(lds_a + (size_t)(wave * A_INSTRS + i) * 8 * BK),
+          16, 0, 0);
+    }
+#pragma unroll
+    for (int i = 0; i < B_INSTRS; ++i) {
+      int row = (wave * B_INSTRS + i) * 8 + lrow;
+      const bf16* src = B + (size_t)(n0 + row) * K + k0 + lk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3)))
+               uint32_t*)(lds_b + (size_t)(wave * B_INSTRS + i) * 8 * BK),
+          16, 0, 0);
+    }
+  };
+
+  stage(0, ks_begin * BK);
+  for (int ks = ks_begin; ks < ks_end; ++ks) {
+    int buf = (ks - ks_begin) & 1;
+    __builtin_amdgcn_s_waitcnt(0);
+    __syncthreads();
+    if (ks + 1 < ks_end) stage(buf ^ 1, (ks + 1) * BK);
+
+    const bf16* lds_a = lds + buf * (BM + BN) * BK;
+    const bf16* lds_b = lds_a + BM * BK;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      int kbase = kk * 32 + (lane >> 4) * 8;
+      bf16x8 afrag[FM], bfrag[FN];
+#pragma unroll
+      for (int i = 0; i < FM; ++i) {
+        int row = wrow * (BM / WM) + i * 16 + (lane & 15);
+        afrag[i] = *reinterpret_cast<const bf16x8*>(
+            lds_a + (size_t)row * BK + kbase);
+      }
+#pragma unroll
+      for (int j = 0; j < FN; ++j) {
+        int col = wcol * (BN / WN) + j * 16 + (lane & 15);
+        bfrag[j] = *reinterpret_cast<const bf16x8*>(
+            lds_b + (size_t)col * BK + kbase);
+      }
+#pragma unroll
+      for (int i = 0; i < FM; ++i)
+#pragma unroll
+        for (int j = 0; j < FN; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // f32 partials: [split][M][N]; lanes 0-15 write 64-byte runs
+  float* out = partials + (size_t)split * M * N;
+#pragma unroll
+  for (int i = 0; i < FM; ++i) {
+#pragma unroll
+    for (int j = 0; j < FN; ++j) {
+      int col = n0 + wcol * (BN / WN) + j * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wrow * (BM / WM) + i * 16 + (lane >> 4) * 4 + r;
+        if (row < M) out[(size_t)row * N + col] = acc[i][j][r];
+      }
+    }
+  }
+}
+
+template <bool RELU, bool RESIDUAL>
+__global__ void __launch_bounds__(256)
+    splitk_reduce_kernel(const float* __restrict__ partials, int splits,
+                         i64 mn, int N, const float* __restrict__ scale,
+                         const float* __restrict__ bias,
+                         const bf16* __restrict__ residual,
+                         bf16* __restrict__ C) {
+  i64 gs = (i64)gridDim.x * blockDim.x;
+  for (i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x; i < mn; i += gs) {
+    float v = 0.f;
+    for (int sp = 0; sp < splits; ++sp) v += partials[(size_t)sp * mn + i];
+    int col = (int)(i % N);
+    if (scale) v *= scale[col];
+    if (bias) v += bias[col];
+    if constexpr (RESIDUAL) v += bf16_to_f32(residual[i]);
+    if constexpr (RELU) v = v > 0.f ? v : 0.f;
+    C[i] = f32_to_bf16(v);
+  }
+}
+
+const bf16* device_zero_chunk();
+
+bool try_splitk(const GemmArgs& g, hipStream_t s,
+                const ConvDesc* dd = nullptr) {
+  constexpr int BM = 128, BN = 128;
+  if (!g.splitk_scratch || g.N % BN != 0 || g.K < 1024) return false;
+  int ntiles_m = (g.M + BM - 1) / BM;
+  int tiles = ntiles_m * (g.N / BN);
+  if (tiles >= 256) return false;  // already fills the chip
+  int ksteps = g.K / 64;
+  int want = std::min({16, ksteps / 4, (2048 + tiles - 1) / tiles});
+  size_t per_split = (size_t)g.M * g.N * 4;
+  int fit = (int)(g.splitk_scratch_bytes / per_split);
+  int splits = std::min(want, fit);
+  if (splits < 2) return false;
+  int ksteps_per_split = (ksteps + splits - 1) / splits;
+  splits = (ksteps + ksteps_per_split - 1) / ksteps_per_split;
+  if (dd) {
+    gemm_bf16_splitk_kernel<BM, BN, 2, 2, true>
+        <<<tiles * splits, 256, 0, s>>>(
+            (const bf16*)g.A, (const bf16*)g.B, g.M, g.N, g.K,
+            ksteps_per_split, (float*)g.splitk_scratch, *dd,
+            device_zero_chunk());
+  } else {
+    gemm_bf16_splitk_kernel<BM, BN, 2, 2, false>
+        <<<tiles * splits, 256, 0, s>>>(
+            (const bf16*)g.A, (const bf16*)g.B, g.M, g.N, g.K,
+            ksteps_per_split, (float*)g.splitk_scratch, ConvDesc{}, nullptr);
+  }
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess) {
+    throw ScannerError(std::string("splitk launch failed: ") +
+                       hipGetErrorString(e));
+  }
+  i64 mn = (i64)g.M * g.N;
+  int grid = (int)std::min<i64>(4096, (mn + 255) / 256);
+  auto disp = [&](auto relu, auto res) {
+    splitk_reduce_kernel<decltype(relu)::value, decltype(res)::value>
+        <<<grid, 256, 0, s>>>((const float*)g.splitk_scratch, splits, mn,
+                              g.N, g.scale, g.bias,
+                              (const bf16*)g.residual, (bf16*)g.C);
+  };
+  if (g.relu && g.residual)
+    disp(std::true_type{}, std::true_type{});
+  else if (g.relu)
+    disp(std::true_type{}, std::false_type{});
+  else if (g.residual)
+    disp(std::false_type{}, std::true_type{});
+  else
+    disp(std::false_type{}, std::false_type{});
+  e = hipGetLastError();
+  if (e != hipSuccess) {
+    throw ScannerError(std::string("splitk reduce launch failed: ") +
+                       hipGetErrorString(e));
+  }
+  return true;
+}
+
 void launch_smallk(const GemmArgs& g, hipStream_t s) {
   constexpr int BM = 128, BN = 128;
   int ntiles_m = (g.M + BM - 1) / BM;
@@ -474,6 +698,9 @@ void conv_gemm_bf16(const GemmArgs& g, const ConvDesc& d, void* stream) {
   SCA_CHECK(g.K % 64 == 0 && g.N % 64 == 0, "conv gemm K/N must be x64");
   SCA_CHECK(d.c % 8 == 0, "implicit conv needs c % 8 == 0");
   SCA_CHECK(g.M == d.n * d.oh * d.ow, "conv gemm M mismatch");
+  if (try_splitk(g, s, &d)) {
+    return;
+  }
   if (g.N % 128 == 0) {
     launch_conv_variant<128, 128, 2, 2>(g, d, s);
   } else {
@@ -487,6 +714,8 @@ void gemm_bf16(const GemmArgs& g, void* stream) {
   SCA_CHECK(g.N % 64 == 0, "gemm N must be a multiple of 64");
   if (g.K == 64 && g.N % 128 == 0 && g.M >= 1024) {
     launch_smallk(g, s);
+  } else if (try_splitk(g, s)) {
+    // handled
   } else if (g.N % 128 == 0 && g.M > 64) {
     launch_variant<128, 128, 2, 2>(g, s);
   } else {
