@@ -130,7 +130,10 @@ PYBIND11_MODULE(_core, m) {
     u8* dB = new_buffer(dev, (size_t)N * K * 2);
     u8* dC = new_buffer(dev, (size_t)M * N * 2);
     u8* dS = new_buffer(dev, (size_t)N * 8);
+    u8* dSK = new_buffer(dev, 64u << 20);
     GemmArgs g;
+    g.splitk_scratch = dSK;
+    g.splitk_scratch_bytes = 64u << 20;
     g.A = dA;
     g.B = dB;
     g.C = dC;
@@ -158,6 +161,7 @@ PYBIND11_MODULE(_core, m) {
     delete_buffer(dev, dB);
     delete_buffer(dev, dC);
     delete_buffer(dev, dS);
+    delete_buffer(dev, dSK);
     return ms / iters;
   });
 

@@ -52,6 +52,30 @@ class NamedStream:
                 yield item
 
 
+def ingest_videos(sc, entries, codec="svc", io_packet_size=128,
+                  inplace=False):
+    """Batch-ingest videos; returns (streams, failures) where failures is
+    a list of (name, error) for inputs that could not be ingested (parity:
+    Client.ingest_videos + FailedVideo reporting, ingest.cpp:867).
+
+    entries: list of (name, frames-or-npy-path).
+    """
+    streams, failures = [], []
+    for name, src in entries:
+        try:
+            if isinstance(src, str):
+                streams.append(NamedVideoStream(sc, name, path=src,
+                                                codec=codec,
+                                                io_packet_size=io_packet_size))
+            else:
+                streams.append(NamedVideoStream(sc, name, frames=src,
+                                                codec=codec,
+                                                io_packet_size=io_packet_size))
+        except Exception as e:
+            failures.append((name, f"{type(e).__name__}: {e}"))
+    return streams, failures
+
+
 class NamedVideoStream(NamedStream):
     """A frame column stored as a (possibly codec-compressed) video table
     (parity: NamedVideoStream storage.py:250-374). With no codec libraries
@@ -91,3 +115,13 @@ class NamedVideoStream(NamedStream):
                      3: np.float64}[ftype]
             arr = np.frombuffer(buf, dtype=dtype).reshape(shape)
             yield fn(arr) if fn is not None else arr
+
+
+    def save_npy(self, path, rows=None):
+        """Export decoded frames to an .npy file (deviation from the
+        reference's save_mp4: this image ships no H.264 encoder; the SVC
+        stream itself is the compressed representation and .npy is the
+        interchange export)."""
+        frames = np.stack(list(self.load(rows=rows)))
+        np.save(path, frames)
+        return path

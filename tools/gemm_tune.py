@@ -12,7 +12,10 @@ SHAPES = [
     ("rn.l1.c3   ", 16 * 56 * 56, 256, 64),
     ("rn.l2.c2   ", 16 * 28 * 28, 128, 1152),
     ("rn.l3.c2   ", 16 * 14 * 14, 256, 2304),
-    ("rn.l4.c2   ", 16 * 7 * 7, 512, 4608),
+    ("rn.l4.c2   ", 16 * 7 * 7, 512, 4608),   # split-K path
+    ("rn.l4.c2b32", 32 * 7 * 7, 512, 4608),
+    ("rn.l4.ds   ", 32 * 7 * 7, 2048, 1024),
+    ("rn.fc      ", 32, 1024, 2048),
     ("pose.b2    ", 8 * 184 * 184, 64, 576),
     ("pose.b4    ", 8 * 92 * 92, 128, 1152),
     ("pose.stage ", 8 * 46 * 46, 128, 1728),
