@@ -599,3 +599,16 @@ def test_overlapping_slices(sc):
     expect = [i + (i % 8) + 1 for i in range(8)] + \
              [4 + i + (i % 8) + 1 for i in range(8)]
     assert vals == expect, vals
+
+
+def test_batch_ingest_and_export(sc, tmp_path):
+    from scanner_amd.storage import ingest_videos
+    good = make_video(n=6)
+    bad = np.zeros((3, 4), np.uint8)  # wrong rank -> ingest failure
+    streams, failures = ingest_videos(
+        sc, [("bi_ok", good), ("bi_bad", bad)], codec="svc")
+    assert len(streams) == 1 and streams[0].name == "bi_ok"
+    assert len(failures) == 1 and failures[0][0] == "bi_bad"
+
+    out = streams[0].save_npy(str(tmp_path / "clip.npy"))
+    np.testing.assert_array_equal(np.load(out), good)
