@@ -314,7 +314,7 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t) {
       const KernelFactory& kf = kernel_registry().get(op.name, op.device);
       inst.kernels[i] = kf.make(cfg);
       auto ait = job.op_args.find((i32)i);
-      const std::vector<u8>& sargs =
+      std::vector<u8> sargs =
           ait == job.op_args.end() ? std::vector<u8>{} : ait->second;
       // fetch-once resources (weights etc.) — first fetch per op name
       // process-wide, then every instance sets up (reference:

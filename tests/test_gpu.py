@@ -200,3 +200,22 @@ def test_gpu_compressed_output(sc):
            cache_mode=sp.CacheMode.Overwrite)
     ref = np.stack(list(sp.NamedVideoStream(sc, "g_cmp_ref").load()))
     assert np.abs(got.astype(int) - ref.astype(int)).max() <= 1
+
+
+def test_gpu_4k_decode_histogram(sc):
+    """4K frames through SVC GPU decode + histogram (the pose pipeline's
+    decode shape; catches large-frame geometry bugs the 1080p tests
+    miss)."""
+    from conftest import make_smooth_video
+    frames = make_smooth_video(n=6, h=2160, w=3840)
+    video = sp.NamedVideoStream(sc, "g_4k", frames=frames, codec="svc")
+    frame = sc.io.Input([video])
+    hist = sc.ops.Histogram(frame=frame, device=sp.DeviceType.GPU)
+    out = sp.NamedStream(sc, "g_4k_out")
+    sc.run(sc.io.Output(hist, [out]), sp.PerfParams.manual(2, 4),
+           cache_mode=sp.CacheMode.Overwrite, gpu_ids=[0])
+    rows = list(out.load())
+    assert len(rows) == 6
+    for i, blob in enumerate(rows):
+        got = np.frombuffer(blob, dtype=np.uint32).reshape(3, 256)
+        np.testing.assert_array_equal(got, ref_histogram(frames[i]))
