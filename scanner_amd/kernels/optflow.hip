@@ -15,8 +15,6 @@
 // kernel never synchronizes the device.
 #include <hip/hip_runtime.h>
 
-#include <climits>
-
 #include <map>
 #include <vector>
 
@@ -162,156 +160,88 @@ __global__ void __launch_bounds__(256, 5)
   __syncthreads();
 
   int x = tx0 + threadIdx.x, y = ty0 + threadIdx.y;
-  bool active = x < w && y < h;  // no early return: barriers follow
+  if (x >= w || y >= h) return;
 
   const float* fin = flow_in + (i64)pair * h * w * 2;
   float* fout = flow_out + (i64)pair * h * w * 2;
-  float u = 0.f, v = 0.f;
-  if (active) {
-    u = fin[((i64)y * w + x) * 2 + 0];
-    v = fin[((i64)y * w + x) * 2 + 1];
-  }
+  float u = fin[((i64)y * w + x) * 2 + 0];
+  float v = fin[((i64)y * w + x) * 2 + 1];
 
+  // The warp offset (u,v) is constant across the window, so the bilinear
+  // fractions are too: the 49 taps read a contiguous (2R+2)^2 region of I1
+  // at one fractional offset. Interior fast path: slide two unrolled row
+  // register arrays down that region — 64 loads/pixel instead of 196, no
+  // per-tap floor/clamp. Border threads take the clamped slow path.
+  float a11 = 0, a12 = 0, a22 = 0, b1 = 0, b2 = 0;
   float wxf = x + u, wyf = y + v;
   int xi = (int)floorf(wxf), yi = (int)floorf(wyf);
   float ax = wxf - xi, ay = wyf - yi;
   constexpr int RW = 2 * RADIUS + 2;
-
-  // Block-uniform I1 staging: flow is locally smooth at the fine levels,
-  // so the union of the block's warped windows usually fits a small LDS
-  // tile. Staging it turns ~64 per-pixel VMEM loads (the dominant stall:
-  // PMC WAIT_ANY 49%) into one cooperative 6 KB fill + LDS reads. The
-  // uniform/divergent decision is computed from shared min/max, so the
-  // branch is block-uniform and the barriers inside are safe.
-  constexpr int B1 = TILE + 2 * RADIUS + 2 + 16;  // staged extent (40 @ R=3)
-  __shared__ float lds_i1[B1 * B1];
-  __shared__ int s_minx, s_maxx, s_miny, s_maxy;
-  if (threadIdx.x == 0 && threadIdx.y == 0) {
-    s_minx = INT_MAX;
-    s_maxx = INT_MIN;
-    s_miny = INT_MAX;
-    s_maxy = INT_MIN;
-  }
-  __syncthreads();
-  if (active) {
-    atomicMin(&s_minx, xi);
-    atomicMax(&s_maxx, xi);
-    atomicMin(&s_miny, yi);
-    atomicMax(&s_maxy, yi);
-  }
-  __syncthreads();
-  int bx0 = s_minx - RADIUS, by0 = s_miny - RADIUS;
-  int bw_ = s_maxx - s_minx + RW, bh_ = s_maxy - s_miny + RW;
-  bool uniform = s_minx <= s_maxx && bw_ <= B1 && bh_ <= B1;
-
-  float a11 = 0, a12 = 0, a22 = 0, b1 = 0, b2 = 0;
-  if (uniform) {
-    for (int i = threadIdx.y * TILE + threadIdx.x; i < bh_ * bw_;
-         i += TILE * TILE) {
-      int ry = i / bw_, rx = i % bw_;
-      int gy = min(max(by0 + ry, 0), h - 1);
-      int gx = min(max(bx0 + rx, 0), w - 1);
-      lds_i1[i] = I1[(i64)gy * w + gx];
+  bool interior = xi - RADIUS >= 0 && xi + RADIUS + 1 < w &&
+                  yi - RADIUS >= 0 && yi + RADIUS + 1 < h;
+  if (interior) {
+    float r0[RW], r1[RW];
+    const float* row = I1 + (i64)(yi - RADIUS) * w + (xi - RADIUS);
+#pragma unroll
+    for (int j = 0; j < RW; ++j) r0[j] = row[j];
+#pragma unroll
+    for (int dy = -RADIUS; dy <= RADIUS; ++dy) {
+      const float* nrow = I1 + (i64)(yi + dy + 1) * w + (xi - RADIUS);
+#pragma unroll
+      for (int j = 0; j < RW; ++j) r1[j] = nrow[j];
+      int ly = (int)threadIdx.y + HALO + dy;
+      int gy_ = (int)threadIdx.y + RADIUS + dy;
+#pragma unroll
+      for (int dx = -RADIUS; dx <= RADIUS; ++dx) {
+        int lx = (int)threadIdx.x + HALO + dx;
+        int gx_ = (int)threadIdx.x + RADIUS + dx;
+        float ix = lds_gx[gy_ * GW + gx_];
+        float iy = lds_gy[gy_ * GW + gx_];
+        int j = dx + RADIUS;
+        // per-tap 2D lerp: a hoisted vertical-lerp row array costs 8
+        // VGPRs and one wave of occupancy — latency, not VALU, is the
+        // bound here (PMC: ACTIVE 26%, WAIT 49%)
+        float top = r0[j] + ax * (r0[j + 1] - r0[j]);
+        float bot = r1[j] + ax * (r1[j + 1] - r1[j]);
+        float it = top + ay * (bot - top) - lds_i0[ly * LW + lx];
+        a11 += ix * ix;
+        a12 += ix * iy;
+        a22 += iy * iy;
+        b1 += ix * it;
+        b2 += iy * it;
+      }
+#pragma unroll
+      for (int j = 0; j < RW; ++j) r0[j] = r1[j];
     }
-    __syncthreads();
-    if (active) {
-      int rx = xi - RADIUS - bx0, ry0 = yi - RADIUS - by0;
-      float r0[RW], r1[RW];
-      const float* row = lds_i1 + ry0 * bw_ + rx;
+  } else {
 #pragma unroll
-      for (int j = 0; j < RW; ++j) r0[j] = row[j];
+    for (int dy = -RADIUS; dy <= RADIUS; ++dy) {
 #pragma unroll
-      for (int dy = -RADIUS; dy <= RADIUS; ++dy) {
-        const float* nrow = lds_i1 + (ry0 + dy + RADIUS + 1) * bw_ + rx;
-#pragma unroll
-        for (int j = 0; j < RW; ++j) r1[j] = nrow[j];
+      for (int dx = -RADIUS; dx <= RADIUS; ++dx) {
+        int lx = (int)threadIdx.x + HALO + dx;
         int ly = (int)threadIdx.y + HALO + dy;
-        int gy_ = (int)threadIdx.y + RADIUS + dy;
-#pragma unroll
-        for (int dx = -RADIUS; dx <= RADIUS; ++dx) {
-          int lx = (int)threadIdx.x + HALO + dx;
-          int gx_ = (int)threadIdx.x + RADIUS + dx;
-          float ix = lds_gx[gy_ * GW + gx_];
-          float iy = lds_gy[gy_ * GW + gx_];
-          int j = dx + RADIUS;
-          float top = r0[j] + ax * (r0[j + 1] - r0[j]);
-          float bot = r1[j] + ax * (r1[j + 1] - r1[j]);
-          float it = top + ay * (bot - top) - lds_i0[ly * LW + lx];
-          a11 += ix * ix;
-          a12 += ix * iy;
-          a22 += iy * iy;
-          b1 += ix * it;
-          b2 += iy * it;
-        }
-#pragma unroll
-        for (int j = 0; j < RW; ++j) r0[j] = r1[j];
-      }
-    }
-  } else if (active) {
-    bool interior = xi - RADIUS >= 0 && xi + RADIUS + 1 < w &&
-                    yi - RADIUS >= 0 && yi + RADIUS + 1 < h;
-    if (interior) {
-      float r0[RW], r1[RW];
-      const float* row = I1 + (i64)(yi - RADIUS) * w + (xi - RADIUS);
-#pragma unroll
-      for (int j = 0; j < RW; ++j) r0[j] = row[j];
-#pragma unroll
-      for (int dy = -RADIUS; dy <= RADIUS; ++dy) {
-        const float* nrow = I1 + (i64)(yi + dy + 1) * w + (xi - RADIUS);
-#pragma unroll
-        for (int j = 0; j < RW; ++j) r1[j] = nrow[j];
-        int ly = (int)threadIdx.y + HALO + dy;
-        int gy_ = (int)threadIdx.y + RADIUS + dy;
-#pragma unroll
-        for (int dx = -RADIUS; dx <= RADIUS; ++dx) {
-          int lx = (int)threadIdx.x + HALO + dx;
-          int gx_ = (int)threadIdx.x + RADIUS + dx;
-          float ix = lds_gx[gy_ * GW + gx_];
-          float iy = lds_gy[gy_ * GW + gx_];
-          int j = dx + RADIUS;
-          float top = r0[j] + ax * (r0[j + 1] - r0[j]);
-          float bot = r1[j] + ax * (r1[j + 1] - r1[j]);
-          float it = top + ay * (bot - top) - lds_i0[ly * LW + lx];
-          a11 += ix * ix;
-          a12 += ix * iy;
-          a22 += iy * iy;
-          b1 += ix * it;
-          b2 += iy * it;
-        }
-#pragma unroll
-        for (int j = 0; j < RW; ++j) r0[j] = r1[j];
-      }
-    } else {
-#pragma unroll
-      for (int dy = -RADIUS; dy <= RADIUS; ++dy) {
-#pragma unroll
-        for (int dx = -RADIUS; dx <= RADIUS; ++dx) {
-          int lx = (int)threadIdx.x + HALO + dx;
-          int ly = (int)threadIdx.y + HALO + dy;
-          float ix = lds_gx[(threadIdx.y + RADIUS + dy) * GW +
-                            threadIdx.x + RADIUS + dx];
-          float iy = lds_gy[(threadIdx.y + RADIUS + dy) * GW +
-                            threadIdx.x + RADIUS + dx];
-          float i0v = lds_i0[ly * LW + lx];
-          int x0c = min(max(xi + dx, 0), w - 1);
-          int x1c = min(max(xi + dx + 1, 0), w - 1);
-          int y0c = min(max(yi + dy, 0), h - 1);
-          int y1c = min(max(yi + dy + 1, 0), h - 1);
-          float w00 = I1[(i64)y0c * w + x0c], w01 = I1[(i64)y0c * w + x1c];
-          float w10 = I1[(i64)y1c * w + x0c], w11 = I1[(i64)y1c * w + x1c];
-          float i1v = w00 * (1 - ay) * (1 - ax) + w01 * (1 - ay) * ax +
-                      w10 * ay * (1 - ax) + w11 * ay * ax;
-          float it = i1v - i0v;
-          a11 += ix * ix;
-          a12 += ix * iy;
-          a22 += iy * iy;
-          b1 += ix * it;
-          b2 += iy * it;
-        }
+        float ix = lds_gx[(threadIdx.y + RADIUS + dy) * GW +
+                          threadIdx.x + RADIUS + dx];
+        float iy = lds_gy[(threadIdx.y + RADIUS + dy) * GW +
+                          threadIdx.x + RADIUS + dx];
+        float i0v = lds_i0[ly * LW + lx];
+        int x0c = min(max(xi + dx, 0), w - 1);
+        int x1c = min(max(xi + dx + 1, 0), w - 1);
+        int y0c = min(max(yi + dy, 0), h - 1);
+        int y1c = min(max(yi + dy + 1, 0), h - 1);
+        float w00 = I1[(i64)y0c * w + x0c], w01 = I1[(i64)y0c * w + x1c];
+        float w10 = I1[(i64)y1c * w + x0c], w11 = I1[(i64)y1c * w + x1c];
+        float i1v = w00 * (1 - ay) * (1 - ax) + w01 * (1 - ay) * ax +
+                    w10 * ay * (1 - ax) + w11 * ay * ax;
+        float it = i1v - i0v;
+        a11 += ix * ix;
+        a12 += ix * iy;
+        a22 += iy * iy;
+        b1 += ix * it;
+        b2 += iy * it;
       }
     }
   }
-  if (!active) return;
   float det = a11 * a22 - a12 * a12;
   if (det > of::kDetEps) {
     u -= (a22 * b1 - a12 * b2) / det;
