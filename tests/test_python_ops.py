@@ -149,3 +149,24 @@ def test_python_op_fetch_resources(sc):
     assert len(list(out.load())) == n
     assert len(ResourceKernel.FETCHES) == 1      # fetched exactly once
     assert len(ResourceKernel.SETUPS) >= 1       # per instance
+
+
+@register_python_op()
+def CountNulls(col: bytes) -> bytes:
+    return (b"null" if col is None else b"val")
+
+
+def test_python_op_null_elements(sc):
+    """RepeatNull gaps reach python kernels as None (reference:
+    NullElement -> None in python kernels)."""
+    n = 4
+    tab = sc.new_table("nul", ["col"],
+                       [[int(i).to_bytes(4, "little")] for i in range(n)])
+    col = sc.io.Input([tab])
+    spaced = sc.streams.RepeatNull(col, [2])
+    out_op = sc.ops.CountNulls(col=spaced)
+    out = sp.NamedStream(sc, "nul_out")
+    sc.run(sc.io.Output(out_op, [out]), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite)
+    rows = list(out.load())
+    assert rows == [b"val", b"null"] * n
