@@ -54,7 +54,7 @@ class PerfParams:
     def __init__(self, work_packet_size=16, io_packet_size=128,
                  cpu_pool=0, gpu_pool=0, pipeline_instances_per_node=None,
                  load_sparsity_threshold=8, queue_size_per_pipeline=4,
-                 profiler_level=1):
+                 profiler_level=1, num_load_workers=0):
         self.work_packet_size = int(work_packet_size)
         self.io_packet_size = int(io_packet_size)
         self.cpu_pool = int(cpu_pool)
@@ -63,6 +63,7 @@ class PerfParams:
         self.load_sparsity_threshold = int(load_sparsity_threshold)
         self.queue_size_per_pipeline = int(queue_size_per_pipeline)
         self.profiler_level = int(profiler_level)
+        self.num_load_workers = int(num_load_workers)  # 0 = auto
 
     @classmethod
     def manual(cls, work_packet_size, io_packet_size, **kw):
@@ -109,6 +110,7 @@ class PerfParams:
             "gpu_pool_size": self.gpu_pool,
             "sparsity_threshold": self.load_sparsity_threshold,
             "profiler_level": self.profiler_level,
+            "load_workers": self.num_load_workers,
         }
 
 

@@ -54,6 +54,8 @@ PerfParams perf_from_dict(const py::dict& d) {
     pp.work_packet_size = d["work_packet_size"].cast<i64>();
   if (d.contains("pipeline_instances"))
     pp.pipeline_instances = d["pipeline_instances"].cast<i32>();
+  if (d.contains("load_workers"))
+    pp.load_workers = d["load_workers"].cast<i32>();
   if (d.contains("cpu_pool_size"))
     pp.cpu_pool_size = d["cpu_pool_size"].cast<size_t>();
   if (d.contains("gpu_pool_size"))

@@ -130,37 +130,70 @@ void LocalExecutor::finalize_job(i32 job) {
 void LocalExecutor::run() {
   prepare();
   auto tasks = all_tasks();
-  BoundedQueue<TaskDesc> q;
-  for (auto& t : tasks) q.push(t);
-  q.close();
+  BoundedQueue<TaskDesc> q_raw;
+  for (auto& t : tasks) q_raw.push(t);
+  q_raw.close();
 
   std::atomic<bool> failed{false};
   std::string fail_msg;
   std::mutex fail_mu;
+  auto record_failure = [&](const std::exception& ex) {
+    std::lock_guard<std::mutex> l(fail_mu);
+    if (!failed.exchange(true)) fail_msg = ex.what();
+  };
+
+  // Load-worker stage (reference: dedicated load threads feeding the
+  // pipeline, worker.cpp:85): derive plans + read video source spans
+  // ahead of the instances, bounded queue for backpressure so at most a
+  // few tasks' pinned bytes are in flight.
+  i32 n_load = pp_.load_workers > 0
+                   ? pp_.load_workers
+                   : std::min<i32>(8, std::max<i32>(2, (i32)instances_.size()));
+  BoundedQueue<std::shared_ptr<PreparedTask>> q_ready(
+      instances_.size() + 2);
+  std::vector<std::unique_ptr<Profiler>> load_profilers;
+  for (i32 i = 0; i < n_load; ++i) {
+    load_profilers.push_back(
+        std::make_unique<Profiler>((ProfilerLevel)pp_.profiler_level));
+  }
+  std::atomic<i32> live_loaders{n_load};
+  auto loader = [&](i32 li) {
+    while (auto t = q_raw.try_pop()) {
+      if (failed.load()) break;
+      try {
+        q_ready.push(prepare_task(*t, load_profilers[li].get()));
+      } catch (const std::exception& ex) {
+        record_failure(ex);
+        break;
+      }
+    }
+    if (live_loaders.fetch_sub(1) == 1) q_ready.close();
+  };
 
   auto worker = [&](i32 idx) {
     Instance& inst = *instances_[idx];
-    while (auto t = q.try_pop()) {
+    while (auto pt = q_ready.pop()) {
       if (failed.load()) return;
       try {
-        process_task(inst, *t);
+        process_task(inst, (*pt)->desc, pt->get());
       } catch (const std::exception& ex) {
-        std::lock_guard<std::mutex> l(fail_mu);
-        failed.store(true);
-        fail_msg = ex.what();
+        record_failure(ex);
+        // Unblock loaders waiting on a full ready-queue.
+        q_ready.close();
         return;
       }
     }
   };
 
-  if (instances_.size() == 1) {
-    worker(0);
-  } else {
-    std::vector<std::thread> threads;
-    for (size_t i = 0; i < instances_.size(); ++i)
-      threads.emplace_back(worker, (i32)i);
-    for (auto& th : threads) th.join();
-  }
+  std::vector<std::thread> threads;
+  for (i32 i = 0; i < n_load; ++i) threads.emplace_back(loader, i);
+  for (size_t i = 0; i < instances_.size(); ++i)
+    threads.emplace_back(worker, (i32)i);
+  for (auto& th : threads) th.join();
+  // Failure paths can leave prepared tasks queued; drain so their pinned
+  // buffers return to the pool before profilers move.
+  while (auto leftover = q_ready.try_pop()) leftover->reset();
+  for (auto& lp : load_profilers) profilers_.push_back(std::move(lp));
   if (failed.load()) throw ScannerError("job failed: " + fail_msg);
   for (size_t j = 0; j < jobs_.size(); ++j) finalize_job((i32)j);
 }
@@ -171,10 +204,83 @@ void LocalExecutor::process_task_public(i32 instance, const TaskDesc& t) {
   process_task(*instances_[instance], t);
 }
 
+PreparedTask::~PreparedTask() {
+  for (auto& sp : spans) {
+    if (sp.buf) delete_buffer(CPU_DEVICE, sp.buf);
+  }
+}
+
+u8* PreparedTask::take_span(i32 op, i32 item, u64 lo, u64 hi) {
+  for (auto& sp : spans) {
+    if (sp.buf && sp.op == op && sp.item == item && sp.lo == lo &&
+        sp.hi == hi) {
+      u8* b = sp.buf;
+      sp.buf = nullptr;  // ownership moves to the caller
+      return b;
+    }
+  }
+  return nullptr;
+}
+
+// Load-worker half of a task (runs on the prefetch pool): derive the plan
+// and read every svc video source span into pinned memory so the pipeline
+// instance starts with its input bytes already resident.
+std::shared_ptr<PreparedTask> LocalExecutor::prepare_task(const TaskDesc& t,
+                                                          Profiler* prof) {
+  auto pt = std::make_shared<PreparedTask>();
+  pt->desc = t;
+  const JobBinding& job = jobs_[t.job];
+  {
+    Profiler::Scope s(prof, "derive_task_plan");
+    pt->plan = derive_task_plan(graph_, analyses_[t.job], job, t.start,
+                                t.end);
+  }
+  for (auto& kv : pt->plan.load_rows) {
+    i32 op_idx = kv.first;
+    const std::vector<i64>& rows = kv.second;
+    const SourceArgsC& src = job.sources.at(op_idx);
+    const OpStaticInfo& si = analyses_[t.job].info[op_idx];
+    if (si.output_types[0] != ColumnType::Video) continue;
+    TableMetadata table = db_->get_table(src.table);
+    auto items = items_for_rows(table, rows);
+    size_t ri = 0;
+    for (auto& ir : items) {
+      VideoMetadata m = read_video_metadata(*db_, table, src.column,
+                                            ir.item);
+      std::vector<i64> local;
+      while (ri < rows.size() && rows[ri] < ir.row_end) {
+        local.push_back(rows[ri] - ir.row_start);
+        ++ri;
+      }
+      if (m.codec != "svc") continue;
+      std::vector<i64> span = svc_decode_span(m, local);
+      if (span.empty()) continue;
+      u64 lo = m.sample_offsets[span.front()];
+      u64 hi = m.sample_offsets[span.back()] + m.sample_sizes[span.back()];
+      PreparedTask::Span sp;
+      sp.op = op_idx;
+      sp.item = ir.item;
+      sp.lo = lo;
+      sp.hi = hi;
+      sp.buf = new_buffer(CPU_DEVICE, hi - lo);
+      {
+        Profiler::Scope sl(prof, "load:video");
+        db_->storage()->read_range(
+            db_->paths().item(table.id, table.column_id(src.column),
+                              ir.item),
+            lo, hi - lo, sp.buf);
+      }
+      pt->spans.push_back(sp);
+    }
+  }
+  return pt;
+}
+
 // Load Input-op rows into the per-op output maps.
 void LocalExecutor::load_inputs(
     Instance& inst, const TaskDesc& t, const TaskPlan& plan,
-    std::vector<std::map<std::string, std::unordered_map<i64, Element>>>& outs) {
+    std::vector<std::map<std::string, std::unordered_map<i64, Element>>>& outs,
+    PreparedTask* prep) {
   const JobBinding& job = jobs_[t.job];
   for (auto& kv : plan.load_rows) {
     i32 op_idx = kv.first;
@@ -215,8 +321,10 @@ void LocalExecutor::load_inputs(
           if (span.empty()) continue;
           u64 lo = m.sample_offsets[span.front()];
           u64 hi = m.sample_offsets[span.back()] + m.sample_sizes[span.back()];
-          u8* stream_buf = new_buffer(CPU_DEVICE, hi - lo);
-          {
+          u8* stream_buf =
+              prep ? prep->take_span(op_idx, ir.item, lo, hi) : nullptr;
+          if (!stream_buf) {
+            stream_buf = new_buffer(CPU_DEVICE, hi - lo);
             Profiler::Scope sl(inst.profiler, "load:video");
             db_->storage()->read_range(
                 db_->paths().item(table.id, table.column_id(src.column),
@@ -283,13 +391,16 @@ void LocalExecutor::load_inputs(
   }
 }
 
-void LocalExecutor::process_task(Instance& inst, const TaskDesc& t) {
+void LocalExecutor::process_task(Instance& inst, const TaskDesc& t,
+                                 PreparedTask* prep) {
   const JobBinding& job = jobs_[t.job];
   const JobAnalysis& ja = analyses_[t.job];
   size_t n = graph_.ops.size();
 
   TaskPlan plan;
-  {
+  if (prep) {
+    plan = std::move(prep->plan);
+  } else {
     Profiler::Scope s(inst.profiler, "derive_task_plan");
     plan = derive_task_plan(graph_, ja, job, t.start, t.end);
   }
@@ -348,7 +459,7 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t) {
   };
 
   try {
-    load_inputs(inst, t, plan, outs);
+    load_inputs(inst, t, plan, outs, prep);
 
     for (size_t i = 0; i < n; ++i) {
       const OpNode& op = graph_.ops[i];

@@ -24,6 +24,7 @@ struct PerfParams {
   i64 io_packet_size = 128;
   i64 work_packet_size = 32;
   i32 pipeline_instances = 1;
+  i32 load_workers = 0;  // 0 = auto (min(8, max(2, instances)))
   i64 queue_size = 4;
   size_t cpu_pool_size = 0;
   size_t gpu_pool_size = 0;
@@ -35,6 +36,24 @@ struct TaskDesc {
   i32 job = 0;
   i32 task = 0;  // item index in the output table
   i64 start = 0, end = 0;
+};
+
+// A task with its plan derived and its video source byte spans already
+// read into pinned memory by a load worker (reference: the dedicated
+// load_worker threads feeding the pipeline, worker.cpp:85 +
+// load_worker.cpp). Owns the span buffers until consumed.
+struct PreparedTask {
+  TaskDesc desc;
+  TaskPlan plan;
+  struct Span {
+    i32 op = 0;
+    i32 item = 0;
+    u64 lo = 0, hi = 0;
+    u8* buf = nullptr;  // pinned (CPU pool)
+  };
+  std::vector<Span> spans;
+  ~PreparedTask();
+  u8* take_span(i32 op, i32 item, u64 lo, u64 hi);
 };
 
 class LocalExecutor {
@@ -64,10 +83,16 @@ class LocalExecutor {
 
  private:
   struct Instance;
-  void process_task(Instance& inst, const TaskDesc& t);
+  void process_task(Instance& inst, const TaskDesc& t,
+                    PreparedTask* prep = nullptr);
+  // Load-worker half of a task: derive the plan + read video source spans
+  // into pinned buffers (runs on prefetch threads in run()).
+  std::shared_ptr<PreparedTask> prepare_task(const TaskDesc& t,
+                                             Profiler* prof);
   void load_inputs(Instance& inst, const TaskDesc& t, const TaskPlan& plan,
                    std::vector<std::map<std::string,
-                                        std::unordered_map<i64, Element>>>& outs);
+                                        std::unordered_map<i64, Element>>>& outs,
+                   PreparedTask* prep = nullptr);
   void make_instance(i32 idx);
 
   std::shared_ptr<Database> db_;
