@@ -52,7 +52,9 @@ std::vector<ConvSpec> pose_specs(int stages) {
     for (const char* br : {"L", "S"}) {
       std::string p = "s" + std::to_string(t) + br;
       int out_c = br[0] == 'L' ? kPafC : kHeatC;
-      int in_c = t == 1 ? 128 : 128 + kPafC + kHeatC;  // concat(F,L,S)
+      // concat(F,L,S) = 185 ch, zero-padded to 192 so the stage convs run
+      // on the implicit-GEMM path (c % 8 == 0)
+      int in_c = t == 1 ? 128 : 192;
       c3(p + "_1", in_c, 128, 1);
       c3(p + "_2", 128, 128, 1);
       c3(p + "_3", 128, 128, 1);
@@ -112,7 +114,7 @@ class PoseKernelGPU : public BatchedKernel {
     u8* brS = new_buffer(dev, featpix * 64 * 2);   // np(19) = 64
     u8* brT = new_buffer(dev, featpix * 512 * 2);  // stage mid buffers
     u8* brU = new_buffer(dev, featpix * 512 * 2);
-    u8* cat = new_buffer(dev, featpix * (128 + kPafC + kHeatC) * 2);
+    u8* cat = new_buffer(dev, featpix * 192 * 2);
 
     auto conv = [&](const std::string& name, const u8* x, int h, int w,
                     u8* y, int& oh, int& ow) {
@@ -175,7 +177,7 @@ class PoseKernelGPU : public BatchedKernel {
         stage_in = feat;
       } else {
         concat3_bf16(feat, 128, 128, brL, kPafC, 64, brS, kHeatC, 64,
-                     (i64)featpix, cat, s);
+                     (i64)featpix, cat, 192, s);
         stage_in = cat;
       }
       for (const char* br : {"L", "S"}) {

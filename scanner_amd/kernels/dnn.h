@@ -69,7 +69,7 @@ void bf16_rows_to_f32(const void* in, int n, int stride_cols, int ncols,
 // channel strides let branches read from GEMM-padded buffers.
 void concat3_bf16(const void* a, int ca, int stride_a, const void* b, int cb,
                   int stride_b, const void* c, int cc, int stride_c, i64 npix,
-                  void* out, void* stream);
+                  void* out, int out_stride, void* stream);
 
 // Per-(frame,channel) spatial argmax of NHWC bf16 maps (padded channel
 // stride c_stride): out[n][nch][3] f32 = {x, y, peak}.

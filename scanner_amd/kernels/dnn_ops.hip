@@ -230,8 +230,9 @@ __global__ void __launch_bounds__(256)
     concat3_kernel(const bf16* __restrict__ a, int ca, int sa,
                    const bf16* __restrict__ b, int cb, int sb,
                    const bf16* __restrict__ c, int cc, int scc, i64 npix,
-                   bf16* __restrict__ out) {
-  int ct = ca + cb + cc;
+                   bf16* __restrict__ out, int ct) {
+  // ct >= ca+cb+cc: channels past the concat are zero-filled so the
+  // consumer can run with a c%8==0 stride (implicit-GEMM requirement)
   i64 total = npix * ct;
   i64 gs = (i64)gridDim.x * blockDim.x;
   for (i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x; i < total;
@@ -243,8 +244,10 @@ __global__ void __launch_bounds__(256)
       v = a[pix * sa + ch];
     else if (ch < ca + cb)
       v = b[pix * sb + (ch - ca)];
-    else
+    else if (ch < ca + cb + cc)
       v = c[pix * scc + (ch - ca - cb)];
+    else
+      v = (bf16)0.f;
     out[i] = v;
   }
 }
@@ -299,11 +302,11 @@ __global__ void __launch_bounds__(256)
 
 void concat3_bf16(const void* a, int ca, int stride_a, const void* b, int cb,
                   int stride_b, const void* c, int cc, int stride_c, i64 npix,
-                  void* out, void* stream) {
-  i64 total = npix * (ca + cb + cc);
+                  void* out, int out_stride, void* stream) {
+  i64 total = npix * out_stride;
   concat3_kernel<<<grid_for(total), 256, 0, (hipStream_t)stream>>>(
       (const bf16*)a, ca, stride_a, (const bf16*)b, cb, stride_b,
-      (const bf16*)c, cc, stride_c, npix, (bf16*)out);
+      (const bf16*)c, cc, stride_c, npix, (bf16*)out, out_stride);
   DNN_CHECK();
 }
 
