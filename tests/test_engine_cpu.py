@@ -635,3 +635,25 @@ def test_multi_instance_stress(sc):
     rows = list(sp.NamedStream(sc, "mi_out", column="histogram").load())
     got = np.frombuffer(rows[0], np.uint32).reshape(3, 256)
     assert got.sum() == 32 * 40 * 3
+
+
+def test_stencil_after_stride(sc):
+    """A stencil op downstream of Stride sees the SAMPLED domain: flow of
+    strided frames pairs sampled neighbors (original rows 3k and 3k+3),
+    not adjacent original frames (reference: stencils apply in each op's
+    own input domain after DomainSampler remapping)."""
+    base = make_textured_pair(h=64, w=96, dx=3, dy=0)[0]
+    # frames shift right 1px per ORIGINAL frame -> 3px per sampled step
+    frames = np.stack([np.roll(base, i, axis=1) for i in range(12)])
+    video = sp.NamedVideoStream(sc, "sas", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    sampled = sc.streams.Stride(frame, [3])
+    flow = sc.ops.OpticalFlow(frame=sampled)
+    stats = sc.ops.FlowStats(flow=flow)
+    out = sp.NamedStream(sc, "sas_out")
+    sc.run(sc.io.Output(stats, [out]), sp.PerfParams.manual(2, 4),
+           cache_mode=sp.CacheMode.Overwrite)
+    rows = [np.frombuffer(b, np.float32) for b in out.load()]
+    assert len(rows) == 4  # ceil(12/3)
+    # mean |u| of the first window ~ 3 px (sampled-domain neighbors)
+    assert 2.0 < rows[0][0] < 4.0, rows[0]
