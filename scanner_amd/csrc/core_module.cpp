@@ -496,6 +496,8 @@ PYBIND11_MODULE(_core, m) {
 
   // ---- execution ----
 
+  py::class_<PreparedTask, std::shared_ptr<PreparedTask>>(m, "PreparedTask");
+
   py::class_<LocalExecutor>(m, "LocalExecutor")
       .def(py::init([](std::shared_ptr<Database> db, const py::bytes& graph_b,
                        const py::bytes& jobs_b, const py::dict& perf,
@@ -526,6 +528,18 @@ PYBIND11_MODULE(_core, m) {
              TaskDesc t{job, task, start, end};
              py::gil_scoped_release rel;
              ex.process_task_public(instance, t);
+           })
+      .def("prepare_task",
+           [](LocalExecutor& ex, i32 job, i32 task, i64 start, i64 end) {
+             TaskDesc t{job, task, start, end};
+             py::gil_scoped_release rel;
+             return ex.prepare_task_public(t);
+           })
+      .def("process_prepared",
+           [](LocalExecutor& ex, i32 instance,
+              std::shared_ptr<PreparedTask> pt) {
+             py::gil_scoped_release rel;
+             ex.process_prepared_public(instance, pt);
            })
       .def("finalize_job", &LocalExecutor::finalize_job)
       .def("total_output_rows", &LocalExecutor::total_output_rows)

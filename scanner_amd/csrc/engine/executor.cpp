@@ -204,6 +204,21 @@ void LocalExecutor::process_task_public(i32 instance, const TaskDesc& t) {
   process_task(*instances_[instance], t);
 }
 
+std::shared_ptr<PreparedTask> LocalExecutor::prepare_task_public(
+    const TaskDesc& t) {
+  prepare();
+  static thread_local Profiler loader_prof(ProfilerLevel::Info);
+  return prepare_task(t, &loader_prof);
+}
+
+void LocalExecutor::process_prepared_public(
+    i32 instance, const std::shared_ptr<PreparedTask>& pt) {
+  prepare();
+  while ((i32)instances_.size() <= instance)
+    make_instance((i32)instances_.size());
+  process_task(*instances_[instance], pt->desc, pt.get());
+}
+
 PreparedTask::~PreparedTask() {
   for (auto& sp : spans) {
     if (sp.buf) delete_buffer(CPU_DEVICE, sp.buf);

@@ -73,6 +73,11 @@ class LocalExecutor {
   // they attach to the existing descriptors (shared storage).
   void prepare(bool create_outputs = true);
   void process_task_public(i32 instance, const TaskDesc& t);
+  // Distributed-worker prefetch: load-worker half on any thread, then the
+  // instance half (same split run() uses internally).
+  std::shared_ptr<PreparedTask> prepare_task_public(const TaskDesc& t);
+  void process_prepared_public(i32 instance,
+                               const std::shared_ptr<PreparedTask>& pt);
   std::vector<TaskDesc> all_tasks() const;
   void finalize_job(i32 job);  // set end_rows + commit output table
 

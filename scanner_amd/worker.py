@@ -43,7 +43,14 @@ class WorkerServer:
         self.worker_id = resp["worker_id"]
 
         self._task_q = queue.Queue()
+        # load-worker stage (parity: dedicated load threads): plans +
+        # pinned source reads happen ahead of the pipeline instances
+        self._prep_q = queue.Queue(maxsize=pipeline_instances + 2)
         self._threads = []
+        for i in range(max(2, pipeline_instances)):
+            t = threading.Thread(target=self._loader_loop, daemon=True)
+            t.start()
+            self._threads.append(t)
         for i in range(pipeline_instances):
             t = threading.Thread(target=self._instance_loop, args=(i,),
                                  daemon=True)
@@ -136,17 +143,45 @@ class WorkerServer:
             for t in resp["tasks"]:
                 self._task_q.put((resp["job_id"], t))
 
-    def _instance_loop(self, idx):
+    def _loader_loop(self):
         while not self._shutdown.is_set():
             try:
                 job_id, t = self._task_q.get(timeout=0.2)
             except queue.Empty:
                 continue
             stream, task, start, end = t
-            ok, err = True, None
+            pt, err = None, None
             try:
                 self._ensure_job(job_id)
-                self._executor.process_task(idx, stream, task, start, end)
+                pt = self._executor.prepare_task(stream, task, start, end)
+            except Exception as e:
+                err = f"{type(e).__name__}: {e}"
+            while not self._shutdown.is_set():
+                try:
+                    self._prep_q.put((job_id, t, pt, err), timeout=0.2)
+                    break
+                except queue.Full:
+                    continue
+
+    def _instance_loop(self, idx):
+        while not self._shutdown.is_set():
+            try:
+                job_id, t, pt, err = self._prep_q.get(timeout=0.2)
+            except queue.Empty:
+                continue
+            stream, task, start, end = t
+            ok = err is None
+            try:
+                if ok:
+                    self._ensure_job(job_id)
+                    if pt is not None and self._job_id == job_id:
+                        self._executor.process_prepared(idx, pt)
+                    else:
+                        # job changed between prepare and process (rare):
+                        # the prepared buffers are dropped and the task
+                        # re-reads its inputs
+                        self._executor.process_task(idx, stream, task,
+                                                    start, end)
             except Exception as e:
                 ok, err = False, f"{type(e).__name__}: {e}"
             try:
