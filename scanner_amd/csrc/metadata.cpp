@@ -1,5 +1,9 @@
 #include "metadata.h"
 
+#include <fcntl.h>
+#include <sys/file.h>
+#include <unistd.h>
+
 #include "serialize.h"
 
 namespace sca {
@@ -166,6 +170,38 @@ void Database::persist() {
   storage_->write_all(paths_.db_metadata(), buf.data(), buf.size());
 }
 
+void Database::refresh() {
+  if (storage_->exists(paths_.db_metadata())) {
+    meta_ = DatabaseMetadata::deserialize(
+        storage_->read_all(paths_.db_metadata()));
+  }
+}
+
+namespace {
+// Inter-process exclusive lock on <db>/db.lock for metadata
+// read-modify-write cycles. Multiple processes mutate the shared db
+// (master job executors, ingesting clients); without this, concurrent
+// persist() calls lose each other's tables. flock is posix-only, like the
+// shipped storage backend; an object-store backend would route metadata
+// mutations through the master instead (reference model).
+class ScopedDbLock {
+ public:
+  explicit ScopedDbLock(const std::string& db_path) {
+    fd_ = ::open((db_path + "/db.lock").c_str(), O_CREAT | O_RDWR, 0644);
+    if (fd_ >= 0) (void)::flock(fd_, LOCK_EX);
+  }
+  ~ScopedDbLock() {
+    if (fd_ >= 0) {
+      (void)::flock(fd_, LOCK_UN);
+      (void)::close(fd_);
+    }
+  }
+
+ private:
+  int fd_ = -1;
+};
+}  // namespace
+
 void Database::recover() {
   std::lock_guard<std::mutex> l(mu_);
   std::vector<std::string> dead;
@@ -186,7 +222,9 @@ TableMetadata Database::new_table(const std::string& name,
                                   const std::vector<ColumnType>& column_types,
                                   bool overwrite) {
   SCA_CHECK(column_names.size() == column_types.size(), "column spec mismatch");
+  ScopedDbLock dbl(paths_.root);
   std::lock_guard<std::mutex> l(mu_);
+  refresh();
   auto it = meta_.table_ids.find(name);
   if (it != meta_.table_ids.end()) {
     if (!overwrite)
@@ -213,7 +251,9 @@ TableMetadata Database::new_table(const std::string& name,
 }
 
 void Database::commit_table(i32 table_id) {
+  ScopedDbLock dbl(paths_.root);
   std::lock_guard<std::mutex> l(mu_);
+  refresh();
   meta_.committed_tables.insert(table_id);
   persist();
 }
@@ -231,7 +271,9 @@ void Database::update_table(const TableMetadata& t) {
 }
 
 void Database::delete_table(const std::string& name) {
+  ScopedDbLock dbl(paths_.root);
   std::lock_guard<std::mutex> l(mu_);
+  refresh();
   auto it = meta_.table_ids.find(name);
   if (it == meta_.table_ids.end()) return;
   i32 id = it->second;
@@ -244,6 +286,8 @@ void Database::delete_table(const std::string& name) {
 
 bool Database::has_table(const std::string& name) {
   std::lock_guard<std::mutex> l(mu_);
+  if (meta_.table_ids.count(name)) return true;
+  refresh();
   return meta_.table_ids.count(name) > 0;
 }
 
@@ -252,8 +296,13 @@ TableMetadata Database::get_table(const std::string& name) {
   {
     std::lock_guard<std::mutex> l(mu_);
     auto it = meta_.table_ids.find(name);
-    if (it == meta_.table_ids.end())
-      throw ScannerError("no table '" + name + "'");
+    if (it == meta_.table_ids.end()) {
+      // another process may have created it since we loaded metadata
+      refresh();
+      it = meta_.table_ids.find(name);
+      if (it == meta_.table_ids.end())
+        throw ScannerError("no table '" + name + "'");
+    }
     id = it->second;
   }
   return get_table(id);
@@ -280,7 +329,9 @@ std::vector<std::string> Database::table_names() {
 }
 
 i32 Database::new_job(const std::string& name) {
+  ScopedDbLock dbl(paths_.root);
   std::lock_guard<std::mutex> l(mu_);
+  refresh();
   i32 id = meta_.next_job_id++;
   meta_.job_ids[name + "#" + std::to_string(id)] = id;
   storage_->make_dirs(paths_.job_dir(id));
@@ -289,7 +340,9 @@ i32 Database::new_job(const std::string& name) {
 }
 
 void Database::commit_job(i32 job_id) {
+  ScopedDbLock dbl(paths_.root);
   std::lock_guard<std::mutex> l(mu_);
+  refresh();
   meta_.committed_jobs.insert(job_id);
   persist();
 }

@@ -112,6 +112,11 @@ class Database {
 
  private:
   void persist();  // caller holds mu_
+  // Re-read db_metadata.bin from storage (caller holds mu_). Mutators call
+  // this under the inter-process file lock so concurrent writers (master
+  // job executors, ingesting clients, workers) merge instead of losing
+  // updates.
+  void refresh();
   std::shared_ptr<StorageBackend> storage_;
   DatabasePaths paths_;
   std::mutex mu_;

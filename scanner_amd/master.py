@@ -61,6 +61,8 @@ class MasterServer:
         self._next_worker_id = 0
         self._lock = threading.Lock()
         self._job = None            # current bulk job
+        self._job_queue = collections.deque()  # queued bulk jobs (FIFO)
+        self._jobs_by_id = {}       # id -> _BulkJob (current + queued + done)
         self._job_counter = 0
         self._no_workers_timeout = no_workers_timeout
         self._task_timeout = task_timeout
@@ -129,8 +131,6 @@ class MasterServer:
 
     def _new_job(self, req):
         with self._lock:
-            if self._job is not None and not self._job.complete.is_set():
-                raise ScannerException("a bulk job is already running")
             jid = self._job_counter
             self._job_counter += 1
             py_ops = [{"name": n, **v}
@@ -154,20 +154,36 @@ class MasterServer:
                 job.tasks.append(t)
                 job.to_assign.append(t)
                 job.stream_tasks[stream] += 1
-            self._job = job
+            self._jobs_by_id[jid] = job
+            if self._job is None or self._job.complete.is_set():
+                self._job = job
+                job.started = time.time()
+            else:
+                # FIFO bulk-job queue (reference: master job_processor
+                # queue, master.cpp:1328-1353)
+                self._job_queue.append(job)
         return {"job_id": jid, "n_tasks": len(job.tasks)}
 
     def _get_job(self, req):
-        job = self._job
-        if job is None or job.id != req["job_id"]:
+        job = self._jobs_by_id.get(req["job_id"])
+        if job is None:
             raise ScannerException("no such job")
         return {"job_id": job.id, "graph": job.graph, "jobs": job.jobs,
                 "perf": job.perf, "py_ops": job.py_ops}
+
+    def _promote_if_done(self):
+        # caller holds lock: activate the next queued bulk job once the
+        # current one completes
+        if (self._job is not None and self._job.complete.is_set()
+                and self._job_queue):
+            self._job = self._job_queue.popleft()
+            self._job.started = time.time()
 
     def _next_work(self, req):
         wid = req["worker_id"]
         n = req.get("max_tasks", 1)
         with self._lock:
+            self._promote_if_done()
             job = self._job
             if job is None or job.complete.is_set():
                 return {"job_id": -1, "tasks": []}
@@ -212,8 +228,9 @@ class MasterServer:
 
     def _job_status(self, req):
         with self._lock:
-            job = self._job
-            if job is None or job.id != req.get("job_id", -1):
+            self._promote_if_done()
+            job = self._jobs_by_id.get(req.get("job_id", -1))
+            if job is None:
                 return {"exists": False}
             return {
                 "exists": True,

@@ -211,3 +211,42 @@ def test_distributed_profile_collection(tmp_path):
         for w in workers:
             w.shutdown()
         master.shutdown()
+
+
+def test_bulk_job_queueing(tmp_path):
+    """A second bulk job submitted while one runs is queued and executes
+    after it (reference: master job_processor queue)."""
+    db = _mk_db(tmp_path)
+    master = MasterServer(db)
+    workers = [start_worker(master.addr, db)]
+    try:
+        sc = sp.Client(db_path=db, master=master.addr)
+        vids = [make_video(n=10, seed=s) for s in range(2)]
+        streams = [sp.NamedVideoStream(sc, f"q{i}", frames=v, codec="raw")
+                   for i, v in enumerate(vids)]
+        results = {}
+
+        def run_one(i):
+            try:
+                sc2 = sp.Client(db_path=db, master=master.addr)
+                frame = sc2.io.Input([streams[i]])
+                hist = sc2.ops.Histogram(frame=frame)
+                out = sp.NamedStream(sc2, f"q{i}_hist")
+                sc2.run(sc2.io.Output(hist, [out]),
+                        sp.PerfParams.manual(2, 4),
+                        cache_mode=sp.CacheMode.Overwrite)
+                results[i] = len(list(out.load()))
+            except Exception as e:
+                results[i] = f"{type(e).__name__}: {e}"
+
+        import threading
+        ts = [threading.Thread(target=run_one, args=(i,)) for i in range(2)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join(timeout=120)
+        assert results == {0: 10, 1: 10}, results
+    finally:
+        for w in workers:
+            w.shutdown()
+        master.shutdown()
