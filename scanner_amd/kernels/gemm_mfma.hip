@@ -141,9 +141,52 @@ __global__ void __launch_bounds__(256, 2)
   // over 4 waves; same for B.
   constexpr int A_INSTRS = BM / 8 / 4;  // per wave
   constexpr int B_INSTRS = BN / 8 / 4;
+  // Implicit-conv staging geometry: the output-pixel part of each lane's
+  // gather address is k-independent, and (cell, cj, dr, ds) advance by a
+  // constant per BK step — precompute the former, step the latter
+  // incrementally so the hot stage() does ZERO integer divisions (3 divs
+  // per LDS-DMA instruction otherwise; K/64 steps x 4 instrs adds up).
+  const int lrow = lane >> 3;          // 0..7 row within instr block
+  const int lk = (lane & 7) * 8;       // k offset (8 bf16 = 16B)
+  i64 imp_rowbase[IMPLICIT ? A_INSTRS : 1];  // nn*h*w*c + cj later
+  int imp_hh0[IMPLICIT ? A_INSTRS : 1];      // p*stride - pad
+  int imp_ww0[IMPLICIT ? A_INSTRS : 1];      // q*stride - pad
+  int imp_cell = 0, imp_cj = 0, imp_dr = 0, imp_ds = 0;
+  if constexpr (IMPLICIT) {
+#pragma unroll
+    for (int i = 0; i < A_INSTRS; ++i) {
+      int row = (wave * A_INSTRS + i) * 8 + lrow;
+      int grow = m0 + row;
+      if (grow >= M) grow = M - 1;
+      int q = grow % d.ow;
+      int t = grow / d.ow;
+      int p = t % d.oh;
+      int nn = t / d.oh;
+      imp_rowbase[i] = (i64)nn * d.h * d.w * d.c;
+      imp_hh0[i] = p * d.stride - d.pad;
+      imp_ww0[i] = q * d.stride - d.pad;
+    }
+    imp_cell = lk / d.c;
+    imp_cj = lk - imp_cell * d.c;
+    imp_dr = imp_cell / d.s;
+    imp_ds = imp_cell - imp_dr * d.s;
+  }
+  auto imp_advance = [&](int dk) {
+    // advance (cell, cj) by dk k-positions; c >= 8 so the wrap loop is
+    // short and branch-uniform across the wave (same cj for all lanes of
+    // one lk — lanes differ only via lk, folded into the initial state)
+    imp_cj += dk;
+    while (imp_cj >= d.c) {
+      imp_cj -= d.c;
+      ++imp_cell;
+      ++imp_ds;
+      if (imp_ds == d.s) {
+        imp_ds = 0;
+        ++imp_dr;
+      }
+    }
+  };
   auto stage = [&](int buf, int k0) {
-    const int lrow = lane >> 3;          // 0..7 row within instr block
-    const int lk = (lane & 7) * 8;       // k offset (8 bf16 = 16B)
     bf16* lds_a = lds + buf * (BM + BN) * BK;
     bf16* lds_b = lds_a + BM * BK;
 #pragma unroll
@@ -153,21 +196,14 @@ __global__ void __launch_bounds__(256, 2)
       if (grow >= M) grow = M - 1;  // clamp: garbage rows masked at store
       const bf16* src;
       if constexpr (IMPLICIT) {
-        int k = k0 + lk;
-        int cell = k / d.c;
-        int cj = k - cell * d.c;
-        if (cell >= d.r * d.s) {
+        if (imp_cell >= d.r * d.s) {
           src = zero;
         } else {
-          int dr = cell / d.s, ds = cell - dr * d.s;
-          int q = grow % d.ow;
-          int t = grow / d.ow;
-          int p = t % d.oh;
-          int nn = t / d.oh;
-          int hh = p * d.stride - d.pad + dr;
-          int ww = q * d.stride - d.pad + ds;
+          int hh = imp_hh0[i] + imp_dr;
+          int ww = imp_ww0[i] + imp_ds;
           src = (hh >= 0 && hh < d.h && ww >= 0 && ww < d.w)
-                    ? A + ((((i64)nn * d.h + hh) * d.w + ww) * d.c + cj)
+                    ? A + (imp_rowbase[i] + ((i64)hh * d.w + ww) * d.c +
+                           imp_cj)
                     : zero;
         }
       } else {
@@ -198,7 +234,10 @@ __global__ void __launch_bounds__(256, 2)
     // prefetch next while computing current
     __builtin_amdgcn_s_waitcnt(/*vmcnt(0) lgkmcnt(0)*/ 0);
     __syncthreads();
-    if (ks + 1 < ksteps) stage(buf ^ 1, (ks + 1) * BK);
+    if (ks + 1 < ksteps) {
+      if constexpr (IMPLICIT) imp_advance(BK);
+      stage(buf ^ 1, (ks + 1) * BK);
+    }
 
     const bf16* lds_a = lds + buf * (BM + BN) * BK;
     const bf16* lds_b = lds_a + BM * BK;
