@@ -96,6 +96,7 @@ PYBIND11_MODULE(_core, m) {
   register_pose_op();
   register_color_gpu();
   register_image_encoder_op();
+  register_detector_op();
 
   // Load a user op plugin .so built with tools/build_op.py (parity:
   // Client.load_op / REGISTER_OP static registrars in user libraries,

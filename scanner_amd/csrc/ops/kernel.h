@@ -158,5 +158,6 @@ void register_optflow_gpu();
 void register_pose_op();
 void register_color_gpu();
 void register_image_encoder_op();
+void register_detector_op();
 
 }  // namespace sca

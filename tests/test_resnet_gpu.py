@@ -102,3 +102,42 @@ def test_pose_op(sc):
     b = run("b")
     for x, y in zip(a, b):
         np.testing.assert_array_equal(x, y)
+
+
+def test_detector_op(sc):
+    from conftest import make_video
+    from scanner_amd import types
+    frames = make_video(n=5, h=360, w=480)
+    video = sp.NamedVideoStream(sc, "det_v", frames=frames, codec="raw")
+
+    def run(tag):
+        frame = sc.io.Input([video])
+        det = sc.ops.Detector(frame=frame, device=sp.DeviceType.GPU)
+        out = sp.NamedStream(sc, f"det_out_{tag}")
+        sc.run(sc.io.Output(det, [out]), sp.PerfParams.manual(4, 8),
+               cache_mode=sp.CacheMode.Overwrite, gpu_ids=[0])
+        return [types.loads("BoundingBoxList", b) for b in out.load()]
+
+    a = run("a")
+    assert len(a) == 5
+    for boxes in a:
+        for b in boxes:
+            assert 0 <= b.x1 < b.x2 <= 480
+            assert 0 <= b.y1 < b.y2 <= 360
+            assert 0.0 <= b.score <= 1.0
+            assert 0 <= b.label < 8
+        # NMS invariant: no kept pair overlaps above threshold
+        for i in range(len(boxes)):
+            for j in range(i + 1, len(boxes)):
+                bi, bj = boxes[i], boxes[j]
+                ix = max(0, min(bi.x2, bj.x2) - max(bi.x1, bj.x1))
+                iy = max(0, min(bi.y2, bj.y2) - max(bi.y1, bj.y1))
+                inter = ix * iy
+                ua = ((bi.x2-bi.x1)*(bi.y2-bi.y1)
+                      + (bj.x2-bj.x1)*(bj.y2-bj.y1) - inter)
+                assert inter / max(ua, 1e-9) < 0.5
+    b = run("b")
+    for x, y in zip(a, b):  # deterministic
+        assert len(x) == len(y)
+        for p, q in zip(x, y):
+            assert p.to_bytes() == q.to_bytes()

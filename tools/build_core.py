@@ -26,6 +26,7 @@ CPP_SOURCES = [
     "csrc/ops/resnet50.cpp",
     "csrc/ops/pose.cpp",
     "csrc/ops/image_encoder.cpp",
+    "csrc/ops/detector.cpp",
     "csrc/engine/table_io.cpp",
     "csrc/engine/executor.cpp",
     "csrc/video/svc_cpu.cpp",
