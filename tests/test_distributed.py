@@ -250,3 +250,26 @@ def test_bulk_job_queueing(tmp_path):
         for w in workers:
             w.shutdown()
         master.shutdown()
+
+
+def test_straggler_task_timeout(tmp_path):
+    """Tasks exceeding the master's task timeout are failed and retried;
+    repeated timeouts blacklist the stream (reference: per-task timeout
+    master.cpp:1750-1776 + blacklist)."""
+    db = _mk_db(tmp_path)
+    master = MasterServer(db, task_timeout=1.0)
+    workers = [start_worker(master.addr, db)]
+    try:
+        sc = sp.Client(db_path=db, master=master.addr)
+        tab = sc.new_table("slow", ["col"],
+                           [[int(i).to_bytes(4, "little")] for i in range(4)])
+        col = sc.io.Input([tab])
+        slow = sc.ops.Sleep(ignore=col, ms=4000)
+        out = sp.NamedStream(sc, "slow_out")
+        with pytest.raises(sp.ScannerException, match="blacklist"):
+            sc.run(sc.io.Output(slow, [out]), sp.PerfParams.manual(2, 4),
+                   cache_mode=sp.CacheMode.Overwrite)
+    finally:
+        for w in workers:
+            w.shutdown()
+        master.shutdown()
