@@ -429,6 +429,7 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t,
       if (is_builtin_op(op.name)) continue;
       KernelConfig cfg;
       cfg.device = op.device == DeviceType::GPU ? inst.gpu : CPU_DEVICE;
+      cfg.output_device = cfg.device;
       SCA_CHECK(op.device != DeviceType::GPU || cfg.device.is_gpu(),
                 "graph requests GPU but executor has no gpu_ids");
       for (auto& e : op.inputs) cfg.input_columns.push_back(e.column);
@@ -438,6 +439,12 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t,
       cfg.max_batch = ja.info[i].batch;
       cfg.profiler = inst.profiler;
       const KernelFactory& kf = kernel_registry().get(op.name, op.device);
+      if (kf.output_device_type >= 0) {
+        cfg.output_device =
+            kf.output_device_type == (i32)DeviceType::GPU
+                ? inst.gpu
+                : CPU_DEVICE;
+      }
       inst.kernels[i] = kf.make(cfg);
       auto ait = job.op_args.find((i32)i);
       std::vector<u8> sargs =
@@ -698,7 +705,7 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t,
           for (size_t r = bi; r < be; ++r) {
             Element& e = output[c][r - bi];
             e.index = otp.compute_rows[r];
-            if (!e.is_null) e.device = kdev;
+            if (!e.is_null) e.device = kernel->config().output_device;
             if (required.count(e.index)) {
               out_map[e.index] = e;
             } else if (e.buffer) {

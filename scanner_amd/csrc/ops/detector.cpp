@@ -268,6 +268,8 @@ void register_detector_op() {
   KernelFactory f;
   f.op_name = "Detector";
   f.device_type = DeviceType::GPU;
+  // box lists are assembled host-side after the D2H of the head maps
+  f.output_device_type = (i32)DeviceType::CPU;
   f.preferred_batch = 16;
   f.make = [](const KernelConfig& c) -> std::unique_ptr<BaseKernel> {
     return std::make_unique<DetectorKernelGPU>(c);

@@ -17,6 +17,11 @@ namespace sca {
 
 struct KernelConfig {
   DeviceHandle device;
+  // Where this kernel's outputs live. Defaults to `device`; kernels
+  // registered with an explicit output device (reference:
+  // KernelBuilder.output_device(), kernel.h:412-475) may differ — e.g. a
+  // GPU detector emitting small CPU-side box lists.
+  DeviceHandle output_device;
   std::vector<std::string> input_columns;
   std::vector<std::string> output_columns;
   std::vector<u8> args;         // op-instance args (msgpack from Python)
@@ -107,6 +112,9 @@ struct OpInfo {
 struct KernelFactory {
   std::string op_name;
   DeviceType device_type = DeviceType::CPU;
+  // -1 = outputs on the kernel device (default); else a DeviceType the
+  // kernel's outputs live on (reference .output_device()).
+  i32 output_device_type = -1;
   i32 preferred_batch = 1;
   // number of devices this kernel wants (reference .num_devices());
   i32 num_devices = 1;
