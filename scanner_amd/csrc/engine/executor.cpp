@@ -150,7 +150,7 @@ void LocalExecutor::run() {
                    ? pp_.load_workers
                    : std::min<i32>(8, std::max<i32>(2, (i32)instances_.size()));
   BoundedQueue<std::shared_ptr<PreparedTask>> q_ready(
-      instances_.size() + 2);
+      std::max<size_t>((size_t)pp_.queue_size, instances_.size() + 2));
   std::vector<std::unique_ptr<Profiler>> load_profilers;
   for (i32 i = 0; i < n_load; ++i) {
     load_profilers.push_back(
