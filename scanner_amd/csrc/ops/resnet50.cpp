@@ -6,6 +6,11 @@
 // Weights are random-init (He) by default — there is no network in this
 // environment — or loaded from a tensor file for numerics tests
 // (tests/test_resnet_gpu.py compares against a PyTorch fp32 reference).
+#include <hip/hip_runtime.h>
+
+#include <cstdlib>
+#include <set>
+
 #include "../../kernels/dnn.h"
 #include "../memory.h"
 #include "../msgpack.h"
@@ -74,86 +79,134 @@ std::shared_ptr<DeviceModel> get_model(DeviceHandle dev,
   });
 }
 
+// The whole forward (~120 kernel launches per batch: preprocess, 53
+// convs incl. split-K pairs, pools, epilogues) is captured into a
+// hipGraph per batch size: per-call launch overhead collapses to one
+// hipGraphLaunch. Requirements engineered for capture: the workspace is
+// persistent per kernel instance (no allocation inside the capture), the
+// input-frame pointer array is a fixed device buffer whose CONTENTS are
+// refreshed before each launch, and the final activations land in a
+// persistent buffer copied out (bf16->f32 cast) outside the graph.
+// SCANNER_NO_HIPGRAPH=1 disables capture (debug escape hatch, like the
+// reference's NO_PIPELINING).
 class ResNet50KernelGPU : public BatchedKernel {
  public:
   explicit ResNet50KernelGPU(const KernelConfig& cfg) : BatchedKernel(cfg) {
     auto a = mp::decode(cfg.args);
     weights_file_ = a.get_str("weights_file", "");
     seed_ = (u64)a.get_int("seed", 1234);
-    batch_ = cfg.max_batch;
+    batch_ = std::max(1, cfg.max_batch);
     model_ = get_model(cfg.device, weights_file_, seed_);
+    const char* env = std::getenv("SCANNER_NO_HIPGRAPH");
+    graphs_enabled_ = !(env && env[0] == '1');
+  }
+
+  ~ResNet50KernelGPU() override {
+    for (auto& kv : graph_exec_) (void)hipGraphExecDestroy(kv.second);
+    if (ws_.d_ptrs && memory_initialized() &&
+        ws_.generation == memory_generation()) {
+      DeviceHandle dev = config_.device;
+      for (u8* b : {ws_.d_ptrs, ws_.act0, ws_.act1, ws_.act2, ws_.resid,
+                    ws_.colbuf, ws_.pre, ws_.skbuf})
+        delete_buffer(dev, b);
+      delete_buffer(CPU_DEVICE, (u8*)ws_.h_ptrs);
+    }
   }
 
   void execute_batch(const BatchedElements& in, BatchedElements& out) override;
 
  private:
+  static constexpr size_t kSplitkBytes = 64u << 20;
+
+  struct Workspace {
+    u8* d_ptrs = nullptr;
+    const u8** h_ptrs = nullptr;  // pinned staging for the pointer array
+    u8* act0 = nullptr;
+    u8* act1 = nullptr;
+    u8* act2 = nullptr;
+    u8* resid = nullptr;
+    u8* colbuf = nullptr;
+    u8* pre = nullptr;
+    u8* skbuf = nullptr;
+    int ih = 0, iw = 0, ic = 0;
+    u64 generation = 0;
+  };
+
+  void ensure_workspace(int ih, int iw, int ic);
+  // Enqueue the full forward for n frames on stream s; returns the buffer
+  // holding the padded fc output rows ([n][1024] bf16).
+  u8* run_forward(int n, void* s);
+
   std::string weights_file_;
   u64 seed_;
   i32 batch_;
+  bool graphs_enabled_;
   std::shared_ptr<DeviceModel> model_;
+  Workspace ws_;
+  std::map<int, hipGraphExec_t> graph_exec_;  // batch size -> graph
+  std::map<int, u8*> cached_final_;           // batch size -> fc output buf
+  std::set<int> warmed_;  // first run per batch size goes uncaptured
 };
 
-void ResNet50KernelGPU::execute_batch(const BatchedElements& in,
-                                      BatchedElements& out) {
-  void* s = per_thread_hip_stream();
+void ResNet50KernelGPU::ensure_workspace(int ih, int iw, int ic) {
   DeviceHandle dev = config_.device;
-  int n = (int)in[0].size();
-  if (n == 0) return;
-  const Element& f0 = in[0][0];
-  SCA_CHECK(f0.is_frame && f0.device.is_gpu(),
-            "ResNet50 needs GPU frame input");
-  int ih = f0.frame_info.shape[0], iw = f0.frame_info.shape[1],
-      ic = f0.frame_info.shape[2];
-
-  // device pointer array for the input frames
-  std::vector<const u8*> ptrs(n);
-  for (int i = 0; i < n; ++i) ptrs[i] = in[0][i].buffer;
-  u8* d_ptrs = new_buffer(dev, n * sizeof(u8*));
-  memcpy_buffer(d_ptrs, dev, (const u8*)ptrs.data(), CPU_DEVICE,
-                n * sizeof(u8*));
-
-  // workspace (act elements peak: n*112*112*64 == n*802816)
+  if (ws_.d_ptrs && (ws_.ih != ih || ws_.iw != iw || ws_.ic != ic)) {
+    // geometry changed: captured preprocess is stale
+    for (auto& kv : graph_exec_) (void)hipGraphExecDestroy(kv.second);
+    graph_exec_.clear();
+    cached_final_.clear();
+    warmed_.clear();
+    ws_.ih = ih;
+    ws_.iw = iw;
+    ws_.ic = ic;
+    return;
+  }
+  if (ws_.d_ptrs) return;
+  int n = batch_;
   size_t act_elems = (size_t)n * 802816;
   size_t im2col_elems = (size_t)n * 112 * 112 * 192;  // conv1 is the max
-  u8* act0 = new_buffer(dev, act_elems * 2);
-  u8* act1 = new_buffer(dev, act_elems * 2);
-  u8* act2 = new_buffer(dev, act_elems * 2);
-  u8* resid = new_buffer(dev, act_elems * 2);
-  u8* colbuf = new_buffer(dev, im2col_elems * 2);
-  u8* pre = new_buffer(dev, (size_t)n * 224 * 224 * 3 * 2);
-  // split-K f32 partials for the launch-bound stage-3/4 convs + fc
-  constexpr size_t kSplitkBytes = 64u << 20;
-  u8* skbuf = new_buffer(dev, kSplitkBytes);
+  ws_.d_ptrs = new_buffer(dev, n * sizeof(u8*));
+  ws_.h_ptrs = (const u8**)new_buffer(CPU_DEVICE, n * sizeof(u8*));
+  ws_.act0 = new_buffer(dev, act_elems * 2);
+  ws_.act1 = new_buffer(dev, act_elems * 2);
+  ws_.act2 = new_buffer(dev, act_elems * 2);
+  ws_.resid = new_buffer(dev, act_elems * 2);
+  ws_.colbuf = new_buffer(dev, im2col_elems * 2);
+  ws_.pre = new_buffer(dev, (size_t)n * 224 * 224 * 3 * 2);
+  ws_.skbuf = new_buffer(dev, kSplitkBytes);
+  ws_.ih = ih;
+  ws_.iw = iw;
+  ws_.ic = ic;
+  ws_.generation = memory_generation();
+}
 
+u8* ResNet50KernelGPU::run_forward(int n, void* s) {
   auto conv = [&](const char* name, const u8* x, int h, int w, u8* y,
-                  const u8* residual, int& oh, int& ow) -> int {
+                  const u8* residual, int& oh, int& ow) {
     const auto& e = model_->convs[model_->by_name.at(name)];
     const ConvSpec& sp = e.spec;
     oh = (h + 2 * sp.pad - sp.r) / sp.stride + 1;
     ow = (w + 2 * sp.pad - sp.s) / sp.stride + 1;
-    int M = n * oh * ow;
     const u8* A = x;
     bool direct = sp.r == 1 && sp.s == 1 && sp.stride == 1 && sp.pad == 0;
-    // Implicit GEMM for spatial convs with c % 8 == 0: the GEMM stages
-    // im2col rows straight from the activation tensor (no HBM round trip).
     bool implicit = !direct && sp.in_c % 8 == 0;
     if (!direct && !implicit) {
       im2col_bf16(x, n, h, w, sp.in_c, sp.r, sp.s, sp.stride, sp.pad,
-                  colbuf, oh, ow, sp.kp(), s);
-      A = colbuf;
+                  ws_.colbuf, oh, ow, sp.kp(), s);
+      A = ws_.colbuf;
     }
     GemmArgs g;
     g.A = A;
     g.B = model_->weights + e.w_off * 2;
     g.C = y;
-    g.M = M;
+    g.M = n * oh * ow;
     g.N = sp.np();
     g.K = sp.kp();
     g.scale = (const float*)model_->scalebias + e.sb_off;
     g.bias = (const float*)model_->scalebias + e.sb_off + sp.np();
     g.residual = residual;
     g.relu = sp.relu;
-    g.splitk_scratch = skbuf;
+    g.splitk_scratch = ws_.skbuf;
     g.splitk_scratch_bytes = kSplitkBytes;
     if (implicit) {
       ConvDesc d{n, h, w, sp.in_c, sp.r, sp.s, sp.stride, sp.pad, oh, ow};
@@ -161,22 +214,21 @@ void ResNet50KernelGPU::execute_batch(const BatchedElements& in,
     } else {
       gemm_bf16(g, s);
     }
-    return sp.np();
   };
 
-  // ---- forward ----
   f32* mean = model_->mean;
-  preprocess_frames_bf16(d_ptrs, n, ih, iw, ic, 224, pre, mean, mean + 3, s);
+  preprocess_frames_bf16(ws_.d_ptrs, n, ws_.ih, ws_.iw, ws_.ic, 224, ws_.pre,
+                         mean, mean + 3, s);
   int h = 224, w = 224, oh, ow;
-  conv("conv1", pre, h, w, act0, nullptr, oh, ow);  // -> n,112,112,64
+  conv("conv1", ws_.pre, h, w, ws_.act0, nullptr, oh, ow);
   h = oh;
   w = ow;
-  maxpool3x3s2_bf16(act0, n, h, w, 64, act1, 56, 56, s);
+  maxpool3x3s2_bf16(ws_.act0, n, h, w, 64, ws_.act1, 56, 56, s);
   h = w = 56;
   // Three rotating activation buffers: x holds the block input; a/b are
   // the two others. conv3 writes into a (its conv1 temp is dead by then),
   // never into the residual source.
-  u8* bufs[3] = {act1, act0, act2};
+  u8* bufs[3] = {ws_.act1, ws_.act0, ws_.act2};
   u8* x = bufs[0];
   ResNet50Config cfg;
   for (size_t b = 0; b < cfg.blocks.size(); ++b) {
@@ -190,8 +242,8 @@ void ResNet50KernelGPU::execute_batch(const BatchedElements& in,
     u8* bbuf = others[1];
     const u8* identity = x;
     if (bk.downsample) {
-      conv((p + ".downsample").c_str(), x, h, w, resid, nullptr, oh, ow);
-      identity = resid;
+      conv((p + ".downsample").c_str(), x, h, w, ws_.resid, nullptr, oh, ow);
+      identity = ws_.resid;
     }
     conv((p + ".conv1").c_str(), x, h, w, a, nullptr, oh, ow);
     conv((p + ".conv2").c_str(), a, h, w, bbuf, nullptr, oh, ow);
@@ -203,7 +255,6 @@ void ResNet50KernelGPU::execute_batch(const BatchedElements& in,
   // x: [n,7,7,2048]
   u8* y = bufs[0] == x ? bufs[1] : bufs[0];
   global_avgpool_bf16(x, n, 7, 7, 2048, y, s);
-  // fc: M=n, K=2048, N=1024 (padded from 1000)
   {
     const auto& e = model_->convs[model_->by_name.at("fc")];
     GemmArgs g;
@@ -216,25 +267,76 @@ void ResNet50KernelGPU::execute_batch(const BatchedElements& in,
     g.scale = (const float*)model_->scalebias + e.sb_off;
     g.bias = (const float*)model_->scalebias + e.sb_off + e.spec.np();
     g.relu = false;
-    g.splitk_scratch = skbuf;
+    g.splitk_scratch = ws_.skbuf;
     g.splitk_scratch_bytes = kSplitkBytes;
     gemm_bf16(g, s);
+  }
+  return x;  // [n][1024] bf16 rows (buffer rotation is n-independent)
+}
+
+void ResNet50KernelGPU::execute_batch(const BatchedElements& in,
+                                      BatchedElements& out) {
+  void* s = per_thread_hip_stream();
+  DeviceHandle dev = config_.device;
+  int n = (int)in[0].size();
+  if (n == 0) return;
+  SCA_CHECK(n <= batch_, "batch exceeds kernel max_batch");
+  const Element& f0 = in[0][0];
+  SCA_CHECK(f0.is_frame && f0.device.is_gpu(),
+            "ResNet50 needs GPU frame input");
+  ensure_workspace(f0.frame_info.shape[0], f0.frame_info.shape[1],
+                   f0.frame_info.shape[2]);
+
+  // Refresh the pointer-array CONTENTS; the captured graph reads from the
+  // fixed ws_.d_ptrs address.
+  for (int i = 0; i < n; ++i) ws_.h_ptrs[i] = in[0][i].buffer;
+  hipError_t he =
+      hipMemcpyAsync(ws_.d_ptrs, ws_.h_ptrs, n * sizeof(u8*),
+                     hipMemcpyHostToDevice, (hipStream_t)s);
+  SCA_CHECK(he == hipSuccess, "resnet pointer upload failed");
+
+  u8* final_buf = nullptr;
+  auto it = graph_exec_.find(n);
+  if (graphs_enabled_ && it != graph_exec_.end()) {
+    SCA_CHECK(hipGraphLaunch(it->second, (hipStream_t)s) == hipSuccess,
+              "hipGraphLaunch failed");
+    final_buf = cached_final_[n];
+  } else if (graphs_enabled_ && warmed_.count(n)) {
+    // second occurrence of this batch size: capture
+    hipGraph_t graph = nullptr;
+    SCA_CHECK(hipStreamBeginCapture((hipStream_t)s,
+                                    hipStreamCaptureModeThreadLocal) ==
+                  hipSuccess,
+              "hipStreamBeginCapture failed");
+    u8* fb = nullptr;
+    try {
+      fb = run_forward(n, s);
+    } catch (...) {
+      (void)hipStreamEndCapture((hipStream_t)s, &graph);
+      if (graph) (void)hipGraphDestroy(graph);
+      throw;
+    }
+    SCA_CHECK(hipStreamEndCapture((hipStream_t)s, &graph) == hipSuccess,
+              "hipStreamEndCapture failed");
+    hipGraphExec_t exec = nullptr;
+    hipError_t ie = hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0);
+    (void)hipGraphDestroy(graph);
+    SCA_CHECK(ie == hipSuccess, "hipGraphInstantiate failed");
+    graph_exec_[n] = exec;
+    cached_final_[n] = fb;
+    SCA_CHECK(hipGraphLaunch(exec, (hipStream_t)s) == hipSuccess,
+              "hipGraphLaunch failed");
+    final_buf = fb;
+  } else {
+    final_buf = run_forward(n, s);
+    warmed_.insert(n);
   }
 
   // logits: first 1000 of each padded row, bf16 -> f32, one block buffer
   size_t logit_bytes = (size_t)n * 1000 * 4;
   u8* out_block = new_block_buffer(dev, logit_bytes, n);
-  bf16_rows_to_f32(x, n, 1024, 1000, out_block, s);
-
+  bf16_rows_to_f32(final_buf, n, 1024, 1000, out_block, s);
   sync_per_thread_stream();
-  delete_buffer(dev, d_ptrs);
-  delete_buffer(dev, act0);
-  delete_buffer(dev, act1);
-  delete_buffer(dev, act2);
-  delete_buffer(dev, resid);
-  delete_buffer(dev, colbuf);
-  delete_buffer(dev, pre);
-  delete_buffer(dev, skbuf);
 
   for (int i = 0; i < n; ++i) {
     Element e;

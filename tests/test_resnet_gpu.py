@@ -141,3 +141,27 @@ def test_detector_op(sc):
         assert len(x) == len(y)
         for p, q in zip(x, y):
             assert p.to_bytes() == q.to_bytes()
+
+
+def test_resnet50_hipgraph_consistent(sc):
+    """Graph-captured forwards must produce identical logits to the eager
+    warm-up runs (first packet runs uncaptured, the second captures, later
+    packets replay the graph)."""
+    from conftest import make_video
+    frames = make_video(n=12, h=240, w=320)
+    video = sp.NamedVideoStream(sc, "hg_v", frames=frames, codec="raw")
+
+    def run(tag):
+        frame = sc.io.Input([video])
+        logits = sc.ops.ResNet50(frame=frame, device=sp.DeviceType.GPU,
+                                 batch=4)
+        out = sp.NamedStream(sc, f"hg_out_{tag}")
+        sc.run(sc.io.Output(logits, [out]), sp.PerfParams.manual(4, 12),
+               cache_mode=sp.CacheMode.Overwrite, gpu_ids=[0],
+               pipeline_instances=1)
+        return np.stack([np.frombuffer(b, np.float32) for b in out.load()])
+
+    a = run("a")  # packets: warm (uncaptured), capture, replay
+    b = run("b")  # all replays
+    assert a.shape == b.shape == (12, 1000)
+    np.testing.assert_array_equal(a, b)
