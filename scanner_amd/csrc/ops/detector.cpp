@@ -87,6 +87,12 @@ std::vector<Box> nms_best(std::vector<Box> boxes, float thresh) {
 
 class DetectorKernelGPU : public BatchedKernel {
  public:
+  ~DetectorKernelGPU() override {
+    if (bufs_[0] && memory_initialized() && gen_ == memory_generation()) {
+      for (u8* b : bufs_) delete_buffer(config_.device, b);
+    }
+  }
+
   explicit DetectorKernelGPU(const KernelConfig& cfg) : BatchedKernel(cfg) {
     auto a = mp::decode(cfg.args);
     weights_file_ = a.get_str("weights_file", "");
@@ -114,20 +120,35 @@ class DetectorKernelGPU : public BatchedKernel {
     int ih = f0.frame_info.shape[0], iw = f0.frame_info.shape[1],
         ic = f0.frame_info.shape[2];
 
+    // Persistent workspace sized for max_batch.
+    int nb = std::max(n, std::max(1, config_.max_batch));
+    size_t hw0 = (size_t)kInHW * kInHW;
+    if (!bufs_[0]) {
+      size_t fpix_b = (size_t)nb * kFeat * kFeat;
+      bufs_[0] = new_buffer(dev, (size_t)nb * sizeof(u8*));   // d_ptrs
+      bufs_[1] = new_buffer(dev, (size_t)nb * hw0 * 3 * 2);   // pre
+      bufs_[2] = new_buffer(dev, (size_t)nb * hw0 * 64 * 2);  // actA
+      bufs_[3] = new_buffer(dev, (size_t)nb * hw0 * 64 * 2);  // actB
+      bufs_[4] = new_buffer(dev, (size_t)nb * hw0 * 64 * 2);  // colbuf (b1)
+      bufs_[5] = new_buffer(dev, fpix_b * 64 * 2);            // loc
+      bufs_[6] = new_buffer(dev, fpix_b * 64 * 2);            // cls
+      bufs_[7] = new_buffer(dev, fpix_b * 64 * 4);            // loc_f
+      bufs_[8] = new_buffer(dev, fpix_b * 64 * 4);            // cls_f
+      gen_ = memory_generation();
+    }
+    u8* d_ptrs = bufs_[0];
+    u8* pre = bufs_[1];
+    u8* actA = bufs_[2];
+    u8* actB = bufs_[3];
+    u8* colbuf = bufs_[4];
+    u8* loc = bufs_[5];
+    u8* cls = bufs_[6];
+    size_t featpix = (size_t)n * kFeat * kFeat;
+
     std::vector<const u8*> ptrs(n);
     for (int i = 0; i < n; ++i) ptrs[i] = in[0][i].buffer;
-    u8* d_ptrs = new_buffer(dev, n * sizeof(u8*));
     memcpy_buffer(d_ptrs, dev, (const u8*)ptrs.data(), CPU_DEVICE,
                   n * sizeof(u8*));
-
-    size_t hw0 = (size_t)kInHW * kInHW;
-    size_t featpix = (size_t)n * kFeat * kFeat;
-    u8* pre = new_buffer(dev, (size_t)n * hw0 * 3 * 2);
-    u8* actA = new_buffer(dev, (size_t)n * hw0 * 64 * 2);
-    u8* actB = new_buffer(dev, (size_t)n * hw0 * 64 * 2);
-    u8* colbuf = new_buffer(dev, (size_t)n * hw0 * 64 * 2);  // b1 only
-    u8* loc = new_buffer(dev, featpix * 64 * 2);   // np(16) = 64
-    u8* cls = new_buffer(dev, featpix * 64 * 2);   // np(32) = 64
 
     auto conv = [&](const std::string& name, const u8* x, int h, int w,
                     u8* y, int& oh, int& ow) {
@@ -176,8 +197,8 @@ class DetectorKernelGPU : public BatchedKernel {
 
     // Small maps: bring both heads to the host as f32 and post-process.
     i64 cells = (i64)n * kFeat * kFeat;
-    u8* loc_f = new_buffer(dev, cells * 64 * 4);
-    u8* cls_f = new_buffer(dev, cells * 64 * 4);
+    u8* loc_f = bufs_[7];
+    u8* cls_f = bufs_[8];
     bf16_rows_to_f32(loc, cells, 64, 64, loc_f, s);
     bf16_rows_to_f32(cls, cells, 64, 64, cls_f, s);
     std::vector<f32> loc_h(cells * 64), cls_h(cells * 64);
@@ -186,9 +207,6 @@ class DetectorKernelGPU : public BatchedKernel {
                   loc_h.size() * 4);
     memcpy_buffer((u8*)cls_h.data(), CPU_DEVICE, cls_f, dev,
                   cls_h.size() * 4);
-
-    for (u8* b : {d_ptrs, pre, actA, actB, colbuf, loc, cls, loc_f, cls_f})
-      delete_buffer(dev, b);
 
     // Per frame: sigmoid scores, decode anchors, NMS, pack.
     float sx = (float)iw / kInHW, sy = (float)ih / kInHW;
@@ -252,6 +270,8 @@ class DetectorKernelGPU : public BatchedKernel {
   u64 seed_;
   float score_thresh_, nms_thresh_;
   std::shared_ptr<DeviceModel> model_;
+  u8* bufs_[9] = {};
+  u64 gen_ = 0;
 };
 
 }  // namespace

@@ -85,6 +85,12 @@ class PoseKernelGPU : public BatchedKernel {
                             });
   }
 
+  ~PoseKernelGPU() override {
+    if (bufs_[0] && memory_initialized() && gen_ == memory_generation()) {
+      for (u8* b : bufs_) delete_buffer(config_.device, b);
+    }
+  }
+
   void execute_batch(const BatchedElements& in, BatchedElements& out) override {
     void* s = per_thread_hip_stream();
     DeviceHandle dev = config_.device;
@@ -95,26 +101,43 @@ class PoseKernelGPU : public BatchedKernel {
     int ih = f0.frame_info.shape[0], iw = f0.frame_info.shape[1],
         ic = f0.frame_info.shape[2];
 
+    // Persistent workspace sized for max_batch (no per-execute pool
+    // traffic). Peak activation: b1 output n x 368^2 x 64; peak im2col:
+    // stage concat convs.
+    int nb = std::max(n, std::max(1, config_.max_batch));
+    size_t hw0 = (size_t)kInHW * kInHW;
+    if (!bufs_[0]) {
+      size_t fpix_b = (size_t)nb * kFeatHW * kFeatHW;
+      bufs_[0] = new_buffer(dev, (size_t)nb * sizeof(u8*));       // d_ptrs
+      bufs_[1] = new_buffer(dev, (size_t)nb * hw0 * 3 * 2);       // pre
+      bufs_[2] = new_buffer(dev, (size_t)nb * hw0 * 64 * 2);      // actA
+      bufs_[3] = new_buffer(dev, (size_t)nb * hw0 * 64 * 2);      // actB
+      bufs_[4] = new_buffer(dev, (size_t)nb * (hw0 / 4) * 640 * 2);
+      bufs_[5] = new_buffer(dev, fpix_b * 128 * 2);               // feat
+      bufs_[6] = new_buffer(dev, fpix_b * 64 * 2);                // brL
+      bufs_[7] = new_buffer(dev, fpix_b * 64 * 2);                // brS
+      bufs_[8] = new_buffer(dev, fpix_b * 512 * 2);               // brT
+      bufs_[9] = new_buffer(dev, fpix_b * 512 * 2);               // brU
+      bufs_[10] = new_buffer(dev, fpix_b * 192 * 2);              // cat
+      gen_ = memory_generation();
+    }
+    u8* d_ptrs = bufs_[0];
+    u8* pre = bufs_[1];
+    u8* actA = bufs_[2];
+    u8* actB = bufs_[3];
+    u8* colbuf = bufs_[4];
+    u8* feat = bufs_[5];
+    u8* brL = bufs_[6];
+    u8* brS = bufs_[7];
+    u8* brT = bufs_[8];
+    u8* brU = bufs_[9];
+    u8* cat = bufs_[10];
+    size_t featpix = (size_t)n * kFeatHW * kFeatHW;
+
     std::vector<const u8*> ptrs(n);
     for (int i = 0; i < n; ++i) ptrs[i] = in[0][i].buffer;
-    u8* d_ptrs = new_buffer(dev, n * sizeof(u8*));
     memcpy_buffer(d_ptrs, dev, (const u8*)ptrs.data(), CPU_DEVICE,
                   n * sizeof(u8*));
-
-    // Workspace. Peak activation: b1 output n x 368^2 x 64; peak im2col:
-    // b2 (downsampled writes: n x 184^2 x kp(64*9)=576).
-    size_t hw0 = (size_t)kInHW * kInHW;
-    size_t featpix = (size_t)n * kFeatHW * kFeatHW;
-    u8* pre = new_buffer(dev, (size_t)n * hw0 * 3 * 2);
-    u8* actA = new_buffer(dev, (size_t)n * hw0 * 64 * 2);
-    u8* actB = new_buffer(dev, (size_t)n * hw0 * 64 * 2);
-    u8* colbuf = new_buffer(dev, (size_t)n * (hw0 / 4) * 640 * 2);
-    u8* feat = new_buffer(dev, featpix * 128 * 2);
-    u8* brL = new_buffer(dev, featpix * 64 * 2);   // np(38) = 64
-    u8* brS = new_buffer(dev, featpix * 64 * 2);   // np(19) = 64
-    u8* brT = new_buffer(dev, featpix * 512 * 2);  // stage mid buffers
-    u8* brU = new_buffer(dev, featpix * 512 * 2);
-    u8* cat = new_buffer(dev, featpix * 192 * 2);
 
     auto conv = [&](const std::string& name, const u8* x, int h, int w,
                     u8* y, int& oh, int& ow) {
@@ -197,10 +220,6 @@ class PoseKernelGPU : public BatchedKernel {
     heatmap_argmax(brS, n, kFeatHW, kFeatHW, 64, kHeatC, out_block, s);
 
     sync_per_thread_stream();
-    for (u8* b : {d_ptrs, pre, actA, actB, colbuf, feat, brL, brS, brT, brU,
-                  cat}) {
-      delete_buffer(dev, b);
-    }
 
     for (int i = 0; i < n; ++i) {
       Element e;
@@ -216,6 +235,8 @@ class PoseKernelGPU : public BatchedKernel {
   u64 seed_;
   int stages_;
   std::shared_ptr<DeviceModel> model_;
+  u8* bufs_[11] = {};
+  u64 gen_ = 0;
 };
 
 }  // namespace
