@@ -657,3 +657,34 @@ def test_stencil_after_stride(sc):
     assert len(rows) == 4  # ceil(12/3)
     # mean |u| of the first window ~ 3 px (sampled-domain neighbors)
     assert 2.0 < rows[0][0] < 4.0, rows[0]
+
+
+def test_recovery_drops_uncommitted(sc, tmp_path):
+    """Startup recovery garbage-collects uncommitted output tables
+    (reference: recover_and_init_database master.cpp:1311-1327) while
+    committed tables survive."""
+    import msgpack
+    from scanner_amd import _core
+
+    frames = make_video(n=8)
+    video = sp.NamedVideoStream(sc, "rec_in", frames=frames, codec="raw")
+
+    # Build an executor whose prepare() creates the uncommitted output
+    # table, then "crash" (drop it without running).
+    frame = sc.io.Input([video])
+    hist = sc.ops.Histogram(frame=frame)
+    out = sp.NamedStream(sc, "rec_out")
+    sink = sc.io.Output(hist, [out])
+    graph_bytes, jobs, _, _ = sc._assemble(sink)
+    ex = _core.LocalExecutor(sc._db, graph_bytes, msgpack.packb(jobs),
+                             sp.PerfParams.manual(4, 8).to_dict(1), [])
+    ex.prepare(True)
+    del ex
+    assert sc.has_table("rec_out")
+    assert not sc._db.table_committed("rec_out")
+
+    # A fresh client with recovery drops the uncommitted table, keeps the
+    # committed input.
+    sc2 = sp.Client(db_path=sc._db_path)
+    assert sc2.has_table("rec_in")
+    assert not sc2.has_table("rec_out")
