@@ -257,7 +257,31 @@ class Client:
                                   or (len(gpu_ids) if gpu_ids else 1))
         ex = _core.LocalExecutor(self._db, graph_bytes, jobs_bytes,
                                  perf.to_dict(pipeline_instances), gpu_ids)
-        ex.run()
+        if show_progress:
+            # run() releases the GIL; poll task completion meanwhile
+            # (parity: wait_on_job progress bar, client.py:1188-1261)
+            import sys
+            import threading
+            ex.prepare(True)
+            total = len(ex.all_tasks())
+            done = threading.Event()
+
+            def _progress():
+                while not done.wait(0.5):
+                    print(f"\r[scanner] {ex.tasks_done()}/{total} tasks",
+                          end="", file=sys.stderr, flush=True)
+                print(f"\r[scanner] {ex.tasks_done()}/{total} tasks",
+                      file=sys.stderr, flush=True)
+
+            t = threading.Thread(target=_progress, daemon=True)
+            t.start()
+            try:
+                ex.run()
+            finally:
+                done.set()
+                t.join()
+        else:
+            ex.run()
         self._last_profilers = ex.profilers()
         return Profile(self._last_profilers)
 

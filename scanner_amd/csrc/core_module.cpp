@@ -543,6 +543,7 @@ PYBIND11_MODULE(_core, m) {
              ex.process_prepared_public(instance, pt);
            })
       .def("finalize_job", &LocalExecutor::finalize_job)
+      .def("tasks_done", &LocalExecutor::tasks_done)
       .def("total_output_rows", &LocalExecutor::total_output_rows)
       .def("profilers",
            [](LocalExecutor& ex) { return profilers_to_py(ex.profilers()); });

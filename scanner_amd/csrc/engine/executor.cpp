@@ -285,6 +285,7 @@ std::shared_ptr<PreparedTask> LocalExecutor::prepare_task(const TaskDesc& t,
                               ir.item),
             lo, hi - lo, sp.buf);
       }
+      prof->increment("io_read_bytes", (i64)(hi - lo));
       pt->spans.push_back(sp);
     }
   }
@@ -345,9 +346,11 @@ void LocalExecutor::load_inputs(
                 db_->paths().item(table.id, table.column_id(src.column),
                                   ir.item),
                 lo, hi - lo, stream_buf);
+            inst.profiler->increment("io_read_bytes", (i64)(hi - lo));
           }
           if (inst.gpu.is_gpu()) {
             Profiler::Scope sd(inst.profiler, "decode:gpu");
+            inst.profiler->increment("decoded_frames", (i64)local.size());
             auto elems =
                 svc_decode_gpu(stream_buf, hi - lo, m, local, inst.gpu, lo);
             for (auto& e : elems) {
@@ -611,12 +614,18 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t,
             std::vector<u8> stream;
             svc_encode_cpu(contig.data(), nf, h, w, ch, /*gop=*/16, stream,
                            vm);
+            inst.profiler->increment("io_write_bytes", (i64)stream.size());
             write_video_item(*db_, table, cname, t.task, stream, vm);
             for (size_t k = 0; k < cols[c].size(); ++k) {
               if (owned[c][k] && cols[c][k].buffer)
                 delete_buffer(CPU_DEVICE, cols[c][k].buffer);
             }
             continue;
+          }
+          {
+            i64 wb = 0;
+            for (auto& e : cols[c]) wb += (i64)e.size;
+            inst.profiler->increment("io_write_bytes", wb);
           }
           write_column_item(*db_, table, table.columns[c].name, t.task,
                             cols[c]);
@@ -729,6 +738,7 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t,
   free_all();
   inst.profiler->increment("tasks", 1);
   inst.profiler->increment("rows", t.end - t.start);
+  tasks_done_.fetch_add(1);
 }
 
 }  // namespace sca

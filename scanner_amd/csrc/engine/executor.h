@@ -85,6 +85,7 @@ class LocalExecutor {
     return profilers_;
   }
   i64 total_output_rows() const;
+  i64 tasks_done() const { return tasks_done_.load(); }
 
  private:
   struct Instance;
@@ -111,6 +112,7 @@ class LocalExecutor {
   std::vector<std::vector<i64>> task_rows_; // per job: cumulative task ends
   bool prepared_ = false;
 
+  std::atomic<i64> tasks_done_{0};
   std::vector<std::unique_ptr<Instance>> instances_;
   std::vector<std::unique_ptr<Profiler>> profilers_;
 };

@@ -688,3 +688,19 @@ def test_recovery_drops_uncommitted(sc, tmp_path):
     sc2 = sp.Client(db_path=sc._db_path)
     assert sc2.has_table("rec_in")
     assert not sc2.has_table("rec_out")
+
+
+def test_profiler_io_counters(sc):
+    """Profiler counters record IO byte totals (reference: profiler
+    increment counters, e.g. io_write column_sink.cpp:198)."""
+    frames = make_video(n=8)
+    video = sp.NamedVideoStream(sc, "ctr", frames=frames, codec="svc")
+    frame = sc.io.Input([video])
+    hist = sc.ops.Histogram(frame=frame)
+    out = sp.NamedStream(sc, "ctr_out")
+    prof = sc.run(sc.io.Output(hist, [out]), sp.PerfParams.manual(4, 8),
+                  cache_mode=sp.CacheMode.Overwrite)
+    counters = prof.counters()
+    assert counters.get("io_read_bytes", 0) > 0
+    assert counters.get("io_write_bytes", 0) >= 8 * 3 * 256 * 4
+    assert counters.get("rows", 0) == 8
