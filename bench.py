@@ -119,8 +119,12 @@ def main():
         io_packet_size=int(os.environ.get("SCANNER_BENCH_IO", "64")),
         gpu_pool=(24 << 30) if have_gpu else 0,
         cpu_pool=(4 << 30) if have_gpu else 0)
+    # measured sweet spots: IO-bound pipelines (hist/full/resnet) want 6
+    # instances to overlap reads/H2D; compute-saturated ones (flow/pose)
+    # want 4 (more just thrash the CUs)
+    default_inst = "4" if args.pipeline in ("flow", "pose") else "6"
     instances = int(os.environ.get("SCANNER_BENCH_INSTANCES",
-                                   "6" if have_gpu else "1"))
+                                   default_inst if have_gpu else "1"))
 
     def one_step(tag):
         sink = build_pipeline(sc, sp, video, args.pipeline, device,
