@@ -11,6 +11,7 @@
 #include <hip/hip_runtime.h>
 
 #include <algorithm>
+#include <cstdlib>
 #include <map>
 #include <mutex>
 
@@ -557,16 +558,37 @@ __global__ void __launch_bounds__(256)
                          const float* __restrict__ bias,
                          const bf16* __restrict__ residual,
                          bf16* __restrict__ C) {
+  // 4 elements per thread via float4 (N is a multiple of 64, so mn % 4 ==
+  // 0 and every row stays 16-byte aligned): the scalar version measured
+  // only ~2.2 TB/s on S x M x N partial streams.
+  i64 quads = mn / 4;
   i64 gs = (i64)gridDim.x * blockDim.x;
-  for (i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x; i < mn; i += gs) {
-    float v = 0.f;
-    for (int sp = 0; sp < splits; ++sp) v += partials[(size_t)sp * mn + i];
-    int col = (int)(i % N);
-    if (scale) v *= scale[col];
-    if (bias) v += bias[col];
-    if constexpr (RESIDUAL) v += bf16_to_f32(residual[i]);
-    if constexpr (RELU) v = v > 0.f ? v : 0.f;
-    C[i] = f32_to_bf16(v);
+  for (i64 q = (i64)blockIdx.x * blockDim.x + threadIdx.x; q < quads;
+       q += gs) {
+    i64 i = q * 4;
+    float4 acc = make_float4(0.f, 0.f, 0.f, 0.f);
+    for (int sp = 0; sp < splits; ++sp) {
+      float4 p = reinterpret_cast<const float4*>(
+          partials + (size_t)sp * mn)[q];
+      acc.x += p.x;
+      acc.y += p.y;
+      acc.z += p.z;
+      acc.w += p.w;
+    }
+    float v4[4] = {acc.x, acc.y, acc.z, acc.w};
+    bf16 o4[4];
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      float v = v4[k];
+      int col = (int)((i + k) % N);
+      if (scale) v *= scale[col];
+      if (bias) v += bias[col];
+      if constexpr (RESIDUAL) v += bf16_to_f32(residual[i + k]);
+      if constexpr (RELU) v = v > 0.f ? v : 0.f;
+      o4[k] = f32_to_bf16(v);
+    }
+    *reinterpret_cast<uint64_t*>(C + i) =
+        *reinterpret_cast<const uint64_t*>(o4);
   }
 }
 
@@ -579,8 +601,12 @@ bool try_splitk(const GemmArgs& g, hipStream_t s,
   int ntiles_m = (g.M + BM - 1) / BM;
   int tiles = ntiles_m * (g.N / BN);
   if (tiles >= 256) return false;  // already fills the chip
+  static const int kMaxSplits = []() {
+    const char* e = std::getenv("SCANNER_SPLITK_MAX");
+    return e ? std::max(2, atoi(e)) : 16;
+  }();
   int ksteps = g.K / 64;
-  int want = std::min({16, ksteps / 4, (2048 + tiles - 1) / tiles});
+  int want = std::min({kMaxSplits, ksteps / 4, (2048 + tiles - 1) / tiles});
   size_t per_split = (size_t)g.M * g.N * 4;
   int fit = (int)(g.splitk_scratch_bytes / per_split);
   int splits = std::min(want, fit);
