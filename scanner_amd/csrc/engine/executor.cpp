@@ -3,6 +3,7 @@
 #include <algorithm>
 #include <set>
 
+#include "../hip_util.h"
 #include "../memory.h"
 #include "../video/svc.h"
 
@@ -411,6 +412,14 @@ void LocalExecutor::load_inputs(
 
 void LocalExecutor::process_task(Instance& inst, const TaskDesc& t,
                                  PreparedTask* prep) {
+  // HIP device selection is per-thread: instance threads (std::thread in
+  // run(), Python threads in the distributed worker) start on device 0,
+  // which is wrong for every rank but rank 0 on a multi-GPU node. Pin this
+  // thread to the instance's GPU before any stream/kernel use (sticky on
+  // purpose — the thread serves this instance for its lifetime).
+  if (inst.gpu.is_gpu()) {
+    (void)hipSetDevice(inst.gpu.id);
+  }
   const JobBinding& job = jobs_[t.job];
   const JobAnalysis& ja = analyses_[t.job];
   size_t n = graph_.ops.size();
