@@ -219,3 +219,23 @@ def test_gpu_4k_decode_histogram(sc):
     for i, blob in enumerate(rows):
         got = np.frombuffer(blob, dtype=np.uint32).reshape(3, 256)
         np.testing.assert_array_equal(got, ref_histogram(frames[i]))
+
+
+def test_gpu_svc_decode_odd_shapes(sc):
+    """GPU SVC decode on odd dimensions (tail groups, non-multiple-of-32
+    frame bytes) must match ingest exactly."""
+    rng = np.random.RandomState(7)
+    for i, (h, w) in enumerate([(13, 17), (37, 51), (9, 127)]):
+        frames = rng.randint(0, 256, size=(11, h, w, 3)).astype(np.uint8)
+        v = sp.NamedVideoStream(sc, f"godd{i}", frames=frames, codec="svc",
+                                io_packet_size=5)
+        frame = sc.io.Input([v])
+        # GPU ColorConvert forces the GPU decode path; planar is lossless
+        cc = sc.ops.ColorConvert(frame=frame, format="planar",
+                                 device=sp.DeviceType.GPU)
+        out = sp.NamedStream(sc, f"godd{i}_out")
+        sc.run(sc.io.Output(cc, [out]), sp.PerfParams.manual(4, 8),
+               cache_mode=sp.CacheMode.Overwrite, gpu_ids=[0])
+        got = np.stack(list(sp.NamedVideoStream(sc, f"godd{i}_out").load()))
+        ref = np.transpose(frames, (0, 3, 1, 2))
+        np.testing.assert_array_equal(got, ref)
