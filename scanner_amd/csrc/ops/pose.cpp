@@ -1,9 +1,11 @@
 // Pose-estimation op: OpenPose-style two-branch multi-stage CNN
 // (capability parity: the reference ecosystem's scannertools pose op /
 // BASELINE.json config 5 "OpenPose-style multi-DNN pose-detection graph").
-// bf16 inference on the hand-written MFMA GEMM (kernels/gemm_mfma.hip) via
-// im2col, like ResNet-50 (resnet50.cpp); weights are He-init random (no
-// network in this environment) or loaded from a TNSR file.
+// bf16 inference on the hand-written MFMA GEMM (kernels/gemm_mfma.hip),
+// implicit-GEMM for every conv with c % 8 == 0 (the 185-ch stage concat is
+// zero-padded to 192 for exactly this reason), like ResNet-50
+// (resnet50.cpp); weights are He-init random (no network in this
+// environment) or loaded from a TNSR file.
 //
 // Topology (input resized to 368x368, feature stride 8 like OpenPose):
 //   backbone: 8 x 3x3 convs (stride-2 at b2/b4/b6 — stride convs instead

@@ -1,8 +1,12 @@
 // ResNet-50 bf16 inference op — the DNN-inference capability of the
 // framework (parity role: the reference's scannertools DNN ops, e.g.
 // caffe-based frame classification). All conv/GEMM work runs on the
-// hand-written MFMA GEMM (kernels/gemm_mfma.hip) via im2col; BN is folded
-// into the GEMM epilogue (scale/bias), ReLU and residual adds are fused.
+// hand-written MFMA GEMM (kernels/gemm_mfma.hip): spatial convs via the
+// implicit-GEMM path (im2col rows gathered from NHWC activations during
+// LDS staging; only conv1's c=3 uses an explicit im2col buffer), deep-K
+// launch-bound shapes via split-K, K=64 1x1 convs via the M-walking
+// small-K kernel. BN is folded into the GEMM epilogue (scale/bias), ReLU
+// and residual adds are fused.
 // Weights are random-init (He) by default — there is no network in this
 // environment — or loaded from a tensor file for numerics tests
 // (tests/test_resnet_gpu.py compares against a PyTorch fp32 reference).
