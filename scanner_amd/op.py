@@ -55,23 +55,16 @@ class Op:
     def outputs(self):
         return list(self._outputs)
 
-    def _sole_output(self):
-        if len(self._outputs) != 1:
-            raise ScannerException(
-                f"op {self._name} has {len(self._outputs)} outputs; pick a "
-                "column with op['name'] first")
-        return self._outputs[0]
-
     # Single-output convenience: compression annotations apply to the sole
     # output column (reference ops hand back OpColumns directly).
     def compress(self, **kw):
-        return self._sole_output().compress(**kw)
+        return self._single().compress(**kw)
 
     def compress_video(self, **kw):
-        return self._sole_output().compress_video(**kw)
+        return self._single().compress_video(**kw)
 
     def lossless(self):
-        return self._sole_output().lossless()
+        return self._single().lossless()
 
     def __getitem__(self, name):
         for c in self._outputs:
