@@ -402,3 +402,23 @@ class ClusterClient:
     def shutdown(self):
         self._heartbeat_stop.set()
         self._rpc.try_call("Shutdown", {}, timeout=5)
+
+
+def main():
+    import argparse
+    ap = argparse.ArgumentParser(
+        description="scanner_amd master server (reference: start_master)")
+    ap.add_argument("--db-path", required=True)
+    ap.add_argument("--addr", default="0.0.0.0:5001")
+    ap.add_argument("--no-workers-timeout", type=float, default=30.0)
+    ap.add_argument("--task-timeout", type=float, default=600.0)
+    args = ap.parse_args()
+    m = MasterServer(args.db_path, args.addr,
+                     no_workers_timeout=args.no_workers_timeout,
+                     task_timeout=args.task_timeout)
+    print(f"master listening on {m.addr} (db: {args.db_path})", flush=True)
+    m._server.wait()
+
+
+if __name__ == "__main__":
+    main()
