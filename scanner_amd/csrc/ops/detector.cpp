@@ -247,6 +247,16 @@ class DetectorKernelGPU : public BatchedKernel {
           cand.push_back(b);
         }
       }
+      // standard SSD pre-NMS top-k: bounds the O(k^2) NMS on dense
+      // candidate sets (random-init heads score half the anchor grid)
+      constexpr size_t kPreNmsTopK = 1000;
+      if (cand.size() > kPreNmsTopK) {
+        std::partial_sort(cand.begin(), cand.begin() + kPreNmsTopK,
+                          cand.end(), [](const Box& a, const Box& b) {
+                            return a.score > b.score;
+                          });
+        cand.resize(kPreNmsTopK);
+      }
       auto kept = nms_best(std::move(cand), nms_thresh_);
       // BoundingBoxList blob (types.py format: '<I' count + '<5fi' boxes)
       size_t blob = 4 + kept.size() * 24;
