@@ -139,6 +139,15 @@ __global__ void __launch_bounds__(256, 5)
   const float* I1 = gray + level_off + (i64)pair_f1[pair] * h * w;
 
   int tx0 = blockIdx.x * TILE, ty0 = blockIdx.y * TILE;
+  int x = tx0 + threadIdx.x, y = ty0 + threadIdx.y;
+  // Issue the flow read before the LDS fill + gradient barriers so its
+  // ~200-900 cycle latency hides under the cooperative staging work.
+  const float* fin = flow_in + (i64)pair * h * w * 2;
+  float u = 0.f, v = 0.f;
+  if (x < w && y < h) {
+    u = fin[((i64)y * w + x) * 2 + 0];
+    v = fin[((i64)y * w + x) * 2 + 1];
+  }
   // Cooperative LDS fill with clamped loads.
   for (int i = threadIdx.y * TILE + threadIdx.x; i < LW * LW;
        i += TILE * TILE) {
@@ -159,13 +168,8 @@ __global__ void __launch_bounds__(256, 5)
   }
   __syncthreads();
 
-  int x = tx0 + threadIdx.x, y = ty0 + threadIdx.y;
   if (x >= w || y >= h) return;
-
-  const float* fin = flow_in + (i64)pair * h * w * 2;
   float* fout = flow_out + (i64)pair * h * w * 2;
-  float u = fin[((i64)y * w + x) * 2 + 0];
-  float v = fin[((i64)y * w + x) * 2 + 1];
 
   // The warp offset (u,v) is constant across the window, so the bilinear
   // fractions are too: the 49 taps read a contiguous (2R+2)^2 region of I1
