@@ -45,12 +45,9 @@ __device__ inline void epilogue_store(f32x4 (&acc)[FM][FN], float* scratch,
                                       int M, int N, int m0, int n0, int wrow,
                                       int wcol, int lane,
                                       const float* __restrict__ scale,
-                                      const float* __restrict__ bias,
-                                      int STRIDE) {
-  // STRIDE: scratch row pitch in floats. COLS+4 keeps 16 B alignment and
-  // breaks bank phase; the small-K kernel passes COLS (overlaying the
-  // consumed 16 KB A staging buffer — occupancy beats the bank conflicts).
+                                      const float* __restrict__ bias) {
   constexpr int COLS = FN * 16;
+  constexpr int STRIDE = COLS + 4;  // keeps 16 B row alignment, breaks banks
 #pragma unroll
   for (int i = 0; i < FM; ++i) {
     asm volatile("s_waitcnt lgkmcnt(0)");  // WAR: prior reads done
@@ -285,7 +282,7 @@ __global__ void __launch_bounds__(256, 2)
       reinterpret_cast<float*>(lds) + (size_t)wave * 16 * EP_STRIDE;
   epilogue_store<FM, FN, RELU, RESIDUAL>(acc, scratch, C, residual, M, N,
                                          m0, n0, wrow, wcol, lane, scale,
-                                         bias, EP_STRIDE);
+                                         bias);
 }
 
 // Small-K variant (K = 64: the DNN ops' 1x1 convolutions out of 64-channel
@@ -310,6 +307,7 @@ __global__ void __launch_bounds__(256, 2)
   constexpr int FN = BN / WN / 16;
   __shared__ bf16 lds_b[BN * K];
   __shared__ bf16 lds_a[2 * BM * K];
+  __shared__ float lds_ep[4 * 16 * (BN / WN + 4)];
 
   int tid = threadIdx.x;
   int lane = tid & 63;
@@ -396,19 +394,12 @@ __global__ void __launch_bounds__(256, 2)
               afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
     }
 
-    // Coalesced store overlaying the CONSUMED half of the A staging LDS
-    // (tile mt's fragments are all read by now): a dedicated 17 KB scratch
-    // cost a whole wave of occupancy (LDS-limited 2 -> 3 blocks/CU). The
-    // raw s_barrier (never __syncthreads: its fence would drain the next
-    // tile's in-flight LDS-DMA, guide G17) makes every wave's fragment
-    // reads complete before any wave overwrites the buffer.
-    asm volatile("s_waitcnt lgkmcnt(0)");
-    asm volatile("s_barrier");
-    float* ep = reinterpret_cast<float*>(lds_a + (size_t)buf * BM * K) +
-                (size_t)wave * 16 * (BN / WN);
-    epilogue_store<FM, FN, RELU, RESIDUAL>(acc, ep, C, residual, M, N,
-                                           mt * BM, n0, wrow, wcol, lane,
-                                           scale, bias, BN / WN);
+    // Coalesced store through dedicated per-wave scratch (lds_a/lds_b stay
+    // live — the next tile's A prefetch is in flight right now, and the
+    // wave-local lgkmcnt waits inside don't drain it).
+    epilogue_store<FM, FN, RELU, RESIDUAL>(
+        acc, lds_ep + (size_t)wave * 16 * (BN / WN + 4), C, residual, M, N,
+        mt * BM, n0, wrow, wcol, lane, scale, bias);
     // No trailing barrier (see main kernel note): next iteration's leading
     // waitcnt+barrier is the only ordering needed, and keeping the A
     // prefetch un-drained across the epilogue is the whole point.
