@@ -423,6 +423,29 @@ class ColorConvertKernelCPU : public BatchedKernel {
   std::string mode_;
 };
 
+// ---- ConcatBytes: variadic test op — concatenates all input columns'
+//      blobs per row (parity: reference variadic ops, op.h:77
+//      variadic_inputs + py_test.py:695) ----
+class ConcatBytesKernel : public BaseKernel {
+ public:
+  using BaseKernel::BaseKernel;
+  void execute(const StenciledElements& in, BatchedElements& out) override {
+    size_t rows = in.empty() ? 0 : in[0].size();
+    for (size_t r = 0; r < rows; ++r) {
+      size_t total = 0;
+      for (auto& col : in) total += col[r][0].size;
+      Element e = alloc_bytes(config_.device, total);
+      size_t off = 0;
+      for (auto& col : in) {
+        const Element& src = col[r][0];
+        std::memcpy(e.buffer + off, src.buffer, src.size);
+        off += src.size;
+      }
+      out[0].push_back(e);
+    }
+  }
+};
+
 // ---- FlowStats: dense flow frame -> {mean|u|, mean|v|, max|u|, max|v|} ----
 class FlowStatsKernelCPU : public Kernel {
  public:
@@ -580,6 +603,21 @@ void register_stdlib_ops() {
     f.preferred_batch = 8;
     f.make = [](const KernelConfig& c) -> std::unique_ptr<BaseKernel> {
       return std::make_unique<ColorConvertKernelCPU>(c);
+    };
+    kernel_registry().add(f);
+  }
+  {
+    OpInfo o;
+    o.name = "ConcatBytes";
+    o.variadic_inputs = true;
+    o.output_columns = {{"out", ColumnType::Bytes}};
+    op_registry().add(o);
+    KernelFactory f;
+    f.op_name = "ConcatBytes";
+    f.device_type = DeviceType::CPU;
+    f.preferred_batch = 8;
+    f.make = [](const KernelConfig& c) -> std::unique_ptr<BaseKernel> {
+      return std::make_unique<ConcatBytesKernel>(c);
     };
     kernel_registry().add(f);
   }

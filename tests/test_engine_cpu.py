@@ -737,3 +737,24 @@ def test_svc_fuzz_roundtrip(sc):
         got_sparse = list(v.load(rows=rows))
         for k, r in enumerate(rows):
             np.testing.assert_array_equal(got_sparse[k], frames[r])
+
+
+def test_variadic_op(sc):
+    """Variadic-input C++ op: sc.ops.X(inputs=[...]) concatenates any
+    number of input columns (reference: variadic_inputs op.h:77)."""
+    n = 5
+    a = sc.new_table("va", ["col"],
+                     [[f"a{i}".encode()] for i in range(n)])
+    b = sc.new_table("vb", ["col"],
+                     [[f"b{i}".encode()] for i in range(n)])
+    c = sc.new_table("vc", ["col"],
+                     [[f"c{i}".encode()] for i in range(n)])
+    ca = sc.io.Input([a])
+    cb = sc.io.Input([b])
+    cc = sc.io.Input([c])
+    cat = sc.ops.ConcatBytes(inputs=[ca, cb, cc])
+    out = sp.NamedStream(sc, "va_out")
+    sc.run(sc.io.Output(cat, [out]), sp.PerfParams.manual(2, 4),
+           cache_mode=sp.CacheMode.Overwrite)
+    rows = list(out.load())
+    assert rows == [f"a{i}b{i}c{i}".encode() for i in range(n)]

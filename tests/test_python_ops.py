@@ -2,6 +2,7 @@
 stencil, stencil+batch, per-stream args, stateful python kernels)."""
 import numpy as np
 import pytest
+import typing
 from typing import Any, Sequence
 
 import scanner_amd as sp
@@ -170,3 +171,31 @@ def test_python_op_null_elements(sc):
            cache_mode=sp.CacheMode.Overwrite)
     rows = list(out.load())
     assert rows == [b"val", b"null"] * n
+
+
+@register_python_op()
+def SplitStats(frame: FrameType) -> typing.Tuple[bytes, bytes]:
+    import numpy as np
+    return (float(np.mean(frame)).hex().encode(),
+            float(np.std(frame)).hex().encode())
+
+
+def test_python_op_multi_output(sc):
+    """Tuple return annotation -> multiple output columns (parity:
+    reference multi-output python ops)."""
+    from conftest import make_video
+    frames = make_video(n=6, h=24, w=32)
+    video = sp.NamedVideoStream(sc, "mo", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    st = sc.ops.SplitStats(frame=frame)
+    out = sp.NamedStream(sc, "mo_out")
+    sc.run(sc.io.Output([st["out0"], st["out1"]], [out]),
+           sp.PerfParams.manual(4, 8), cache_mode=sp.CacheMode.Overwrite)
+    import numpy as np
+    means = [float.fromhex(b.decode())
+             for b in sp.NamedStream(sc, "mo_out", column="out0").load()]
+    stds = [float.fromhex(b.decode())
+            for b in sp.NamedStream(sc, "mo_out", column="out1").load()]
+    for i in range(6):
+        assert abs(means[i] - float(np.mean(frames[i]))) < 1e-6
+        assert abs(stds[i] - float(np.std(frames[i]))) < 1e-5
