@@ -175,8 +175,7 @@ void validate_graph(const JobGraph& graph) {
   for (size_t i = 0; i < graph.ops.size(); ++i) {
     const OpNode& op = graph.ops[i];
     if (op.inputs.empty()) continue;
-    i32 l0 = level[op.inputs[0].op] +
-             (is_slice_op(graph.ops[op.inputs[0].op].name) ? 0 : 0);
+    i32 l0 = level[op.inputs[0].op];
     for (auto& e : op.inputs) {
       SCA_CHECK(level[e.op] == level[op.inputs[0].op],
                 "op '" + op.name + "' mixes inputs at different slice levels");
