@@ -82,50 +82,22 @@ __global__ void __launch_bounds__(256)
   }
 }
 
-// Bilinear-upsample flow from (sh,sw) to (dh,dw) for npairs pairs, scaling
-// vector components by the dimension ratio. Layout: [pair][h*w*2].
-__global__ void __launch_bounds__(256)
-    of_upsample_kernel(const float* __restrict__ src, int sh, int sw,
-                       float* __restrict__ dst, int dh, int dw, int npairs) {
-  i64 total = (i64)npairs * dh * dw;
-  i64 stride = (i64)gridDim.x * blockDim.x;
-  float fx_scale = (float)dw / sw, fy_scale = (float)dh / sh;
-  for (i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += stride) {
-    int p = (int)(i / ((i64)dh * dw));
-    i64 pix = i % ((i64)dh * dw);
-    int y = (int)(pix / dw), x = (int)(pix % dw);
-    float sxf = (x + 0.5f) / fx_scale - 0.5f;
-    float syf = (y + 0.5f) / fy_scale - 0.5f;
-    int x0 = (int)floorf(sxf), y0 = (int)floorf(syf);
-    float ax = sxf - x0, ay = syf - y0;
-    int x0c = min(max(x0, 0), sw - 1), x1c = min(max(x0 + 1, 0), sw - 1);
-    int y0c = min(max(y0, 0), sh - 1), y1c = min(max(y0 + 1, 0), sh - 1);
-    const float* s = src + (i64)p * sh * sw * 2;
-    float* d = dst + ((i64)p * dh * dw + (i64)y * dw + x) * 2;
-#pragma unroll
-    for (int ch = 0; ch < 2; ++ch) {
-      float v00 = s[((i64)y0c * sw + x0c) * 2 + ch];
-      float v01 = s[((i64)y0c * sw + x1c) * 2 + ch];
-      float v10 = s[((i64)y1c * sw + x0c) * 2 + ch];
-      float v11 = s[((i64)y1c * sw + x1c) * 2 + ch];
-      float v = v00 * (1 - ay) * (1 - ax) + v01 * (1 - ay) * ax +
-                v10 * ay * (1 - ax) + v11 * ay * ax;
-      d[ch] = v * (ch == 0 ? fx_scale : fy_scale);
-    }
-  }
-}
 
 // One LK iteration at one pyramid level for all pairs. Block = 16x16 pixel
 // tile of one pair's plane (blockIdx.z = pair); I0 tile + halo staged in
 // LDS. RADIUS is a compile-time template so the window loops fully unroll.
+// When `coarse` is non-null, the input flow is read from the NEXT
+// pyramid level (ch x cw) with the upsample bilerp fused in — the
+// standalone of_upsample_kernel pass and its flow round-trip through HBM
+// are skipped for the first iteration of every level.
 template <int RADIUS>
 __global__ void __launch_bounds__(256, 5)
     of_lk_kernel(const float* __restrict__ gray, i64 level_off, int h, int w,
                  const int* __restrict__ pair_f0,
                  const int* __restrict__ pair_f1,
                  const float* __restrict__ flow_in,
-                 float* __restrict__ flow_out) {
+                 float* __restrict__ flow_out,
+                 const float* __restrict__ coarse, int ch, int cw) {
   constexpr int TILE = 16;
   constexpr int HALO = RADIUS + 1;           // gradient needs +-1 past window
   constexpr int LW = TILE + 2 * HALO;        // staged tile width
@@ -142,11 +114,38 @@ __global__ void __launch_bounds__(256, 5)
   int x = tx0 + threadIdx.x, y = ty0 + threadIdx.y;
   // Issue the flow read before the LDS fill + gradient barriers so its
   // ~200-900 cycle latency hides under the cooperative staging work.
-  const float* fin = flow_in + (i64)pair * h * w * 2;
   float u = 0.f, v = 0.f;
   if (x < w && y < h) {
-    u = fin[((i64)y * w + x) * 2 + 0];
-    v = fin[((i64)y * w + x) * 2 + 1];
+    if (coarse != nullptr) {
+      // fused upsample from the coarser level (same math as
+      // of_upsample_kernel: bilerp + per-axis magnitude scaling)
+      const float* src = coarse + (i64)pair * ch * cw * 2;
+      float fx_scale = (float)w / cw, fy_scale = (float)h / ch;
+      float sxf = (x + 0.5f) / fx_scale - 0.5f;
+      float syf = (y + 0.5f) / fy_scale - 0.5f;
+      int x0 = (int)floorf(sxf), y0 = (int)floorf(syf);
+      float axc = sxf - x0, ayc = syf - y0;
+      int x0c = min(max(x0, 0), cw - 1), x1c = min(max(x0 + 1, 0), cw - 1);
+      int y0c = min(max(y0, 0), ch - 1), y1c = min(max(y0 + 1, 0), ch - 1);
+#pragma unroll
+      for (int chn = 0; chn < 2; ++chn) {
+        float v00 = src[((i64)y0c * cw + x0c) * 2 + chn];
+        float v01 = src[((i64)y0c * cw + x1c) * 2 + chn];
+        float v10 = src[((i64)y1c * cw + x0c) * 2 + chn];
+        float v11 = src[((i64)y1c * cw + x1c) * 2 + chn];
+        float vv = v00 * (1 - ayc) * (1 - axc) + v01 * (1 - ayc) * axc +
+                   v10 * ayc * (1 - axc) + v11 * ayc * axc;
+        float sc = chn == 0 ? fx_scale : fy_scale;
+        if (chn == 0)
+          u = vv * sc;
+        else
+          v = vv * sc;
+      }
+    } else {
+      const float* fin = flow_in + (i64)pair * h * w * 2;
+      u = fin[((i64)y * w + x) * 2 + 0];
+      v = fin[((i64)y * w + x) * 2 + 1];
+    }
   }
   // Cooperative LDS fill with clamped loads.
   for (int i = threadIdx.y * TILE + threadIdx.x; i < LW * LW;
@@ -475,17 +474,17 @@ class OpticalFlowKernelGPU : public BaseKernel {
       dim3 block(16, 16);
       for (int it = 0; it < iters_; ++it) {
         float* dst = (l == 0 && it == iters_ - 1) ? (float*)out_block : alt;
-        launch_lk(grid, block, s, gbase, loff[l], lh[l], lw[l], d_pf0, d_pf1,
-                  cur, dst);
+        if (l < top && it == 0) {
+          // first iteration of each finer level: fused upsample read from
+          // the coarser level's result (cur holds level l+1 flow)
+          launch_lk(grid, block, s, gbase, loff[l], lh[l], lw[l], d_pf0,
+                    d_pf1, nullptr, dst, cur, lh[l + 1], lw[l + 1]);
+        } else {
+          launch_lk(grid, block, s, gbase, loff[l], lh[l], lw[l], d_pf0,
+                    d_pf1, cur, dst);
+        }
         std::swap(cur, alt);
         if (dst == (float*)out_block) cur = (float*)out_block;
-      }
-      if (l > 0) {
-        of_upsample_kernel<<<grid_1d((i64)n * lh[l - 1] * lw[l - 1]), 256, 0,
-                             s>>>(cur, lh[l], lw[l], alt, lh[l - 1],
-                                  lw[l - 1], (int)n);
-        OF_CHECK(hipGetLastError());
-        std::swap(cur, alt);
       }
     }
 
@@ -520,19 +519,20 @@ class OpticalFlowKernelGPU : public BaseKernel {
 
   void launch_lk(dim3 grid, dim3 block, hipStream_t s, const float* gray,
                  i64 off, int h, int w, const int* pf0, const int* pf1,
-                 const float* fin, float* fout) {
+                 const float* fin, float* fout,
+                 const float* coarse = nullptr, int ch = 0, int cw = 0) {
     switch (radius_) {
       case 2:
         of_lk_kernel<2><<<grid, block, 0, s>>>(gray, off, h, w, pf0, pf1,
-                                               fin, fout);
+                                               fin, fout, coarse, ch, cw);
         break;
       case 3:
         of_lk_kernel<3><<<grid, block, 0, s>>>(gray, off, h, w, pf0, pf1,
-                                               fin, fout);
+                                               fin, fout, coarse, ch, cw);
         break;
       default:
         of_lk_kernel<4><<<grid, block, 0, s>>>(gray, off, h, w, pf0, pf1,
-                                               fin, fout);
+                                               fin, fout, coarse, ch, cw);
         break;
     }
     OF_CHECK(hipGetLastError());
