@@ -758,3 +758,43 @@ def test_variadic_op(sc):
            cache_mode=sp.CacheMode.Overwrite)
     rows = list(out.load())
     assert rows == [f"a{i}b{i}c{i}".encode() for i in range(n)]
+
+
+def test_per_stream_sampling_args(sc):
+    """Different sampling args per stream in one bulk job (reference:
+    per-stream Job args, py_test.py:339)."""
+    vids = [make_video(n=20, seed=s) for s in range(2)]
+    streams = [sp.NamedVideoStream(sc, f"ps{i}", frames=v, codec="raw")
+               for i, v in enumerate(vids)]
+    frame = sc.io.Input(streams)
+    # stream 0 takes rows 0..5, stream 1 rows 10..20
+    ranged = sc.streams.Range(frame, [(0, 5), (10, 20)])
+    hist = sc.ops.Histogram(frame=ranged)
+    outs = [sp.NamedStream(sc, f"ps{i}_h") for i in range(2)]
+    sc.run(sc.io.Output(hist, outs), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite)
+    assert len(list(outs[0].load())) == 5
+    rows1 = list(outs[1].load())
+    assert len(rows1) == 10
+    got = np.frombuffer(rows1[0], np.uint32).reshape(3, 256)
+    np.testing.assert_array_equal(got, ref_histogram(vids[1][10]))
+
+
+def test_gather_over_svc(sc):
+    """Gather with duplicate + unsorted-window rows over the codec path
+    (keyframe-span decode must serve repeated rows)."""
+    frames = make_video(n=40)
+    video = sp.NamedVideoStream(sc, "gsv", frames=frames, codec="svc")
+    frame = sc.io.Input([video])
+    rows_wanted = [2, 2, 17, 17, 39]
+    g = sc.streams.Gather(frame, [rows_wanted])
+    hist = sc.ops.Histogram(frame=g)
+    out = sp.NamedStream(sc, "gsv_out")
+    sc.run(sc.io.Output(hist, [out]), sp.PerfParams.manual(2, 4),
+           cache_mode=sp.CacheMode.Overwrite)
+    rows = list(out.load())
+    assert len(rows) == len(rows_wanted)
+    for k, blob in enumerate(rows):
+        got = np.frombuffer(blob, np.uint32).reshape(3, 256)
+        np.testing.assert_array_equal(got,
+                                      ref_histogram(frames[rows_wanted[k]]))
