@@ -461,16 +461,18 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t,
       auto ait = job.op_args.find((i32)i);
       std::vector<u8> sargs =
           ait == job.op_args.end() ? std::vector<u8>{} : ait->second;
-      // fetch-once resources (weights etc.) — first fetch per op name
-      // process-wide, then every instance sets up (reference:
+      // fetch-once resources (weights etc.) — first fetch per
+      // (op name, args) within THIS executor (so a later job using the same
+      // op with different args re-fetches; reference fetches per job,
       // evaluate_worker.cpp:493-550 worker-0 fetch + setup barrier).
       {
-        static std::mutex fetch_mu;
-        static std::set<std::string> fetched;
-        std::lock_guard<std::mutex> fl(fetch_mu);
-        if (!fetched.count(op.name)) {
+        std::string fetch_key(op.name);
+        fetch_key.append(1, '\0');
+        fetch_key.append(op.args.begin(), op.args.end());
+        std::lock_guard<std::mutex> fl(fetch_mu_);
+        if (!fetched_.count(fetch_key)) {
           inst.kernels[i]->fetch_resources(op.args);
-          fetched.insert(op.name);
+          fetched_.insert(fetch_key);
         }
       }
       inst.kernels[i]->setup_with_resources(op.args);

@@ -10,6 +10,8 @@
 #pragma once
 
 #include <atomic>
+#include <mutex>
+#include <set>
 #include <thread>
 
 #include "../dag/graph.h"
@@ -115,6 +117,11 @@ class LocalExecutor {
   std::atomic<i64> tasks_done_{0};
   std::vector<std::unique_ptr<Instance>> instances_;
   std::vector<std::unique_ptr<Profiler>> profilers_;
+
+  // fetch_resources dedup, scoped to this executor: keyed by
+  // op name + args so same-op-different-args jobs each fetch.
+  std::mutex fetch_mu_;
+  std::set<std::string> fetched_;
 };
 
 // Move an element to the target device; returns the original element if

@@ -119,6 +119,16 @@ class DetectorKernelGPU : public BatchedKernel {
     SCA_CHECK(f0.is_frame && f0.device.is_gpu(), "Detector needs GPU frames");
     int ih = f0.frame_info.shape[0], iw = f0.frame_info.shape[1],
         ic = f0.frame_info.shape[2];
+    // Geometry is taken from frame 0 for the whole batch (preprocess letterbox
+    // + box rescale), so every element must match it (same check
+    // the OpticalFlow kernel does).
+    for (int k = 1; k < n; ++k) {
+      const Element& fk = in[0][k];
+      SCA_CHECK(fk.is_frame && fk.frame_info.shape[0] == ih &&
+                    fk.frame_info.shape[1] == iw &&
+                    fk.frame_info.shape[2] == ic,
+                "Detector batch has mixed frame geometries");
+    }
 
     // Persistent workspace sized for max_batch.
     int nb = std::max(n, std::max(1, config_.max_batch));

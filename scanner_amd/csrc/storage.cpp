@@ -6,6 +6,7 @@
 #include <sys/types.h>
 #include <unistd.h>
 
+#include <atomic>
 #include <cerrno>
 #include <cstdio>
 #include <cstring>
@@ -49,13 +50,13 @@ class PosixStorage : public StorageBackend {
     // too — load_worker + storehouse).
     constexpr u64 kParallelMin = 16 << 20;
     constexpr u64 kChunk = 8 << 20;
-    bool fail = false;
+    std::atomic<bool> fail{false};  // written from reader threads
     auto read_span = [&](u64 lo, u64 hi) {
       u64 off = lo;
       while (off < hi) {
         ssize_t n = ::pread(fd, out + off, hi - off, offset + off);
         if (n <= 0) {
-          fail = true;
+          fail.store(true, std::memory_order_relaxed);
           return;
         }
         off += n;

@@ -155,7 +155,12 @@ class MasterServer:
                 job.to_assign.append(t)
                 job.stream_tasks[stream] += 1
             self._jobs_by_id[jid] = job
-            if self._job is None or self._job.complete.is_set():
+            # promote any queued job first so a newly submitted bulk job
+            # cannot jump ahead of jobs already waiting in the FIFO queue
+            # 
+            self._promote_if_done()
+            if (self._job is None or self._job.complete.is_set()) \
+                    and not self._job_queue:
                 self._job = job
                 job.started = time.time()
             else:
