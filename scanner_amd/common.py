@@ -54,7 +54,7 @@ class PerfParams:
     def __init__(self, work_packet_size=16, io_packet_size=128,
                  cpu_pool=0, gpu_pool=0, pipeline_instances_per_node=None,
                  load_sparsity_threshold=8, queue_size_per_pipeline=4,
-                 profiler_level=1, num_load_workers=0):
+                 profiler_level=1, num_load_workers=0, span_cache=0):
         self.work_packet_size = int(work_packet_size)
         self.io_packet_size = int(io_packet_size)
         self.cpu_pool = int(cpu_pool)
@@ -64,6 +64,11 @@ class PerfParams:
         self.queue_size_per_pipeline = int(queue_size_per_pipeline)
         self.profiler_level = int(profiler_level)
         self.num_load_workers = int(num_load_workers)  # 0 = auto
+        # HBM span-cache budget in bytes: compressed input GOP spans stay
+        # resident in device memory across tasks/jobs, so each span is read
+        # from storage and crosses PCIe once. 0 = auto (gpu_pool/3, max
+        # 16 GB); 1 = disabled.
+        self.span_cache = int(span_cache)
 
     @classmethod
     def manual(cls, work_packet_size, io_packet_size, **kw):
@@ -111,6 +116,7 @@ class PerfParams:
             "sparsity_threshold": self.load_sparsity_threshold,
             "profiler_level": self.profiler_level,
             "load_workers": self.num_load_workers,
+            "span_cache_size": self.span_cache,
         }
 
 

@@ -64,6 +64,8 @@ PerfParams perf_from_dict(const py::dict& d) {
     pp.sparsity_threshold = d["sparsity_threshold"].cast<i32>();
   if (d.contains("profiler_level"))
     pp.profiler_level = d["profiler_level"].cast<i32>();
+  if (d.contains("span_cache_size"))
+    pp.span_cache_size = d["span_cache_size"].cast<size_t>();
   return pp;
 }
 
@@ -229,6 +231,17 @@ PYBIND11_MODULE(_core, m) {
     init_memory_allocators(cfg);
   });
   m.def("destroy_memory", &destroy_memory_allocators);
+  // HBM span-cache introspection (tests + profiling)
+  m.def("span_cache_stats", [] {
+    py::dict d;
+    d["hits"] = span_cache_hits();
+    d["misses"] = span_cache_misses();
+    d["bytes"] = span_cache_bytes_live();
+    d["budget"] = span_cache_budget();
+    return d;
+  });
+  m.def("span_cache_clear", &span_cache_clear);
+  m.def("span_cache_set_budget", &span_cache_set_budget);
 
   m.def("registered_ops", [] { return op_registry().names(); });
   m.def("op_info", [](const std::string& name) {

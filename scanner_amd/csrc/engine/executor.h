@@ -18,6 +18,7 @@
 #include "../ops/kernel.h"
 #include "../profiler.h"
 #include "../queue.h"
+#include "../video/span_cache.h"
 #include "table_io.h"
 
 namespace sca {
@@ -32,6 +33,10 @@ struct PerfParams {
   size_t gpu_pool_size = 0;
   i32 sparsity_threshold = 8;
   i32 profiler_level = 1;
+  // HBM span-cache byte budget (compressed input GOP spans stay resident in
+  // device memory; video/span_cache.h). 0 = auto: gpu_pool/3 capped at
+  // 16 GB when a GPU pool exists, else 4 GB. Set to 1 to disable.
+  size_t span_cache_size = 0;
 };
 
 struct TaskDesc {
@@ -51,11 +56,13 @@ struct PreparedTask {
     i32 op = 0;
     i32 item = 0;
     u64 lo = 0, hi = 0;
-    u8* buf = nullptr;  // pinned (CPU pool)
+    u8* buf = nullptr;     // pinned (CPU pool); null when cached on-device
+    SpanHandle cached;     // HBM-resident encoded bytes (span cache)
   };
   std::vector<Span> spans;
   ~PreparedTask();
   u8* take_span(i32 op, i32 item, u64 lo, u64 hi);
+  SpanHandle take_cached(i32 op, i32 item, u64 lo, u64 hi);
 };
 
 class LocalExecutor {
@@ -102,6 +109,14 @@ class LocalExecutor {
                                         std::unordered_map<i64, Element>>>& outs,
                    PreparedTask* prep = nullptr);
   void make_instance(i32 idx);
+  // Lookup-or-fill the HBM span cache for one encoded byte range; returns a
+  // ready handle (device-resident bytes) or null when caching declined.
+  // `host_bytes`, when non-null, provides the bytes (skips the storage
+  // read on a fresh insert).
+  SpanHandle acquire_ready_span(const TableMetadata& table,
+                                const std::string& column, i32 item, u64 lo,
+                                u64 hi, DeviceHandle gpu, Profiler* prof,
+                                const u8* host_bytes);
 
   std::shared_ptr<Database> db_;
   JobGraph graph_;
@@ -109,6 +124,7 @@ class LocalExecutor {
   PerfParams pp_;
   std::vector<i32> gpu_ids_;
 
+  u64 db_key_ = 0;  // hash of db root, part of span-cache keys
   std::vector<JobAnalysis> analyses_;
   std::vector<TableMetadata> out_tables_;   // per job
   std::vector<std::vector<i64>> task_rows_; // per job: cumulative task ends

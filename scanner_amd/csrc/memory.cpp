@@ -131,7 +131,15 @@ BlockAllocator::BlockAllocator(std::unique_ptr<Allocator> base)
 
 u8* BlockAllocator::allocate(size_t size, i32 refs) {
   SCA_CHECK(refs > 0, "block buffer needs >=1 ref");
-  u8* p = base_->allocate(size);
+  u8* p = nullptr;
+  try {
+    p = base_->allocate(size);
+  } catch (const std::exception&) {
+    // Pool pressure: give registered caches (HBM span cache) a chance to
+    // release idle entries, then retry once before propagating.
+    if (!run_memory_pressure_callbacks(size)) throw;
+    p = base_->allocate(size);
+  }
   std::lock_guard<std::mutex> l(mu_);
   blocks_.emplace(p, Block{p, size == 0 ? 1 : size, refs});
   return p;
@@ -286,11 +294,28 @@ std::vector<std::function<void()>>& teardown_callbacks() {
   static std::vector<std::function<void()>> cbs;
   return cbs;
 }
+std::mutex g_pressure_mu;
+std::vector<std::function<bool(size_t)>>& pressure_callbacks() {
+  static std::vector<std::function<bool(size_t)>> cbs;
+  return cbs;
+}
 }  // namespace
 
 void register_memory_teardown_callback(std::function<void()> cb) {
   std::lock_guard<std::mutex> l(g_teardown_mu);
   teardown_callbacks().push_back(std::move(cb));
+}
+
+void register_memory_pressure_callback(std::function<bool(size_t)> cb) {
+  std::lock_guard<std::mutex> l(g_pressure_mu);
+  pressure_callbacks().push_back(std::move(cb));
+}
+
+bool run_memory_pressure_callbacks(size_t want) {
+  std::lock_guard<std::mutex> l(g_pressure_mu);
+  bool any = false;
+  for (auto& cb : pressure_callbacks()) any |= cb(want);
+  return any;
 }
 
 void destroy_memory_allocators() {

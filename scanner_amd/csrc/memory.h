@@ -126,6 +126,12 @@ u64 memory_generation();
 // them while the allocators still exist. Survives re-init.
 void register_memory_teardown_callback(std::function<void()> cb);
 
+// Pressure callbacks run when a pool allocation fails; a cache (HBM span
+// cache) frees idle entries and returns true if it released anything, and
+// the allocation is retried once. Survives re-init.
+void register_memory_pressure_callback(std::function<bool(size_t)> cb);
+bool run_memory_pressure_callbacks(size_t want);
+
 u8* new_buffer(DeviceHandle device, size_t size);
 u8* new_block_buffer(DeviceHandle device, size_t size, i32 refs);
 void add_buffer_ref(DeviceHandle device, u8* buffer, i32 n = 1);

@@ -67,6 +67,15 @@ std::vector<Element> svc_decode_gpu(const u8* stream_host, size_t size,
                                     const std::vector<i64>& want,
                                     DeviceHandle dev, u64 stream_offset = 0);
 
+// Decode from an already device-resident encoded stream (the HBM span
+// cache path — video/span_cache.h): `stream_dev` holds the item's bytes
+// starting at stream offset `dev_lo` and must cover the keyframe-aligned
+// span of `want`. No host access, no H2D.
+std::vector<Element> svc_decode_gpu_dev(const u8* stream_dev, u64 dev_lo,
+                                        const VideoMetadata& vm,
+                                        const std::vector<i64>& want,
+                                        DeviceHandle dev);
+
 struct SvcPacketView {
   bool is_key = false;
   u32 nbytes = 0;

@@ -146,29 +146,22 @@ __global__ void __launch_bounds__(128)
 
 }  // namespace
 
-std::vector<Element> svc_decode_gpu(const u8* stream_host, size_t size,
-                                    const VideoMetadata& vm,
-                                    const std::vector<i64>& want,
-                                    DeviceHandle dev, u64 stream_offset) {
+// Shared decode core: `d_stream` holds device-resident encoded bytes
+// covering [lo, hi) of the item's stream. Packet geometry (nbytes, ngroups,
+// nsuper, keyframe-ness) is fully determined by the VideoMetadata, so no
+// host-side packet parsing is needed — this is what lets the HBM span cache
+// (video/span_cache.h) feed decode without the bytes ever revisiting the
+// host. `up_ev`, when non-null, is an event the upload was recorded on;
+// decode chains wait on it.
+static std::vector<Element> svc_decode_gpu_impl(
+    const u8* d_stream, u64 lo, const VideoMetadata& vm,
+    const std::vector<i64>& span, const std::vector<i64>& want,
+    DeviceHandle dev, hipEvent_t up_ev) {
   hipStream_t s = (hipStream_t)per_thread_hip_stream();
   u32 nbytes = (u32)((i64)vm.height * vm.width * vm.channels);
-  std::vector<i64> span = svc_decode_span(vm, want);
+  u32 ngroups = (nbytes + 31) / 32;
+  u32 nsuper = (ngroups + 127) / 128;
   std::vector<Element> out;
-  if (span.empty()) return out;
-
-  // Upload the byte range covering the span asynchronously on the main
-  // stream; decode chains fork off it via an event. stream_host is pinned
-  // (CPU pool = hipHostMalloc) and outlives the end-of-function sync.
-  u64 lo = vm.sample_offsets[span.front()];
-  u64 hi = vm.sample_offsets[span.back()] + vm.sample_sizes[span.back()];
-  SCA_CHECK(lo >= stream_offset && hi <= stream_offset + size,
-            "svc stream range does not cover decode span");
-  u8* d_stream = new_buffer(dev, hi - lo);
-  SVC_CHECK(hipMemcpyAsync(d_stream, stream_host + (lo - stream_offset),
-                           hi - lo, hipMemcpyHostToDevice, s));
-  hipEvent_t up_ev;
-  SVC_CHECK(hipEventCreateWithFlags(&up_ev, hipEventDisableTiming));
-  SVC_CHECK(hipEventRecord(up_ev, s));
 
   // Frames inside a GOP are a serial prediction chain, but GOPs are
   // independent: each keyframe starts a new chain on one of 4 auxiliary
@@ -185,10 +178,17 @@ std::vector<Element> svc_decode_gpu(const u8* stream_host, size_t size,
   u8* prev = nullptr;
   u8* chain_scratch[2] = {nullptr, nullptr};
   int flip = 0;
+  size_t ki = 0;  // cursor into sorted keyframe_indices
   for (i64 f : span) {
-    const u8* pkt_h = stream_host + (vm.sample_offsets[f] - stream_offset);
-    SvcPacketView v = svc_parse_packet(pkt_h, vm.sample_sizes[f]);
-    SCA_CHECK(v.nbytes == nbytes, "svc frame size mismatch");
+    while (ki < vm.keyframe_indices.size() && vm.keyframe_indices[ki] < f)
+      ++ki;
+    bool is_key =
+        ki < vm.keyframe_indices.size() && vm.keyframe_indices[ki] == f;
+    SvcPacketView v;
+    v.is_key = is_key;
+    v.nbytes = nbytes;
+    v.ngroups = ngroups;
+    v.nsuper = nsuper;
     if (v.is_key) {
       // new chain
       chain = (chain + 1);
@@ -197,7 +197,7 @@ std::vector<Element> svc_decode_gpu(const u8* stream_host, size_t size,
       if (!chain_used[ci]) {
         chain_used[ci] = true;
         chain_stream[ci] = cs;
-        SVC_CHECK(hipStreamWaitEvent(cs, up_ev, 0));
+        if (up_ev) SVC_CHECK(hipStreamWaitEvent(cs, up_ev, 0));
       }
       prev = nullptr;
       chain_scratch[0] = chain_scratch[1] = nullptr;
@@ -252,12 +252,64 @@ std::vector<Element> svc_decode_gpu(const u8* stream_host, size_t size,
     SVC_CHECK(hipStreamWaitEvent(s, ev, 0));
     SVC_CHECK(hipEventDestroy(ev));
   }
-  SVC_CHECK(hipEventDestroy(up_ev));
   SVC_CHECK(hipStreamSynchronize(s));
-  delete_buffer(dev, d_stream);
   for (u8* sc : scratches) delete_buffer(dev, sc);
   SCA_CHECK(wi == want.size(), "svc gpu decode: not all frames produced");
   return out;
+}
+
+std::vector<Element> svc_decode_gpu(const u8* stream_host, size_t size,
+                                    const VideoMetadata& vm,
+                                    const std::vector<i64>& want,
+                                    DeviceHandle dev, u64 stream_offset) {
+  std::vector<i64> span = svc_decode_span(vm, want);
+  if (span.empty()) return {};
+  hipStream_t s = (hipStream_t)per_thread_hip_stream();
+
+  // Upload the byte range covering the span asynchronously on the main
+  // stream; decode chains fork off it via an event. stream_host is pinned
+  // (CPU pool = hipHostMalloc) and outlives the end-of-function sync.
+  u64 lo = vm.sample_offsets[span.front()];
+  u64 hi = vm.sample_offsets[span.back()] + vm.sample_sizes[span.back()];
+  SCA_CHECK(lo >= stream_offset && hi <= stream_offset + size,
+            "svc stream range does not cover decode span");
+  // Sanity-check the first packet against metadata while host bytes exist
+  // (the device path below trusts VideoMetadata alone).
+  SvcPacketView v0 = svc_parse_packet(stream_host + (lo - stream_offset),
+                                      vm.sample_sizes[span.front()]);
+  SCA_CHECK(v0.nbytes == (u32)((i64)vm.height * vm.width * vm.channels),
+            "svc frame size mismatch");
+  u8* d_stream = new_buffer(dev, hi - lo);
+  SVC_CHECK(hipMemcpyAsync(d_stream, stream_host + (lo - stream_offset),
+                           hi - lo, hipMemcpyHostToDevice, s));
+  hipEvent_t up_ev;
+  SVC_CHECK(hipEventCreateWithFlags(&up_ev, hipEventDisableTiming));
+  SVC_CHECK(hipEventRecord(up_ev, s));
+  std::vector<Element> out;
+  try {
+    out = svc_decode_gpu_impl(d_stream, lo, vm, span, want, dev, up_ev);
+  } catch (...) {
+    (void)hipEventDestroy(up_ev);
+    delete_buffer(dev, d_stream);
+    throw;
+  }
+  SVC_CHECK(hipEventDestroy(up_ev));
+  delete_buffer(dev, d_stream);
+  return out;
+}
+
+std::vector<Element> svc_decode_gpu_dev(const u8* stream_dev, u64 dev_lo,
+                                        const VideoMetadata& vm,
+                                        const std::vector<i64>& want,
+                                        DeviceHandle dev) {
+  std::vector<i64> span = svc_decode_span(vm, want);
+  if (span.empty()) return {};
+  u64 lo = vm.sample_offsets[span.front()];
+  u64 hi = vm.sample_offsets[span.back()] + vm.sample_sizes[span.back()];
+  SCA_CHECK(lo >= dev_lo, "svc device stream range does not cover span");
+  (void)hi;
+  return svc_decode_gpu_impl(stream_dev + (lo - dev_lo), lo, vm, span, want,
+                             dev, nullptr);
 }
 
 }  // namespace sca

@@ -239,3 +239,58 @@ def test_gpu_svc_decode_odd_shapes(sc):
         got = np.stack(list(sp.NamedVideoStream(sc, f"godd{i}_out").load()))
         ref = np.transpose(frames, (0, 3, 1, 2))
         np.testing.assert_array_equal(got, ref)
+
+
+def test_gpu_span_cache_hits_and_numerics(sc):
+    """Second run of the same pipeline decodes from HBM-cached spans (no
+    storage re-read, no H2D) and produces identical results (VERDICT r01
+    #1: span cache)."""
+    from scanner_amd import _core
+    frames = make_smooth_video(n=40, h=72, w=96)
+    video = sp.NamedVideoStream(sc, "g_sc", frames=frames, codec="svc")
+
+    def run(name):
+        frame = sc.io.Input([video])
+        hist = sc.ops.Histogram(frame=frame, device=sp.DeviceType.GPU)
+        out = sp.NamedStream(sc, name)
+        sc.run(sc.io.Output(hist, [out]),
+               sp.PerfParams.manual(8, 16, span_cache=256 << 20),
+               cache_mode=sp.CacheMode.Overwrite, gpu_ids=[0])
+        return list(out.load())
+
+    _core.span_cache_clear()
+    r1 = run("g_sc_1")
+    s1 = _core.span_cache_stats()
+    r2 = run("g_sc_2")
+    s2 = _core.span_cache_stats()
+    assert r1 == r2
+    for i, blob in enumerate(r1):
+        got = np.frombuffer(blob, dtype=np.uint32).reshape(3, 256)
+        np.testing.assert_array_equal(got, ref_histogram(frames[i]))
+    # run 2 must be served from the cache
+    assert s2["hits"] > s1["hits"]
+    assert s2["misses"] == s1["misses"]
+    assert s2["bytes"] > 0
+
+
+def test_gpu_span_cache_eviction():
+    """A tiny budget forces LRU eviction instead of unbounded growth."""
+    from scanner_amd import _core
+    _core.span_cache_clear()
+    _core.span_cache_set_budget(1 << 20)  # 1 MB
+    import tempfile, os
+    import scanner_amd as spp
+    sc2 = spp.Client(db_path=os.path.join(
+        tempfile.mkdtemp(prefix="scse_"), "db"))
+    frames = make_smooth_video(n=64, h=72, w=96)  # ~0.4 MB/frame raw
+    video = spp.NamedVideoStream(sc2, "g_ev", frames=frames, codec="svc")
+    frame = sc2.io.Input([video])
+    hist = sc2.ops.Histogram(frame=frame, device=spp.DeviceType.GPU)
+    out = spp.NamedStream(sc2, "g_ev_out")
+    sc2.run(sc2.io.Output(hist, [out]),
+            spp.PerfParams.manual(8, 16, span_cache=1 << 20),
+            cache_mode=spp.CacheMode.Overwrite, gpu_ids=[0])
+    s = _core.span_cache_stats()
+    assert s["bytes"] <= 1 << 20
+    rows = list(out.load())
+    assert len(rows) == 64

@@ -30,6 +30,7 @@ CPP_SOURCES = [
     "csrc/engine/table_io.cpp",
     "csrc/engine/executor.cpp",
     "csrc/video/svc_cpu.cpp",
+    "csrc/video/span_cache.cpp",
     "csrc/core_module.cpp",
 ]
 HIP_SOURCES = [
