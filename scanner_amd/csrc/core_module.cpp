@@ -242,6 +242,20 @@ PYBIND11_MODULE(_core, m) {
   });
   m.def("span_cache_clear", &span_cache_clear);
   m.def("span_cache_set_budget", &span_cache_set_budget);
+  // engine memory accounting (device -1 = CPU)
+  m.def("mem_stats", [](int device) {
+    DeviceHandle d = device < 0 ? CPU_DEVICE
+                                : DeviceHandle{DeviceType::GPU, device};
+    py::dict r;
+    r["live"] = mem_bytes_live(d);
+    r["peak"] = mem_bytes_peak(d);
+    return r;
+  });
+  m.def("mem_reset_peak", [](int device) {
+    DeviceHandle d = device < 0 ? CPU_DEVICE
+                                : DeviceHandle{DeviceType::GPU, device};
+    mem_reset_peak(d);
+  });
 
   m.def("registered_ops", [] { return op_registry().names(); });
   m.def("op_info", [](const std::string& name) {

@@ -366,11 +366,14 @@ std::shared_ptr<PreparedTask> LocalExecutor::prepare_task(const TaskDesc& t,
   return pt;
 }
 
-// Load Input-op rows into the per-op output maps.
-void LocalExecutor::load_inputs(
-    Instance& inst, const TaskDesc& t, const TaskPlan& plan,
-    std::vector<std::map<std::string, std::unordered_map<i64, Element>>>& outs,
-    PreparedTask* prep) {
+// ---- source loading -------------------------------------------------
+// Non-video and raw-codec columns load eagerly (cheap byte reads); svc
+// video sources get an incremental decode state per item so the streaming
+// packet loop pulls frames as needed instead of materializing the whole
+// task's decode output up front.
+void LocalExecutor::init_sources(
+    Instance& inst, const TaskDesc& t, const TaskPlan& plan, OutsVec& outs,
+    std::map<i32, std::vector<SvcItemState>>& svc_states, PreparedTask* prep) {
   const JobBinding& job = jobs_[t.job];
   for (auto& kv : plan.load_rows) {
     i32 op_idx = kv.first;
@@ -382,9 +385,6 @@ void LocalExecutor::load_inputs(
     Profiler::Scope s(inst.profiler, "load:" + src.table);
 
     if (si.output_types[0] == ColumnType::Video) {
-      // Raw-codec video: frames are stored as plain elements; geometry from
-      // the item's VideoMetadata. (svc-codec spans are decoded by the
-      // decode stage — wired in video/decoder.cpp.)
       auto items = items_for_rows(table, rows);
       std::unordered_map<i32, VideoMetadata> vm;
       for (auto& ir : items)
@@ -393,94 +393,56 @@ void LocalExecutor::load_inputs(
       bool any_svc = false;
       for (auto& kv2 : vm) any_svc |= kv2.second.codec == "svc";
       if (any_svc) {
-        // Codec path (decoder-automaton parity): per item, decode exactly
-        // the wanted frames from their keyframe-aligned GOP spans, on the
-        // instance's GPU when it has one (frames land in HBM and feed GPU
-        // ops with zero copies), else on CPU.
+        // Codec path (decoder-automaton parity): set up per-item decode
+        // states; frames decode per work packet in advance_svc_source, on
+        // the instance's GPU when it has one (straight into HBM).
+        auto& states = svc_states[op_idx];
         size_t ri = 0;
         for (auto& ir : items) {
           const VideoMetadata& m = vm[ir.item];
-          std::vector<i64> local;
+          SvcItemState st;
+          st.item = ir.item;
+          st.row_start = ir.row_start;
+          st.vm = m;
           while (ri < rows.size() && rows[ri] < ir.row_end) {
-            local.push_back(rows[ri] - ir.row_start);
+            st.local.push_back(rows[ri] - ir.row_start);
             ++ri;
           }
-          // Read only the byte range covering the keyframe-aligned decode
-          // span, into pinned memory (fast DMA H2D).
-          std::vector<i64> span = svc_decode_span(m, local);
+          std::vector<i64> span = svc_decode_span(m, st.local);
           if (span.empty()) continue;
-          u64 lo = m.sample_offsets[span.front()];
-          u64 hi = m.sample_offsets[span.back()] + m.sample_sizes[span.back()];
-          // Device-resident bytes first: handed over by the load worker or
-          // already in the HBM span cache from an earlier task.
-          SpanHandle ch =
-              prep ? prep->take_cached(op_idx, ir.item, lo, hi) : nullptr;
-          if (ch && (!inst.gpu.is_gpu() || ch->dev.id != inst.gpu.id)) {
-            ch.reset();  // prefetched for a different GPU — fall back
+          st.lo = m.sample_offsets[span.front()];
+          st.hi = m.sample_offsets[span.back()] + m.sample_sizes[span.back()];
+          // Device-resident bytes first (prefetched by the load worker or
+          // resident from an earlier task), then prefetched host bytes,
+          // then cache-fill / storage read.
+          st.cached =
+              prep ? prep->take_cached(op_idx, ir.item, st.lo, st.hi)
+                   : nullptr;
+          if (st.cached &&
+              (!inst.gpu.is_gpu() || st.cached->dev.id != inst.gpu.id)) {
+            st.cached.reset();  // prefetched for a different GPU
           }
-          u8* stream_buf =
-              prep ? prep->take_span(op_idx, ir.item, lo, hi) : nullptr;
-          if (!ch && inst.gpu.is_gpu()) {
-            // cold path / direct process_task call: fill the cache from the
-            // instance thread (reuses host bytes when we hold them)
-            ch = acquire_ready_span(table, src.column, ir.item, lo, hi,
-                                    inst.gpu, inst.profiler, stream_buf);
+          st.host =
+              prep ? prep->take_span(op_idx, ir.item, st.lo, st.hi) : nullptr;
+          if (!st.cached && inst.gpu.is_gpu()) {
+            st.cached = acquire_ready_span(table, src.column, ir.item, st.lo,
+                                           st.hi, inst.gpu, inst.profiler,
+                                           st.host);
           }
-          if (ch) {
-            Profiler::Scope sd(inst.profiler, "decode:gpu");
-            inst.profiler->increment("decoded_frames", (i64)local.size());
-            auto elems = svc_decode_gpu_dev(ch->ptr, lo, m, local, inst.gpu);
-            for (auto& e : elems) {
-              i64 grow = e.index + ir.row_start;
-              Element ge = e;
-              ge.index = grow;
-              outs[op_idx][col_name][grow] = ge;
-            }
-            // decode synced its stream before returning; safe to unpin
-            ch.reset();
-            if (stream_buf) delete_buffer(CPU_DEVICE, stream_buf);
-            continue;
+          if (st.cached && st.host) {
+            delete_buffer(CPU_DEVICE, st.host);
+            st.host = nullptr;
           }
-          if (!stream_buf) {
-            stream_buf = new_buffer(CPU_DEVICE, hi - lo);
+          if (!st.cached && !st.host) {
+            st.host = new_buffer(CPU_DEVICE, st.hi - st.lo);
             Profiler::Scope sl(inst.profiler, "load:video");
             db_->storage()->read_range(
                 db_->paths().item(table.id, table.column_id(src.column),
                                   ir.item),
-                lo, hi - lo, stream_buf);
-            inst.profiler->increment("io_read_bytes", (i64)(hi - lo));
+                st.lo, st.hi - st.lo, st.host);
+            inst.profiler->increment("io_read_bytes", (i64)(st.hi - st.lo));
           }
-          if (inst.gpu.is_gpu()) {
-            Profiler::Scope sd(inst.profiler, "decode:gpu");
-            inst.profiler->increment("decoded_frames", (i64)local.size());
-            auto elems =
-                svc_decode_gpu(stream_buf, hi - lo, m, local, inst.gpu, lo);
-            for (auto& e : elems) {
-              i64 grow = e.index + ir.row_start;
-              Element ge = e;
-              ge.index = grow;
-              outs[op_idx][col_name][grow] = ge;
-            }
-            delete_buffer(CPU_DEVICE, stream_buf);
-          } else {
-            Profiler::Scope sd(inst.profiler, "decode:cpu");
-            std::vector<std::vector<u8>> frames;
-            svc_decode_cpu(stream_buf, hi - lo, m, local, frames, lo);
-            delete_buffer(CPU_DEVICE, stream_buf);
-            for (size_t k = 0; k < local.size(); ++k) {
-              Element e;
-              e.is_frame = true;
-              e.frame_info.shape[0] = m.height;
-              e.frame_info.shape[1] = m.width;
-              e.frame_info.shape[2] = m.channels;
-              e.frame_info.type = m.frame_type;
-              e.size = frames[k].size();
-              e.buffer = new_buffer(CPU_DEVICE, e.size);
-              std::memcpy(e.buffer, frames[k].data(), e.size);
-              e.index = local[k] + ir.row_start;
-              outs[op_idx][col_name][e.index] = e;
-            }
-          }
+          states.push_back(std::move(st));
         }
         continue;
       }
@@ -510,6 +472,67 @@ void LocalExecutor::load_inputs(
     }
   }
 }
+
+void LocalExecutor::advance_svc_source(Instance& inst,
+                                       std::vector<SvcItemState>& states,
+                                       i64 max_row, const std::string& col,
+                                       OutsVec& outs, i32 op_idx) {
+  for (auto& st : states) {
+    size_t k = st.cursor;
+    while (k < st.local.size() && st.local[k] + st.row_start <= max_row) ++k;
+    if (k == st.cursor) continue;
+    std::vector<i64> want(st.local.begin() + st.cursor,
+                          st.local.begin() + k);
+    st.cursor = k;
+    if (st.cached) {
+      Profiler::Scope sd(inst.profiler, "decode:gpu");
+      inst.profiler->increment("decoded_frames", (i64)want.size());
+      auto elems = svc_decode_gpu_dev(st.cached->ptr, st.lo, st.vm, want,
+                                      inst.gpu);
+      for (auto& e : elems) {
+        e.index += st.row_start;
+        outs[op_idx][col][e.index] = e;
+      }
+    } else if (inst.gpu.is_gpu()) {
+      Profiler::Scope sd(inst.profiler, "decode:gpu");
+      inst.profiler->increment("decoded_frames", (i64)want.size());
+      auto elems =
+          svc_decode_gpu(st.host, st.hi - st.lo, st.vm, want, inst.gpu,
+                         st.lo);
+      for (auto& e : elems) {
+        e.index += st.row_start;
+        outs[op_idx][col][e.index] = e;
+      }
+    } else {
+      Profiler::Scope sd(inst.profiler, "decode:cpu");
+      std::vector<std::vector<u8>> frames;
+      svc_decode_cpu(st.host, st.hi - st.lo, st.vm, want, frames, st.lo);
+      for (size_t q = 0; q < want.size(); ++q) {
+        Element e;
+        e.is_frame = true;
+        e.frame_info.shape[0] = st.vm.height;
+        e.frame_info.shape[1] = st.vm.width;
+        e.frame_info.shape[2] = st.vm.channels;
+        e.frame_info.type = st.vm.frame_type;
+        e.size = frames[q].size();
+        e.buffer = new_buffer(CPU_DEVICE, e.size);
+        std::memcpy(e.buffer, frames[q].data(), e.size);
+        e.index = want[q] + st.row_start;
+        outs[op_idx][col][e.index] = e;
+      }
+    }
+    // release the encoded bytes once the item is fully decoded (decode
+    // synced its stream before returning)
+    if (st.cursor == st.local.size()) {
+      st.cached.reset();
+      if (st.host) {
+        delete_buffer(CPU_DEVICE, st.host);
+        st.host = nullptr;
+      }
+    }
+  }
+}
+
 
 void LocalExecutor::process_task(Instance& inst, const TaskDesc& t,
                                  PreparedTask* prep) {
@@ -582,272 +605,448 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t,
     inst.kernels_job = t.job;
   }
 
-  // Per-op, per-column row->element maps.
-  std::vector<std::map<std::string, std::unordered_map<i64, Element>>> outs(n);
-  auto free_all = [&]() {
+  // ---- streaming packet execution with liveness-driven frees ----
+  //
+  // The task no longer materializes every op's whole-task output: sink rows
+  // are processed in work_packet_size packets; a backward pass per packet
+  // computes how far each op's cursor must advance (stencils/warmup/remaps
+  // included — the per-task plan already resolved exact row sets), sources
+  // decode incrementally, and every element is freed as soon as its last
+  // planned consumer has read it. Peak engine memory is therefore bounded
+  // by the live stencil window, independent of io_packet_size (reference
+  // analogue: work packets flowing through stage threads,
+  // worker.cpp:1663-1722, + liveness dag_analysis.cpp:1145-1326).
+  OutsVec outs(n);
+  // planned read counts per (op, column, row)
+  std::vector<std::map<std::string, std::unordered_map<i64, i32>>> reads(n);
+  const OpNode& out_op = graph_.ops.back();
+  for (size_t j = 0; j < n; ++j) {
+    const OpNode& op = graph_.ops[j];
+    const OpTaskPlan& otp = plan.ops[j];
+    if (is_input_op(op.name)) continue;
+    if (is_output_op(op.name)) {
+      for (auto& e : op.inputs) {
+        auto& rm = reads[e.op][e.column];
+        for (i64 row = t.start; row < t.end; ++row) rm[row]++;
+      }
+    } else if (is_sample_op(op.name) || is_slice_op(op.name) ||
+               is_unslice_op(op.name)) {
+      if (op.inputs.empty()) continue;
+      auto& rm = reads[op.inputs[0].op][op.inputs[0].column];
+      for (i64 u : otp.remap)
+        if (u >= 0) rm[u]++;
+    } else {
+      for (auto& e : op.inputs) {
+        auto& rm = reads[e.op][e.column];
+        for (auto& win : otp.windows)
+          for (i64 w : win) rm[w]++;
+      }
+    }
+  }
+
+  std::vector<std::pair<DeviceHandle, u8*>> pending_free;
+  auto dec_read = [&](i32 opi, const std::string& col, i64 row) {
+    auto& rm = reads[opi][col];
+    auto it = rm.find(row);
+    SCA_CHECK(it != rm.end(), "liveness underflow (op " +
+                                  graph_.ops[opi].name + " row " +
+                                  std::to_string(row) + ")");
+    if (--it->second == 0) {
+      rm.erase(it);
+      auto& om = outs[opi][col];
+      auto oit = om.find(row);
+      if (oit != om.end()) {
+        if (oit->second.buffer)
+          pending_free.emplace_back(oit->second.device, oit->second.buffer);
+        om.erase(oit);
+      }
+    }
+  };
+  // Frees are deferred to packet boundaries: same-stream ordering covers
+  // this instance's later work, but the pool is shared across instances,
+  // so wait for in-flight kernels before returning buffers.
+  auto flush_frees = [&]() {
+    if (pending_free.empty()) return;
+    if (inst.gpu.is_gpu()) sync_per_thread_stream();
+    for (auto& pf : pending_free) delete_buffer(pf.first, pf.second);
+    pending_free.clear();
+  };
+
+  // Sink accumulation (one storage item per task, written at task end;
+  // D2H happens per packet so it overlaps later compute).
+  std::vector<std::vector<Element>> sink_cols(out_op.inputs.size());
+  std::vector<std::vector<u8>> sink_owned(out_op.inputs.size());
+
+  std::map<i32, std::vector<SvcItemState>> svc_states;
+  std::vector<size_t> cursor(n, 0);
+
+  auto cleanup_all = [&]() {
+    if (inst.gpu.is_gpu()) {
+      try {
+        sync_per_thread_stream();
+      } catch (...) {
+      }
+    }
+    for (auto& pf : pending_free) delete_buffer(pf.first, pf.second);
+    pending_free.clear();
     for (auto& per_op : outs) {
       for (auto& per_col : per_op) {
         for (auto& kv : per_col.second) {
           Element& e = kv.second;
           if (e.buffer) delete_buffer(e.device, e.buffer);
         }
+        per_col.second.clear();
+      }
+    }
+    for (size_t c = 0; c < sink_cols.size(); ++c) {
+      for (size_t k = 0; k < sink_cols[c].size(); ++k) {
+        if (sink_owned[c][k] && sink_cols[c][k].buffer)
+          delete_buffer(CPU_DEVICE, sink_cols[c][k].buffer);
+      }
+      sink_cols[c].clear();
+      sink_owned[c].clear();
+    }
+    for (auto& kv : svc_states) {
+      for (auto& st : kv.second) {
+        st.cached.reset();
+        if (st.host) {
+          delete_buffer(CPU_DEVICE, st.host);
+          st.host = nullptr;
+        }
       }
     }
   };
 
   try {
-    load_inputs(inst, t, plan, outs, prep);
+    init_sources(inst, t, plan, outs, svc_states, prep);
 
-    for (size_t i = 0; i < n; ++i) {
-      const OpNode& op = graph_.ops[i];
-      const OpStaticInfo& si = ja.info[i];
-      OpTaskPlan& otp = plan.ops[i];
-      if (otp.required_rows.empty() && !is_output_op(op.name)) continue;
+    i64 W = std::max<i64>(1, pp_.work_packet_size);
+    for (i64 ps = t.start; ps < t.end; ps += W) {
+      i64 pe = std::min(t.end, ps + W);
 
-      if (is_input_op(op.name)) continue;
-
-      if (is_sample_op(op.name) || is_slice_op(op.name) ||
-          is_unslice_op(op.name)) {
-        // remap: alias parent elements (zero-copy, extra ref per alias)
-        auto& parent_map = outs[op.inputs[0].op][op.inputs[0].column];
-        auto& out_map = outs[i][si.output_columns[0]];
-        for (size_t k = 0; k < otp.required_rows.size(); ++k) {
-          i64 row = otp.required_rows[k];
-          i64 up = otp.remap[k];
-          Element e;
-          if (up < 0) {
-            e.is_null = true;
-          } else {
-            auto pit = parent_map.find(up);
-            SCA_CHECK(pit != parent_map.end(),
-                      "remap source row missing (op " + op.name + ")");
-            e = pit->second;
-            if (e.buffer) add_buffer_ref(e.device, e.buffer);
+      // backward pass: per-op cursor targets for this packet
+      std::vector<i64> need(n, -1);
+      std::vector<size_t> target(n, 0);
+      for (auto& e : out_op.inputs)
+        need[e.op] = std::max(need[e.op], pe - 1);
+      for (i64 i = (i64)n - 2; i >= 0; --i) {
+        const OpNode& op = graph_.ops[i];
+        const OpTaskPlan& otp = plan.ops[i];
+        if (is_input_op(op.name)) continue;
+        if (is_sample_op(op.name) || is_slice_op(op.name) ||
+            is_unslice_op(op.name)) {
+          const auto& rr = otp.required_rows;
+          size_t tgt =
+              std::upper_bound(rr.begin(), rr.end(), need[i]) - rr.begin();
+          if (tgt < cursor[i]) tgt = cursor[i];
+          for (size_t k = cursor[i]; k < tgt; ++k) {
+            if (otp.remap[k] >= 0)
+              need[op.inputs[0].op] =
+                  std::max(need[op.inputs[0].op], otp.remap[k]);
           }
-          e.index = row;
-          out_map[row] = e;
+          target[i] = tgt;
+        } else {
+          const auto& cr = otp.compute_rows;
+          size_t tgt =
+              std::upper_bound(cr.begin(), cr.end(), need[i]) - cr.begin();
+          if (tgt < cursor[i]) tgt = cursor[i];
+          if (tgt > cursor[i]) {
+            i64 mx = -1;
+            for (size_t k = cursor[i]; k < tgt; ++k)
+              for (i64 w : otp.windows[k]) mx = std::max(mx, w);
+            if (mx >= 0)
+              for (auto& e : op.inputs)
+                need[e.op] = std::max(need[e.op], mx);
+          }
+          target[i] = tgt;
         }
-        continue;
       }
 
-      if (is_output_op(op.name)) {
-        // Sink: gather required rows from parents, move to CPU, write item.
-        Profiler::Scope s(inst.profiler, "save");
-        const TableMetadata& table = out_tables_[t.job];
-        std::vector<std::vector<Element>> cols(op.inputs.size());
-        std::vector<std::vector<u8>> owned(op.inputs.size());
-        // Batched D2H: gather all device-resident outputs into one
-        // memcpy_vec (async on the instance stream, one sync).
-        std::vector<u8*> d2h_dst;
-        std::vector<const u8*> d2h_src;
-        std::vector<size_t> d2h_sz;
-        DeviceHandle gpu_src = CPU_DEVICE;
-        for (size_t c = 0; c < op.inputs.size(); ++c) {
-          auto& parent_map = outs[op.inputs[c].op][op.inputs[c].column];
-          for (i64 row = t.start; row < t.end; ++row) {
-            auto pit = parent_map.find(row);
-            SCA_CHECK(pit != parent_map.end(), "sink missing row");
-            Element e = pit->second;
-            if (e.is_null) {
-              Element ne;
-              ne.is_null = true;
-              cols[c].push_back(ne);
-              owned[c].push_back(0);
-              continue;
-            }
-            if (!e.device.is_gpu()) {
-              e.index = row;
-              cols[c].push_back(e);
-              owned[c].push_back(0);
-              continue;
-            }
-            Element ce = e;
-            ce.device = CPU_DEVICE;
-            ce.buffer = new_buffer(CPU_DEVICE, e.size);
-            ce.index = row;
-            cols[c].push_back(ce);
-            owned[c].push_back(1);
-            d2h_dst.push_back(ce.buffer);
-            d2h_src.push_back(e.buffer);
-            d2h_sz.push_back(e.size);
-            gpu_src = e.device;
+      // forward pass
+      for (size_t i = 0; i < n; ++i) {
+        const OpNode& op = graph_.ops[i];
+        const OpStaticInfo& si = ja.info[i];
+        OpTaskPlan& otp = plan.ops[i];
+
+        if (is_input_op(op.name)) {
+          if (need[i] >= 0) {
+            auto sit = svc_states.find((i32)i);
+            if (sit != svc_states.end())
+              advance_svc_source(inst, sit->second, need[i],
+                                 si.output_columns[0], outs, (i32)i);
+            // eager sources already materialized in init_sources
           }
+          continue;
         }
-        if (!d2h_dst.empty()) {
-          memcpy_vec(d2h_dst, CPU_DEVICE, d2h_src, gpu_src, d2h_sz);
-        }
-        // Sink-side codec annotations (OpColumn.compress_video; parity:
-        // reference compressed-output columns + PostEvaluateWorker encode,
-        // evaluate_worker.cpp:1329-1560).
-        std::map<std::string, std::string> compress;
-        if (!op.args.empty()) {
-          auto a = mp::decode(op.args);
-          auto& am = a.as_map();
-          auto cit = am.find("compress");
-          if (cit != am.end()) {
-            for (auto& kv : cit->second.as_map())
-              compress[kv.first] = kv.second.as_str();
+
+        if (is_sample_op(op.name) || is_slice_op(op.name) ||
+            is_unslice_op(op.name)) {
+          // remap: alias parent elements (zero-copy, extra ref per alias)
+          auto& parent_map = outs[op.inputs[0].op][op.inputs[0].column];
+          auto& out_map = outs[i][si.output_columns[0]];
+          for (size_t k = cursor[i]; k < target[i]; ++k) {
+            i64 row = otp.required_rows[k];
+            i64 up = otp.remap[k];
+            Element e;
+            if (up < 0) {
+              e.is_null = true;
+            } else {
+              auto pit = parent_map.find(up);
+              SCA_CHECK(pit != parent_map.end(),
+                        "remap source row missing (op " + op.name + ")");
+              e = pit->second;
+              if (e.buffer) add_buffer_ref(e.device, e.buffer);
+            }
+            e.index = row;
+            out_map[row] = e;
+            if (up >= 0) dec_read(op.inputs[0].op, op.inputs[0].column, up);
           }
+          cursor[i] = target[i];
+          continue;
         }
-        for (size_t c = 0; c < op.inputs.size(); ++c) {
-          const std::string& cname = table.columns[c].name;
-          auto cmp = compress.find(cname);
-          if (cmp != compress.end() &&
-              table.columns[c].type == ColumnType::Video) {
-            SCA_CHECK(cmp->second == "svc",
-                      "unknown sink codec '" + cmp->second + "'");
-            i32 h = 0, w = 0, ch = 0;
-            size_t fsize = 0;
-            std::vector<u8> contig;
-            i64 nf = 0;
-            for (auto& e : cols[c]) {
-              SCA_CHECK(!e.is_null,
-                        "cannot codec-compress a column with null rows");
-              SCA_CHECK(e.is_frame && e.frame_info.type == FrameType::U8,
-                        "svc compression needs u8 frames");
-              if (nf == 0) {
-                h = e.frame_info.shape[0];
-                w = e.frame_info.shape[1];
-                ch = e.frame_info.shape[2];
-                fsize = e.size;
-                contig.reserve(fsize * cols[c].size());
+
+        if (is_output_op(op.name)) {
+          // gather packet rows, batched D2H, hold refs for the task-end
+          // write
+          Profiler::Scope s(inst.profiler, "save");
+          std::vector<u8*> d2h_dst;
+          std::vector<const u8*> d2h_src;
+          std::vector<size_t> d2h_sz;
+          DeviceHandle gpu_src = CPU_DEVICE;
+          for (size_t c = 0; c < op.inputs.size(); ++c) {
+            auto& parent_map = outs[op.inputs[c].op][op.inputs[c].column];
+            for (i64 row = ps; row < pe; ++row) {
+              auto pit = parent_map.find(row);
+              SCA_CHECK(pit != parent_map.end(), "sink missing row");
+              Element e = pit->second;
+              if (e.is_null) {
+                Element ne;
+                ne.is_null = true;
+                sink_cols[c].push_back(ne);
+                sink_owned[c].push_back(0);
+                continue;
               }
-              SCA_CHECK(e.size == fsize,
-                        "svc compression needs uniform frame sizes");
-              contig.insert(contig.end(), e.buffer, e.buffer + e.size);
-              ++nf;
+              if (!e.device.is_gpu()) {
+                // producer element may be freed before the task-end write;
+                // the sink keeps its own block reference
+                add_buffer_ref(e.device, e.buffer);
+                e.index = row;
+                sink_cols[c].push_back(e);
+                sink_owned[c].push_back(1);
+                continue;
+              }
+              Element ce = e;
+              ce.device = CPU_DEVICE;
+              ce.buffer = new_buffer(CPU_DEVICE, e.size);
+              ce.index = row;
+              sink_cols[c].push_back(ce);
+              sink_owned[c].push_back(1);
+              d2h_dst.push_back(ce.buffer);
+              d2h_src.push_back(e.buffer);
+              d2h_sz.push_back(e.size);
+              gpu_src = e.device;
             }
-            VideoMetadata vm;
-            vm.width = w;
-            vm.height = h;
-            vm.channels = ch;
-            vm.frame_type = FrameType::U8;
-            vm.num_frames = nf;
-            std::vector<u8> stream;
-            svc_encode_cpu(contig.data(), nf, h, w, ch, /*gop=*/16, stream,
-                           vm);
-            inst.profiler->increment("io_write_bytes", (i64)stream.size());
-            write_video_item(*db_, table, cname, t.task, stream, vm);
-            for (size_t k = 0; k < cols[c].size(); ++k) {
-              if (owned[c][k] && cols[c][k].buffer)
-                delete_buffer(CPU_DEVICE, cols[c][k].buffer);
-            }
-            continue;
           }
+          if (!d2h_dst.empty()) {
+            memcpy_vec(d2h_dst, CPU_DEVICE, d2h_src, gpu_src, d2h_sz);
+          }
+          for (size_t c = 0; c < op.inputs.size(); ++c)
+            for (i64 row = ps; row < pe; ++row)
+              dec_read(op.inputs[c].op, op.inputs[c].column, row);
+          continue;
+        }
+
+        // ---- kernel op: advance [cursor, target) in batches ----
+        BaseKernel* kernel = inst.kernels[i].get();
+        DeviceHandle kdev = kernel->config().device;
+        auto& out_maps = outs[i];
+        auto rit = reads[i].end();  // re-looked-up per column below
+
+        size_t bi = cursor[i];
+        while (bi < target[i]) {
+          size_t be = bi + 1;
+          while (be < target[i] && (i64)(be - bi) < (i64)si.batch &&
+                 !otp.reset_before[be]) {
+            ++be;
+          }
+          if (otp.reset_before[bi]) kernel->reset();
+
+          StenciledElements input(op.inputs.size());
+          std::vector<Element> scratch;  // owned device copies to free
+          for (size_t c = 0; c < op.inputs.size(); ++c) {
+            auto& parent_map = outs[op.inputs[c].op][op.inputs[c].column];
+            input[c].resize(be - bi);
+            for (size_t r = bi; r < be; ++r) {
+              for (i64 wrow : otp.windows[r]) {
+                auto pit = parent_map.find(wrow);
+                SCA_CHECK(pit != parent_map.end(),
+                          "kernel input row missing for op '" + op.name +
+                              "'");
+                bool copied;
+                Element e = element_to_device(pit->second, kdev, copied);
+                if (copied) scratch.push_back(e);
+                input[c][r - bi].push_back(e);
+              }
+            }
+          }
+
+          BatchedElements output(si.output_columns.size());
           {
-            i64 wb = 0;
-            for (auto& e : cols[c]) wb += (i64)e.size;
-            inst.profiler->increment("io_write_bytes", wb);
+            Profiler::Scope s(inst.profiler, "op:" + op.name);
+            kernel->execute(input, output);
           }
-          write_column_item(*db_, table, table.columns[c].name, t.task,
-                            cols[c]);
-          if (table.columns[c].type == ColumnType::Video) {
-            // Raw-stored frame column: record geometry so readers can
-            // reconstruct frames.
-            VideoMetadata vm;
-            vm.codec = "raw";
-            vm.num_frames = (i64)cols[c].size();
-            for (auto& e : cols[c]) {
-              if (e.is_null) continue;
-              vm.height = e.frame_info.shape[0];
-              vm.width = e.frame_info.shape[1];
-              vm.channels = e.frame_info.shape[2];
-              vm.frame_type = e.frame_info.type;
-              break;
+          for (size_t c = 0; c < si.output_columns.size(); ++c) {
+            SCA_CHECK(output[c].size() == be - bi,
+                      "op '" + op.name + "' produced " +
+                          std::to_string(output[c].size()) +
+                          " rows, expected " + std::to_string(be - bi));
+            rit = reads[i].find(si.output_columns[c]);
+            auto& out_map = out_maps[si.output_columns[c]];
+            for (size_t r = bi; r < be; ++r) {
+              Element& e = output[c][r - bi];
+              e.index = otp.compute_rows[r];
+              if (!e.is_null) e.device = kernel->config().output_device;
+              bool wanted =
+                  rit != reads[i].end() && rit->second.count(e.index);
+              if (wanted) {
+                out_map[e.index] = e;
+              } else if (e.buffer) {
+                // warmup-only row or unused output column — freed at the
+                // packet boundary (after the stream sync)
+                pending_free.emplace_back(e.device, e.buffer);
+              }
             }
-            for (i64 k = 0; k < vm.num_frames; ++k) {
-              vm.keyframe_indices.push_back(k);
-              vm.sample_sizes.push_back(cols[c][k].size);
-              vm.sample_offsets.push_back(
-                  k == 0 ? 0 : vm.sample_offsets[k - 1] +
-                                   vm.sample_sizes[k - 1]);
-            }
-            auto vbuf = vm.serialize();
-            db_->storage()->write_all(
-                db_->paths().video_metadata(
-                    table.id, table.column_id(table.columns[c].name), t.task),
-                vbuf.data(), vbuf.size());
           }
+          // GPU kernels are async on this thread's stream; staging copies
+          // can only be released after the kernels that read them complete.
+          if (!scratch.empty() && kdev.is_gpu()) sync_per_thread_stream();
+          for (auto& e : scratch) delete_buffer(e.device, e.buffer);
+          // consume the stencil reads of this batch
+          for (size_t c = 0; c < op.inputs.size(); ++c)
+            for (size_t r = bi; r < be; ++r)
+              for (i64 wrow : otp.windows[r])
+                dec_read(op.inputs[c].op, op.inputs[c].column, wrow);
+          bi = be;
+        }
+        cursor[i] = target[i];
+      }
+
+      flush_frees();
+    }
+
+    // ---- task-end write: one storage item per (column, task) ----
+    {
+      Profiler::Scope s(inst.profiler, "save");
+      const TableMetadata& table = out_tables_[t.job];
+      auto& cols = sink_cols;
+      auto& owned = sink_owned;
+      // Sink-side codec annotations (OpColumn.compress_video; parity:
+      // reference compressed-output columns + PostEvaluateWorker encode,
+      // evaluate_worker.cpp:1329-1560).
+      std::map<std::string, std::string> compress;
+      if (!out_op.args.empty()) {
+        auto a = mp::decode(out_op.args);
+        auto& am = a.as_map();
+        auto cit = am.find("compress");
+        if (cit != am.end()) {
+          for (auto& kv : cit->second.as_map())
+            compress[kv.first] = kv.second.as_str();
+        }
+      }
+      for (size_t c = 0; c < out_op.inputs.size(); ++c) {
+        const std::string& cname = table.columns[c].name;
+        auto cmp = compress.find(cname);
+        if (cmp != compress.end() &&
+            table.columns[c].type == ColumnType::Video) {
+          SCA_CHECK(cmp->second == "svc",
+                    "unknown sink codec '" + cmp->second + "'");
+          i32 h = 0, w = 0, ch = 0;
+          size_t fsize = 0;
+          std::vector<u8> contig;
+          i64 nf = 0;
+          for (auto& e : cols[c]) {
+            SCA_CHECK(!e.is_null,
+                      "cannot codec-compress a column with null rows");
+            SCA_CHECK(e.is_frame && e.frame_info.type == FrameType::U8,
+                      "svc compression needs u8 frames");
+            if (nf == 0) {
+              h = e.frame_info.shape[0];
+              w = e.frame_info.shape[1];
+              ch = e.frame_info.shape[2];
+              fsize = e.size;
+              contig.reserve(fsize * cols[c].size());
+            }
+            SCA_CHECK(e.size == fsize,
+                      "svc compression needs uniform frame sizes");
+            contig.insert(contig.end(), e.buffer, e.buffer + e.size);
+            ++nf;
+          }
+          VideoMetadata vm;
+          vm.width = w;
+          vm.height = h;
+          vm.channels = ch;
+          vm.frame_type = FrameType::U8;
+          vm.num_frames = nf;
+          std::vector<u8> stream;
+          svc_encode_cpu(contig.data(), nf, h, w, ch, /*gop=*/16, stream,
+                         vm);
+          inst.profiler->increment("io_write_bytes", (i64)stream.size());
+          write_video_item(*db_, table, cname, t.task, stream, vm);
           for (size_t k = 0; k < cols[c].size(); ++k) {
             if (owned[c][k] && cols[c][k].buffer)
               delete_buffer(CPU_DEVICE, cols[c][k].buffer);
           }
+          cols[c].clear();
+          owned[c].clear();
+          continue;
         }
-        continue;
-      }
-
-      // ---- kernel op ----
-      BaseKernel* kernel = inst.kernels[i].get();
-      DeviceHandle kdev = kernel->config().device;
-      std::set<i64> required(otp.required_rows.begin(),
-                             otp.required_rows.end());
-      auto& out_maps = outs[i];
-
-      size_t bi = 0;
-      while (bi < otp.compute_rows.size()) {
-        // batch boundaries: batch size limit + reset markers
-        size_t be = bi + 1;
-        while (be < otp.compute_rows.size() &&
-               (i64)(be - bi) < (i64)si.batch && !otp.reset_before[be]) {
-          ++be;
-        }
-        if (otp.reset_before[bi]) kernel->reset();
-
-        // Gather stenciled inputs, moving to kernel device as needed.
-        StenciledElements input(op.inputs.size());
-        std::vector<Element> scratch;  // owned device copies to free
-        for (size_t c = 0; c < op.inputs.size(); ++c) {
-          auto& parent_map = outs[op.inputs[c].op][op.inputs[c].column];
-          input[c].resize(be - bi);
-          for (size_t r = bi; r < be; ++r) {
-            for (i64 wrow : otp.windows[r]) {
-              auto pit = parent_map.find(wrow);
-              SCA_CHECK(pit != parent_map.end(),
-                        "kernel input row missing for op '" + op.name + "'");
-              bool copied;
-              Element e = element_to_device(pit->second, kdev, copied);
-              if (copied) scratch.push_back(e);
-              input[c][r - bi].push_back(e);
-            }
-          }
-        }
-
-        BatchedElements output(si.output_columns.size());
         {
-          Profiler::Scope s(inst.profiler, "op:" + op.name);
-          kernel->execute(input, output);
+          i64 wb = 0;
+          for (auto& e : cols[c]) wb += (i64)e.size;
+          inst.profiler->increment("io_write_bytes", wb);
         }
-        for (size_t c = 0; c < si.output_columns.size(); ++c) {
-          SCA_CHECK(output[c].size() == be - bi,
-                    "op '" + op.name + "' produced " +
-                        std::to_string(output[c].size()) + " rows, expected " +
-                        std::to_string(be - bi));
-          auto& out_map = out_maps[si.output_columns[c]];
-          for (size_t r = bi; r < be; ++r) {
-            Element& e = output[c][r - bi];
-            e.index = otp.compute_rows[r];
-            if (!e.is_null) e.device = kernel->config().output_device;
-            if (required.count(e.index)) {
-              out_map[e.index] = e;
-            } else if (e.buffer) {
-              delete_buffer(e.device, e.buffer);  // warmup-only row
-            }
+        write_column_item(*db_, table, table.columns[c].name, t.task,
+                          cols[c]);
+        if (table.columns[c].type == ColumnType::Video) {
+          // Raw-stored frame column: record geometry so readers can
+          // reconstruct frames.
+          VideoMetadata vm;
+          vm.codec = "raw";
+          vm.num_frames = (i64)cols[c].size();
+          for (auto& e : cols[c]) {
+            if (e.is_null) continue;
+            vm.height = e.frame_info.shape[0];
+            vm.width = e.frame_info.shape[1];
+            vm.channels = e.frame_info.shape[2];
+            vm.frame_type = e.frame_info.type;
+            break;
           }
+          for (i64 k = 0; k < vm.num_frames; ++k) {
+            vm.keyframe_indices.push_back(k);
+            vm.sample_sizes.push_back(cols[c][k].size);
+            vm.sample_offsets.push_back(
+                k == 0 ? 0 : vm.sample_offsets[k - 1] +
+                                 vm.sample_sizes[k - 1]);
+          }
+          auto vbuf = vm.serialize();
+          db_->storage()->write_all(
+              db_->paths().video_metadata(
+                  table.id, table.column_id(table.columns[c].name), t.task),
+              vbuf.data(), vbuf.size());
         }
-        // GPU kernels are async on this thread's stream; staging copies
-        // can only be released after the kernels that read them complete.
-        // Everything else stays async — same-stream ordering covers all
-        // later consumers (kernels and D2H copies alike).
-        if (!scratch.empty() && kdev.is_gpu()) sync_per_thread_stream();
-        for (auto& e : scratch) delete_buffer(e.device, e.buffer);
-        bi = be;
+        for (size_t k = 0; k < cols[c].size(); ++k) {
+          if (owned[c][k] && cols[c][k].buffer)
+            delete_buffer(CPU_DEVICE, cols[c][k].buffer);
+        }
+        cols[c].clear();
+        owned[c].clear();
       }
     }
   } catch (...) {
-    free_all();
+    cleanup_all();
     throw;
   }
-  free_all();
+  cleanup_all();
   inst.profiler->increment("tasks", 1);
   inst.profiler->increment("rows", t.end - t.start);
   tasks_done_.fetch_add(1);

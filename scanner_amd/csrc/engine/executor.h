@@ -98,16 +98,39 @@ class LocalExecutor {
 
  private:
   struct Instance;
+  using OutsVec =
+      std::vector<std::map<std::string, std::unordered_map<i64, Element>>>;
+  // Incremental decode state for one svc video item of one Input op: the
+  // encoded byte span (device-cached or pinned host) plus a cursor over the
+  // wanted item-local rows, so decode advances one work packet at a time
+  // as the streaming executor pulls rows (reference analogue: the
+  // feeder/retriever DecoderAutomata, decoder_automata.cpp:126-230).
+  struct SvcItemState {
+    i32 item = 0;
+    i64 row_start = 0;
+    VideoMetadata vm;
+    std::vector<i64> local;  // item-local wanted rows, ascending
+    size_t cursor = 0;       // decoded prefix of `local`
+    SpanHandle cached;       // HBM-resident encoded bytes [lo, hi)
+    u8* host = nullptr;      // pinned encoded bytes [lo, hi)
+    u64 lo = 0, hi = 0;
+  };
   void process_task(Instance& inst, const TaskDesc& t,
                     PreparedTask* prep = nullptr);
   // Load-worker half of a task: derive the plan + read video source spans
   // into pinned buffers (runs on prefetch threads in run()).
   std::shared_ptr<PreparedTask> prepare_task(const TaskDesc& t,
                                              Profiler* prof);
-  void load_inputs(Instance& inst, const TaskDesc& t, const TaskPlan& plan,
-                   std::vector<std::map<std::string,
-                                        std::unordered_map<i64, Element>>>& outs,
-                   PreparedTask* prep = nullptr);
+  // Eagerly load non-video / raw-codec source rows into `outs`; set up
+  // incremental decode states for svc video sources.
+  void init_sources(Instance& inst, const TaskDesc& t, const TaskPlan& plan,
+                    OutsVec& outs,
+                    std::map<i32, std::vector<SvcItemState>>& svc_states,
+                    PreparedTask* prep);
+  // Decode all not-yet-decoded rows with global index <= max_row.
+  void advance_svc_source(Instance& inst, std::vector<SvcItemState>& states,
+                          i64 max_row, const std::string& col, OutsVec& outs,
+                          i32 op_idx);
   void make_instance(i32 idx);
   // Lookup-or-fill the HBM span cache for one encoded byte range; returns a
   // ready handle (device-resident bytes) or null when caching declined.

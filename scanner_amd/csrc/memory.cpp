@@ -142,6 +142,8 @@ u8* BlockAllocator::allocate(size_t size, i32 refs) {
   }
   std::lock_guard<std::mutex> l(mu_);
   blocks_.emplace(p, Block{p, size == 0 ? 1 : size, refs});
+  live_bytes_ += size == 0 ? 1 : size;
+  if (live_bytes_ > peak_bytes_) peak_bytes_ = live_bytes_;
   return p;
 }
 
@@ -177,6 +179,7 @@ void BlockAllocator::release(const u8* ptr) {
     SCA_CHECK(it != blocks_.end(), "release of unknown buffer");
     if (--it->second.refs == 0) {
       to_free = it->second.base;
+      live_bytes_ -= it->second.size;
       blocks_.erase(it);
     }
   }
@@ -186,6 +189,21 @@ void BlockAllocator::release(const u8* ptr) {
 size_t BlockAllocator::num_live() const {
   std::lock_guard<std::mutex> l(mu_);
   return blocks_.size();
+}
+
+size_t BlockAllocator::bytes_live() const {
+  std::lock_guard<std::mutex> l(mu_);
+  return live_bytes_;
+}
+
+size_t BlockAllocator::bytes_peak() const {
+  std::lock_guard<std::mutex> l(mu_);
+  return peak_bytes_;
+}
+
+void BlockAllocator::reset_peak() {
+  std::lock_guard<std::mutex> l(mu_);
+  peak_bytes_ = live_bytes_;
 }
 
 // ---------------- global state ----------------
@@ -416,6 +434,16 @@ size_t gpu_pool_bytes_in_use(i32 gpu_id) {
   auto it = g_mem.gpus.find(gpu_id);
   if (it == g_mem.gpus.end() || !it->second.pool) return 0;
   return it->second.pool->bytes_in_use();
+}
+
+size_t mem_bytes_live(DeviceHandle dev) {
+  return allocators_for(dev).block->bytes_live();
+}
+size_t mem_bytes_peak(DeviceHandle dev) {
+  return allocators_for(dev).block->bytes_peak();
+}
+void mem_reset_peak(DeviceHandle dev) {
+  allocators_for(dev).block->reset_peak();
 }
 
 }  // namespace sca

@@ -86,6 +86,10 @@ class BlockAllocator {
   void release(const u8* ptr);
 
   size_t num_live() const;
+  // Live/peak byte accounting (streaming-executor memory tests).
+  size_t bytes_live() const;
+  size_t bytes_peak() const;
+  void reset_peak();
 
  private:
   struct Block {
@@ -93,6 +97,8 @@ class BlockAllocator {
     size_t size;
     i32 refs;
   };
+  size_t live_bytes_ = 0;
+  size_t peak_bytes_ = 0;
   // Find block containing ptr; caller holds mu_. Returns iterator into
   // blocks_ (keyed by base address).
   std::map<const u8*, Block>::iterator find_block(const u8* ptr);
@@ -149,6 +155,12 @@ void memcpy_vec(const std::vector<u8*>& dests, DeviceHandle dest_device,
                 const std::vector<size_t>& sizes);
 
 size_t gpu_pool_bytes_in_use(i32 gpu_id);
+
+// Live/peak block-allocated bytes for a device (peak resets via
+// mem_reset_peak; used by the streaming-executor memory-bound tests).
+size_t mem_bytes_live(DeviceHandle dev);
+size_t mem_bytes_peak(DeviceHandle dev);
+void mem_reset_peak(DeviceHandle dev);
 
 // The calling thread's HIP stream (lazily created). All of a pipeline
 // instance's kernels and copies are issued on its thread's stream, so
