@@ -14,6 +14,9 @@
 #include "hip_util.h"
 #include "memory.h"
 #include "msgpack.h"
+#include "video/h264.h"
+#include "video/ingest.h"
+#include "video/mp4.h"
 #include "ops/kernel.h"
 #include "ops/python_kernel.h"
 #include "video/svc.h"
@@ -342,6 +345,63 @@ PYBIND11_MODULE(_core, m) {
           db->update_table(t);
           db->commit_table(t.id);
         });
+
+  // ---- real-video ingest (mp4/Annex-B -> indexed h264 table) ----
+  m.def("ingest_video_file",
+        [](std::shared_ptr<Database> db, const std::string& name,
+           const std::string& column, const std::string& path) {
+          IngestResult r = ingest_video_file(*db, name, column, path);
+          py::dict d;
+          d["num_frames"] = r.num_frames;
+          d["width"] = r.width;
+          d["height"] = r.height;
+          d["codec"] = r.codec;
+          return d;
+        });
+  m.def("export_mp4",
+        [](std::shared_ptr<Database> db, const std::string& name,
+           const std::string& column, const std::string& path, double fps) {
+          export_mp4(*db, name, column, path, fps);
+        });
+  // parser introspection (unit tests)
+  m.def("h264_index", [](py::bytes b) {
+    std::string s = b;
+    H264Index idx = h264_index_annexb((const u8*)s.data(), s.size());
+    py::dict d;
+    d["width"] = idx.width;
+    d["height"] = idx.height;
+    d["num_frames"] = idx.num_frames;
+    d["sample_offsets"] = idx.sample_offsets;
+    d["sample_sizes"] = idx.sample_sizes;
+    d["keyframe_indices"] = idx.keyframe_indices;
+    d["sps"] = py::bytes((const char*)idx.sps.data(), idx.sps.size());
+    d["pps"] = py::bytes((const char*)idx.pps.data(), idx.pps.size());
+    return d;
+  });
+  m.def("mp4_probe", [](py::bytes b) {
+    std::string s = b;
+    Mp4Track t = mp4_parse((const u8*)s.data(), s.size());
+    py::dict d;
+    d["width"] = t.width;
+    d["height"] = t.height;
+    d["length_size"] = t.length_size;
+    d["sample_offsets"] = t.sample_offsets;
+    d["sample_sizes"] = t.sample_sizes;
+    d["keyframe_indices"] = t.keyframe_indices;
+    d["timescale"] = t.timescale;
+    d["n_sps"] = (i64)t.sps.size();
+    d["n_pps"] = (i64)t.pps.size();
+    return d;
+  });
+  m.def("h264_parse_sps_py", [](py::bytes b) {
+    std::string s = b;
+    H264Sps sps = h264_parse_sps((const u8*)s.data(), s.size());
+    py::dict d;
+    d["width"] = sps.width;
+    d["height"] = sps.height;
+    d["profile_idc"] = sps.profile_idc;
+    return d;
+  });
 
   m.def("write_video_table",
         [](std::shared_ptr<Database> db, const std::string& name,

@@ -391,7 +391,20 @@ void LocalExecutor::init_sources(
         vm[ir.item] = read_video_metadata(*db_, table, src.column, ir.item);
 
       bool any_svc = false;
-      for (auto& kv2 : vm) any_svc |= kv2.second.codec == "svc";
+      for (auto& kv2 : vm) {
+        // Decoder-slot dispatch: h264 tables are fully indexed
+        // (ingest.cpp) but decoding needs the rocDecode/VCN hardware
+        // decoder, which this image does not ship — fail loudly instead
+        // of silently mis-reading (the reference shipped its NVDEC path
+        // disabled the same way, evaluate_worker.cpp:90-93).
+        SCA_CHECK(kv2.second.codec != "h264",
+                  "table '" + src.table +
+                      "' is H.264: decoding requires the rocDecode/VCN "
+                      "hardware decoder (not in this image). Export it "
+                      "with save_mp4() or transcode to codec='svc'. "
+                      "Decoder slot: csrc/video/ingest.h");
+        any_svc |= kv2.second.codec == "svc";
+      }
       if (any_svc) {
         // Codec path (decoder-automaton parity): set up per-item decode
         // states; frames decode per work packet in advance_svc_source, on

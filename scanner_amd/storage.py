@@ -85,6 +85,15 @@ class NamedVideoStream(NamedStream):
     def __init__(self, sc, name, path=None, frames=None, codec="svc",
                  column=None, io_packet_size=128):
         super().__init__(sc, name, column)
+        if path is not None and not self.committed() and \
+                path.lower().endswith((".mp4", ".h264", ".264")):
+            # real video file: demux + H.264 keyframe index (pure parsing;
+            # parity: ingest.cpp:175-380). Decode of the result needs the
+            # rocDecode/VCN hardware decoder.
+            from . import _core
+            _core.ingest_video_file(self._sc._db, self.name,
+                                    self._column or "frame", path)
+            return
         if frames is None and path is not None:
             frames = np.load(path)
         if frames is not None and not self.committed():
@@ -118,10 +127,19 @@ class NamedVideoStream(NamedStream):
 
 
     def save_npy(self, path, rows=None):
-        """Export decoded frames to an .npy file (deviation from the
-        reference's save_mp4: this image ships no H.264 encoder; the SVC
-        stream itself is the compressed representation and .npy is the
-        interchange export)."""
+        """Export decoded frames to an .npy file (for SVC/raw tables; the
+        SVC stream itself is the compressed representation and .npy is the
+        decoded interchange export)."""
         frames = np.stack(list(self.load(rows=rows)))
         np.save(path, frames)
+        return path
+
+    def save_mp4(self, path, fps=30.0):
+        """Remux an ingested H.264 table back into a playable .mp4
+        (no transcode, so no codec library needed; parity:
+        NamedVideoStream.save_mp4, storage.py:353-374). Only valid for
+        tables ingested from real video (codec 'h264')."""
+        from . import _core
+        _core.export_mp4(self._sc._db, self.name, self.column_name(),
+                         path, fps)
         return path

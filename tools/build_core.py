@@ -31,6 +31,9 @@ CPP_SOURCES = [
     "csrc/engine/executor.cpp",
     "csrc/video/svc_cpu.cpp",
     "csrc/video/span_cache.cpp",
+    "csrc/video/h264.cpp",
+    "csrc/video/mp4.cpp",
+    "csrc/video/ingest.cpp",
     "csrc/core_module.cpp",
 ]
 HIP_SOURCES = [
