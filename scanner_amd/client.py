@@ -78,6 +78,7 @@ class Client:
         """rows: list (per row) of list (per column) of bytes."""
         per_col = [[r[c] for r in rows] for c in range(len(columns))]
         _core.write_bytes_table(self._db, name, columns, per_col, 128)
+        return NamedStream(self, name)
 
     def ingest_video_file(self, path, name, column="frame"):
         """Ingest a real video file (.mp4 with an AVC track, or an Annex-B
@@ -85,7 +86,6 @@ class Client:
         pure parsing (parity: Client.ingest_videos -> ingest.cpp:175-380).
         Returns {num_frames, width, height, codec}."""
         return _core.ingest_video_file(self._db, name, column, path)
-        return NamedStream(self, name)
 
     def sequence(self, name):
         return NamedStream(self, name)
