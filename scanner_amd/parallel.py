@@ -30,7 +30,6 @@ CPU fallback uses the gloo backend so the whole module is testable without
 a GPU (tests/test_parallel.py runs world_size=2 over gloo).
 """
 import os
-import pickle
 
 import torch
 import torch.distributed as dist
@@ -74,53 +73,79 @@ def gather_column(blobs, device, dst=0, group=None,
                   max_bytes=_DEFAULT_MAX_BYTES):
     """Gather a list of per-row byte blobs from every rank to `dst`.
 
-    On GPU the payload moves as uint8 CUDA tensors over RCCL/xGMI (single
-    hop on the full mesh); on CPU over gloo. Returns the concatenated
-    per-rank lists (rank order) on `dst`, None elsewhere. Payloads larger
-    than `max_bytes` per rank are sent in multiple rounds so one link is
-    never saturated by a single message.
+    Size-exact point-to-point: after one tiny all_gather of per-rank
+    [n_blobs, total_bytes] headers, every non-dst rank sends its blob-length
+    table and raw concatenated bytes straight to `dst` (isend), and `dst`
+    posts matching irecvs for all ranks at once — on the xGMI full mesh
+    each transfer rides its own direct link concurrently. No pickling, no
+    pad-to-max, and no O(world x max_payload) all_gather fan-out (each byte
+    crosses exactly one link once). `max_bytes` chunks the wire messages so
+    a huge column doesn't sit behind one monolithic send.
+
+    Returns the concatenated per-rank lists (rank order) on `dst`, None
+    elsewhere.
     """
     if not dist.is_initialized() or dist.get_world_size(group) == 1:
-        return list(blobs)
+        return [bytes(b) for b in blobs]
     rank = dist.get_rank(group)
     world = dist.get_world_size(group)
+    dev = device if device.type == "cuda" else None
+    ddev = device if device.type == "cuda" else torch.device("cpu")
 
-    payload = pickle.dumps(blobs, protocol=pickle.HIGHEST_PROTOCOL)
-    chunks = [payload[i:i + max_bytes]
-              for i in range(0, len(payload), max_bytes)] or [b""]
-    n_rounds = torch.tensor([len(chunks)], dtype=torch.int64)
-    if device.type == "cuda":
-        n_rounds = n_rounds.to(device)
-    dist.all_reduce(n_rounds, op=dist.ReduceOp.MAX, group=group)
-    rounds = int(n_rounds.item())
+    lens = [len(b) for b in blobs]
+    total = sum(lens)
+    hdr = torch.tensor([len(blobs), total], dtype=torch.int64, device=dev)
+    hdrs = [torch.zeros(2, dtype=torch.int64, device=dev)
+            for _ in range(world)]
+    dist.all_gather(hdrs, hdr, group=group)
 
-    received = [bytearray() for _ in range(world)]
-    for r in range(rounds):
-        chunk = chunks[r] if r < len(chunks) else b""
-        t = _to_tensor(chunk, device)
-        size = torch.tensor([t.numel()], dtype=torch.int64,
-                            device=device if device.type == "cuda" else None)
-        sizes = [torch.zeros_like(size) for _ in range(world)]
-        dist.all_gather(sizes, size, group=group)
-        maxn = max(int(s.item()) for s in sizes)
-        pad = torch.zeros(maxn, dtype=torch.uint8, device=device)
-        pad[:t.numel()] = t
-        # all_gather rather than gather: supported uniformly on RCCL and
-        # gloo (plain gather is not implemented by the nccl backend), and
-        # on the xGMI full mesh the extra fan-out is one hop per link.
-        outs = [torch.empty(maxn, dtype=torch.uint8, device=device)
-                for _ in range(world)]
-        dist.all_gather(outs, pad, group=group)
-        if rank == dst:
-            for i in range(world):
-                n = int(sizes[i].item())
-                if n:
-                    received[i] += outs[i][:n].cpu().numpy().tobytes()
     if rank != dst:
+        works = []
+        if lens:
+            meta = torch.tensor(lens, dtype=torch.int64, device=dev)
+            works.append(dist.isend(meta, dst, group=group))
+        if total:
+            data = torch.cat([
+                torch.frombuffer(bytearray(b), dtype=torch.uint8)
+                for b in blobs if len(b)]).to(ddev)
+            for off in range(0, total, max_bytes):
+                works.append(dist.isend(data[off:off + max_bytes], dst,
+                                        group=group))
+        for w in works:
+            w.wait()
         return None
+
+    # dst: post size-exact receives for every rank concurrently (per-pair
+    # FIFO matches the sender's meta-then-chunks order)
+    metas, datas, works = {}, {}, []
+    for src in range(world):
+        if src == dst:
+            continue
+        nb, nd = int(hdrs[src][0]), int(hdrs[src][1])
+        if nb:
+            metas[src] = torch.zeros(nb, dtype=torch.int64, device=dev)
+            works.append(dist.irecv(metas[src], src, group=group))
+        if nd:
+            datas[src] = torch.zeros(nd, dtype=torch.uint8, device=ddev)
+            for off in range(0, nd, max_bytes):
+                works.append(dist.irecv(datas[src][off:off + max_bytes],
+                                        src, group=group))
+    for w in works:
+        w.wait()
+
     result = []
-    for i in range(world):
-        result.extend(pickle.loads(bytes(received[i])))
+    for src in range(world):
+        if src == dst:
+            result.extend(bytes(b) for b in blobs)
+            continue
+        if src not in metas:
+            continue
+        src_lens = metas[src].cpu().tolist()
+        raw = datas[src].cpu().numpy().tobytes() if src in datas else b""
+        off = 0
+        for n in src_lens:
+            result.append(raw[off:off + n])
+            off += n
     return result
 
 

@@ -104,3 +104,92 @@ def test_bench_cpu_world2(tmp_path):
     assert result["value"] > 0
     # only rank 0 prints the result line (gloo chatter may appear)
     assert not any(ln.startswith("{") for ln in outs[1][1].splitlines())
+
+
+WORKER8 = r"""
+import os, sys, tempfile
+sys.path.insert(0, os.environ["SCA_REPO"])
+import numpy as np
+import torch
+from scanner_amd import parallel
+
+rank, world, device = parallel.init_from_env(device_type="cpu")
+assert world == 8
+
+import scanner_amd as sp
+
+# Frame-shard DP as an engine mode (SURVEY 2.7): one logical 64-frame clip
+# sharded across 8 ranks; each rank runs the engine on its shard and the
+# per-row result columns come back to rank 0 over the collective plane
+# (gloo here, RCCL/xGMI on GPU boxes).
+n_total, h, w = 64, 48, 64
+yy, xx = np.mgrid[0:h, 0:w]
+all_frames = np.zeros((n_total, h, w, 3), np.uint8)
+for i in range(n_total):
+    all_frames[i, :, :, 0] = (xx + 2 * i) % 256
+    all_frames[i, :, :, 1] = (yy + i) % 256
+    all_frames[i, :, :, 2] = (xx + yy + 3 * i) % 256
+
+s, e = parallel.shard_rows(n_total, world, rank)
+tmp = tempfile.mkdtemp(prefix=f"sca_dp8_r{rank}_")
+sc = sp.Client(db_path=os.path.join(tmp, "db"))
+video = sp.NamedVideoStream(sc, "shard", frames=all_frames[s:e],
+                            codec="svc")
+frame = sc.io.Input([video])
+hist = sc.ops.Histogram(frame=frame)
+out = sp.NamedStream(sc, "shard_out")
+sc.run(sc.io.Output(hist, [out]), sp.PerfParams.manual(4, 8),
+       cache_mode=sp.CacheMode.Overwrite)
+blobs = list(out.load())
+assert len(blobs) == e - s
+
+got = parallel.gather_column(blobs, device)
+if rank == 0:
+    assert len(got) == n_total
+    for i in range(n_total):
+        histo = np.frombuffer(got[i], np.uint32).reshape(3, 256)
+        expect = np.stack([
+            np.bincount(all_frames[i][:, :, c].ravel(), minlength=256)
+            for c in range(3)]).astype(np.uint32)
+        np.testing.assert_array_equal(histo, expect)
+else:
+    assert got is None
+
+# uneven/empty payloads: rank k sends k blobs of k*17 bytes (rank 0 none)
+blobs = [bytes([rank]) * (rank * 17) for _ in range(rank)]
+got = parallel.gather_column(blobs, device, max_bytes=64)
+if rank == 0:
+    expect = []
+    for r in range(world):
+        expect.extend([bytes([r]) * (r * 17)] * r)
+    assert got == expect, (len(got), len(expect))
+print(f"rank {rank} OK8", flush=True)
+"""
+
+
+def test_parallel_gloo_world8_engine_gather(tmp_path):
+    """8-rank frame-shard DP: engine jobs on real column data per rank,
+    size-exact P2P gather to rank 0 (VERDICT r01 #4)."""
+    script = tmp_path / "worker8.py"
+    script.write_text(WORKER8)
+    procs = []
+    for rank in range(8):
+        env = dict(os.environ)
+        env.update({
+            "SCA_REPO": REPO,
+            "RANK": str(rank),
+            "LOCAL_RANK": str(rank),
+            "WORLD_SIZE": "8",
+            "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": "29621",
+        })
+        procs.append(subprocess.Popen(
+            [sys.executable, str(script)], env=env,
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+    outs = []
+    for p in procs:
+        out, _ = p.communicate(timeout=300)
+        outs.append(out.decode())
+    for rank, (p, out) in enumerate(zip(procs, outs)):
+        assert p.returncode == 0, f"rank {rank} failed:\n{out[-3000:]}"
+        assert f"rank {rank} OK8" in out
