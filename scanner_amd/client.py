@@ -40,6 +40,9 @@ class Client:
         self.streams = StreamsGenerator(self)
         self.partitioner = PartitionerGenerator()
         self.io = IOGenerator(self)
+        from .sources import SinksGenerator, SourcesGenerator
+        self.sources = SourcesGenerator(self)
+        self.sinks = SinksGenerator(self)
         self._op_info_cache = {}
         self._last_profilers = None
         # distributed mode
@@ -167,6 +170,12 @@ class Client:
             op_args = {}
             for i, op in enumerate(ops):
                 if op._name == "Input":
+                    if getattr(op, "_source", None):
+                        sources[str(i)] = {
+                            "source": op._source,
+                            "args": msgpack.packb(op._source_args[j]),
+                        }
+                        continue
                     s = op._streams[j]
                     sources[str(i)] = {"table": s.name,
                                        "column": s.column_name()}
@@ -194,12 +203,18 @@ class Client:
                 elif op._stream_args is not None:
                     op_args[str(i)] = msgpack.packb(
                         stream_arg(op._stream_args, j))
-            jobs.append({
+            job = {
                 "sources": sources,
                 "sampling": sampling,
                 "op_args": op_args,
-                "sink_table": out_streams[j].name,
-            })
+            }
+            if getattr(sink_op, "_sink", None):
+                job["sink_table"] = ""
+                job["sink_name"] = sink_op._sink
+                job["sink_args"] = msgpack.packb(sink_op._sink_args[j])
+            else:
+                job["sink_table"] = out_streams[j].name
+            jobs.append(job)
         return graph_bytes, jobs, out_streams, n_jobs
 
     # ---- run ----
@@ -221,6 +236,9 @@ class Client:
         keep = []
         for j, js in enumerate(jobs):
             s = out_streams[j]
+            if s is None:  # user sink: no output table to cache-check
+                keep.append(js)
+                continue
             if s.exists() and self._db.table_committed(s.name):
                 if cache_mode == CacheMode.Error:
                     raise ScannerException(

@@ -14,6 +14,7 @@
 #include "hip_util.h"
 #include "memory.h"
 #include "msgpack.h"
+#include "ops/source.h"
 #include "video/h264.h"
 #include "video/ingest.h"
 #include "video/mp4.h"
@@ -102,6 +103,7 @@ PYBIND11_MODULE(_core, m) {
   register_color_gpu();
   register_image_encoder_op();
   register_detector_op();
+  register_files_source_sink();
 
   // Load a user op plugin .so built with tools/build_op.py (parity:
   // Client.load_op / REGISTER_OP static registrars in user libraries,
@@ -261,6 +263,8 @@ PYBIND11_MODULE(_core, m) {
   });
 
   m.def("registered_ops", [] { return op_registry().names(); });
+  m.def("registered_sources", [] { return source_registry().names(); });
+  m.def("registered_sinks", [] { return sink_registry().names(); });
   m.def("op_info", [](const std::string& name) {
     const OpInfo& o = op_registry().get(name);
     py::dict d;

@@ -64,6 +64,10 @@ mp::Value JobBinding::to_msgpack() const {
     mp::Map sm;
     sm["table"] = kv.second.table;
     sm["column"] = kv.second.column;
+    if (!kv.second.source.empty()) {
+      sm["source"] = kv.second.source;
+      sm["args"] = kv.second.args;
+    }
     srcs[std::to_string(kv.first)] = mp::Value(std::move(sm));
   }
   m["sources"] = std::move(srcs);
@@ -75,6 +79,10 @@ mp::Value JobBinding::to_msgpack() const {
   for (auto& kv : op_args) oa[std::to_string(kv.first)] = kv.second;
   m["op_args"] = std::move(oa);
   m["sink_table"] = sink.table;
+  if (!sink.sink.empty()) {
+    m["sink_name"] = sink.sink;
+    m["sink_args"] = sink.args;
+  }
   return mp::Value(std::move(m));
 }
 
@@ -84,6 +92,9 @@ JobBinding JobBinding::from_msgpack(const mp::Value& v) {
     SourceArgsC s;
     s.table = kv.second.get_str("table", "");
     s.column = kv.second.get_str("column", "");
+    s.source = kv.second.get_str("source", "");
+    auto& sm = kv.second.as_map();
+    if (sm.count("args")) s.args = sm.at("args").as_bin();
     b.sources[std::stoi(kv.first)] = s;
   }
   for (auto& kv : v.as_map().at("sampling").as_map()) {
@@ -93,6 +104,10 @@ JobBinding JobBinding::from_msgpack(const mp::Value& v) {
     b.op_args[std::stoi(kv.first)] = kv.second.as_bin();
   }
   b.sink.table = v.get_str("sink_table", "");
+  b.sink.sink = v.get_str("sink_name", "");
+  if (v.as_map().count("sink_args")) {
+    b.sink.args = v.as_map().at("sink_args").as_bin();
+  }
   return b;
 }
 

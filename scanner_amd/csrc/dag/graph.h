@@ -66,10 +66,18 @@ struct JobGraph {
 struct SourceArgsC {
   std::string table;
   std::string column;
+  // Registered user source (ops/source.h): when non-empty, `source` names
+  // a SourceFactory and `args` are its per-stream msgpack args; the
+  // table/column fields are unused.
+  std::string source;
+  std::vector<u8> args;
 };
 
 struct SinkArgsC {
   std::string table;
+  // Registered user sink: when non-empty, output bypasses table storage.
+  std::string sink;
+  std::vector<u8> args;
 };
 
 struct JobBinding {

@@ -5,6 +5,7 @@
 
 #include "../hip_util.h"
 #include "../memory.h"
+#include "../ops/source.h"
 #include "../video/svc.h"
 
 namespace sca {
@@ -74,6 +75,13 @@ void LocalExecutor::prepare(bool create_outputs) {
   if (prepared_) return;
   validate_graph(graph_);
   SourceRowsFn source_rows = [this](const SourceArgsC& s) -> i64 {
+    if (!s.source.empty()) {
+      // user source: the registered enumerator names the row domain
+      SourceConfig cfg;
+      cfg.args = s.args;
+      return source_registry().get(s.source).make_enumerator(cfg)
+          ->total_elements();
+    }
     return db_->get_table(s.table).num_rows();
   };
   for (auto& job : jobs_) {
@@ -94,11 +102,13 @@ void LocalExecutor::prepare(bool create_outputs) {
                 "Output references unknown column '" + e.column + "'");
       col_types.push_back(psi.output_types[ci]);
     }
-    TableMetadata t =
-        create_outputs
-            ? db_->new_table(jobs_[j].sink.table, col_names, col_types,
-                             /*overwrite=*/true)
-            : db_->get_table(jobs_[j].sink.table);
+    TableMetadata t;
+    if (jobs_[j].sink.sink.empty()) {
+      t = create_outputs
+              ? db_->new_table(jobs_[j].sink.table, col_names, col_types,
+                               /*overwrite=*/true)
+              : db_->get_table(jobs_[j].sink.table);
+    }  // user sink: no output table
     out_tables_.push_back(t);
     // task boundaries
     i64 rows = analyses_[j].output_rows;
@@ -132,6 +142,7 @@ i64 LocalExecutor::total_output_rows() const {
 }
 
 void LocalExecutor::finalize_job(i32 job) {
+  if (!jobs_[job].sink.sink.empty()) return;  // user sink: nothing to commit
   TableMetadata t = out_tables_[job];
   t.end_rows = task_rows_[job];
   db_->update_table(t);
@@ -380,6 +391,22 @@ void LocalExecutor::init_sources(
     const std::vector<i64>& rows = kv.second;
     const SourceArgsC& src = job.sources.at(op_idx);
     const OpStaticInfo& si = analyses_[t.job].info[op_idx];
+    if (!src.source.empty()) {
+      // user source (ops/source.h): reads exact rows, CPU-resident
+      Profiler::Scope s(inst.profiler, "load:source:" + src.source);
+      SourceConfig cfg;
+      cfg.args = src.args;
+      cfg.profiler = inst.profiler;
+      auto src_inst = source_registry().get(src.source).make(cfg);
+      ElementVector elems;
+      src_inst->read(rows, elems);
+      SCA_CHECK(elems.size() == rows.size(),
+                "source '" + src.source + "' produced " +
+                    std::to_string(elems.size()) + " rows, wanted " +
+                    std::to_string(rows.size()));
+      for (auto& e : elems) outs[op_idx][si.output_columns[0]][e.index] = e;
+      continue;
+    }
     TableMetadata table = db_->get_table(src.table);
     const std::string& col_name = si.output_columns[0];
     Profiler::Scope s(inst.profiler, "load:" + src.table);
@@ -835,6 +862,7 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t,
               if (e.is_null) {
                 Element ne;
                 ne.is_null = true;
+                ne.index = row;
                 sink_cols[c].push_back(ne);
                 sink_owned[c].push_back(0);
                 continue;
@@ -948,7 +976,25 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t,
     }
 
     // ---- task-end write: one storage item per (column, task) ----
-    {
+    if (!job.sink.sink.empty()) {
+      // user sink (ops/source.h): bypasses table storage entirely
+      Profiler::Scope s(inst.profiler, "save:sink:" + job.sink.sink);
+      SinkConfig cfg;
+      cfg.args = job.sink.args;
+      cfg.profiler = inst.profiler;
+      auto sink_inst = sink_registry().get(job.sink.sink).make(cfg);
+      sink_inst->new_task(t.task);
+      sink_inst->write(sink_cols);
+      sink_inst->finished();
+      for (size_t c = 0; c < sink_cols.size(); ++c) {
+        for (size_t k = 0; k < sink_cols[c].size(); ++k) {
+          if (sink_owned[c][k] && sink_cols[c][k].buffer)
+            delete_buffer(CPU_DEVICE, sink_cols[c][k].buffer);
+        }
+        sink_cols[c].clear();
+        sink_owned[c].clear();
+      }
+    } else {
       Profiler::Scope s(inst.profiler, "save");
       const TableMetadata& table = out_tables_[t.job];
       auto& cols = sink_cols;

@@ -27,6 +27,7 @@ CPP_SOURCES = [
     "csrc/ops/pose.cpp",
     "csrc/ops/image_encoder.cpp",
     "csrc/ops/detector.cpp",
+    "csrc/ops/files_source.cpp",
     "csrc/engine/table_io.cpp",
     "csrc/engine/executor.cpp",
     "csrc/video/svc_cpu.cpp",
