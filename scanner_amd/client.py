@@ -21,7 +21,7 @@ from .streams import PartitionerGenerator, StreamsGenerator
 class Client:
     def __init__(self, db_path=None, master=None, workers=None,
                  start_cluster=True, recover=True, config=None,
-                 config_path=None):
+                 config_path=None, storage_type=None, bucket=None):
         if config is None and (config_path is not None
                                or db_path is None or master is None):
             from .config import Config
@@ -30,10 +30,20 @@ class Client:
         if config is not None:
             db_path = db_path or config.db_path
             master = master or config.master_address
+            storage_type = storage_type or config.storage_type
+            bucket = bucket or config.bucket
         self._db_path = db_path or os.path.join(
             tempfile.gettempdir(), "scanner_amd_db")
-        os.makedirs(self._db_path, exist_ok=True)
-        self._db = _core.Database(self._db_path)
+        self._storage_type = storage_type or "posix"
+        self._bucket = bucket or ""
+        if self._storage_type == "posix":
+            os.makedirs(self._db_path, exist_ok=True)
+            self._db = _core.Database(self._db_path)
+        else:
+            # object-store backend (S3 semantics emulated over a local
+            # bucket dir; config.py [storage] type="s3" bucket="...")
+            self._db = _core.Database(self._db_path, self._storage_type,
+                                      self._bucket)
         if recover:
             self._db.recover()
         self.ops = OpGenerator(self)

@@ -40,10 +40,16 @@ class Config:
                 raise ScannerException(
                     f"failed to parse {self.config_path}: {e}")
         storage = data.get("storage", {})
-        if storage.get("type", "posix") != "posix":
+        self.storage_type = storage.get("type", "posix")
+        if self.storage_type not in ("posix", "s3", "gcs", "object"):
             raise ScannerException(
-                "only storage type 'posix' is built in; other backends "
-                "implement StorageBackend (csrc/storage.h)")
+                f"unknown storage type '{self.storage_type}' (posix, or "
+                "s3/gcs/object via the object-store backend, "
+                "csrc/storage.h)")
+        self.bucket = os.path.expanduser(storage.get("bucket", ""))
+        if self.storage_type != "posix" and not self.bucket:
+            raise ScannerException(
+                f"storage type '{self.storage_type}' needs a bucket path")
         network = data.get("network", {})
 
         self.db_path = db_path or os.path.expanduser(

@@ -291,6 +291,22 @@ PYBIND11_MODULE(_core, m) {
       .def(py::init([](const std::string& path) {
         return std::make_shared<Database>(StorageBackend::make_posix(), path);
       }))
+      .def(py::init([](const std::string& path,
+                       const std::string& storage_type,
+                       const std::string& bucket) {
+        std::unique_ptr<StorageBackend> s;
+        if (storage_type == "posix") {
+          s = StorageBackend::make_posix();
+        } else if (storage_type == "s3" || storage_type == "gcs" ||
+                   storage_type == "object") {
+          SCA_CHECK(!bucket.empty(),
+                    "object storage needs a bucket path");
+          s = StorageBackend::make_object_store(bucket);
+        } else {
+          throw ScannerError("unknown storage type '" + storage_type + "'");
+        }
+        return std::make_shared<Database>(std::move(s), path);
+      }))
       .def("recover", &Database::recover)
       .def("table_names", &Database::table_names)
       .def("has_table", &Database::has_table)

@@ -10,6 +10,7 @@
 #include <cerrno>
 #include <cstdio>
 #include <cstring>
+#include <set>
 #include <thread>
 
 namespace sca {
@@ -162,10 +163,127 @@ class PosixStorage : public StorageBackend {
   }
 };
 
+// S3-semantics emulation over a local "bucket" directory (capability
+// parity: the reference's GCS/S3 storehouse configs, scannerpy
+// config.py:75-89). Contract differences from POSIX that callers must not
+// assume away — this backend enforces them:
+//   * flat keyspace: keys map to url-encoded filenames in ONE directory;
+//     there are no directories, make_dirs is a no-op
+//   * whole-object PUT: write_all replaces the object atomically (no
+//     rename contract exposed, no partial writes visible)
+//   * range GET: read_range serves [offset, offset+size)
+//   * prefix scan: list_dir/remove_tree operate on key prefixes
+// A networked S3/GCS client slots in behind the same class by swapping the
+// file operations for HTTP calls.
+class ObjectStorage : public StorageBackend {
+ public:
+  explicit ObjectStorage(std::string bucket) : bucket_(std::move(bucket)) {
+    // create the bucket itself (the only real directory)
+    std::string cur;
+    size_t i = 0;
+    while (i < bucket_.size()) {
+      size_t j = bucket_.find('/', i + 1);
+      if (j == std::string::npos) j = bucket_.size();
+      cur = bucket_.substr(0, j);
+      if (!cur.empty() && cur != "/") (void)mkdir(cur.c_str(), 0755);
+      i = j;
+    }
+  }
+
+  std::vector<u8> read_all(const std::string& key) override {
+    return posix_.read_all(keyfile(key));
+  }
+  void read_range(const std::string& key, u64 offset, u64 size,
+                  u8* out) override {
+    posix_.read_range(keyfile(key), offset, size, out);
+  }
+  u64 file_size(const std::string& key) override {
+    return posix_.file_size(keyfile(key));
+  }
+  void write_all(const std::string& key, const u8* data,
+                 size_t size) override {
+    posix_.write_all(keyfile(key), data, size);  // PUT (atomic replace)
+  }
+  bool exists(const std::string& key) override {
+    return posix_.exists(keyfile(key));
+  }
+  void remove(const std::string& key) override {
+    posix_.remove(keyfile(key));
+  }
+  void remove_tree(const std::string& prefix) override {
+    for (auto& name : bucket_keys()) {
+      std::string key = decode(name);
+      if (key == prefix || key.rfind(prefix + "/", 0) == 0) {
+        posix_.remove(bucket_ + "/" + name);
+      }
+    }
+  }
+  void make_dirs(const std::string&) override {}  // no directories
+  std::vector<std::string> list_dir(const std::string& prefix) override {
+    // next path segment under prefix/, deduped (S3 delimiter listing)
+    std::set<std::string> segs;
+    std::string p = prefix + "/";
+    for (auto& name : bucket_keys()) {
+      std::string key = decode(name);
+      if (key.rfind(p, 0) != 0) continue;
+      std::string rest = key.substr(p.size());
+      size_t cut = rest.find('/');
+      segs.insert(cut == std::string::npos ? rest : rest.substr(0, cut));
+    }
+    return std::vector<std::string>(segs.begin(), segs.end());
+  }
+
+ private:
+  std::string keyfile(const std::string& key) const {
+    std::string out;
+    out.reserve(key.size());
+    for (char c : key) {
+      if (c == '/') {
+        out += "%2F";
+      } else if (c == '%') {
+        out += "%25";
+      } else {
+        out.push_back(c);
+      }
+    }
+    return bucket_ + "/" + out;
+  }
+  static std::string decode(const std::string& name) {
+    std::string out;
+    for (size_t i = 0; i < name.size(); ++i) {
+      if (name[i] == '%' && i + 2 < name.size()) {
+        if (name.compare(i, 3, "%2F") == 0) {
+          out.push_back('/');
+          i += 2;
+          continue;
+        }
+        if (name.compare(i, 3, "%25") == 0) {
+          out.push_back('%');
+          i += 2;
+          continue;
+        }
+      }
+      out.push_back(name[i]);
+    }
+    return out;
+  }
+  std::vector<std::string> bucket_keys() {
+    return posix_.list_dir(bucket_);
+  }
+
+  std::string bucket_;
+  PosixStorage posix_;
+};
+
 }  // namespace
 
 std::unique_ptr<StorageBackend> StorageBackend::make_posix() {
   return std::make_unique<PosixStorage>();
+}
+
+std::unique_ptr<StorageBackend> StorageBackend::make_object_store(
+    const std::string& bucket_dir) {
+  return std::make_unique<ObjectStorage>(bucket_dir);
 }
 
 }  // namespace sca

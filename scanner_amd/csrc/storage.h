@@ -30,6 +30,12 @@ class StorageBackend {
   virtual std::vector<std::string> list_dir(const std::string& path) = 0;
 
   static std::unique_ptr<StorageBackend> make_posix();
+  // S3-semantics object store (flat keyspace, whole-object PUT, range GET,
+  // prefix listing) emulated over a local bucket directory; a networked
+  // S3/GCS client implements the same class (reference: storehouse GCS/S3
+  // configs, scannerpy config.py:75-89).
+  static std::unique_ptr<StorageBackend> make_object_store(
+      const std::string& bucket_dir);
 };
 
 // Path scheme (mirrors metadata.h:37-86):
