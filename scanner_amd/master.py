@@ -130,30 +130,35 @@ class MasterServer:
         return {"ok": True}
 
     def _new_job(self, req):
+        # id allocation under the lock; the heavy half (graph analysis +
+        # output-table creation) runs OUTSIDE it so pings/NextWork from 8+
+        # workers are never blocked behind a large job graph (ref: the
+        # async master runs process_job on its own thread,
+        # master.cpp:1328-1353)
         with self._lock:
             jid = self._job_counter
             self._job_counter += 1
             py_ops = [{"name": n, **v}
                       for n, v in self._registered_ops.items()]
-            job = _BulkJob(jid, req["graph"], req["jobs"], req["perf"],
-                           py_ops)
-            # analysis + output-table creation on the master; keep the
-            # executor for stream finalization (same task boundaries)
-            perf_m = dict(job.perf)
-            perf_m["cpu_pool_size"] = 0
-            perf_m["gpu_pool_size"] = 0
-            perf_m["pipeline_instances"] = 1
-            # reload metadata: the client process may have ingested tables
-            # since this Database handle was opened (shared storage)
-            self._db = _core.Database(self._db_path)
-            job.ex = _core.LocalExecutor(
-                self._db, job.graph, msgpack.packb(job.jobs), perf_m, [])
-            job.ex.prepare(True)
-            for (stream, task, start, end) in job.ex.all_tasks():
-                t = (stream, task, start, end)
-                job.tasks.append(t)
-                job.to_assign.append(t)
-                job.stream_tasks[stream] += 1
+        job = _BulkJob(jid, req["graph"], req["jobs"], req["perf"], py_ops)
+        perf_m = dict(job.perf)
+        perf_m["cpu_pool_size"] = 0
+        perf_m["gpu_pool_size"] = 0
+        perf_m["pipeline_instances"] = 1
+        # fresh Database handle: the client process may have ingested
+        # tables since this one was opened (shared storage); table
+        # creation itself is serialized by the db file lock
+        db = _core.Database(self._db_path)
+        job.ex = _core.LocalExecutor(
+            db, job.graph, msgpack.packb(job.jobs), perf_m, [])
+        job.ex.prepare(True)
+        for (stream, task, start, end) in job.ex.all_tasks():
+            t = (stream, task, start, end)
+            job.tasks.append(t)
+            job.to_assign.append(t)
+            job.stream_tasks[stream] += 1
+        with self._lock:
+            self._db = db
             self._jobs_by_id[jid] = job
             # promote any queued job first so a newly submitted bulk job
             # cannot jump ahead of jobs already waiting in the FIFO queue
@@ -287,7 +292,14 @@ class MasterServer:
             time.sleep(PING_INTERVAL)
             with self._lock:
                 workers = list(self._workers.values())
-            for w in workers:
+            if not workers:
+                continue
+
+            # ping all workers concurrently: a dead worker's 5 s timeout
+            # must not delay liveness detection of the other 7+ (reference
+            # pings each worker on the async completion queue,
+            # master.cpp:1837-1965)
+            def ping_one(w):
                 ok = w.client.try_call("Ping", {}, timeout=5)
                 if ok is None:
                     w.failed_pings += 1
@@ -296,6 +308,13 @@ class MasterServer:
                             self._remove_worker(w.id)
                 else:
                     w.failed_pings = 0
+
+            threads = [threading.Thread(target=ping_one, args=(w,),
+                                        daemon=True) for w in workers]
+            for t in threads:
+                t.start()
+            for t in threads:
+                t.join()
 
     def _monitor_loop(self):
         while not self._shutdown.is_set():
