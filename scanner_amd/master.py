@@ -241,7 +241,8 @@ class MasterServer:
             self._promote_if_done()
             job = self._jobs_by_id.get(req.get("job_id", -1))
             if job is None:
-                return {"exists": False}
+                # n_workers always reported (cluster bootstrap polls it)
+                return {"exists": False, "n_workers": len(self._workers)}
             return {
                 "exists": True,
                 "complete": job.complete.is_set(),

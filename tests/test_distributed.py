@@ -327,3 +327,44 @@ def test_master_stress_8workers_2jobs_churn(tmp_path):
             except ProcessLookupError:
                 pass
         master.shutdown()
+
+
+def test_cluster_bootstrap_localhost(tmp_path):
+    """ssh-style cluster bootstrap (parity: reference client.py:596-783)
+    with the transport pointed at local bash — exercises the full remote
+    launch path (nohup, pid capture, readiness polling, teardown) without
+    sshd."""
+    from scanner_amd.cluster import bootstrap_cluster
+    db = _mk_db(tmp_path)
+    os.environ.setdefault("PYTHONPATH", "")
+    old_pp = os.environ["PYTHONPATH"]
+    os.environ["PYTHONPATH"] = REPO + os.pathsep + old_pp
+    try:
+        import socket
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+        s.close()
+        cluster = bootstrap_cluster(
+            db, master_host="", worker_hosts=["", ""],
+            master_port=port, master_advertise="127.0.0.1",
+            ssh_cmd=["bash", "-c"], log_dir=str(tmp_path))
+        try:
+            sc = sp.Client(db_path=db, master=cluster.master_addr)
+            vid = make_video(n=10)
+            stream = sp.NamedVideoStream(sc, "cb", frames=vid, codec="raw")
+            frame = sc.io.Input([stream])
+            hist = sc.ops.Histogram(frame=frame)
+            out = sp.NamedStream(sc, "cb_out")
+            sc.run(sc.io.Output(hist, [out]), sp.PerfParams.manual(2, 4),
+                   cache_mode=sp.CacheMode.Overwrite)
+            rows = list(out.load())
+            assert len(rows) == 10
+            for r, blob in enumerate(rows):
+                got = np.frombuffer(blob, np.uint32).reshape(3, 256)
+                np.testing.assert_array_equal(got, ref_histogram(vid[r]))
+            sc.shutdown()
+        finally:
+            cluster.stop()
+    finally:
+        os.environ["PYTHONPATH"] = old_pp
