@@ -103,6 +103,11 @@ class ResNet50KernelGPU : public BatchedKernel {
     auto a = mp::decode(cfg.args);
     weights_file_ = a.get_str("weights_file", "");
     seed_ = (u64)a.get_int("seed", 1234);
+    // Numerics-test tap: when set ("conv1", "maxpool", "block<N>",
+    // "avgpool"), the op outputs that point's NHWC activation (f32) instead
+    // of logits, so tests can bound bf16 error per stage instead of only
+    // at the logits.
+    debug_tap_ = a.get_str("debug_tap", "");
     batch_ = std::max(1, cfg.max_batch);
     model_ = get_model(cfg.device, weights_file_, seed_);
     const char* env = std::getenv("SCANNER_HIPGRAPH");
@@ -147,6 +152,9 @@ class ResNet50KernelGPU : public BatchedKernel {
 
   std::string weights_file_;
   u64 seed_;
+  std::string debug_tap_;
+  u8* tap_buf_ = nullptr;
+  int tap_h_ = 0, tap_w_ = 0, tap_c_ = 0;
   i32 batch_;
   bool graphs_enabled_;
   std::shared_ptr<DeviceModel> model_;
@@ -224,6 +232,21 @@ u8* ResNet50KernelGPU::run_forward(int n, void* s) {
     }
   };
 
+  auto tap = [&](const std::string& name, u8* buf, int th, int tw,
+                 int tc) -> bool {
+    if (debug_tap_ != name) return false;
+    tap_buf_ = buf;
+    tap_h_ = th;
+    tap_w_ = tw;
+    tap_c_ = tc;
+    return true;
+  };
+  // Mutation seam for the numerics tests: SCANNER_RESNET_SKIP_RESIDUAL=
+  // block<N> drops that block's residual add — the per-stage activation
+  // test must catch this where a logits-only top-1 check may not.
+  const char* mut = std::getenv("SCANNER_RESNET_SKIP_RESIDUAL");
+  std::string skip_residual = mut ? mut : "";
+
   f32* mean = model_->mean;
   preprocess_frames_bf16(ws_.d_ptrs, n, ws_.ih, ws_.iw, ws_.ic, 224, ws_.pre,
                          mean, mean + 3, s);
@@ -231,8 +254,10 @@ u8* ResNet50KernelGPU::run_forward(int n, void* s) {
   conv("conv1", ws_.pre, h, w, ws_.act0, nullptr, oh, ow);
   h = oh;
   w = ow;
+  if (tap("conv1", ws_.act0, h, w, 64)) return ws_.act0;
   maxpool3x3s2_bf16(ws_.act0, n, h, w, 64, ws_.act1, 56, 56, s);
   h = w = 56;
+  if (tap("maxpool", ws_.act1, h, w, 64)) return ws_.act1;
   // Three rotating activation buffers: x holds the block input; a/b are
   // the two others. conv3 writes into a (its conv1 temp is dead by then),
   // never into the residual source.
@@ -257,12 +282,15 @@ u8* ResNet50KernelGPU::run_forward(int n, void* s) {
     conv((p + ".conv2").c_str(), a, h, w, bbuf, nullptr, oh, ow);
     h = oh;
     w = ow;
-    conv((p + ".conv3").c_str(), bbuf, h, w, a, identity, oh, ow);
+    conv((p + ".conv3").c_str(), bbuf, h, w, a,
+         skip_residual == p ? nullptr : identity, oh, ow);
     x = a;
+    if (tap(p, x, h, w, bk.out_c)) return x;
   }
   // x: [n,7,7,2048]
   u8* y = bufs[0] == x ? bufs[1] : bufs[0];
   global_avgpool_bf16(x, n, 7, 7, 2048, y, s);
+  if (tap("avgpool", y, 1, 1, 2048)) return y;
   {
     const auto& e = model_->convs[model_->by_name.at("fc")];
     GemmArgs g;
@@ -302,6 +330,26 @@ void ResNet50KernelGPU::execute_batch(const BatchedElements& in,
       hipMemcpyAsync(ws_.d_ptrs, ws_.h_ptrs, n * sizeof(u8*),
                      hipMemcpyHostToDevice, (hipStream_t)s);
   SCA_CHECK(he == hipSuccess, "resnet pointer upload failed");
+
+  if (!debug_tap_.empty()) {
+    // tap path: uncaptured forward, output the tapped activation as f32
+    tap_buf_ = nullptr;
+    u8* fb = run_forward(n, s);
+    SCA_CHECK(tap_buf_ == fb && tap_buf_,
+              "unknown debug_tap '" + debug_tap_ + "'");
+    size_t elems = (size_t)tap_h_ * tap_w_ * tap_c_;
+    u8* out_block = new_block_buffer(dev, (size_t)n * elems * 4, n);
+    bf16_rows_to_f32(tap_buf_, n, (int)elems, (int)elems, out_block, s);
+    sync_per_thread_stream();
+    for (int i = 0; i < n; ++i) {
+      Element e;
+      e.buffer = out_block + (size_t)i * elems * 4;
+      e.size = elems * 4;
+      e.device = dev;
+      out[0].push_back(e);
+    }
+    return;
+  }
 
   u8* final_buf = nullptr;
   auto it = graph_exec_.find(n);

@@ -71,11 +71,16 @@ def write_tensor_file(path, tensors):
             f.write(data.tobytes())
 
 
-def torch_reference(tensors, frames_u8):
+def torch_reference(tensors, frames_u8, tap=None):
     """fp32 reference forward on CPU via torch.nn.functional. frames_u8:
-    [N,H,W,C] u8. Returns [N,1000] f32 logits."""
+    [N,H,W,C] u8. Returns [N,1000] f32 logits — or, with `tap` set
+    ("conv1", "maxpool", "block<N>", "avgpool"), that point's activation
+    as NHWC f32 (matching the op's debug_tap output layout)."""
     import torch
     import torch.nn.functional as F
+
+    def nhwc(t):
+        return t.permute(0, 2, 3, 1).contiguous().numpy()
 
     mean = torch.tensor([0.485, 0.456, 0.406]).view(1, 3, 1, 1)
     std = torch.tensor([0.229, 0.224, 0.225]).view(1, 3, 1, 1)
@@ -99,7 +104,11 @@ def torch_reference(tensors, frames_u8):
         return y
 
     x = conv("conv1", x, 2, 3, True)
+    if tap == "conv1":
+        return nhwc(x)
     x = F.max_pool2d(x, 3, stride=2, padding=1)
+    if tap == "maxpool":
+        return nhwc(x)
     for b, bk in enumerate(BLOCKS):
         p = f"block{b}"
         identity = x
@@ -108,7 +117,13 @@ def torch_reference(tensors, frames_u8):
         y = conv(f"{p}.conv1", x, 1, 0, True)
         y = conv(f"{p}.conv2", y, bk["stride"], 1, True)
         x = conv(f"{p}.conv3", y, 1, 0, True, residual=identity)
+        if tap == p:
+            return nhwc(x)
     x = x.mean(dim=(2, 3))
+    if tap == "avgpool":
+        return x.numpy().reshape(x.shape[0], 1, 1, 2048)
+    if tap is not None:
+        raise ValueError(f"unknown tap '{tap}'")
     w = torch.from_numpy(
         np.ascontiguousarray(tensors["fc.weight"])).view(1000, 2048)
     b = torch.from_numpy(tensors["fc.bias"])
