@@ -12,7 +12,17 @@ from .client import Client
 from .job import Job
 from .op import Kernel, register_python_op
 from .storage import NamedStream, NamedVideoStream
-from . import parallel, types
+from . import types
+
+
+def __getattr__(name):
+    # `parallel` imports torch (~1.5 s cold): load it lazily so clients,
+    # workers and kernel subprocesses that never touch the RCCL data plane
+    # don't pay for it (PEP 562).
+    if name == "parallel":
+        from . import parallel
+        return parallel
+    raise AttributeError(f"module 'scanner_amd' has no attribute '{name}'")
 
 __version__ = "0.1.0"
 

@@ -152,10 +152,18 @@ _PY_OP_PICKLES = {}
 
 def register_python_op(name=None, device_type=DeviceType.CPU, batch=0,
                        stencil=None, bounded_state=False, warmup=0,
-                       unbounded_state=False):
+                       unbounded_state=False, isolation="thread"):
     """Decorator registering a Python function or Kernel class as an op
     (parity: @scannerpy.register_python_op op.py:317). Input/output columns
-    derive from the annotations of `execute` (class) or the function."""
+    derive from the annotations of `execute` (class) or the function.
+
+    isolation="process" runs each kernel instance in its own child Python
+    process connected by pipes (parity: the reference's PythonKernel
+    subprocess model, python_kernel.cpp:30-103): N pipeline instances of a
+    CPU-heavy Python op then compute on N cores instead of serializing on
+    this process's GIL. The parent blocks in pipe reads (GIL released)
+    while children work; inputs/outputs cross the pipe pickled. Default
+    "thread" runs in-process (no marshaling cost — right for cheap ops)."""
 
     def deco(fn_or_cls):
         from . import _core
@@ -193,6 +201,12 @@ def register_python_op(name=None, device_type=DeviceType.CPU, batch=0,
         eff_batch = batch if batch > 0 else (1024 if batched else 1)
 
         def factory(args_bytes):
+            if isolation == "process":
+                import cloudpickle
+                payload = cloudpickle.dumps(
+                    (fn_or_cls, is_cls, args_bytes, batched, stenciled,
+                     len(in_cols), out_cols))
+                return _SubprocessKernel(payload)
             return _PyKernelAdapter(fn_or_cls, is_cls, args_bytes, batched,
                                     stenciled, len(in_cols), out_cols)
 
@@ -315,3 +329,120 @@ class _PyKernelAdapter:
                 for c in range(self._n_out):
                     outs[c].append(result[c])
         return outs
+
+
+# ---- subprocess kernel isolation (parity: PythonKernel child process,
+# reference python_kernel.cpp:30-103 + kernel.py python_kernel_fn) ----
+
+def _pipe_send(f, obj):
+    import pickle
+    import struct
+    data = pickle.dumps(obj, protocol=pickle.HIGHEST_PROTOCOL)
+    f.write(struct.pack("<Q", len(data)))
+    f.write(data)
+    f.flush()
+
+
+def _pipe_recv(f):
+    import pickle
+    import struct
+    hdr = f.read(8)
+    if len(hdr) < 8:
+        raise EOFError("kernel subprocess pipe closed")
+    (n,) = struct.unpack("<Q", hdr)
+    return pickle.loads(f.read(n))
+
+
+def _child_main():
+    """Entry point of a kernel child process: builds the adapter from the
+    cloudpickled payload, then serves method calls over stdin/stdout."""
+    import sys
+
+    import cloudpickle
+    inp = sys.stdin.buffer
+    outp = sys.stdout.buffer
+    # anything the user op prints must not corrupt the pipe
+    sys.stdout = sys.stderr
+    kind, payload = _pipe_recv(inp)
+    assert kind == "init"
+    adapter = _PyKernelAdapter(*cloudpickle.loads(payload))
+    _pipe_send(outp, ("ok", None))
+    while True:
+        try:
+            msg = _pipe_recv(inp)
+        except EOFError:
+            return
+        kind = msg[0]
+        if kind == "exit":
+            return
+        try:
+            result = getattr(adapter, kind)(*msg[1:])
+            _pipe_send(outp, ("ok", result))
+        except Exception as e:
+            import traceback
+            _pipe_send(outp, ("err", f"{type(e).__name__}: {e}\n"
+                              + traceback.format_exc()))
+
+
+class _SubprocessKernel:
+    """Runs a _PyKernelAdapter in a child process; this proxy mirrors its
+    interface for the C++ bridge. The parent blocks in pipe reads — the
+    GIL is released during the read syscall — so N pipeline instances of a
+    CPU-heavy kernel genuinely use N cores."""
+
+    def __init__(self, payload):
+        import os
+        import subprocess
+        import sys
+        env = dict(os.environ)
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+        self._p = subprocess.Popen(
+            [sys.executable, "-c",
+             "from scanner_amd.op import _child_main; _child_main()"],
+            stdin=subprocess.PIPE, stdout=subprocess.PIPE, env=env)
+        self._call_raw("init", payload)
+
+    def _call_raw(self, kind, *args):
+        if self._p.poll() is not None:
+            raise ScannerException("kernel subprocess died "
+                                   f"(rc={self._p.returncode})")
+        _pipe_send(self._p.stdin, (kind,) + args)
+        status, value = _pipe_recv(self._p.stdout)
+        if status == "err":
+            raise ScannerException(f"python kernel (subprocess): {value}")
+        return value
+
+    def new_stream(self, args_bytes):
+        return self._call_raw("new_stream", args_bytes)
+
+    def reset(self):
+        return self._call_raw("reset")
+
+    def fetch_resources(self, args_bytes):
+        return self._call_raw("fetch_resources", args_bytes)
+
+    def setup_with_resources(self, args_bytes):
+        return self._call_raw("setup_with_resources", args_bytes)
+
+    def execute(self, cols):
+        # numpy views from C++ pickle as copies — exactly the marshaling
+        # the reference pays on its pipes too
+        return self._call_raw("execute", cols)
+
+    def close(self):
+        p = getattr(self, "_p", None)
+        if p is None or p.poll() is not None:
+            return
+        try:
+            _pipe_send(p.stdin, ("exit",))
+            p.stdin.close()
+            p.wait(timeout=5)
+        except Exception:
+            p.kill()
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
