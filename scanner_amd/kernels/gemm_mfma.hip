@@ -413,7 +413,7 @@ __global__ void __launch_bounds__(256, 2)
 // scratch and a reduce kernel applies the epilogue. No inter-workgroup
 // hand-off inside a launch (guide correctness boundary G16) — the split
 // and reduce are separate kernels on the same stream.
-template <int BM, int BN, int WM, int WN, bool IMPLICIT>
+template <int BM, int BN, int WM, int WN, bool IMPLICIT, bool ATOMIC>
 __global__ void __launch_bounds__(256, 2)
     gemm_bf16_splitk_kernel(const bf16* __restrict__ A,
                             const bf16* __restrict__ B, int M, int N, int K,
@@ -535,19 +535,70 @@ __global__ void __launch_bounds__(256, 2)
     __syncthreads();
   }
 
-  // f32 partials: [split][M][N]; lanes 0-15 write 64-byte runs
-  float* out = partials + (size_t)split * M * N;
+  if constexpr (ATOMIC) {
+    // Atomic split-K: all splits add into ONE pre-zeroed f32 [M][N]
+    // buffer (f32 atomic adds are commutative — no inter-WG ordering,
+    // G16-safe); a finalize kernel applies the epilogue. Vs the
+    // partials+reduce pair this removes the S x M x N reduce read
+    // (measured ~12% of all flagship GPU time).
 #pragma unroll
-  for (int i = 0; i < FM; ++i) {
+    for (int i = 0; i < FM; ++i) {
 #pragma unroll
-    for (int j = 0; j < FN; ++j) {
-      int col = n0 + wcol * (BN / WN) + j * 16 + (lane & 15);
+      for (int j = 0; j < FN; ++j) {
+        int col = n0 + wcol * (BN / WN) + j * 16 + (lane & 15);
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int row = m0 + wrow * (BM / WM) + i * 16 + (lane >> 4) * 4 + r;
-        if (row < M) out[(size_t)row * N + col] = acc[i][j][r];
+        for (int r = 0; r < 4; ++r) {
+          int row = m0 + wrow * (BM / WM) + i * 16 + (lane >> 4) * 4 + r;
+          if (row < M)
+            atomicAdd(&partials[(size_t)row * N + col], acc[i][j][r]);
+        }
       }
     }
+  } else {
+    // f32 partials: [split][M][N]; lanes 0-15 write 64-byte runs
+    float* out = partials + (size_t)split * M * N;
+#pragma unroll
+    for (int i = 0; i < FM; ++i) {
+#pragma unroll
+      for (int j = 0; j < FN; ++j) {
+        int col = n0 + wcol * (BN / WN) + j * 16 + (lane & 15);
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = m0 + wrow * (BM / WM) + i * 16 + (lane >> 4) * 4 + r;
+          if (row < M) out[(size_t)row * N + col] = acc[i][j][r];
+        }
+      }
+    }
+  }
+}
+
+template <bool RELU, bool RESIDUAL>
+__global__ void __launch_bounds__(256)
+    splitk_finalize_kernel(const float* __restrict__ acc, i64 mn, int N,
+                           const float* __restrict__ scale,
+                           const float* __restrict__ bias,
+                           const bf16* __restrict__ residual,
+                           bf16* __restrict__ C) {
+  i64 quads = mn / 4;
+  i64 gs = (i64)gridDim.x * blockDim.x;
+  for (i64 q = (i64)blockIdx.x * blockDim.x + threadIdx.x; q < quads;
+       q += gs) {
+    i64 i = q * 4;
+    float4 a4 = reinterpret_cast<const float4*>(acc)[q];
+    float v4[4] = {a4.x, a4.y, a4.z, a4.w};
+    bf16 o4[4];
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      float v = v4[k];
+      int col = (int)((i + k) % N);
+      if (scale) v *= scale[col];
+      if (bias) v += bias[col];
+      if constexpr (RESIDUAL) v += bf16_to_f32(residual[i + k]);
+      if constexpr (RELU) v = v > 0.f ? v : 0.f;
+      o4[k] = f32_to_bf16(v);
+    }
+    *reinterpret_cast<uint64_t*>(C + i) =
+        *reinterpret_cast<const uint64_t*>(o4);
   }
 }
 
@@ -613,30 +664,64 @@ bool try_splitk(const GemmArgs& g, hipStream_t s,
   if (splits < 2) return false;
   int ksteps_per_split = (ksteps + splits - 1) / splits;
   splits = (ksteps + ksteps_per_split - 1) / ksteps_per_split;
-  if (dd) {
-    gemm_bf16_splitk_kernel<BM, BN, 2, 2, true>
+  i64 mn = (i64)g.M * g.N;
+  // Atomic accumulation (default): one pre-zeroed f32 [M][N] buffer, no
+  // S x M x N reduce read. SCANNER_SPLITK_ATOMIC=0 restores the
+  // partials+reduce pair for A/B.
+  static const bool kAtomic = []() {
+    const char* e = std::getenv("SCANNER_SPLITK_ATOMIC");
+    return !(e && e[0] == '0');
+  }();
+  hipError_t e;
+  if (kAtomic) {
+    e = hipMemsetAsync(g.splitk_scratch, 0, (size_t)mn * 4, s);
+    if (e != hipSuccess) {
+      throw ScannerError(std::string("splitk memset failed: ") +
+                         hipGetErrorString(e));
+    }
+    if (dd) {
+      gemm_bf16_splitk_kernel<BM, BN, 2, 2, true, true>
+          <<<tiles * splits, 256, 0, s>>>(
+              (const bf16*)g.A, (const bf16*)g.B, g.M, g.N, g.K,
+              ksteps_per_split, (float*)g.splitk_scratch, *dd,
+              device_zero_chunk());
+    } else {
+      gemm_bf16_splitk_kernel<BM, BN, 2, 2, false, true>
+          <<<tiles * splits, 256, 0, s>>>(
+              (const bf16*)g.A, (const bf16*)g.B, g.M, g.N, g.K,
+              ksteps_per_split, (float*)g.splitk_scratch, ConvDesc{},
+              nullptr);
+    }
+  } else if (dd) {
+    gemm_bf16_splitk_kernel<BM, BN, 2, 2, true, false>
         <<<tiles * splits, 256, 0, s>>>(
             (const bf16*)g.A, (const bf16*)g.B, g.M, g.N, g.K,
             ksteps_per_split, (float*)g.splitk_scratch, *dd,
             device_zero_chunk());
   } else {
-    gemm_bf16_splitk_kernel<BM, BN, 2, 2, false>
+    gemm_bf16_splitk_kernel<BM, BN, 2, 2, false, false>
         <<<tiles * splits, 256, 0, s>>>(
             (const bf16*)g.A, (const bf16*)g.B, g.M, g.N, g.K,
             ksteps_per_split, (float*)g.splitk_scratch, ConvDesc{}, nullptr);
   }
-  hipError_t e = hipGetLastError();
+  e = hipGetLastError();
   if (e != hipSuccess) {
     throw ScannerError(std::string("splitk launch failed: ") +
                        hipGetErrorString(e));
   }
-  i64 mn = (i64)g.M * g.N;
   int grid = (int)std::min<i64>(4096, (mn + 255) / 256);
   auto disp = [&](auto relu, auto res) {
-    splitk_reduce_kernel<decltype(relu)::value, decltype(res)::value>
-        <<<grid, 256, 0, s>>>((const float*)g.splitk_scratch, splits, mn,
-                              g.N, g.scale, g.bias,
-                              (const bf16*)g.residual, (bf16*)g.C);
+    if (kAtomic) {
+      splitk_finalize_kernel<decltype(relu)::value, decltype(res)::value>
+          <<<grid, 256, 0, s>>>((const float*)g.splitk_scratch, mn, g.N,
+                                g.scale, g.bias, (const bf16*)g.residual,
+                                (bf16*)g.C);
+    } else {
+      splitk_reduce_kernel<decltype(relu)::value, decltype(res)::value>
+          <<<grid, 256, 0, s>>>((const float*)g.splitk_scratch, splits, mn,
+                                g.N, g.scale, g.bias,
+                                (const bf16*)g.residual, (bf16*)g.C);
+    }
   };
   if (g.relu && g.residual)
     disp(std::true_type{}, std::true_type{});
