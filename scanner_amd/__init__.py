@@ -18,10 +18,11 @@ from . import types
 def __getattr__(name):
     # `parallel` imports torch (~1.5 s cold): load it lazily so clients,
     # workers and kernel subprocesses that never touch the RCCL data plane
-    # don't pay for it (PEP 562).
+    # don't pay for it (PEP 562). importlib, not `from . import`: the
+    # latter re-enters this __getattr__ through _handle_fromlist.
     if name == "parallel":
-        from . import parallel
-        return parallel
+        import importlib
+        return importlib.import_module(".parallel", __name__)
     raise AttributeError(f"module 'scanner_amd' has no attribute '{name}'")
 
 __version__ = "0.1.0"
