@@ -308,6 +308,7 @@ PYBIND11_MODULE(_core, m) {
         return std::make_shared<Database>(std::move(s), path);
       }))
       .def("recover", &Database::recover)
+      .def("write_megafile", &Database::write_megafile)
       .def("table_names", &Database::table_names)
       .def("has_table", &Database::has_table)
       .def("delete_table", &Database::delete_table)

@@ -163,7 +163,32 @@ Database::Database(std::shared_ptr<StorageBackend> storage,
     std::lock_guard<std::mutex> l(mu_);
     persist();
   }
+  // Megafile pre-warm: one read covers every committed table's descriptor
+  // (committed tables are immutable; deleted ids are unreachable through
+  // the name map). A corrupt/stale megafile degrades to per-table reads.
+  if (storage_->exists(paths_.megafile())) {
+    try {
+      auto buf = storage_->read_all(paths_.megafile());
+      BinReader r(buf);
+      u32 magic = r.u32v();
+      u32 ver = r.u32v();
+      if (magic == 0x5343544d && ver == kVersion) {  // "SCTM"
+        u64 n = r.u64v();
+        std::lock_guard<std::mutex> l(mu_);
+        for (u64 i = 0; i < n; ++i) {
+          auto blob = r.bytes();
+          TableMetadata t = TableMetadata::deserialize(blob);
+          if (meta_.committed_tables.count(t.id)) {
+            table_cache_.emplace(t.id, std::move(t));
+          }
+        }
+      }
+    } catch (const std::exception&) {
+      // degrade silently to per-table descriptor reads
+    }
+  }
 }
+
 
 void Database::persist() {
   auto buf = meta_.serialize();
@@ -256,6 +281,29 @@ void Database::commit_table(i32 table_id) {
   refresh();
   meta_.committed_tables.insert(table_id);
   persist();
+}
+
+void Database::write_megafile() {
+  ScopedDbLock dbl(paths_.root);
+  std::lock_guard<std::mutex> l(mu_);
+  refresh();
+  BinWriter w;
+  w.u32v(0x5343544d);  // "SCTM"
+  w.u32v(kVersion);
+  std::vector<std::vector<u8>> blobs;
+  for (i32 id : meta_.committed_tables) {
+    auto it = table_cache_.find(id);
+    if (it != table_cache_.end()) {
+      blobs.push_back(it->second.serialize());
+      continue;
+    }
+    if (!storage_->exists(paths_.table_descriptor(id))) continue;
+    blobs.push_back(storage_->read_all(paths_.table_descriptor(id)));
+  }
+  w.u64v(blobs.size());
+  for (auto& b : blobs) w.bytes(b);
+  auto buf = w.take();
+  storage_->write_all(paths_.megafile(), buf.data(), buf.size());
 }
 
 bool Database::table_committed(i32 table_id) {

@@ -107,6 +107,14 @@ class Database {
   i32 new_job(const std::string& name);
   void commit_job(i32 job_id);
 
+  // Table megafile (reference: write_table_megafile metadata.cpp:441-530,
+  // a cloud-storage round-trip optimization): batch every COMMITTED
+  // table's descriptor into one object. On open, a present megafile
+  // pre-warms the descriptor cache so get_table() needs zero per-table
+  // reads; committed tables are immutable, so cached entries cannot go
+  // stale (deletes are filtered by the name->id map).
+  void write_megafile();
+
   StorageBackend* storage() { return storage_.get(); }
   const DatabasePaths& paths() const { return paths_; }
 

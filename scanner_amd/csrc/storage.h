@@ -49,6 +49,7 @@ struct DatabasePaths {
   std::string root;
   explicit DatabasePaths(std::string r) : root(std::move(r)) {}
   std::string db_metadata() const { return root + "/db_metadata.bin"; }
+  std::string megafile() const { return root + "/table_megafile.bin"; }
   std::string table_dir(i32 table_id) const {
     return root + "/tables/" + std::to_string(table_id);
   }

@@ -66,6 +66,8 @@ class MasterServer:
         self._job_counter = 0
         self._no_workers_timeout = no_workers_timeout
         self._task_timeout = task_timeout
+        self._checkpoint_frequency = max(1, checkpoint_frequency)
+        self._jobs_completed = 0
         self._registered_ops = {}   # name -> {spec, pickled}
         self._shutdown = threading.Event()
         self._server = RpcServer(addr, {
@@ -273,6 +275,19 @@ class MasterServer:
                      and t[0] not in job.blacklisted]
         if not remaining and not job.complete.is_set():
             job.complete.set()
+            # checkpoint: batch committed-table descriptors into the
+            # megafile every `checkpoint_frequency` bulk jobs (reference:
+            # master.cpp:1109-1112 + write_table_megafile) — off the lock
+            self._jobs_completed += 1
+            if self._jobs_completed % self._checkpoint_frequency == 0:
+                threading.Thread(target=self._write_megafile,
+                                 daemon=True).start()
+
+    def _write_megafile(self):
+        try:
+            self._db.write_megafile()
+        except Exception:
+            pass  # a failed checkpoint degrades to per-table reads
 
     def _remove_worker(self, wid):
         # caller holds lock
