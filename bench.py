@@ -123,10 +123,11 @@ def main():
         # clip is ~2 GB encoded; 288 GB HBM) — storage+PCIe are paid once
         span_cache=int(os.environ.get("SCANNER_BENCH_SPANCACHE",
                                       str(8 << 30))) if have_gpu else 0)
-    # measured sweet spots: IO-bound pipelines (hist/full/resnet) want 6
-    # instances to overlap reads/H2D; compute-saturated ones (flow/pose)
-    # want 4 (more just thrash the CUs)
-    default_inst = "4" if args.pipeline in ("flow", "pose") else "6"
+    # measured sweet spots (r02 A/B): compute-saturated pipelines
+    # (full/resnet/flow/pose) peak at 4 instances — more thrash the CUs
+    # (inst=4 16.4k vs inst=8 14.7k f/s on the flagship); the decode-bound
+    # hist pipeline still wants 6 to overlap decode chains.
+    default_inst = "6" if args.pipeline == "hist" else "4"
     instances = int(os.environ.get("SCANNER_BENCH_INSTANCES",
                                    default_inst if have_gpu else "1"))
 

@@ -23,15 +23,23 @@ using bf16 = __bf16;
 
 __global__ void __launch_bounds__(256)
     preprocess_kernel(const u8* const* __restrict__ frames, int n, int ih,
-                      int iw, int ic, int ohw, bf16* __restrict__ out,
+                      int iw, int ic, int ohw, int oc,
+                      bf16* __restrict__ out,
                       const float* __restrict__ mean,
                       const float* __restrict__ stdv) {
-  i64 total = (i64)n * ohw * ohw * 3;
+  // oc > 3 writes zero-padded channels: the consumer is the implicit-GEMM
+  // conv path, whose 16-byte LDS-DMA gather needs c % 8 == 0 — padding at
+  // preprocess removes conv1's explicit im2col HBM round trip entirely.
+  i64 total = (i64)n * ohw * ohw * oc;
   i64 stride = (i64)gridDim.x * blockDim.x;
   for (i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += stride) {
-    int c = (int)(i % 3);
-    i64 pix = i / 3;
+    int c = (int)(i % oc);
+    if (c >= 3) {
+      out[i] = (bf16)0.f;
+      continue;
+    }
+    i64 pix = i / oc;
     int x = (int)(pix % ohw);
     i64 t = pix / ohw;
     int y = (int)(t % ohw);
@@ -163,11 +171,11 @@ inline int grid_for(i64 total, int per_thread = 1) {
 void preprocess_frames_bf16(const void* frames_ptr_array, int n, int in_h,
                             int in_w, int in_c, int out_hw, void* out,
                             const float* mean, const float* std_,
-                            void* stream) {
-  i64 total = (i64)n * out_hw * out_hw * 3;
+                            void* stream, int out_c) {
+  i64 total = (i64)n * out_hw * out_hw * out_c;
   preprocess_kernel<<<grid_for(total), 256, 0, (hipStream_t)stream>>>(
       (const u8* const*)frames_ptr_array, n, in_h, in_w, in_c, out_hw,
-      (bf16*)out, mean, std_);
+      out_c, (bf16*)out, mean, std_);
   DNN_CHECK();
 }
 
