@@ -45,7 +45,7 @@ void conv_gemm_bf16(const GemmArgs& g, const ConvDesc& d, void* stream);
 void preprocess_frames_bf16(const void* frames_ptr_array, int n, int in_h,
                             int in_w, int in_c, int out_hw, void* out,
                             const float* mean, const float* std_,
-                            void* stream);
+                            void* stream, int out_c = 3);
 
 // NHWC bf16 -> im2col rows [n*out_h*out_w][k_padded] (zero-padded past
 // r*s*c). stride/pad symmetric.
