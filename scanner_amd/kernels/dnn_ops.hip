@@ -11,6 +11,7 @@ namespace sca {
 namespace {
 
 using bf16 = __bf16;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 
 #define DNN_CHECK()                                                      \
   do {                                                                   \
@@ -144,6 +145,45 @@ __global__ void __launch_bounds__(256)
   }
 }
 
+// Vectorized variant for c % 8 == 0 (the resnet path, c=64): one thread
+// owns 8 channels, 16-byte loads per window tap — the scalar kernel's
+// 2-byte gathers ran at 2.3 TB/s.
+__global__ void __launch_bounds__(256)
+    maxpool8_kernel(const bf16* __restrict__ in, int n, int h, int w, int c,
+                    bf16* __restrict__ out, int oh, int ow) {
+  int c8 = c / 8;
+  i64 total = (i64)n * oh * ow * c8;
+  i64 stride = (i64)gridDim.x * blockDim.x;
+  for (i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    int cc = (int)(i % c8) * 8;
+    i64 t = i / c8;
+    int x = (int)(t % ow);
+    t /= ow;
+    int y = (int)(t % oh);
+    int f = (int)(t / oh);
+    float best[8];
+#pragma unroll
+    for (int k = 0; k < 8; ++k) best[k] = -1e30f;
+    for (int dy = 0; dy < 3; ++dy) {
+      int yy = y * 2 - 1 + dy;
+      if (yy < 0 || yy >= h) continue;
+      for (int dx = 0; dx < 3; ++dx) {
+        int xx = x * 2 - 1 + dx;
+        if (xx < 0 || xx >= w) continue;
+        bf16x8 v = *reinterpret_cast<const bf16x8*>(
+            in + (((i64)f * h + yy) * w + xx) * c + cc);
+#pragma unroll
+        for (int k = 0; k < 8; ++k) best[k] = fmaxf(best[k], (float)v[k]);
+      }
+    }
+    bf16x8 o;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) o[k] = (bf16)best[k];
+    *reinterpret_cast<bf16x8*>(out + i * 8) = o;
+  }
+}
+
 __global__ void __launch_bounds__(256)
     avgpool_kernel(const bf16* __restrict__ in, int n, int h, int w, int c,
                    bf16* __restrict__ out) {
@@ -191,9 +231,15 @@ void im2col_bf16(const void* in, int n, int h, int w, int c, int r, int s,
 
 void maxpool3x3s2_bf16(const void* in, int n, int h, int w, int c, void* out,
                        int out_h, int out_w, void* stream) {
-  i64 total = (i64)n * out_h * out_w * c;
-  maxpool_kernel<<<grid_for(total), 256, 0, (hipStream_t)stream>>>(
-      (const bf16*)in, n, h, w, c, (bf16*)out, out_h, out_w);
+  if (c % 8 == 0) {
+    i64 total = (i64)n * out_h * out_w * (c / 8);
+    maxpool8_kernel<<<grid_for(total), 256, 0, (hipStream_t)stream>>>(
+        (const bf16*)in, n, h, w, c, (bf16*)out, out_h, out_w);
+  } else {
+    i64 total = (i64)n * out_h * out_w * c;
+    maxpool_kernel<<<grid_for(total), 256, 0, (hipStream_t)stream>>>(
+        (const bf16*)in, n, h, w, c, (bf16*)out, out_h, out_w);
+  }
   DNN_CHECK();
 }
 
