@@ -4,10 +4,14 @@
 // engine elements.
 //
 // Kernel shape: one workgroup per supergroup (128 groups x 32 bytes = 4 KiB
-// of payload). The block loads the 128 group bit-widths, prefix-sums them in
-// LDS for packed offsets, then each lane unpacks its group's 32 residuals
-// and applies the predictor (delta: previous frame byte — fully parallel;
-// key: serial 32-byte chain within the lane).
+// of payload) decoding a WHOLE GOP chain segment in one launch: the block
+// loads each frame's 128 group bit-widths, prefix-sums them for packed
+// offsets, unpacks its lane's 32 residuals and applies the predictor —
+// with the inter-frame prediction chain carried in REGISTERS across the
+// frame loop. Versus the earlier one-launch-per-frame design this removes
+// 15/16 of the kernel launches and every HBM round trip of intermediate
+// chain state (an unwanted frame between two wanted ones never touches
+// memory at all).
 #include <hip/hip_runtime.h>
 
 #include "../csrc/memory.h"
@@ -31,117 +35,156 @@ __device__ inline u8 dev_unzigzag(u8 z) {
   return (u8)v;
 }
 
+constexpr int kGopBatch = 16;
+
+struct SvcGopArgs {
+  u64 pkt_off[kGopBatch];  // packet offsets within the device stream
+  u8* out[kGopBatch];      // frame output; null = keep in registers only
+  const u8* prev;          // chain state entering the launch (delta start)
+  u32 key_mask;            // bit f = frame f is a keyframe
+  i32 nframes;
+  u32 nbytes, ngroups, nsuper;
+};
+
 // One block per supergroup, 128 threads (one lane per group). Group
 // payloads are 4*width bytes and 4-byte aligned, so the unpack runs on u32
-// loads; the delta path reads prev and writes cur as u32s (the frame base
-// of a group is 32-byte aligned).
+// loads. The frame loop keeps the 32 decoded bytes per lane in registers
+// as the next frame's prediction source.
 __global__ void __launch_bounds__(128)
-    svc_decode_frame_kernel(const u8* __restrict__ pkt_widths,
-                            const u32* __restrict__ pkt_super_off,
-                            const u8* __restrict__ pkt_packed,
-                            const u8* __restrict__ prev, bool is_key,
-                            u32 nbytes, u32 ngroups,
-                            u8* __restrict__ cur) {
+    svc_decode_gop_kernel(const u8* __restrict__ stream, SvcGopArgs a) {
   u32 s = blockIdx.x;
   u32 g0 = s * 128;
-  u32 local_n = min(128u, ngroups - g0);
+  u32 local_n = min(128u, a.ngroups - g0);
   u32 tid = threadIdx.x;
   u32 lane = tid & 63;
   u32 wave = tid >> 6;
-
-  u32 w = tid < local_n ? (u32)pkt_widths[g0 + tid] : 0;
-  // Packed offsets by prefix sum of 4*w: wave64 shuffle scan (no LDS
-  // traffic), one barrier to carry wave 0's total into wave 1 — vs the 14
-  // barriers of a 128-wide Hillis-Steele LDS scan (measured 35 us/frame
-  // before, decode-bound histogram pipeline).
-  u32 val = 4u * w;
-  u32 x = val;
-#pragma unroll
-  for (u32 d = 1; d < 64; d <<= 1) {
-    u32 y = __shfl_up(x, d, 64);
-    if (lane >= d) x += y;
-  }
-  __shared__ u32 wave_total;
-  if (wave == 0 && lane == 63) wave_total = x;
-  __syncthreads();
-  u32 my_off = x - val + (wave ? wave_total : 0);
-  if (tid >= local_n) return;
-
+  bool valid = tid < local_n;
   u32 g = g0 + tid;
-  const u32* q =
-      reinterpret_cast<const u32*>(pkt_packed + pkt_super_off[s] + my_off);
-  // Fully unrolled unpack straight into packed u32 output words: res bytes
-  // never touch memory (a partially-unrolled byte array spills to
-  // scratch).
-  u32 res32[8];
-  if (w == 0) {
-#pragma unroll
-    for (int k = 0; k < 8; ++k) res32[k] = 0;
-  } else {
-    u64 acc = 0;
-    u32 nacc = 0;
-    u32 qi = 0;
-    u32 mask = (1u << w) - 1;
-#pragma unroll
-    for (int k = 0; k < 32; ++k) {
-      if (nacc < w) {
-        acc |= ((u64)q[qi++]) << nacc;
-        nacc += 32;
-      }
-      u32 r = (u32)(acc & mask);
-      if ((k & 3) == 0)
-        res32[k / 4] = r;
-      else
-        res32[k / 4] |= r << (8 * (k & 3));
-      acc >>= w;
-      nacc -= w;
-    }
-  }
-  auto res_at = [&](int k) -> u8 {
-    return (u8)((res32[k / 4] >> (8 * (k & 3))) & 0xff);
-  };
   u32 base = g * 32;
-  if (base + 32 > nbytes) {  // scalar tail group
-    u32 n = nbytes - base;
-    if (is_key) {
-      u8 p = 128;
-      for (u32 k = 0; k < n; ++k) {
-        p = (u8)(p + dev_unzigzag(res_at(k)));
-        cur[base + k] = p;
-      }
+  bool tail = valid && base + 32 > a.nbytes;
+  u32 ntail = tail ? a.nbytes - base : 32;
+
+  // double-buffered wave-carry slot: one barrier per frame instead of two
+  __shared__ u32 wave_total[2];
+
+  u32 cur32[8];
+#pragma unroll
+  for (int k = 0; k < 8; ++k) cur32[k] = 0;
+  if (valid && a.prev && !(a.key_mask & 1)) {
+    if (!tail) {
+      const uint4* p4 = reinterpret_cast<const uint4*>(a.prev + base);
+      uint4 v0 = p4[0], v1 = p4[1];
+      cur32[0] = v0.x; cur32[1] = v0.y; cur32[2] = v0.z; cur32[3] = v0.w;
+      cur32[4] = v1.x; cur32[5] = v1.y; cur32[6] = v1.z; cur32[7] = v1.w;
     } else {
-      for (u32 k = 0; k < n; ++k)
-        cur[base + k] = (u8)(prev[base + k] + dev_unzigzag(res_at(k)));
-    }
-    return;
-  }
-  u32 out[8];
-  if (is_key) {
-    u8 p = 128;
-#pragma unroll
-    for (int k = 0; k < 32; ++k) {
-      p = (u8)(p + dev_unzigzag(res_at(k)));
-      out[k / 4] = (k % 4 == 0) ? p : (out[k / 4] | ((u32)p << (8 * (k % 4))));
-    }
-  } else {
-    const u32* prev32 = reinterpret_cast<const u32*>(prev + base);
-#pragma unroll
-    for (int k = 0; k < 8; ++k) {
-      u32 pv = prev32[k];
-      u32 rz = res32[k];
-      u32 o = 0;
-#pragma unroll
-      for (int b = 0; b < 4; ++b) {
-        u8 byte = (u8)((pv >> (8 * b)) & 0xff);
-        byte = (u8)(byte + dev_unzigzag((u8)((rz >> (8 * b)) & 0xff)));
-        o |= (u32)byte << (8 * b);
+      for (u32 k = 0; k < ntail; ++k) {
+        cur32[k / 4] |= (u32)a.prev[base + k] << (8 * (k & 3));
       }
-      out[k] = o;
     }
   }
-  uint4* dst = reinterpret_cast<uint4*>(cur + base);
-  dst[0] = make_uint4(out[0], out[1], out[2], out[3]);
-  dst[1] = make_uint4(out[4], out[5], out[6], out[7]);
+
+  for (i32 f = 0; f < a.nframes; ++f) {
+    const u8* pkt = stream + a.pkt_off[f];
+    const u32* super_off = reinterpret_cast<const u32*>(pkt + 20);
+    const u8* widths = pkt + 20 + a.nsuper * 4;
+    const u8* packed = widths + (a.ngroups + 3) / 4 * 4;
+
+    u32 w = valid ? (u32)widths[g0 + tid] : 0;
+    // Packed offsets by prefix sum of 4*w: wave64 shuffle scan, one
+    // barrier to carry wave 0's total into wave 1.
+    u32 val = 4u * w;
+    u32 x = val;
+#pragma unroll
+    for (u32 d = 1; d < 64; d <<= 1) {
+      u32 y = __shfl_up(x, d, 64);
+      if (lane >= d) x += y;
+    }
+    if (wave == 0 && lane == 63) wave_total[f & 1] = x;
+    __syncthreads();
+    u32 my_off = x - val + (wave ? wave_total[f & 1] : 0);
+
+    // Fully unrolled unpack straight into packed u32 words: residual bytes
+    // never touch memory.
+    u32 res32[8];
+    if (!valid || w == 0) {
+#pragma unroll
+      for (int k = 0; k < 8; ++k) res32[k] = 0;
+    } else {
+      const u32* q =
+          reinterpret_cast<const u32*>(packed + super_off[s] + my_off);
+      u64 acc = 0;
+      u32 nacc = 0;
+      u32 qi = 0;
+      u32 mask = (1u << w) - 1;
+#pragma unroll
+      for (int k = 0; k < 32; ++k) {
+        if (nacc < w) {
+          acc |= ((u64)q[qi++]) << nacc;
+          nacc += 32;
+        }
+        u32 r = (u32)(acc & mask);
+        if ((k & 3) == 0)
+          res32[k / 4] = r;
+        else
+          res32[k / 4] |= r << (8 * (k & 3));
+        acc >>= w;
+        nacc -= w;
+      }
+    }
+
+    if (valid) {
+      bool is_key = (a.key_mask >> f) & 1;
+      u32 nb = tail ? ntail : 32;
+      if (is_key) {
+        u8 p = 128;
+        u32 out[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+        for (u32 k = 0; k < nb; ++k) {
+          u8 rz = (u8)((res32[k / 4] >> (8 * (k & 3))) & 0xff);
+          p = (u8)(p + dev_unzigzag(rz));
+          out[k / 4] |= (u32)p << (8 * (k & 3));
+        }
+#pragma unroll
+        for (int k = 0; k < 8; ++k) cur32[k] = out[k];
+      } else if (!tail) {
+#pragma unroll
+        for (int k = 0; k < 8; ++k) {
+          u32 pv = cur32[k];
+          u32 rz = res32[k];
+          u32 o = 0;
+#pragma unroll
+          for (int b = 0; b < 4; ++b) {
+            u8 byte = (u8)((pv >> (8 * b)) & 0xff);
+            byte = (u8)(byte + dev_unzigzag((u8)((rz >> (8 * b)) & 0xff)));
+            o |= (u32)byte << (8 * b);
+          }
+          cur32[k] = o;
+        }
+      } else {
+        u32 out[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+        for (u32 k = 0; k < nb; ++k) {
+          u8 pv = (u8)((cur32[k / 4] >> (8 * (k & 3))) & 0xff);
+          u8 rz = (u8)((res32[k / 4] >> (8 * (k & 3))) & 0xff);
+          out[k / 4] |= (u32)((u8)(pv + dev_unzigzag(rz))) << (8 * (k & 3));
+        }
+#pragma unroll
+        for (int k = 0; k < 8; ++k) cur32[k] = out[k];
+      }
+
+      if (a.out[f]) {
+        if (!tail) {
+          uint4* dst = reinterpret_cast<uint4*>(a.out[f] + base);
+          dst[0] = make_uint4(cur32[0], cur32[1], cur32[2], cur32[3]);
+          dst[1] = make_uint4(cur32[4], cur32[5], cur32[6], cur32[7]);
+        } else {
+          for (u32 k = 0; k < ntail; ++k) {
+            a.out[f][base + k] =
+                (u8)((cur32[k / 4] >> (8 * (k & 3))) & 0xff);
+          }
+        }
+      }
+    }
+    // wave_total slot f&1 is consumed; the other slot is free next frame
+  }
 }
 
 }  // namespace
@@ -162,36 +205,75 @@ static std::vector<Element> svc_decode_gpu_impl(
   u32 ngroups = (nbytes + 31) / 32;
   u32 nsuper = (ngroups + 127) / 128;
   std::vector<Element> out;
+  if (span.empty()) return out;
 
-  // Frames inside a GOP are a serial prediction chain, but GOPs are
-  // independent: each keyframe starts a new chain on one of 4 auxiliary
-  // streams (fork/join with events), so a work packet's GOPs decode
-  // concurrently. (The reference's decoder automaton is serial per item —
-  // decoder_automata.cpp; GOP concurrency is the MI355X-native upgrade.)
+  // GOP chains are independent: each keyframe starts a new chain on one of
+  // 4 auxiliary streams (fork/join with events) so a work packet's GOPs
+  // decode concurrently; within a chain, kGopBatch frames go down in ONE
+  // launch with the prediction chain in registers. (The reference's
+  // decoder automaton is serial per item — decoder_automata.cpp.)
   constexpr int kChains = 4;
   hipStream_t chain_stream[kChains];
   bool chain_used[kChains] = {false, false, false, false};
   std::vector<u8*> scratches;
-  size_t wi = 0;
+
+  auto make_elem = [&](i64 f) {
+    Element e;
+    e.is_frame = true;
+    e.frame_info.shape[0] = vm.height;
+    e.frame_info.shape[1] = vm.width;
+    e.frame_info.shape[2] = vm.channels;
+    e.frame_info.type = vm.frame_type;
+    e.size = nbytes;
+    e.buffer = new_buffer(dev, nbytes);
+    e.device = dev;
+    e.index = f;
+    return e;
+  };
+
+  SvcGopArgs a{};
+  a.nbytes = nbytes;
+  a.ngroups = ngroups;
+  a.nsuper = nsuper;
+  a.nframes = 0;
+  a.prev = nullptr;
+  a.key_mask = 0;
+
   int chain = -1;
   hipStream_t cs = s;
-  u8* prev = nullptr;
-  u8* chain_scratch[2] = {nullptr, nullptr};
-  int flip = 0;
+  u8* chain_scratch = nullptr;
+  size_t wi = 0;
   size_t ki = 0;  // cursor into sorted keyframe_indices
+
+  auto flush = [&](bool chain_continues) {
+    if (a.nframes == 0) return;
+    int last = a.nframes - 1;
+    if (chain_continues && !a.out[last]) {
+      // the next launch needs this frame as its prediction source
+      if (!chain_scratch) {
+        chain_scratch = new_buffer(dev, nbytes);
+        scratches.push_back(chain_scratch);
+      }
+      a.out[last] = chain_scratch;
+    }
+    svc_decode_gop_kernel<<<nsuper, 128, 0, cs>>>(d_stream, a);
+    SVC_CHECK(hipGetLastError());
+    const u8* next_prev = a.out[last];
+    a.nframes = 0;
+    a.key_mask = 0;
+    a.prev = next_prev;
+  };
+
   for (i64 f : span) {
     while (ki < vm.keyframe_indices.size() && vm.keyframe_indices[ki] < f)
       ++ki;
     bool is_key =
         ki < vm.keyframe_indices.size() && vm.keyframe_indices[ki] == f;
-    SvcPacketView v;
-    v.is_key = is_key;
-    v.nbytes = nbytes;
-    v.ngroups = ngroups;
-    v.nsuper = nsuper;
-    if (v.is_key) {
-      // new chain
-      chain = (chain + 1);
+    if (is_key) {
+      flush(false);
+      a.prev = nullptr;
+      chain_scratch = nullptr;
+      ++chain;
       int ci = chain % kChains;
       cs = (hipStream_t)per_thread_aux_stream(ci);
       if (!chain_used[ci]) {
@@ -199,51 +281,28 @@ static std::vector<Element> svc_decode_gpu_impl(
         chain_stream[ci] = cs;
         if (up_ev) SVC_CHECK(hipStreamWaitEvent(cs, up_ev, 0));
       }
-      prev = nullptr;
-      chain_scratch[0] = chain_scratch[1] = nullptr;
-      flip = 0;
+    } else if (a.nframes == kGopBatch) {
+      flush(true);
     }
-    u64 pkt_off = vm.sample_offsets[f] - lo;
-    const u8* pkt_d = d_stream + pkt_off;
-    const u32* super_off_d = reinterpret_cast<const u32*>(pkt_d + 20);
-    const u8* widths_d = pkt_d + 20 + v.nsuper * 4;
-    const u8* packed_d = widths_d + (v.ngroups + 3) / 4 * 4;
-
+    SCA_CHECK(is_key || a.nframes > 0 || a.prev,
+              "svc chain does not start at a keyframe");
+    int fi = a.nframes++;
+    a.pkt_off[fi] = vm.sample_offsets[f] - lo;
+    if (is_key) a.key_mask |= 1u << fi;
     bool wanted = wi < want.size() && want[wi] == f;
-    u8* cur;
-    Element e;
     if (wanted) {
-      e.is_frame = true;
-      e.frame_info.shape[0] = vm.height;
-      e.frame_info.shape[1] = vm.width;
-      e.frame_info.shape[2] = vm.channels;
-      e.frame_info.type = vm.frame_type;
-      e.size = nbytes;
-      e.buffer = new_buffer(dev, nbytes);
-      e.device = dev;
-      e.index = f;
-      cur = e.buffer;
-    } else {
-      if (!chain_scratch[flip]) {
-        chain_scratch[flip] = new_buffer(dev, nbytes);
-        scratches.push_back(chain_scratch[flip]);
-      }
-      cur = chain_scratch[flip];
-    }
-    u32 blocks = v.nsuper;
-    svc_decode_frame_kernel<<<blocks, 128, 0, cs>>>(
-        widths_d, super_off_d, packed_d, prev, v.is_key, nbytes, v.ngroups,
-        cur);
-    SVC_CHECK(hipGetLastError());
-    if (wanted) {
+      Element e = make_elem(f);
       out.push_back(e);
+      a.out[fi] = e.buffer;
       ++wi;
+    } else {
+      a.out[fi] = nullptr;
     }
-    prev = cur;
-    flip ^= 1;
   }
+  flush(false);
+
   // Join: main stream waits every used chain, then sync before returning
-  // scratch + stream buffers to the shared pool.
+  // scratch buffers to the shared pool.
   for (int ci = 0; ci < kChains; ++ci) {
     if (!chain_used[ci]) continue;
     hipEvent_t ev;
