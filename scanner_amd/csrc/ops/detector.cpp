@@ -42,7 +42,8 @@ std::vector<ConvSpec> detector_specs() {
                 bool relu = true) {
     sp.push_back({n, ic, oc, 3, 3, stride, 1, relu});
   };
-  c3("b1", 3, 64, 1);
+  // b1 consumes 8-channel zero-padded input (implicit-GEMM entry conv)
+  c3("b1", 8, 64, 1);
   c3("b2", 64, 64, 2);
   c3("b3", 64, 128, 1);
   c3("b4", 128, 128, 2);
@@ -104,6 +105,16 @@ class DetectorKernelGPU : public BatchedKernel {
                               Tensors ts;
                               if (!weights_file_.empty())
                                 ts = dnn::load_tensor_file(weights_file_);
+                              if (ts.has("b1.weight")) {
+                                auto& w = ts["b1.weight"];
+                                if ((i64)w.size() == 64LL * 9 * 3) {
+                                  std::vector<f32> e(64LL * 9 * 8, 0.f);
+                                  for (i64 o = 0; o < 64 * 9; ++o)
+                                    for (i64 c = 0; c < 3; ++c)
+                                      e[o * 8 + c] = w[o * 3 + c];
+                                  w = std::move(e);
+                                }
+                              }
                               return dnn::build_device_model(
                                   cfg.device, detector_specs(),
                                   std::move(ts), seed_);
@@ -136,7 +147,7 @@ class DetectorKernelGPU : public BatchedKernel {
     if (!bufs_[0]) {
       size_t fpix_b = (size_t)nb * kFeat * kFeat;
       bufs_[0] = new_buffer(dev, (size_t)nb * sizeof(u8*));   // d_ptrs
-      bufs_[1] = new_buffer(dev, (size_t)nb * hw0 * 3 * 2);   // pre
+      bufs_[1] = new_buffer(dev, (size_t)nb * hw0 * 8 * 2);   // pre (c=8)
       bufs_[2] = new_buffer(dev, (size_t)nb * hw0 * 64 * 2);  // actA
       bufs_[3] = new_buffer(dev, (size_t)nb * hw0 * 64 * 2);  // actB
       bufs_[4] = new_buffer(dev, (size_t)nb * hw0 * 64 * 2);  // colbuf (b1)
@@ -189,7 +200,7 @@ class DetectorKernelGPU : public BatchedKernel {
 
     f32* mean = model_->mean;
     preprocess_frames_bf16(d_ptrs, n, ih, iw, ic, kInHW, pre, mean, mean + 3,
-                           s);
+                           s, /*out_c=*/8);
     int h = kInHW, w = kInHW, oh, ow;
     u8* x = pre;
     u8* bufs[2] = {actA, actB};

@@ -42,7 +42,10 @@ std::vector<ConvSpec> pose_specs(int stages) {
   auto c3 = [&](const std::string& n, int ic, int oc, int stride) {
     sp.push_back({n, ic, oc, 3, 3, stride, 1, true});
   };
-  c3("b1", 3, 64, 1);
+  // b1 consumes 8-channel input: preprocess zero-pads RGB so the entry
+  // conv takes the implicit-GEMM path (no explicit im2col round trip);
+  // 3-channel weight tensors are expanded at load.
+  c3("b1", 8, 64, 1);
   c3("b2", 64, 64, 2);
   c3("b3", 64, 128, 1);
   c3("b4", 128, 128, 2);
@@ -81,6 +84,16 @@ class PoseKernelGPU : public BatchedKernel {
                               Tensors ts;
                               if (!weights_file_.empty())
                                 ts = dnn::load_tensor_file(weights_file_);
+                              if (ts.has("b1.weight")) {
+                                auto& w = ts["b1.weight"];
+                                if ((i64)w.size() == 64LL * 9 * 3) {
+                                  std::vector<f32> e(64LL * 9 * 8, 0.f);
+                                  for (i64 o = 0; o < 64 * 9; ++o)
+                                    for (i64 c = 0; c < 3; ++c)
+                                      e[o * 8 + c] = w[o * 3 + c];
+                                  w = std::move(e);
+                                }
+                              }
                               return dnn::build_device_model(
                                   cfg.device, pose_specs(stages_),
                                   std::move(ts), seed_);
@@ -111,7 +124,7 @@ class PoseKernelGPU : public BatchedKernel {
     if (!bufs_[0]) {
       size_t fpix_b = (size_t)nb * kFeatHW * kFeatHW;
       bufs_[0] = new_buffer(dev, (size_t)nb * sizeof(u8*));       // d_ptrs
-      bufs_[1] = new_buffer(dev, (size_t)nb * hw0 * 3 * 2);       // pre
+      bufs_[1] = new_buffer(dev, (size_t)nb * hw0 * 8 * 2);       // pre (c=8)
       bufs_[2] = new_buffer(dev, (size_t)nb * hw0 * 64 * 2);      // actA
       bufs_[3] = new_buffer(dev, (size_t)nb * hw0 * 64 * 2);      // actB
       bufs_[4] = new_buffer(dev, (size_t)nb * (hw0 / 4) * 640 * 2);
@@ -178,7 +191,7 @@ class PoseKernelGPU : public BatchedKernel {
     // ---- backbone ----
     f32* mean = model_->mean;
     preprocess_frames_bf16(d_ptrs, n, ih, iw, ic, kInHW, pre, mean, mean + 3,
-                           s);
+                           s, /*out_c=*/8);
     int h = kInHW, w = kInHW, oh, ow;
     const char* bb[] = {"b1", "b2", "b3", "b4", "b5", "b6", "b7"};
     u8* x = pre;
