@@ -245,6 +245,68 @@ PYBIND11_MODULE(_core, m) {
     d["budget"] = span_cache_budget();
     return d;
   });
+  // kernel-isolation test: CPU-encode -> GPU-decode -> bytes back (no
+  // engine, no cache, no chunking)
+  m.def("svc_gpu_debug",
+        [](py::array_t<u8, py::array::c_style | py::array::forcecast> frames) {
+          i64 n = frames.shape(0);
+          i32 h = (i32)frames.shape(1), w = (i32)frames.shape(2),
+              c = (i32)frames.shape(3);
+          VideoMetadata vm;
+          std::vector<u8> stream;
+          svc_encode_cpu(frames.data(), n, h, w, c, 16, stream, vm);
+          py::dict r;
+          r["dev"] = svc_gpu_debug_dump(stream, vm);
+          // host-side expectations for supers 255..257, frame 0
+          u32 nbytes = (u32)((i64)h) * w * c;
+          u32 ngroups = (nbytes + 31) / 32;
+          u32 nsuper = (ngroups + 127) / 128;
+          const u8* pkt = stream.data() + vm.sample_offsets[0];
+          const u32* so = (const u32*)(pkt + 20);
+          u64 widths_off = 20ull + (u64)nsuper * 4;
+          u64 packed_off = widths_off + (ngroups + 3) / 4 * 4;
+          py::list host;
+          for (u32 s = 255; s <= 257; ++s) {
+            u32 g0 = s * 128;
+            u32 wv = pkt[widths_off + g0];
+            const u32* q = (const u32*)(pkt + packed_off + so[s]);
+            py::dict e;
+            e["super_off"] = so[s];
+            e["w_lane0"] = wv;
+            e["q0"] = q[0];
+            e["q1"] = q[1];
+            e["packed_off"] = (u32)packed_off;
+            host.append(e);
+          }
+          r["host"] = host;
+          return r;
+        });
+  m.def("svc_gpu_roundtrip",
+        [](py::array_t<u8, py::array::c_style | py::array::forcecast> frames,
+           i32 gop, std::vector<i64> want) {
+          SCA_CHECK(frames.ndim() == 4, "frames must be [N,H,W,C]");
+          i64 n = frames.shape(0);
+          i32 h = (i32)frames.shape(1), w = (i32)frames.shape(2),
+              c = (i32)frames.shape(3);
+          VideoMetadata vm;
+          std::vector<u8> stream;
+          svc_encode_cpu(frames.data(), n, h, w, c, gop, stream, vm);
+          if (want.empty())
+            for (i64 i = 0; i < n; ++i) want.push_back(i);
+          DeviceHandle dev{DeviceType::GPU, 0};
+          auto elems = svc_decode_gpu(stream.data(), stream.size(), vm,
+                                      want, dev, 0);
+          py::array_t<u8> out({(i64)elems.size(), (i64)h, (i64)w, (i64)c});
+          size_t fsize = (size_t)h * w * c;
+          for (size_t i = 0; i < elems.size(); ++i) {
+            memcpy_buffer(out.mutable_data() + i * fsize, CPU_DEVICE,
+                          elems[i].buffer, dev, fsize);
+            delete_buffer(dev, elems[i].buffer);
+          }
+          return out;
+        },
+        py::arg("frames"), py::arg("gop") = 16,
+        py::arg("want") = std::vector<i64>{});
   m.def("span_cache_clear", &span_cache_clear);
   m.def("span_cache_set_budget", &span_cache_set_budget);
   // engine memory accounting (device -1 = CPU)

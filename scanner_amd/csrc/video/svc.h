@@ -88,4 +88,8 @@ struct SvcPacketView {
 
 SvcPacketView svc_parse_packet(const u8* pkt, size_t size);
 
+// debug: launch the decode kernel with a diagnostic dump (tests only)
+std::vector<u32> svc_gpu_debug_dump(const std::vector<u8>& stream,
+                                    const VideoMetadata& vm);
+
 }  // namespace sca

@@ -14,6 +14,8 @@
 // memory at all).
 #include <hip/hip_runtime.h>
 
+#include <cstdlib>
+
 #include "../csrc/memory.h"
 #include "../csrc/video/svc.h"
 
@@ -51,7 +53,8 @@ struct SvcGopArgs {
 // loads. The frame loop keeps the 32 decoded bytes per lane in registers
 // as the next frame's prediction source.
 __global__ void __launch_bounds__(128)
-    svc_decode_gop_kernel(const u8* __restrict__ stream, SvcGopArgs a) {
+    svc_decode_gop_kernel(const u8* __restrict__ stream, SvcGopArgs a,
+                          u32* __restrict__ dbg = nullptr) {
   u32 s = blockIdx.x;
   u32 g0 = s * 128;
   u32 local_n = min(128u, a.ngroups - g0);
@@ -64,8 +67,7 @@ __global__ void __launch_bounds__(128)
   bool tail = valid && base + 32 > a.nbytes;
   u32 ntail = tail ? a.nbytes - base : 32;
 
-  // double-buffered wave-carry slot: one barrier per frame instead of two
-  __shared__ u32 wave_total[2];
+  __shared__ u32 wave_total;
 
   u32 cur32[8];
 #pragma unroll
@@ -83,13 +85,14 @@ __global__ void __launch_bounds__(128)
     }
   }
 
+  // widths/packed offsets relative to the packet start, fully u64
+  const u64 widths_off = 20ull + (u64)a.nsuper * 4;
+  const u64 packed_off = widths_off + (u64)((a.ngroups + 3) / 4 * 4);
   for (i32 f = 0; f < a.nframes; ++f) {
     const u8* pkt = stream + a.pkt_off[f];
     const u32* super_off = reinterpret_cast<const u32*>(pkt + 20);
-    const u8* widths = pkt + 20 + a.nsuper * 4;
-    const u8* packed = widths + (a.ngroups + 3) / 4 * 4;
 
-    u32 w = valid ? (u32)widths[g0 + tid] : 0;
+    u32 w = valid ? (u32)pkt[widths_off + g0 + tid] : 0;
     // Packed offsets by prefix sum of 4*w: wave64 shuffle scan, one
     // barrier to carry wave 0's total into wave 1.
     u32 val = 4u * w;
@@ -99,9 +102,10 @@ __global__ void __launch_bounds__(128)
       u32 y = __shfl_up(x, d, 64);
       if (lane >= d) x += y;
     }
-    if (wave == 0 && lane == 63) wave_total[f & 1] = x;
+    if (wave == 0 && lane == 63) wave_total = x;
     __syncthreads();
-    u32 my_off = x - val + (wave ? wave_total[f & 1] : 0);
+    u32 my_off = x - val + (wave ? wave_total : 0);
+    __syncthreads();  // wave_total consumed before next frame's write
 
     // Fully unrolled unpack straight into packed u32 words: residual bytes
     // never touch memory.
@@ -110,8 +114,19 @@ __global__ void __launch_bounds__(128)
 #pragma unroll
       for (int k = 0; k < 8; ++k) res32[k] = 0;
     } else {
-      const u32* q =
-          reinterpret_cast<const u32*>(packed + super_off[s] + my_off);
+      const u32* q = reinterpret_cast<const u32*>(
+          pkt + (packed_off + (u64)super_off[s] + my_off));
+      if (dbg && f == 0 && tid == 0 && s >= 255 && s <= 257) {
+        u32* d = dbg + (s - 255) * 8;
+        d[0] = super_off[s];
+        d[1] = my_off;
+        d[2] = w;
+        d[3] = q[0];
+        d[4] = q[1];
+        d[5] = (u32)packed_off;
+        d[6] = (u32)a.pkt_off[0];
+        d[7] = 0xd00dfeed;
+      }
       u64 acc = 0;
       u32 nacc = 0;
       u32 qi = 0;
@@ -183,7 +198,6 @@ __global__ void __launch_bounds__(128)
         }
       }
     }
-    // wave_total slot f&1 is consumed; the other slot is free next frame
   }
 }
 
@@ -269,6 +283,11 @@ static std::vector<Element> svc_decode_gpu_impl(
       ++ki;
     bool is_key =
         ki < vm.keyframe_indices.size() && vm.keyframe_indices[ki] == f;
+    static const int kBatchLimit = []() {
+      const char* e = std::getenv("SCANNER_SVC_BATCH");
+      int v = e ? atoi(e) : kGopBatch;
+      return v < 1 ? 1 : (v > kGopBatch ? kGopBatch : v);
+    }();
     if (is_key) {
       flush(false);
       a.prev = nullptr;
@@ -281,7 +300,7 @@ static std::vector<Element> svc_decode_gpu_impl(
         chain_stream[ci] = cs;
         if (up_ev) SVC_CHECK(hipStreamWaitEvent(cs, up_ev, 0));
       }
-    } else if (a.nframes == kGopBatch) {
+    } else if (a.nframes == kBatchLimit) {
       flush(true);
     }
     SCA_CHECK(is_key || a.nframes > 0 || a.prev,
@@ -369,6 +388,39 @@ std::vector<Element> svc_decode_gpu_dev(const u8* stream_dev, u64 dev_lo,
   (void)hi;
   return svc_decode_gpu_impl(stream_dev + (lo - dev_lo), lo, vm, span, want,
                              dev, nullptr);
+}
+
+std::vector<u32> svc_gpu_debug_dump(const std::vector<u8>& stream,
+                                    const VideoMetadata& vm) {
+  DeviceHandle dev{DeviceType::GPU, 0};
+  hipStream_t s = (hipStream_t)per_thread_hip_stream();
+  u32 nbytes = (u32)((i64)vm.height * vm.width * vm.channels);
+  u32 ngroups = (nbytes + 31) / 32;
+  u32 nsuper = (ngroups + 127) / 128;
+  u8* d_stream = new_buffer(dev, stream.size());
+  SVC_CHECK(hipMemcpy(d_stream, stream.data(), stream.size(),
+                      hipMemcpyHostToDevice));
+  u8* d_out = new_buffer(dev, nbytes);
+  u8* d_dbg = new_buffer(dev, 3 * 8 * 4);
+  SVC_CHECK(hipMemset(d_dbg, 0, 3 * 8 * 4));
+  SvcGopArgs a{};
+  a.nbytes = nbytes;
+  a.ngroups = ngroups;
+  a.nsuper = nsuper;
+  a.nframes = 1;
+  a.key_mask = 1;
+  a.pkt_off[0] = vm.sample_offsets[0];
+  a.out[0] = d_out;
+  svc_decode_gop_kernel<<<nsuper, 128, 0, s>>>(d_stream, a, (u32*)d_dbg);
+  SVC_CHECK(hipGetLastError());
+  SVC_CHECK(hipStreamSynchronize(s));
+  std::vector<u32> dbg(3 * 8);
+  SVC_CHECK(hipMemcpy(dbg.data(), d_dbg, 3 * 8 * 4,
+                      hipMemcpyDeviceToHost));
+  delete_buffer(dev, d_stream);
+  delete_buffer(dev, d_out);
+  delete_buffer(dev, d_dbg);
+  return dbg;
 }
 
 }  // namespace sca

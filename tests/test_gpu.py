@@ -294,3 +294,70 @@ def test_gpu_span_cache_eviction():
     assert s["bytes"] <= 1 << 20
     rows = list(out.load())
     assert len(rows) == 64
+
+
+def test_gpu_histogram_4k_raw(sc):
+    """Histogram alone on raw-codec 4K frames (no SVC decode in the path)
+    — isolates the histogram kernels at large frame sizes."""
+    from conftest import make_smooth_video
+    frames = make_smooth_video(n=3, h=2160, w=3840)
+    video = sp.NamedVideoStream(sc, "g4kr", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    hist = sc.ops.Histogram(frame=frame, device=sp.DeviceType.GPU)
+    out = sp.NamedStream(sc, "g4kr_out")
+    sc.run(sc.io.Output(hist, [out]), sp.PerfParams.manual(2, 4),
+           cache_mode=sp.CacheMode.Overwrite, gpu_ids=[0])
+    for i, blob in enumerate(out.load()):
+        got = np.frombuffer(blob, dtype=np.uint32).reshape(3, 256)
+        np.testing.assert_array_equal(got, ref_histogram(frames[i]))
+
+
+def test_gpu_svc_decode_4k_exact(sc):
+    """GPU SVC decode of 4K frames compared byte-exactly against the
+    ingested frames (lossless codec): isolates the GOP-batched decode
+    kernel at large geometry."""
+    from conftest import make_smooth_video
+    frames = make_smooth_video(n=6, h=2160, w=3840)
+    video = sp.NamedVideoStream(sc, "g4kd", frames=frames, codec="svc")
+    frame = sc.io.Input([video])
+    out = sp.NamedStream(sc, "g4kd_out")
+    # sink the decoded frame column itself (raw storage)
+    sc.run(sc.io.Output(frame, [out]), sp.PerfParams.manual(2, 4),
+           cache_mode=sp.CacheMode.Overwrite, gpu_ids=[0])
+    got = np.stack(list(sp.NamedVideoStream(sc, "g4kd_out").load()))
+    np.testing.assert_array_equal(got, frames)
+
+
+def test_gpu_svc_decode_multibatch_chain(sc):
+    """A >16-frame GOP forces the chain across multiple GOP-batch kernel
+    launches (register chain state handed over via the scratch buffer);
+    sparse wants make interior frames registers-only."""
+    from conftest import make_smooth_video
+    frames = make_smooth_video(n=40, h=120, w=160)
+    video = sp.NamedVideoStream(sc, "gmb", frames=frames, codec="svc",
+                                io_packet_size=40)
+    # gop defaults to 16 at ingest; wants span several batches
+    frame = sc.io.Input([video])
+    g = sc.streams.Gather(frame, [[0, 7, 18, 19, 33, 39]])
+    out = sp.NamedStream(sc, "gmb_out")
+    sc.run(sc.io.Output(g, [out]), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite, gpu_ids=[0])
+    got = np.stack(list(sp.NamedVideoStream(sc, "gmb_out").load()))
+    np.testing.assert_array_equal(got, frames[[0, 7, 18, 19, 33, 39]])
+
+
+def test_gpu_svc_decode_1080p_exact_repeated():
+    """Regression for the GOP-batch decode corruption seen past byte 2^20:
+    direct kernel round-trip at 1080p, byte-exact, repeated (the failure
+    was content/size-conditional and schedule-sensitive)."""
+    from conftest import make_smooth_video
+    from scanner_amd import _core
+    frames = make_smooth_video(n=4, h=1080, w=1920)
+    ref = None
+    for it in range(5):
+        got = _core.svc_gpu_roundtrip(frames, 16, [])
+        np.testing.assert_array_equal(got, frames, err_msg=f"iter {it}")
+        if ref is None:
+            ref = got
+        else:
+            np.testing.assert_array_equal(got, ref)  # run-to-run identical
