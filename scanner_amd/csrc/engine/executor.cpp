@@ -1012,7 +1012,7 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t,
             compress[kv.first] = kv.second.as_str();
         }
       }
-      for (size_t c = 0; c < out_op.inputs.size(); ++c) {
+      auto write_col = [&](size_t c) {
         const std::string& cname = table.columns[c].name;
         auto cmp = compress.find(cname);
         if (cmp != compress.end() &&
@@ -1057,7 +1057,7 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t,
           }
           cols[c].clear();
           owned[c].clear();
-          continue;
+          return;
         }
         {
           i64 wb = 0;
@@ -1099,6 +1099,28 @@ void LocalExecutor::process_task(Instance& inst, const TaskDesc& t,
         }
         cols[c].clear();
         owned[c].clear();
+      };
+      // Save fan-out (reference: SaveWorker multi-sink thread pool,
+      // save_worker.cpp:104-140): independent output columns write and
+      // codec-compress in parallel.
+      if (out_op.inputs.size() <= 1) {
+        if (!out_op.inputs.empty()) write_col(0);
+      } else {
+        std::vector<std::thread> savers;
+        std::exception_ptr save_err;
+        std::mutex err_mu;
+        for (size_t c = 0; c < out_op.inputs.size(); ++c) {
+          savers.emplace_back([&, c] {
+            try {
+              write_col(c);
+            } catch (...) {
+              std::lock_guard<std::mutex> el(err_mu);
+              if (!save_err) save_err = std::current_exception();
+            }
+          });
+        }
+        for (auto& th : savers) th.join();
+        if (save_err) std::rethrow_exception(save_err);
       }
     }
   } catch (...) {
