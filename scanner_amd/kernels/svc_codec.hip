@@ -67,8 +67,6 @@ __global__ void __launch_bounds__(128)
   bool tail = valid && base + 32 > a.nbytes;
   u32 ntail = tail ? a.nbytes - base : 32;
 
-  __shared__ u32 wave_total;
-
   u32 cur32[8];
 #pragma unroll
   for (int k = 0; k < 8; ++k) cur32[k] = 0;
@@ -93,8 +91,10 @@ __global__ void __launch_bounds__(128)
     const u32* super_off = reinterpret_cast<const u32*>(pkt + 20);
 
     u32 w = valid ? (u32)pkt[widths_off + g0 + tid] : 0;
-    // Packed offsets by prefix sum of 4*w: wave64 shuffle scan, one
-    // barrier to carry wave 0's total into wave 1.
+    // Packed offsets by prefix sum of 4*w: wave64 shuffle scan. The
+    // cross-wave carry is NOT passed through LDS: wave 1 re-loads wave
+    // 0's 64 widths (one byte per lane) and butterfly-reduces them — a
+    // barrier-free kernel, so a block's 16-frame loop never synchronizes.
     u32 val = 4u * w;
     u32 x = val;
 #pragma unroll
@@ -102,10 +102,17 @@ __global__ void __launch_bounds__(128)
       u32 y = __shfl_up(x, d, 64);
       if (lane >= d) x += y;
     }
-    if (wave == 0 && lane == 63) wave_total = x;
-    __syncthreads();
-    u32 my_off = x - val + (wave ? wave_total : 0);
-    __syncthreads();  // wave_total consumed before next frame's write
+    u32 carry = 0;
+    if (wave == 1) {
+      u32 w0 = (g0 + lane < a.ngroups)
+                   ? (u32)pkt[widths_off + g0 + lane]
+                   : 0;
+      u32 sum = 4u * w0;
+#pragma unroll
+      for (u32 d = 32; d >= 1; d >>= 1) sum += __shfl_xor(sum, d, 64);
+      carry = sum;
+    }
+    u32 my_off = x - val + carry;
 
     // Fully unrolled unpack straight into packed u32 words: residual bytes
     // never touch memory.
