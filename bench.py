@@ -113,7 +113,11 @@ def main():
                                 codec="svc", io_packet_size=128)
     del clip
 
-    gpu_ids = [local_rank] if have_gpu else []
+    # rank -> GPU modulo device count: the driver runs one rank per GPU
+    # (identity mapping on an 8-GPU node); oversubscribed runs (2 ranks on
+    # a 1-GPU box) share device 0 for RCCL-path shakeout
+    gpu_ids = [local_rank % max(1, _core.gpu_device_count())] \
+        if have_gpu else []
     # Pools: steady-state allocation must never hit the driver (hipMalloc
     # synchronizes the device; hipHostMalloc is ~ms per call).
     perf = sp.PerfParams.manual(
@@ -172,7 +176,8 @@ def main():
         gathered = parallel.gather_column(sample, dist_device)
         if rank == 0:
             log(f"gathered {len(gathered)} result rows over "
-                f"{'RCCL/xGMI' if have_gpu else 'gloo'}")
+                f"{dist.get_backend()}"
+                f"{' (RCCL/xGMI)' if dist.get_backend() == 'nccl' else ''}")
 
     total_frames = args.steps * n_frames * world
     fps = total_frames / elapsed

@@ -46,13 +46,26 @@ def init_from_env(device_type="auto"):
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     use_gpu = (device_type == "gpu" or
                (device_type == "auto" and torch.cuda.is_available()))
-    device = torch.device("cuda", local_rank) if use_gpu \
+    # rank -> device modulo device count: identity on a full node (one
+    # rank per GPU), and oversubscribed shakeout runs (2 ranks on a 1-GPU
+    # box) share device 0 instead of failing on an invalid ordinal
+    dev_idx = local_rank % max(1, torch.cuda.device_count()) if use_gpu \
+        else 0
+    device = torch.device("cuda", dev_idx) if use_gpu \
         else torch.device("cpu")
+    # RCCL communicators require one DISTINCT device per rank; when ranks
+    # oversubscribe the GPUs (shakeout runs: 2 ranks on a 1-GPU box) the
+    # collective plane falls back to gloo/CPU while compute stays on the
+    # shared GPU. Full-node runs (one rank per GPU) stay on RCCL/xGMI.
+    oversub = use_gpu and world > max(1, torch.cuda.device_count())
     if world > 1 and not dist.is_initialized():
-        dist.init_process_group(backend="nccl" if use_gpu else "gloo")
-        if use_gpu:
-            torch.cuda.set_device(device)
-    return rank, world, device
+        dist.init_process_group(
+            backend="nccl" if (use_gpu and not oversub) else "gloo")
+    if use_gpu:
+        torch.cuda.set_device(device)
+    coll_device = torch.device("cpu") if (oversub or not use_gpu) \
+        else device
+    return rank, world, coll_device
 
 
 def shard_rows(n_rows, world, rank):
