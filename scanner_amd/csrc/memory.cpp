@@ -343,10 +343,31 @@ void destroy_memory_allocators() {
     std::lock_guard<std::mutex> l(g_teardown_mu);
     for (auto& cb : teardown_callbacks()) cb();
   }
-  std::lock_guard<std::mutex> l(g_mem.mu);
-  g_mem.gpus.clear();
-  g_mem.cpu = DeviceAllocators{};
-  g_mem.initialized = false;
+  std::vector<i32> gpu_ids;
+  {
+    std::lock_guard<std::mutex> l(g_mem.mu);
+    for (auto& kv : g_mem.gpus) gpu_ids.push_back(kv.first);
+    g_mem.gpus.clear();
+    g_mem.cpu = DeviceAllocators{};
+    g_mem.initialized = false;
+  }
+  // Sanitizer affordance (parity: the reference's SystemAllocator dtor
+  // calls cudaDeviceReset so cuda-memcheck can flush its state,
+  // memory.cpp:110-117): SCANNER_DEVICE_RESET=1 resets every device at
+  // teardown so rocprof-compute / sanitizer runs see a clean shutdown.
+  static const bool kReset = []() {
+    const char* e = std::getenv("SCANNER_DEVICE_RESET");
+    return e && e[0] == '1';
+  }();
+  if (kReset) {
+    int prev = 0;
+    (void)hipGetDevice(&prev);
+    for (i32 id : gpu_ids) {
+      (void)hipSetDevice(id);
+      (void)hipDeviceReset();
+    }
+    (void)hipSetDevice(prev);
+  }
 }
 
 bool memory_initialized() { return g_mem.initialized; }
