@@ -654,10 +654,10 @@ bool try_splitk(const GemmArgs& g, hipStream_t s,
   if (tiles >= 256) return false;  // already fills the chip
   static const int kMaxSplits = []() {
     const char* e = std::getenv("SCANNER_SPLITK_MAX");
-    // 8 beats 16 on the flagship A/B (16.2k vs 15.4k f/s): fewer splits
-    // halve the partial+reduce HBM traffic and the chip stays fed by the
-    // co-running pipeline instances.
-    return e ? std::max(2, atoi(e)) : 8;
+    // A/B on the flagship: 6 > 8 > 16 (18.0k / 17.95k / 15.4k f/s) —
+    // fewer splits cut the partial+reduce HBM traffic and the chip stays
+    // fed by the co-running pipeline instances.
+    return e ? std::max(2, atoi(e)) : 6;
   }();
   int ksteps = g.K / 64;
   int want = std::min({kMaxSplits, ksteps / 4, (2048 + tiles - 1) / tiles});
