@@ -166,6 +166,13 @@ def main():
         for k in sorted(stats, key=lambda k: -stats[k]["total_ms"]):
             log(f"[prof] {k}: {stats[k]['total_ms']:.1f} ms "
                 f"x{stats[k]['count']}")
+    trace_path = os.environ.get("SCANNER_PROFILE_TRACE")
+    if trace_path and rank == 0:
+        # Chrome trace of the last step's pipeline stages (open in
+        # chrome://tracing or Perfetto; parity: reference profiler
+        # write_trace, profiler.py:57-198)
+        sc.profile().write_trace(trace_path)
+        log(f"[prof] chrome trace -> {trace_path}")
     elapsed = parallel.allreduce_max_time(elapsed, dist_device)
     if distributed:
         # Post-timing RCCL gather over xGMI: each rank ships the first

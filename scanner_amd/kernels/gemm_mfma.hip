@@ -648,6 +648,14 @@ const bf16* device_zero_chunk();
 bool try_splitk(const GemmArgs& g, hipStream_t s,
                 const ConvDesc* dd = nullptr) {
   constexpr int BM = 128, BN = 128;
+  // SCANNER_SPLITK_OFF=1: skip split-K entirely (A/B: with multiple
+  // pipeline instances co-running, their kernels may fill the chip
+  // without paying the partial+reduce HBM traffic).
+  static const bool kOff = []() {
+    const char* e = std::getenv("SCANNER_SPLITK_OFF");
+    return e && e[0] == '1';
+  }();
+  if (kOff) return false;
   if (!g.splitk_scratch || g.N % BN != 0 || g.K < 1024) return false;
   int ntiles_m = (g.M + BM - 1) / BM;
   int tiles = ntiles_m * (g.N / BN);
