@@ -330,3 +330,79 @@ def test_save_mp4_roundtrip(sc, tmp_path):
     r2 = sc.ingest_video_file(out_path, "rt2")
     assert r2["num_frames"] == 12
     assert (r2["width"], r2["height"]) == (64, 48)
+
+
+def make_sps_high(width, height, scaling_lists=False):
+    """High-profile SPS (profile_idc=100): exercises chroma_format_idc,
+    bit-depth and scaling-list parsing in h264_parse_sps."""
+    w = BitWriter()
+    w.u(100, 8)    # profile_idc high
+    w.u(0, 8)
+    w.u(40, 8)     # level 4.0
+    w.ue(0)        # sps_id
+    w.ue(1)        # chroma_format_idc 4:2:0
+    w.ue(0)        # bit_depth_luma_minus8
+    w.ue(0)        # bit_depth_chroma_minus8
+    w.u(0, 1)      # qpprime
+    if scaling_lists:
+        w.u(1, 1)  # seq_scaling_matrix_present
+        for i in range(8):
+            if i == 0:
+                w.u(1, 1)  # scaling list present
+                # 16 delta_scales of +1 (se code k=1): next_scale stays
+                # nonzero, so the decode process consumes exactly 16 se's
+                for _ in range(16):
+                    w.ue(1)
+            else:
+                w.u(0, 1)
+    else:
+        w.u(0, 1)
+    w.ue(0)        # log2_max_frame_num_minus4
+    w.ue(1)        # pic_order_cnt_type 1
+    w.u(1, 1)      # delta_pic_order_always_zero
+    w.ue(2)        # offset_for_non_ref_pic (se -1): code index 2
+    w.ue(0)        # offset_for_top_to_bottom (se 0)
+    w.ue(2)        # num_ref_frames_in_pic_order_cnt_cycle = 2
+    w.ue(2)        # se entries
+    w.ue(4)
+    w.ue(2)        # max_num_ref_frames
+    w.u(0, 1)      # gaps
+    w.ue(width // 16 - 1)
+    w.ue(height // 16 - 1)
+    w.u(1, 1)      # frame_mbs_only
+    w.u(1, 1)      # direct_8x8
+    w.u(0, 1)      # no cropping
+    w.u(0, 1)      # no vui
+    w.rbsp_trailing()
+    return b"\x67" + escape(w.bytes())
+
+
+def test_sps_high_profile():
+    info = _core.h264_parse_sps_py(make_sps_high(1280, 720))
+    assert (info["width"], info["height"]) == (1280, 720)
+    assert info["profile_idc"] == 100
+    info = _core.h264_parse_sps_py(make_sps_high(640, 480,
+                                                 scaling_lists=True))
+    assert (info["width"], info["height"]) == (640, 480)
+
+
+def test_annexb_high_profile_index():
+    sps, pps = make_sps_high(640, 368), make_pps()
+    stream = bytearray()
+    stream += SC + sps + SC + pps
+    stream += SC + make_slice(idr=True)
+    stream += SC + make_slice(idr=False)
+    idx = _core.h264_index(bytes(stream))
+    assert idx["num_frames"] == 2
+    assert (idx["width"], idx["height"]) == (640, 368)
+    assert idx["keyframe_indices"] == [0]
+
+
+def test_emulation_prevention_in_sps():
+    """An SPS whose RBSP contains 00 00 0x runs must round-trip through
+    escape/unescape (emulation_prevention_three_byte)."""
+    # width 4096: pic_width_in_mbs_minus1 = 255 -> long zero runs in the
+    # exp-golomb bits
+    sps = make_sps(4096, 48)
+    info = _core.h264_parse_sps_py(sps)
+    assert (info["width"], info["height"]) == (4096, 48)
