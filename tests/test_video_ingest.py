@@ -406,3 +406,76 @@ def test_emulation_prevention_in_sps():
     sps = make_sps(4096, 48)
     info = _core.h264_parse_sps_py(sps)
     assert (info["width"], info["height"]) == (4096, 48)
+
+
+def make_mp4_co64_multitrak(n_frames=4):
+    """mp4 variant: a non-video (audio-shaped) trak FIRST, then the AVC
+    trak using co64 (64-bit chunk offsets) — the demuxer must skip the
+    audio trak and read co64."""
+    sps, pps = make_sps(64, 48), make_pps()
+    samples = []
+    for i in range(n_frames):
+        nal = make_slice(idr=(i == 0), pad=5 + i)
+        samples.append(struct.pack(">I", len(nal)) + nal)
+    ftyp = box(b"ftyp", b"isom" + struct.pack(">I", 0x200) + b"isomavc1")
+    mdat = box(b"mdat", b"".join(samples))
+    data_off = len(ftyp) + 8
+
+    # decoy audio trak (hdlr 'soun', no stbl contents we care about)
+    a_hdlr = full(b"hdlr", struct.pack(">I", 0) + b"soun" + b"\x00" * 12 +
+                  b"a\x00")
+    a_mdia = box(b"mdia", full(b"mdhd", struct.pack(">IIIIHH", 0, 0, 48000,
+                                                    0, 0x55c4, 0)) + a_hdlr)
+    a_trak = box(b"trak", full(b"tkhd", b"\x00" * 80, flags=7) + a_mdia)
+
+    avcc = (bytes([1, sps[1], sps[2], sps[3], 0xFC | 3, 0xE0 | 1]) +
+            struct.pack(">H", len(sps)) + sps +
+            bytes([1]) + struct.pack(">H", len(pps)) + pps)
+    avc1 = box(b"avc1",
+               b"\x00" * 6 + struct.pack(">H", 1) + b"\x00" * 16 +
+               struct.pack(">HH", 64, 48) +
+               struct.pack(">II", 0x480000, 0x480000) +
+               struct.pack(">I", 0) + struct.pack(">H", 1) +
+               b"\x00" * 32 + struct.pack(">Hh", 24, -1) +
+               box(b"avcC", avcc))
+    stsd = full(b"stsd", struct.pack(">I", 1) + avc1)
+    stts = full(b"stts", struct.pack(">III", 1, n_frames, 3000))
+    stss = full(b"stss", struct.pack(">II", 1, 1))
+    stsc = full(b"stsc", struct.pack(">IIII", 1, 1, n_frames, 1))
+    stsz = full(b"stsz", struct.pack(">II", 0, n_frames) +
+                b"".join(struct.pack(">I", len(s)) for s in samples))
+    co64 = full(b"co64", struct.pack(">IQ", 1, data_off))
+    stbl = box(b"stbl", stsd + stts + stss + stsc + stsz + co64)
+    url = full(b"url ", b"", flags=1)
+    dinf = box(b"dinf", full(b"dref", struct.pack(">I", 1) + url))
+    minf = box(b"minf", full(b"vmhd", b"\x00" * 8, flags=1) + dinf + stbl)
+    hdlr = full(b"hdlr", struct.pack(">I", 0) + b"vide" + b"\x00" * 12 +
+                b"h\x00")
+    mdhd = full(b"mdhd", struct.pack(">IIIIHH", 0, 0, 90000,
+                                     3000 * n_frames, 0x55c4, 0))
+    mdia = box(b"mdia", mdhd + hdlr + minf)
+    trak = box(b"trak", full(b"tkhd", b"\x00" * 80, flags=7) + mdia)
+    mvhd = full(b"mvhd", struct.pack(">IIII", 0, 0, 90000, 0) +
+                struct.pack(">I", 0x10000) + struct.pack(">H", 0x100) +
+                b"\x00" * 10 + struct.pack(">9I", 0x10000, 0, 0, 0,
+                                           0x10000, 0, 0, 0, 0x40000000) +
+                b"\x00" * 24 + struct.pack(">I", 3))
+    moov = box(b"moov", mvhd + a_trak + trak)
+    sample_offsets = []
+    off = data_off
+    for s in samples:
+        sample_offsets.append(off)
+        off += len(s)
+    return ftyp + mdat + moov, sample_offsets
+
+
+def test_mp4_co64_and_multitrak(sc, tmp_path):
+    f, offsets = make_mp4_co64_multitrak(4)
+    t = _core.mp4_probe(f)
+    assert t["sample_offsets"] == offsets
+    assert (t["width"], t["height"]) == (64, 48)
+    assert t["keyframe_indices"] == [0]
+    p = tmp_path / "multi.mp4"
+    p.write_bytes(f)
+    r = sc.ingest_video_file(str(p), "ing_co64")
+    assert r["num_frames"] == 4
