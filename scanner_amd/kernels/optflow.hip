@@ -16,6 +16,7 @@
 #include <hip/hip_runtime.h>
 
 #include <map>
+#include <type_traits>
 #include <vector>
 
 #include "../csrc/memory.h"
@@ -90,8 +91,8 @@ __global__ void __launch_bounds__(256)
 // pyramid level (ch x cw) with the upsample bilerp fused in — the
 // standalone of_upsample_kernel pass and its flow round-trip through HBM
 // are skipped for the first iteration of every level.
-template <int RADIUS>
-__global__ void __launch_bounds__(256, 5)
+template <int RADIUS, int PY>
+__global__ void __launch_bounds__(256, 4)
     of_lk_kernel(const float* __restrict__ gray, i64 level_off, int h, int w,
                  const int* __restrict__ pair_f0,
                  const int* __restrict__ pair_f1,
@@ -99,23 +100,33 @@ __global__ void __launch_bounds__(256, 5)
                  float* __restrict__ flow_out,
                  const float* __restrict__ coarse, int ch, int cw) {
   constexpr int TILE = 16;
+  constexpr int TH = TILE * PY;              // output rows per block
   constexpr int HALO = RADIUS + 1;           // gradient needs +-1 past window
   constexpr int LW = TILE + 2 * HALO;        // staged tile width
+  constexpr int LH = TH + 2 * HALO;
   constexpr int GW = TILE + 2 * RADIUS;      // gradient tile width
-  __shared__ float lds_i0[LW * LW];
-  __shared__ float lds_gx[GW * GW];
-  __shared__ float lds_gy[GW * GW];
+  constexpr int GH = TH + 2 * RADIUS;
+  __shared__ float lds_i0[LH * LW];
+  __shared__ float lds_gx[GH * GW];
+  __shared__ float lds_gy[GH * GW];
 
   int pair = blockIdx.z;
   const float* I0 = gray + level_off + (i64)pair_f0[pair] * h * w;
   const float* I1 = gray + level_off + (i64)pair_f1[pair] * h * w;
 
-  int tx0 = blockIdx.x * TILE, ty0 = blockIdx.y * TILE;
-  int x = tx0 + threadIdx.x, y = ty0 + threadIdx.y;
-  // Issue the flow read before the LDS fill + gradient barriers so its
-  // ~200-900 cycle latency hides under the cooperative staging work.
-  float u = 0.f, v = 0.f;
-  if (x < w && y < h) {
+  int tx0 = blockIdx.x * TILE, ty0 = blockIdx.y * TH;
+  int x = tx0 + threadIdx.x;
+  // Issue the flow reads (all PY pixels) before the LDS fill + gradient
+  // barriers so their ~200-900 cycle latency hides under the cooperative
+  // staging work. PY > 1 amortizes the fill/barriers over PY output rows
+  // and doubles the per-wave load chains in the window loop.
+  float u[PY], v[PY];
+#pragma unroll
+  for (int py = 0; py < PY; ++py) {
+    u[py] = 0.f;
+    v[py] = 0.f;
+    int y = ty0 + py * TILE + (int)threadIdx.y;
+    if (x >= w || y >= h) continue;
     if (coarse != nullptr) {
       // fused upsample from the coarser level (same math as
       // of_upsample_kernel: bilerp + per-axis magnitude scaling)
@@ -137,18 +148,18 @@ __global__ void __launch_bounds__(256, 5)
                    v10 * ayc * (1 - axc) + v11 * ayc * axc;
         float sc = chn == 0 ? fx_scale : fy_scale;
         if (chn == 0)
-          u = vv * sc;
+          u[py] = vv * sc;
         else
-          v = vv * sc;
+          v[py] = vv * sc;
       }
     } else {
       const float* fin = flow_in + (i64)pair * h * w * 2;
-      u = fin[((i64)y * w + x) * 2 + 0];
-      v = fin[((i64)y * w + x) * 2 + 1];
+      u[py] = fin[((i64)y * w + x) * 2 + 0];
+      v[py] = fin[((i64)y * w + x) * 2 + 1];
     }
   }
   // Cooperative LDS fill with clamped loads.
-  for (int i = threadIdx.y * TILE + threadIdx.x; i < LW * LW;
+  for (int i = threadIdx.y * TILE + threadIdx.x; i < LH * LW;
        i += TILE * TILE) {
     int ly = i / LW, lx = i % LW;
     int gy = min(max(ty0 + ly - HALO, 0), h - 1);
@@ -158,7 +169,7 @@ __global__ void __launch_bounds__(256, 5)
   __syncthreads();
   // Gradient tile computed once per block (each cell is read by up to 49
   // window taps otherwise).
-  for (int i = threadIdx.y * TILE + threadIdx.x; i < GW * GW;
+  for (int i = threadIdx.y * TILE + threadIdx.x; i < GH * GW;
        i += TILE * TILE) {
     int gy = i / GW + 1, gx = i % GW + 1;  // offsets into lds_i0
     lds_gx[i] = 0.5f * (lds_i0[gy * LW + gx + 1] - lds_i0[gy * LW + gx - 1]);
@@ -167,102 +178,103 @@ __global__ void __launch_bounds__(256, 5)
   }
   __syncthreads();
 
-  if (x >= w || y >= h) return;
   float* fout = flow_out + (i64)pair * h * w * 2;
-
-  // The warp offset (u,v) is constant across the window, so the bilinear
-  // fractions are too: the 49 taps read a contiguous (2R+2)^2 region of I1
-  // at one fractional offset. Interior fast path: slide two unrolled row
-  // register arrays down that region — 64 loads/pixel instead of 196, no
-  // per-tap floor/clamp. Border threads take the clamped slow path.
-  float a11 = 0, a12 = 0, a22 = 0, b1 = 0, b2 = 0;
-  float wxf = x + u, wyf = y + v;
-  int xi = (int)floorf(wxf), yi = (int)floorf(wyf);
-  float ax = wxf - xi, ay = wyf - yi;
   constexpr int RW = 2 * RADIUS + 2;
-  bool interior = xi - RADIUS >= 0 && xi + RADIUS + 1 < w &&
-                  yi - RADIUS >= 0 && yi + RADIUS + 1 < h;
-  if (interior) {
-    // Software pipeline: row dy+2 is fetched while rows (dy, dy+1) are
-    // consumed, so each row's ~400-cycle L1/L2 latency hides under a full
-    // iteration of VALU work (PMC before: 49% of wave time in waits).
-    float r0[RW], r1[RW], r2[RW];
-    const float* row = I1 + (i64)(yi - RADIUS) * w + (xi - RADIUS);
+
 #pragma unroll
-    for (int j = 0; j < RW; ++j) r0[j] = row[j];
-    const float* row1 = row + w;
+  for (int py = 0; py < PY; ++py) {
+    int lyb = py * TILE + (int)threadIdx.y;  // row within the tile space
+    int y = ty0 + lyb;
+    if (x >= w || y >= h) continue;
+    float uu = u[py], vv = v[py];
+
+    // The warp offset (u,v) is constant across the window, so the bilinear
+    // fractions are too: the 49 taps read a contiguous (2R+2)^2 region of
+    // I1 at one fractional offset. Interior fast path: pipeline three row
+    // register arrays down that region (row dy+2 fetched while rows
+    // (dy,dy+1) are consumed). Border threads take the clamped slow path.
+    float a11 = 0, a12 = 0, a22 = 0, b1 = 0, b2 = 0;
+    float wxf = x + uu, wyf = y + vv;
+    int xi = (int)floorf(wxf), yi = (int)floorf(wyf);
+    float ax = wxf - xi, ay = wyf - yi;
+    bool interior = xi - RADIUS >= 0 && xi + RADIUS + 1 < w &&
+                    yi - RADIUS >= 0 && yi + RADIUS + 1 < h;
+    if (interior) {
+      float r0[RW], r1[RW], r2[RW];
+      const float* row = I1 + (i64)(yi - RADIUS) * w + (xi - RADIUS);
 #pragma unroll
-    for (int j = 0; j < RW; ++j) r1[j] = row1[j];
+      for (int j = 0; j < RW; ++j) r0[j] = row[j];
+      const float* row1 = row + w;
 #pragma unroll
-    for (int dy = -RADIUS; dy <= RADIUS; ++dy) {
-      if (dy < RADIUS) {
-        const float* p2 = I1 + (i64)(yi + dy + 2) * w + (xi - RADIUS);
+      for (int j = 0; j < RW; ++j) r1[j] = row1[j];
 #pragma unroll
-        for (int j = 0; j < RW; ++j) r2[j] = p2[j];
+      for (int dy = -RADIUS; dy <= RADIUS; ++dy) {
+        if (dy < RADIUS) {
+          const float* p2 = I1 + (i64)(yi + dy + 2) * w + (xi - RADIUS);
+#pragma unroll
+          for (int j = 0; j < RW; ++j) r2[j] = p2[j];
+        }
+        int ly = lyb + HALO + dy;
+        int gy_ = lyb + RADIUS + dy;
+#pragma unroll
+        for (int dx = -RADIUS; dx <= RADIUS; ++dx) {
+          int lx = (int)threadIdx.x + HALO + dx;
+          int gx_ = (int)threadIdx.x + RADIUS + dx;
+          float ix = lds_gx[gy_ * GW + gx_];
+          float iy = lds_gy[gy_ * GW + gx_];
+          int j = dx + RADIUS;
+          float top = r0[j] + ax * (r0[j + 1] - r0[j]);
+          float bot = r1[j] + ax * (r1[j + 1] - r1[j]);
+          float it = top + ay * (bot - top) - lds_i0[ly * LW + lx];
+          a11 += ix * ix;
+          a12 += ix * iy;
+          a22 += iy * iy;
+          b1 += ix * it;
+          b2 += iy * it;
+        }
+#pragma unroll
+        for (int j = 0; j < RW; ++j) {
+          r0[j] = r1[j];
+          r1[j] = r2[j];
+        }
       }
-      int ly = (int)threadIdx.y + HALO + dy;
-      int gy_ = (int)threadIdx.y + RADIUS + dy;
+    } else {
 #pragma unroll
-      for (int dx = -RADIUS; dx <= RADIUS; ++dx) {
-        int lx = (int)threadIdx.x + HALO + dx;
-        int gx_ = (int)threadIdx.x + RADIUS + dx;
-        float ix = lds_gx[gy_ * GW + gx_];
-        float iy = lds_gy[gy_ * GW + gx_];
-        int j = dx + RADIUS;
-        // per-tap 2D lerp: a hoisted vertical-lerp row array costs 8
-        // VGPRs and one wave of occupancy — latency, not VALU, is the
-        // bound here (PMC: ACTIVE 26%, WAIT 49%)
-        float top = r0[j] + ax * (r0[j + 1] - r0[j]);
-        float bot = r1[j] + ax * (r1[j + 1] - r1[j]);
-        float it = top + ay * (bot - top) - lds_i0[ly * LW + lx];
-        a11 += ix * ix;
-        a12 += ix * iy;
-        a22 += iy * iy;
-        b1 += ix * it;
-        b2 += iy * it;
-      }
+      for (int dy = -RADIUS; dy <= RADIUS; ++dy) {
 #pragma unroll
-      for (int j = 0; j < RW; ++j) {
-        r0[j] = r1[j];
-        r1[j] = r2[j];
+        for (int dx = -RADIUS; dx <= RADIUS; ++dx) {
+          int lx = (int)threadIdx.x + HALO + dx;
+          int ly = lyb + HALO + dy;
+          float ix = lds_gx[(lyb + RADIUS + dy) * GW +
+                            threadIdx.x + RADIUS + dx];
+          float iy = lds_gy[(lyb + RADIUS + dy) * GW +
+                            threadIdx.x + RADIUS + dx];
+          float i0v = lds_i0[ly * LW + lx];
+          int x0c = min(max(xi + dx, 0), w - 1);
+          int x1c = min(max(xi + dx + 1, 0), w - 1);
+          int y0c = min(max(yi + dy, 0), h - 1);
+          int y1c = min(max(yi + dy + 1, 0), h - 1);
+          float w00 = I1[(i64)y0c * w + x0c], w01 = I1[(i64)y0c * w + x1c];
+          float w10 = I1[(i64)y1c * w + x0c], w11 = I1[(i64)y1c * w + x1c];
+          float i1v = w00 * (1 - ay) * (1 - ax) + w01 * (1 - ay) * ax +
+                      w10 * ay * (1 - ax) + w11 * ay * ax;
+          float it = i1v - i0v;
+          a11 += ix * ix;
+          a12 += ix * iy;
+          a22 += iy * iy;
+          b1 += ix * it;
+          b2 += iy * it;
+        }
       }
     }
-  } else {
-#pragma unroll
-    for (int dy = -RADIUS; dy <= RADIUS; ++dy) {
-#pragma unroll
-      for (int dx = -RADIUS; dx <= RADIUS; ++dx) {
-        int lx = (int)threadIdx.x + HALO + dx;
-        int ly = (int)threadIdx.y + HALO + dy;
-        float ix = lds_gx[(threadIdx.y + RADIUS + dy) * GW +
-                          threadIdx.x + RADIUS + dx];
-        float iy = lds_gy[(threadIdx.y + RADIUS + dy) * GW +
-                          threadIdx.x + RADIUS + dx];
-        float i0v = lds_i0[ly * LW + lx];
-        int x0c = min(max(xi + dx, 0), w - 1);
-        int x1c = min(max(xi + dx + 1, 0), w - 1);
-        int y0c = min(max(yi + dy, 0), h - 1);
-        int y1c = min(max(yi + dy + 1, 0), h - 1);
-        float w00 = I1[(i64)y0c * w + x0c], w01 = I1[(i64)y0c * w + x1c];
-        float w10 = I1[(i64)y1c * w + x0c], w11 = I1[(i64)y1c * w + x1c];
-        float i1v = w00 * (1 - ay) * (1 - ax) + w01 * (1 - ay) * ax +
-                    w10 * ay * (1 - ax) + w11 * ay * ax;
-        float it = i1v - i0v;
-        a11 += ix * ix;
-        a12 += ix * iy;
-        a22 += iy * iy;
-        b1 += ix * it;
-        b2 += iy * it;
-      }
+    float det = a11 * a22 - a12 * a12;
+    if (det > of::kDetEps) {
+      uu -= (a22 * b1 - a12 * b2) / det;
+      vv -= (a11 * b2 - a12 * b1) / det;
     }
+    fout[((i64)y * w + x) * 2 + 0] = uu;
+    fout[((i64)y * w + x) * 2 + 1] = vv;
   }
-  float det = a11 * a22 - a12 * a12;
-  if (det > of::kDetEps) {
-    u -= (a22 * b1 - a12 * b2) / det;
-    v -= (a11 * b2 - a12 * b1) / det;
-  }
-  fout[((i64)y * w + x) * 2 + 0] = u;
-  fout[((i64)y * w + x) * 2 + 1] = v;
 }
 
 inline int grid_1d(i64 total, int block = 256, int cap = 4096) {
@@ -481,18 +493,23 @@ class OpticalFlowKernelGPU : public BaseKernel {
     const int* d_pf0 = (const int*)d_pf;
     const int* d_pf1 = (const int*)(d_pf + n * sizeof(int));
     for (int l = top; l >= 0; --l) {
-      dim3 grid((lw[l] + 15) / 16, (lh[l] + 15) / 16, (u32)n);
+      // 2 output rows per thread when the level is tall enough to keep
+      // the grid chip-filling; tiny coarse levels stay at 1
+      int py = (lh[l] >= 64 && (i64)n * ((lw[l] + 15) / 16) *
+                                   ((lh[l] + 31) / 32) >= 512) ? 2 : 1;
+      dim3 grid((lw[l] + 15) / 16, (lh[l] + 16 * py - 1) / (16 * py),
+                (u32)n);
       dim3 block(16, 16);
       for (int it = 0; it < iters_; ++it) {
         float* dst = (l == 0 && it == iters_ - 1) ? (float*)out_block : alt;
         if (l < top && it == 0) {
           // first iteration of each finer level: fused upsample read from
           // the coarser level's result (cur holds level l+1 flow)
-          launch_lk(grid, block, s, gbase, loff[l], lh[l], lw[l], d_pf0,
-                    d_pf1, nullptr, dst, cur, lh[l + 1], lw[l + 1]);
+          launch_lk(py, grid, block, s, gbase, loff[l], lh[l], lw[l],
+                    d_pf0, d_pf1, nullptr, dst, cur, lh[l + 1], lw[l + 1]);
         } else {
-          launch_lk(grid, block, s, gbase, loff[l], lh[l], lw[l], d_pf0,
-                    d_pf1, cur, dst);
+          launch_lk(py, grid, block, s, gbase, loff[l], lh[l], lw[l],
+                    d_pf0, d_pf1, cur, dst);
         }
         std::swap(cur, alt);
         if (dst == (float*)out_block) cur = (float*)out_block;
@@ -528,24 +545,26 @@ class OpticalFlowKernelGPU : public BaseKernel {
     return of::num_levels(h, w, max_levels_);
   }
 
-  void launch_lk(dim3 grid, dim3 block, hipStream_t s, const float* gray,
-                 i64 off, int h, int w, const int* pf0, const int* pf1,
-                 const float* fin, float* fout,
+  void launch_lk(int py, dim3 grid, dim3 block, hipStream_t s,
+                 const float* gray, i64 off, int h, int w, const int* pf0,
+                 const int* pf1, const float* fin, float* fout,
                  const float* coarse = nullptr, int ch = 0, int cw = 0) {
-    switch (radius_) {
-      case 2:
-        of_lk_kernel<2><<<grid, block, 0, s>>>(gray, off, h, w, pf0, pf1,
-                                               fin, fout, coarse, ch, cw);
-        break;
-      case 3:
-        of_lk_kernel<3><<<grid, block, 0, s>>>(gray, off, h, w, pf0, pf1,
-                                               fin, fout, coarse, ch, cw);
-        break;
-      default:
-        of_lk_kernel<4><<<grid, block, 0, s>>>(gray, off, h, w, pf0, pf1,
-                                               fin, fout, coarse, ch, cw);
-        break;
-    }
+    auto go = [&](auto rad, auto pyc) {
+      of_lk_kernel<decltype(rad)::value, decltype(pyc)::value>
+          <<<grid, block, 0, s>>>(gray, off, h, w, pf0, pf1, fin, fout,
+                                  coarse, ch, cw);
+    };
+    auto by_r = [&](auto pyc) {
+      switch (radius_) {
+        case 2: go(std::integral_constant<int, 2>{}, pyc); break;
+        case 3: go(std::integral_constant<int, 3>{}, pyc); break;
+        default: go(std::integral_constant<int, 4>{}, pyc); break;
+      }
+    };
+    if (py == 2)
+      by_r(std::integral_constant<int, 2>{});
+    else
+      by_r(std::integral_constant<int, 1>{});
     OF_CHECK(hipGetLastError());
   }
 

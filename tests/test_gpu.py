@@ -105,20 +105,22 @@ def test_gpu_pool_allocator():
 def test_gpu_optical_flow_matches_cpu(sc):
     from test_engine_cpu import make_textured_pair
     pair = make_textured_pair(h=96, w=128, dx=2, dy=1)
-    # 5 frames: repeat the pair pattern so several stencil windows run
-    frames = np.concatenate([pair, pair[::-1], pair[:1]])
+    # 24 frames: enough stencil windows that the LK launch takes the
+    # 2-rows-per-thread (PY=2) path as well as PY=1 on the coarse levels
+    frames = np.concatenate([pair, pair[::-1]] * 6)
+    n = frames.shape[0]
     video = sp.NamedVideoStream(sc, "g_of", frames=frames, codec="raw")
     for dev, name in ((sp.DeviceType.CPU, "g_of_cpu"),
                       (sp.DeviceType.GPU, "g_of_gpu")):
         frame = sc.io.Input([video])
         flow = sc.ops.OpticalFlow(frame=frame, device=dev)
         out = sp.NamedStream(sc, name)
-        sc.run(sc.io.Output(flow, [out]), sp.PerfParams.manual(4, 8),
+        sc.run(sc.io.Output(flow, [out]), sp.PerfParams.manual(24, 24),
                cache_mode=sp.CacheMode.Overwrite,
                gpu_ids=[0] if dev == sp.DeviceType.GPU else [])
     cpu = np.stack(list(sp.NamedVideoStream(sc, "g_of_cpu").load()))
     gpu = np.stack(list(sp.NamedVideoStream(sc, "g_of_gpu").load()))
-    assert cpu.shape == gpu.shape == (5, 96, 128, 2)
+    assert cpu.shape == gpu.shape == (n, 96, 128, 2)
     # same algorithm in f32; differences only from fma contraction
     diff = np.abs(cpu - gpu)
     assert np.median(diff) < 1e-3, np.median(diff)
