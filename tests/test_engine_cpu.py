@@ -885,3 +885,24 @@ def test_table_megafile(tmp_path):
     for k in range(3):
         rows = list(sp.NamedStream(sc2, f"mf{k}").load())
         assert rows == [bytes([k, i]) for i in range(4 + k)]
+
+
+def test_gather_unsorted_streaming(sc):
+    """Out-of-order Gather ([39, 5, 20, 2]) through the STREAMING packet
+    executor: early output packets need late input rows, so the backward
+    watermark pass must pull them forward while liveness counts keep every
+    element alive until its last (possibly much later) read."""
+    frames = make_video(n=40)
+    video = sp.NamedVideoStream(sc, "guns", frames=frames, codec="svc")
+    wanted = [39, 5, 20, 2, 20]
+    frame = sc.io.Input([video])
+    g = sc.streams.Gather(frame, [wanted])
+    hist = sc.ops.Histogram(frame=g)
+    out = sp.NamedStream(sc, "guns_out")
+    sc.run(sc.io.Output(hist, [out]), sp.PerfParams.manual(2, 3),
+           cache_mode=sp.CacheMode.Overwrite)
+    rows = list(out.load())
+    assert len(rows) == len(wanted)
+    for k, blob in enumerate(rows):
+        got = np.frombuffer(blob, np.uint32).reshape(3, 256)
+        np.testing.assert_array_equal(got, ref_histogram(frames[wanted[k]]))
