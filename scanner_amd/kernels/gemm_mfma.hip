@@ -759,7 +759,10 @@ void launch_smallk(const GemmArgs& g, hipStream_t s) {
   int ntiles_n = (g.N + BN - 1) / BN;
   // Enough strips to fill the chip (>=2048 workgroups when the shape
   // allows), each walking a run of consecutive M-tiles.
-  int target_wgs = 2048;
+  static const int target_wgs = []() {
+    const char* e = std::getenv("SCANNER_SMALLK_WGS");
+    return e ? std::max(256, atoi(e)) : 2048;
+  }();
   int strips = std::max(1, std::min(ntiles_m, target_wgs / ntiles_n));
   int tiles_per_wg = (ntiles_m + strips - 1) / strips;
   strips = (ntiles_m + tiles_per_wg - 1) / tiles_per_wg;
