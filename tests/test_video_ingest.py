@@ -479,3 +479,28 @@ def test_mp4_co64_and_multitrak(sc, tmp_path):
     p.write_bytes(f)
     r = sc.ingest_video_file(str(p), "ing_co64")
     assert r["num_frames"] == 4
+
+
+def test_export_mp4_rejects_svc_table(sc):
+    frames = np.random.RandomState(0).randint(
+        0, 255, size=(4, 32, 32, 3)).astype(np.uint8)
+    sp.NamedVideoStream(sc, "svc_only", frames=frames, codec="svc")
+    with pytest.raises(Exception, match="h264"):
+        sp.NamedVideoStream(sc, "svc_only").save_mp4("/tmp/nope.mp4")
+
+
+def test_ingest_videos_batch_with_file(sc, tmp_path):
+    """storage.ingest_videos accepts real video file paths alongside
+    frame arrays, reporting failures per entry (FailedVideo parity)."""
+    from scanner_amd.storage import ingest_videos
+    stream, _, _ = make_annexb(gops=2, frames_per_gop=3)
+    p = tmp_path / "c.h264"
+    p.write_bytes(stream)
+    frames = np.zeros((4, 16, 16, 3), np.uint8)
+    streams, failures = ingest_videos(
+        sc, [("bv_file", str(p)), ("bv_frames", frames),
+             ("bv_bad", str(tmp_path / "missing.mp4"))])
+    assert len(streams) == 2 and len(failures) == 1
+    assert failures[0][0] == "bv_bad"
+    assert sc.table_info("bv_file")["num_rows"] == 6
+    assert sc.table_info("bv_frames")["num_rows"] == 4
