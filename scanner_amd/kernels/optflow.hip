@@ -493,10 +493,20 @@ class OpticalFlowKernelGPU : public BaseKernel {
     const int* d_pf0 = (const int*)d_pf;
     const int* d_pf1 = (const int*)(d_pf + n * sizeof(int));
     for (int l = top; l >= 0; --l) {
-      // 2 output rows per thread when the level is tall enough to keep
-      // the grid chip-filling; tiny coarse levels stay at 1
+      // multiple output rows per thread when the level is tall enough
+      // to keep the grid chip-filling; tiny coarse levels stay at 1
+      // (SCANNER_LK_PY overrides for A/B)
+      static const int py_max = []() {
+        const char* e = std::getenv("SCANNER_LK_PY");
+        int v = e ? atoi(e) : 2;
+        return v == 1 || v == 2 || v == 4 ? v : 2;
+      }();
       int py = (lh[l] >= 64 && (i64)n * ((lw[l] + 15) / 16) *
-                                   ((lh[l] + 31) / 32) >= 512) ? 2 : 1;
+                                   ((lh[l] + 31) / 32) >= 512) ? py_max : 1;
+      if (py == 4 && (lh[l] < 128 || (i64)n * ((lw[l] + 15) / 16) *
+                                         ((lh[l] + 63) / 64) < 512)) {
+        py = 2;
+      }
       dim3 grid((lw[l] + 15) / 16, (lh[l] + 16 * py - 1) / (16 * py),
                 (u32)n);
       dim3 block(16, 16);
@@ -561,7 +571,9 @@ class OpticalFlowKernelGPU : public BaseKernel {
         default: go(std::integral_constant<int, 4>{}, pyc); break;
       }
     };
-    if (py == 2)
+    if (py == 4)
+      by_r(std::integral_constant<int, 4>{});
+    else if (py == 2)
       by_r(std::integral_constant<int, 2>{});
     else
       by_r(std::integral_constant<int, 1>{});
