@@ -95,9 +95,13 @@ def main():
 
     have_gpu = _core.have_gpu()
     device = sp.DeviceType.GPU if have_gpu else sp.DeviceType.CPU
-    if args.pipeline in ("resnet", "full") and "ResNet50" not in \
-            _core.registered_ops():
-        log("ResNet50 op not available; falling back to hist pipeline")
+    if args.pipeline in ("resnet", "full", "pose") and not (
+            have_gpu and "ResNet50" in _core.registered_ops()):
+        log("DNN ops need a GPU; falling back to hist pipeline")
+        args.pipeline = "hist"
+    if args.pipeline == "flow" and not _core.has_kernel(
+            "OpticalFlow", int(device)):
+        log("OpticalFlow kernel missing; falling back to hist")
         args.pipeline = "hist"
 
     tmp = tempfile.mkdtemp(prefix=f"scanner_bench_r{rank}_")
