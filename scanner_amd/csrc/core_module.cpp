@@ -141,6 +141,7 @@ PYBIND11_MODULE(_core, m) {
     u8* dC = new_buffer(dev, (size_t)M * N * 2);
     u8* dS = new_buffer(dev, (size_t)N * 8);
     u8* dSK = new_buffer(dev, 64u << 20);
+    splitk_scratch_init(dSK, 64u << 20, nullptr);
     GemmArgs g;
     g.splitk_scratch = dSK;
     g.splitk_scratch_bytes = 64u << 20;

@@ -204,6 +204,7 @@ void ResNet50KernelGPU::ensure_workspace(int ih, int iw, int ic) {
   ws_.colbuf = nullptr;  // every resnet conv runs direct or implicit
   ws_.pre = new_buffer(dev, (size_t)n * 224 * 224 * 8 * 2);
   ws_.skbuf = new_buffer(dev, kSplitkBytes);
+  splitk_scratch_init(ws_.skbuf, kSplitkBytes, nullptr);
   ws_.ih = ih;
   ws_.iw = iw;
   ws_.ic = ic;

@@ -27,6 +27,13 @@ struct GemmArgs {
 
 void gemm_bf16(const GemmArgs& g, void* stream);
 
+// Zero the fused-reduce tile-counter region at the TAIL of a split-K
+// scratch buffer. Owners must call this ONCE when they allocate the
+// scratch (the kernel self-restores the counters to zero after each
+// launch, so no per-call work is needed). stream may be null (synchronous
+// memset).
+void splitk_scratch_init(void* scratch, size_t bytes, void* stream);
+
 // Implicit-GEMM convolution: the GEMM's A operand is gathered straight
 // from the NHWC activation tensor during LDS staging (no im2col buffer in
 // HBM). g.A = activations (NHWC bf16); g.M = n*oh*ow; g.K = kp (padded

@@ -410,16 +410,31 @@ __global__ void __launch_bounds__(256, 2)
 // For launch-bound shapes (few output tiles, deep K — e.g. ResNet stage-4
 // conv2: M-tiles x N-tiles = 52 workgroups on a 256-CU chip, measured
 // ~59 TF/s), S workgroup groups each compute a K-slice partial into f32
-// scratch and a reduce kernel applies the epilogue. No inter-workgroup
-// hand-off inside a launch (guide correctness boundary G16) — the split
-// and reduce are separate kernels on the same stream.
+// scratch and a reduce kernel applies the epilogue. An OPT-IN fused
+// variant (SCANNER_SPLITK_FUSED=1) instead has the LAST workgroup to
+// finish a tile (tile-completion counter) reduce the partials in-kernel —
+// G16-safe since no workgroup ever WAITS on another (each increments its
+// tile's counter exactly once and exits; only the one observing
+// count == splits-1 does the extra work) — but it measured 2.2x SLOWER on
+// the flagship (see the kFused comment in try_splitk), so the separate
+// reduce launch stays the default.
+struct SplitkEp {
+  unsigned int* counters = nullptr;  // null => unfused (write partials only)
+  int splits = 0;
+  const float* scale = nullptr;
+  const float* bias = nullptr;
+  const bf16* residual = nullptr;
+  bf16* C = nullptr;
+  int relu = 0;
+};
+
 template <int BM, int BN, int WM, int WN, bool IMPLICIT, bool ATOMIC>
 __global__ void __launch_bounds__(256, 2)
     gemm_bf16_splitk_kernel(const bf16* __restrict__ A,
                             const bf16* __restrict__ B, int M, int N, int K,
                             int ksteps_per_split,
                             float* __restrict__ partials, ConvDesc d,
-                            const bf16* __restrict__ zero) {
+                            const bf16* __restrict__ zero, SplitkEp ep) {
   constexpr int BK = 64;
   constexpr int FM = BM / WM / 16;
   constexpr int FN = BN / WN / 16;
@@ -569,6 +584,53 @@ __global__ void __launch_bounds__(256, 2)
         }
       }
     }
+    if (ep.counters) {
+      // Fused reduce: release this split's partial (per-thread fence,
+      // then a block barrier so tid 0's atomic is ordered after EVERY
+      // thread's stores), count the tile's completions, and if this
+      // workgroup is the last one, sum the partials in fixed split order
+      // (bitwise identical to splitk_reduce_kernel) + epilogue + bf16
+      // store. The counter is restored to 0 for the next launch that
+      // reuses this scratch (launches on the stream are ordered, and
+      // hipGraph replays see the same all-zero state).
+      __threadfence();
+      __syncthreads();
+      __shared__ unsigned int s_old;
+      if (tid == 0) s_old = atomicAdd(ep.counters + tile, 1u);
+      __syncthreads();
+      if (s_old != (unsigned int)ep.splits - 1) return;
+      __threadfence();  // acquire: other splits' partials now visible
+      int rows = min(BM, M - m0);
+      constexpr int Q = BN / 4;  // float4 quads per tile row
+      for (int idx = tid; idx < rows * Q; idx += 256) {
+        int r = idx / Q, q = idx - r * Q;
+        size_t base = (size_t)(m0 + r) * N + n0 + q * 4;
+        float4 av = make_float4(0.f, 0.f, 0.f, 0.f);
+        for (int sp = 0; sp < ep.splits; ++sp) {
+          float4 p = *reinterpret_cast<const float4*>(
+              partials + (size_t)sp * M * N + base);
+          av.x += p.x;
+          av.y += p.y;
+          av.z += p.z;
+          av.w += p.w;
+        }
+        float v4[4] = {av.x, av.y, av.z, av.w};
+        bf16 o4[4];
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+          float v = v4[k];
+          int col = n0 + q * 4 + k;
+          if (ep.scale) v *= ep.scale[col];
+          if (ep.bias) v += ep.bias[col];
+          if (ep.residual) v += bf16_to_f32(ep.residual[base + k]);
+          if (ep.relu) v = v > 0.f ? v : 0.f;
+          o4[k] = f32_to_bf16(v);
+        }
+        *reinterpret_cast<uint64_t*>(ep.C + base) =
+            *reinterpret_cast<const uint64_t*>(o4);
+      }
+      if (tid == 0) ep.counters[tile] = 0;
+    }
   }
 }
 
@@ -670,7 +732,14 @@ bool try_splitk(const GemmArgs& g, hipStream_t s,
   int ksteps = g.K / 64;
   int want = std::min({kMaxSplits, ksteps / 4, (2048 + tiles - 1) / tiles});
   size_t per_split = (size_t)g.M * g.N * 4;
-  int fit = (int)(g.splitk_scratch_bytes / per_split);
+  // The scratch tail holds the fused-reduce tile counters (tiles < 256 so
+  // 4 KB is ample); the owner zeroes it once at allocation
+  // (splitk_scratch_init) and the kernel self-restores it after each use.
+  constexpr size_t kCtrBytes = 4096;
+  size_t avail = g.splitk_scratch_bytes > kCtrBytes
+                     ? g.splitk_scratch_bytes - kCtrBytes
+                     : 0;
+  int fit = (int)(avail / per_split);
   int splits = std::min(want, fit);
   if (splits < 2) return false;
   int ksteps_per_split = (ksteps + splits - 1) / splits;
@@ -686,6 +755,32 @@ bool try_splitk(const GemmArgs& g, hipStream_t s,
     const char* e = std::getenv("SCANNER_SPLITK_ATOMIC");
     return e && e[0] == '1';
   }();
+  // Fused reduce (opt-in, SCANNER_SPLITK_FUSED=1; default OFF). Measured
+  // NEGATIVE on MI355X: flagship 7.6k vs 16.6k f/s, resnet 7.9k vs 18.0k
+  // (same box, 3-step A/B, profiles/r02_results.md update 12). Numerics
+  // are correct (27 GPU tests green with it on), but the device-scope
+  // release fence every workgroup needs before bumping its tile counter
+  // forces an L2 writeback per WG — on 8 per-XCD L2s that evicts the
+  // B-operand and co-running instances' working sets — and the tail
+  // reduce runs with one 256-thread WG per tile vs the standalone
+  // kernel's chip-wide parallelism. Kept as a documented experiment; the
+  // separate splitk_reduce_kernel launch is the default.
+  static const bool kFused = []() {
+    const char* e = std::getenv("SCANNER_SPLITK_FUSED");
+    return e && e[0] == '1';
+  }();
+  bool fused = kFused && !kAtomic;
+  SplitkEp ep{};
+  if (fused) {
+    ep.counters = reinterpret_cast<unsigned int*>(
+        (u8*)g.splitk_scratch + g.splitk_scratch_bytes - kCtrBytes);
+    ep.splits = splits;
+    ep.scale = g.scale;
+    ep.bias = g.bias;
+    ep.residual = (const bf16*)g.residual;
+    ep.C = (bf16*)g.C;
+    ep.relu = g.relu ? 1 : 0;
+  }
   hipError_t e;
   if (kAtomic) {
     e = hipMemsetAsync(g.splitk_scratch, 0, (size_t)mn * 4, s);
@@ -698,31 +793,36 @@ bool try_splitk(const GemmArgs& g, hipStream_t s,
           <<<tiles * splits, 256, 0, s>>>(
               (const bf16*)g.A, (const bf16*)g.B, g.M, g.N, g.K,
               ksteps_per_split, (float*)g.splitk_scratch, *dd,
-              device_zero_chunk());
+              device_zero_chunk(), SplitkEp{});
     } else {
       gemm_bf16_splitk_kernel<BM, BN, 2, 2, false, true>
           <<<tiles * splits, 256, 0, s>>>(
               (const bf16*)g.A, (const bf16*)g.B, g.M, g.N, g.K,
               ksteps_per_split, (float*)g.splitk_scratch, ConvDesc{},
-              nullptr);
+              nullptr, SplitkEp{});
     }
   } else if (dd) {
     gemm_bf16_splitk_kernel<BM, BN, 2, 2, true, false>
         <<<tiles * splits, 256, 0, s>>>(
             (const bf16*)g.A, (const bf16*)g.B, g.M, g.N, g.K,
             ksteps_per_split, (float*)g.splitk_scratch, *dd,
-            device_zero_chunk());
+            device_zero_chunk(), ep);
   } else {
     gemm_bf16_splitk_kernel<BM, BN, 2, 2, false, false>
         <<<tiles * splits, 256, 0, s>>>(
             (const bf16*)g.A, (const bf16*)g.B, g.M, g.N, g.K,
-            ksteps_per_split, (float*)g.splitk_scratch, ConvDesc{}, nullptr);
+            ksteps_per_split, (float*)g.splitk_scratch, ConvDesc{}, nullptr,
+            ep);
   }
   e = hipGetLastError();
   if (e != hipSuccess) {
     throw ScannerError(std::string("splitk launch failed: ") +
                        hipGetErrorString(e));
   }
+  if (fused) return true;  // epilogue ran in-kernel; no reduce launch
+  // 4x more threads than quads: the extra waves are pure memory-level
+  // parallelism for this latency-bound pass (measured 18.2 us avg vs a
+  // ~1.3 us bandwidth bound)
   int grid = (int)std::min<i64>(4096, (mn + 255) / 256);
   auto disp = [&](auto relu, auto res) {
     if (kAtomic) {
@@ -873,6 +973,15 @@ void conv_gemm_bf16(const GemmArgs& g, const ConvDesc& d, void* stream) {
   } else {
     launch_conv_variant<64, 64, 2, 2>(g, d, s);
   }
+}
+
+void splitk_scratch_init(void* scratch, size_t bytes, void* stream) {
+  if (!scratch || bytes < 4096) return;
+  u8* tail = (u8*)scratch + bytes - 4096;
+  hipError_t e = stream
+                     ? hipMemsetAsync(tail, 0, 4096, (hipStream_t)stream)
+                     : hipMemset(tail, 0, 4096);
+  SCA_CHECK(e == hipSuccess, "splitk_scratch_init memset failed");
 }
 
 void gemm_bf16(const GemmArgs& g, void* stream) {
