@@ -84,8 +84,15 @@ struct SvcPacketView {
   const u32* super_off = nullptr;
   const u8* widths = nullptr;
   const u8* packed = nullptr;
+  size_t packed_size = 0;  // bytes available in the packed region
 };
 
+// Parse + validate a packet header. All size arithmetic is 64-bit and the
+// geometry invariants (ngroups = ceil(nbytes/32), nsuper = ceil(ngroups/
+// 128)) are enforced, so a corrupt header cannot make the derived
+// pointers overrun `size` (tests/cpp/asan_parsers.cpp fuzzes this under
+// AddressSanitizer). Per-group width/offset payloads are validated lazily
+// by the CPU decoder as it walks them.
 SvcPacketView svc_parse_packet(const u8* pkt, size_t size);
 
 // debug: launch the decode kernel with a diagnostic dump (tests only)

@@ -10,9 +10,13 @@ namespace {
 constexpr u32 kMagic = 0x31435653;  // 'SVC1' LE
 
 inline u8 zigzag(i32 r) {
-  // r in [-255, 255] reduced mod 256 as i8
+  // r in [-255, 255] reduced mod 256 as i8; computed in unsigned
+  // arithmetic ((v<<1) on a negative int is UB before C++20 — UBSan
+  // caught the signed form via tests/cpp/asan_parsers.cpp)
   i8 v = (i8)(u8)(r & 0xff);
-  return (u8)((v << 1) ^ (v >> 7));
+  u8 doubled = (u8)((u32)(u8)v << 1);  // (v << 1) mod 256
+  u8 sign = v < 0 ? 0xff : 0x00;       // arithmetic v >> 7
+  return (u8)(doubled ^ sign);
 }
 inline u8 unzigzag(u8 z) {
   i8 v = (i8)((z >> 1) ^ (-(i32)(z & 1)));
@@ -110,10 +114,17 @@ void encode_frame(const u8* cur, const u8* prev, u32 nbytes, bool key,
 void decode_frame(const SvcPacketView& v, const u8* prev, u8* cur) {
   for (u32 g = 0; g < v.ngroups; ++g) {
     u32 w = v.widths[g];
+    // corrupt-stream guards: widths > 8 would shift out of range and
+    // read past the group's 4*w payload; arbitrary super_off values
+    // would walk the read cursor outside the packet (the fuzz test
+    // drives both — tests/cpp/asan_parsers.cpp)
+    SCA_CHECK(w <= 8, "svc packet: group width > 8");
     // packed offset: supergroup base + local prefix
     u32 s = g / 128;
-    u32 off = v.super_off[s];
+    u64 off = v.super_off[s];
     for (u32 gg = s * 128; gg < g; ++gg) off += 4u * v.widths[gg];
+    SCA_CHECK(off + 4u * w <= v.packed_size,
+              "svc packet: packed region overrun");
     const u8* q = v.packed + off;
     u8 res[32];
     if (w == 0) {
@@ -158,13 +169,21 @@ SvcPacketView svc_parse_packet(const u8* pkt, size_t size) {
   std::memcpy(&v.nbytes, pkt + 8, 4);
   std::memcpy(&v.ngroups, pkt + 12, 4);
   std::memcpy(&v.nsuper, pkt + 16, 4);
-  size_t header = 20;
-  size_t widths_padded = (v.ngroups + 3) / 4 * 4;
-  SCA_CHECK(size >= header + v.nsuper * 4 + widths_padded,
-            "svc packet truncated");
+  // geometry invariants in 64-bit: a corrupt header must not be able to
+  // wrap the size check (e.g. nsuper=2^30 makes nsuper*4 tiny in u32) or
+  // claim more groups than nbytes implies
+  SCA_CHECK(v.ngroups == ((u64)v.nbytes + 31) / 32,
+            "svc packet: ngroups inconsistent with nbytes");
+  SCA_CHECK(v.nsuper == ((u64)v.ngroups + 127) / 128,
+            "svc packet: nsuper inconsistent with ngroups");
+  u64 header = 20;
+  u64 widths_padded = ((u64)v.ngroups + 3) / 4 * 4;
+  u64 need = header + (u64)v.nsuper * 4 + widths_padded;
+  SCA_CHECK((u64)size >= need, "svc packet truncated");
   v.super_off = reinterpret_cast<const u32*>(pkt + header);
-  v.widths = pkt + header + v.nsuper * 4;
+  v.widths = pkt + header + (size_t)v.nsuper * 4;
   v.packed = v.widths + widths_padded;
+  v.packed_size = size - (size_t)need;
   return v;
 }
 
@@ -230,6 +249,9 @@ void svc_decode_cpu(const u8* stream, size_t size, const VideoMetadata& vm,
               "svc stream range does not cover decode span");
     const u8* pkt = stream + (vm.sample_offsets[f] - stream_offset);
     SvcPacketView v = svc_parse_packet(pkt, vm.sample_sizes[f]);
+    // the output buffers are sized from the TABLE metadata; a corrupt
+    // packet must not be able to claim a larger frame and write past them
+    SCA_CHECK(v.nbytes == nbytes, "svc packet: frame size mismatch");
     SCA_CHECK(v.is_key || f == last + 1,
               "svc decode: non-contiguous delta frame");
     decode_frame(v, prev.data(), cur.data());

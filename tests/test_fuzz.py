@@ -46,13 +46,14 @@ def test_asan_parser_fuzz(tmp_path):
         paths[name] = str(p)
 
     src = os.path.join(REPO, "tests", "cpp", "asan_parsers.cpp")
-    mp4_tu = os.path.join(REPO, "scanner_amd", "csrc", "video", "mp4.cpp")
-    h264_tu = os.path.join(REPO, "scanner_amd", "csrc", "video", "h264.cpp")
+    vid = os.path.join(REPO, "scanner_amd", "csrc", "video")
+    tus = [os.path.join(vid, f) for f in ("mp4.cpp", "h264.cpp",
+                                          "svc_cpu.cpp")]
     binp = str(tmp_path / "asan_fuzz")
     r = subprocess.run(
         ["g++", "-O1", "-g", "-std=c++17",
          "-fsanitize=address,undefined", "-fno-sanitize-recover=all",
-         src, mp4_tu, h264_tu, "-o", binp],
+         src, *tus, "-o", binp],
         capture_output=True, text=True, timeout=300)
     if r.returncode != 0:
         pytest.fail(f"asan build failed:\n{r.stderr[-2000:]}")
