@@ -180,17 +180,6 @@ def main():
         sc.profile().write_trace(trace_path)
         log(f"[prof] chrome trace -> {trace_path}")
     elapsed = parallel.allreduce_max_time(elapsed, dist_device)
-    if distributed:
-        # Post-timing RCCL gather over xGMI: each rank ships the first
-        # result rows of its last step's output column to rank 0 (the
-        # reference round-trips results through shared storage instead).
-        sample = list(sp.NamedStream(
-            sc, f"bench_out_s{args.steps-1}").load(rows=range(4)))
-        gathered = parallel.gather_column(sample, dist_device)
-        if rank == 0:
-            log(f"gathered {len(gathered)} result rows over "
-                f"{dist.get_backend()}"
-                f"{' (RCCL/xGMI)' if dist.get_backend() == 'nccl' else ''}")
 
     total_frames = args.steps * n_frames * world
     fps = total_frames / elapsed
@@ -226,6 +215,21 @@ def main():
         print(json.dumps(result), flush=True)
 
     if distributed:
+        # Post-timing RCCL gather over xGMI: each rank ships the first
+        # result rows of its last step's output column to rank 0 (the
+        # reference round-trips results through shared storage instead).
+        # Runs AFTER the JSON line so a data-plane hiccup can never lose
+        # the measurement; failures are logged, not fatal.
+        try:
+            sample = list(sp.NamedStream(
+                sc, f"bench_out_s{args.steps-1}").load(rows=range(4)))
+            gathered = parallel.gather_column(sample, dist_device)
+            if rank == 0:
+                log(f"gathered {len(gathered)} result rows over "
+                    f"{dist.get_backend()}"
+                    f"{' (RCCL/xGMI)' if dist.get_backend() == 'nccl' else ''}")
+        except Exception as e:  # pragma: no cover
+            log(f"[rank {rank}] post-bench gather failed: {e}")
         dist.destroy_process_group()
     shutil.rmtree(tmp, ignore_errors=True)
 
