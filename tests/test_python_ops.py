@@ -33,6 +33,14 @@ def StencilSum(cols: Sequence[bytes]) -> bytes:
     return s.to_bytes(8, "little")
 
 
+@register_python_op(stencil=[-1, 0, 1], batch=3)
+def StencilBatchSum(cols: Sequence[Sequence[bytes]]) -> Sequence[bytes]:
+    """Depth-2 annotation: stencil-within-batch (reference: stencil+batch
+    python kernels, py_test.py:558-729 / op.py:389-535). cols[b][s]."""
+    return [sum(int.from_bytes(c, "little") for c in win).to_bytes(
+        8, "little") for win in cols]
+
+
 @register_python_op()
 class StatefulCounter(sp.Kernel):
     def __init__(self, config, **kwargs):
@@ -100,6 +108,23 @@ def test_python_op_stencil(sc):
            cache_mode=sp.CacheMode.Overwrite)
     vals = [int.from_bytes(b, "little") for b in out.load()]
     # REPEAT_EDGE clamping at both ends
+    expect = [min(max(i - 1, 0), n - 1) + i + min(i + 1, n - 1)
+              for i in range(n)]
+    assert vals == expect
+
+
+def test_python_op_stencil_within_batch(sc):
+    """Batched execution where each batch element receives its full
+    stencil window, across task boundaries (REPEAT_EDGE at the ends)."""
+    n = 11
+    tab = int_table(sc, "p4b", n)
+    col = sc.io.Input([tab])
+    summed = sc.ops.StencilBatchSum(cols=col)
+    out = sp.NamedStream(sc, "p4b_out")
+    # tiny packets force windows to span work/io packet boundaries
+    sc.run(sc.io.Output(summed, [out]), sp.PerfParams.manual(2, 4),
+           cache_mode=sp.CacheMode.Overwrite)
+    vals = [int.from_bytes(b, "little") for b in out.load()]
     expect = [min(max(i - 1, 0), n - 1) + i + min(i + 1, n - 1)
               for i in range(n)]
     assert vals == expect
