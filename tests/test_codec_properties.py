@@ -69,6 +69,37 @@ def test_svc_roundtrip_property(n, h, w, content, seed):
     np.testing.assert_array_equal(got, frames)
 
 
+@settings(max_examples=12, deadline=None, derandomize=True,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(gops=st.integers(1, 4), fpg=st.integers(1, 6),
+       fps=st.sampled_from([24.0, 30.0, 60.0]))
+def test_mp4_export_ingest_duality(gops, fpg, fps, tmp_path_factory):
+    """H.264 ingest -> mp4 export -> probe/re-ingest keeps the full index
+    (sample count, keyframe structure, dimensions) for every GOP shape.
+    Exercises the writer (mp4_write) against the parser (mp4_parse) as
+    inverse functions on the index level."""
+    from test_video_ingest import make_annexb
+
+    sc = client()
+    tmp = tmp_path_factory.mktemp("mp4prop")
+    stream, _, keyframes = make_annexb(gops=gops, frames_per_gop=fpg)
+    n = gops * fpg
+    p = tmp / "clip.h264"
+    p.write_bytes(stream)
+    name = fresh("mp4prop")
+    r = sc.ingest_video_file(str(p), name)
+    assert r["num_frames"] == n
+    out_path = str(tmp / "out.mp4")
+    sp.NamedVideoStream(sc, name).save_mp4(out_path, fps=fps)
+    from scanner_amd import _core
+    t = _core.mp4_probe(open(out_path, "rb").read())
+    assert len(t["sample_offsets"]) == n
+    assert t["keyframe_indices"] == keyframes
+    r2 = sc.ingest_video_file(out_path, fresh("mp4prop_re"))
+    assert r2["num_frames"] == n
+    assert (r2["width"], r2["height"]) == (r["width"], r["height"])
+
+
 @settings(max_examples=15, deadline=None, derandomize=True,
           suppress_health_check=[HealthCheck.too_slow])
 @given(rows=st.lists(st.integers(0, 19), min_size=1, max_size=12),
