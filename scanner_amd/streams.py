@@ -89,8 +89,18 @@ class StreamsGenerator:
             for iv in intervals])
 
     def Gather(self, col, rows):
-        """rows: per-stream list of row-index lists."""
-        return self._sample(col, "Gather", [{"rows": list(r)} for r in rows])
+        """rows: per-stream list of row-index lists (each >= 0; unsorted
+        and duplicated rows are allowed)."""
+        rows = [list(r) for r in rows]
+        for r in rows:
+            for x in r:
+                if x < 0:
+                    # -1 is the ENGINE's null-element sentinel (RepeatNull
+                    # gaps); letting it through Gather would silently turn
+                    # a user indexing bug into null output rows
+                    raise ScannerException(
+                        f"Gather rows must be >= 0, got {x}")
+        return self._sample(col, "Gather", [{"rows": r} for r in rows])
 
     def Repeat(self, col, spacings):
         return self._space(col, "Repeat", [

@@ -1,0 +1,83 @@
+"""Negative-path coverage: invalid DAGs, bad sampling args, missing
+tables, kernel input validation — every misuse must raise a clean Python
+exception (reference: graph validation dag_analysis.cpp:43 + client-side
+checks in scannerpy client.py run()); silent wrong output or a crash is
+the failure mode these tests guard against."""
+import pytest
+
+import scanner_amd as sp
+
+
+def int_table(sc, name, n, width=8):
+    return sc.new_table(name, ["col"],
+                        [[int(i).to_bytes(width, "little")]
+                         for i in range(n)])
+
+
+def test_unknown_op(sc):
+    tab = int_table(sc, "ep1", 4)
+    with pytest.raises(Exception, match="unknown op"):
+        sc.ops.NoSuchOp(col=sc.io.Input([tab]))
+
+
+def test_missing_input_column(sc):
+    tab = int_table(sc, "ep2", 4)
+    with pytest.raises(sp.ScannerException, match="missing input column"):
+        sc.ops.TestIncrement(bogus=sc.io.Input([tab]))
+
+
+def test_out_of_domain_range(sc):
+    tab = int_table(sc, "ep3", 10)
+    col = sc.io.Input([tab])
+    r = sc.streams.Range(col, [(5, 100)])
+    out = sp.NamedStream(sc, "ep3_out")
+    with pytest.raises(Exception, match="out of op domain"):
+        sc.run(sc.io.Output(r, [out]), sp.PerfParams.manual(4, 8),
+               cache_mode=sp.CacheMode.Overwrite)
+
+
+def test_negative_gather_rejected(sc):
+    tab = int_table(sc, "ep4", 10)
+    col = sc.io.Input([tab])
+    with pytest.raises(sp.ScannerException, match="Gather rows"):
+        sc.streams.Gather(col, [[-1, 2]])
+
+
+def test_load_missing_table(sc):
+    with pytest.raises(Exception, match="no table"):
+        list(sp.NamedStream(sc, "ep_nope").load())
+
+
+def test_stream_count_mismatch(sc):
+    a = int_table(sc, "ep5a", 4)
+    b = int_table(sc, "ep5b", 4)
+    col = sc.io.Input([a, b])
+    out = sp.NamedStream(sc, "ep5_out")  # 1 output for 2 input streams
+    with pytest.raises(sp.ScannerException, match="stream count mismatch"):
+        sc.run(sc.io.Output(sc.ops.TestIncrement(ignore=col), [out]),
+               sp.PerfParams.manual(4, 8),
+               cache_mode=sp.CacheMode.Overwrite)
+
+
+def test_kernel_input_validation_fails_job(sc):
+    """A kernel's own input check (TestIncrement wants 8-byte rows)
+    surfaces as a job failure with the kernel's message, not a crash."""
+    tab = int_table(sc, "ep6", 4, width=2)
+    col = sc.io.Input([tab])
+    out = sp.NamedStream(sc, "ep6_out")
+    with pytest.raises(Exception, match="i64 input"):
+        sc.run(sc.io.Output(sc.ops.TestIncrement(ignore=col), [out]),
+               sp.PerfParams.manual(4, 8),
+               cache_mode=sp.CacheMode.Overwrite)
+
+
+def test_zero_row_job(sc):
+    """An empty sampling result is a valid job: zero output rows, no
+    error (the engine must handle tasks with nothing to produce)."""
+    tab = int_table(sc, "ep7", 10)
+    col = sc.io.Input([tab])
+    r = sc.streams.Range(col, [(3, 3)])
+    out = sp.NamedStream(sc, "ep7_out")
+    sc.run(sc.io.Output(r, [out]), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite)
+    assert list(out.load()) == []
