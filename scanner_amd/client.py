@@ -242,6 +242,21 @@ class Client:
         perf = perf_params or PerfParams.estimate()
         graph_bytes, jobs, out_streams, n_jobs = self._assemble(outputs)
 
+        # An output stream naming a table that is also an INPUT of the
+        # same job would be deleted (Overwrite) or half-written before the
+        # job reads it — destroying the user's data with a confusing
+        # downstream error. Reject before touching storage.
+        for j, js in enumerate(jobs):
+            s = out_streams[j]
+            if s is None:
+                continue
+            in_tables = {src["table"] for src in js["sources"].values()
+                         if "table" in src}
+            if s.name in in_tables:
+                raise ScannerException(
+                    f"output stream '{s.name}' is also an input table of "
+                    "the same job; write to a different table")
+
         # CacheMode handling (parity: client.py:1386-1432)
         keep = []
         for j, js in enumerate(jobs):

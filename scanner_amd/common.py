@@ -57,6 +57,17 @@ class PerfParams:
                  profiler_level=1, num_load_workers=0, span_cache=0):
         self.work_packet_size = int(work_packet_size)
         self.io_packet_size = int(io_packet_size)
+        if self.work_packet_size < 1 or self.io_packet_size < 1:
+            # 0 used to reach the engine and die as std::bad_alloc deep in
+            # task partitioning; fail at construction with a real message
+            raise ScannerException(
+                "work_packet_size and io_packet_size must be >= 1 "
+                f"(got {self.work_packet_size}, {self.io_packet_size})")
+        if self.io_packet_size < self.work_packet_size:
+            raise ScannerException(
+                "io_packet_size must be >= work_packet_size "
+                f"(got io={self.io_packet_size} < "
+                f"work={self.work_packet_size})")
         self.cpu_pool = int(cpu_pool)
         self.gpu_pool = int(gpu_pool)
         self.pipeline_instances_per_node = pipeline_instances_per_node

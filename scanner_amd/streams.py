@@ -33,6 +33,15 @@ class StreamsGenerator:
     def __init__(self, client):
         self._client = client
 
+    @staticmethod
+    def _positive(v, what):
+        """The engine's samplers clamp stride/spacing to 1 defensively; at
+        the API a non-positive value is a caller bug — reject it instead
+        of silently reinterpreting (same policy as negative Gather rows)."""
+        if v < 1:
+            raise ScannerException(f"{what} must be >= 1, got {v}")
+        return v
+
     def _sample(self, col, kind, per_stream_args):
         col = _as_column(col)
         name = _frame_variant(col, "Sample")
@@ -57,7 +66,8 @@ class StreamsGenerator:
     def Stride(self, col, strides):
         """strides: int or per-stream list of ints."""
         return self._sample(col, "Strided", [
-            {"stride": s} for s in self._bcast(strides)])
+            {"stride": self._positive(s, "stride")}
+            for s in self._bcast(strides)])
 
     def Range(self, col, ranges):
         """ranges: (start, end) or per-stream list of (start, end) or
@@ -77,13 +87,14 @@ class StreamsGenerator:
     def StridedRange(self, col, ranges):
         """ranges: per-stream list of (start, end, stride)."""
         return self._sample(col, "StridedRange", [
-            {"stride": r[2], "starts": [r[0]], "ends": [r[1]]}
+            {"stride": self._positive(r[2], "stride"),
+             "starts": [r[0]], "ends": [r[1]]}
             for r in self._bcast(ranges, tup=True)])
 
     def StridedRanges(self, col, intervals=None, stride=1):
         """intervals: per-stream list of [(s,e),...], one stride."""
         return self._sample(col, "StridedRanges", [
-            {"stride": stride,
+            {"stride": self._positive(stride, "stride"),
              "starts": [s for s, _ in iv],
              "ends": [e for _, e in iv]}
             for iv in intervals])
@@ -104,11 +115,13 @@ class StreamsGenerator:
 
     def Repeat(self, col, spacings):
         return self._space(col, "Repeat", [
-            {"spacing": s} for s in self._bcast(spacings)])
+            {"spacing": self._positive(s, "spacing")}
+            for s in self._bcast(spacings)])
 
     def RepeatNull(self, col, spacings):
         return self._space(col, "RepeatNull", [
-            {"spacing": s} for s in self._bcast(spacings)])
+            {"spacing": self._positive(s, "spacing")}
+            for s in self._bcast(spacings)])
 
     def Slice(self, col, partitions):
         """partitions: per-stream partitioner spec from sc.partitioner.*"""
@@ -142,13 +155,19 @@ class StreamsGenerator:
 class PartitionerGenerator:
     """sc.partitioner.* (parity: python/scannerpy/partitioner.py)."""
 
+    @staticmethod
+    def _group(v):
+        if int(v) < 1:
+            raise ScannerException(f"slice group size must be >= 1, got {v}")
+        return int(v)
+
     def all(self, group_size=None):
         if group_size is None:
             return {"kind": "All"}
-        return {"kind": "Strided", "stride": int(group_size)}
+        return {"kind": "Strided", "stride": self._group(group_size)}
 
     def strided(self, group_size):
-        return {"kind": "Strided", "stride": int(group_size)}
+        return {"kind": "Strided", "stride": self._group(group_size)}
 
     def ranges(self, intervals):
         return {"kind": "Ranges",

@@ -71,6 +71,42 @@ def test_kernel_input_validation_fails_job(sc):
                cache_mode=sp.CacheMode.Overwrite)
 
 
+def test_nonpositive_sampling_args_rejected(sc):
+    tab = int_table(sc, "ep8", 10)
+    for make in (
+        lambda c: sc.streams.Stride(c, [0]),
+        lambda c: sc.streams.Stride(c, [-2]),
+        lambda c: sc.streams.StridedRange(c, [(0, 5, 0)]),
+        lambda c: sc.streams.StridedRanges(c, [[(0, 5)]], stride=0),
+        lambda c: sc.streams.Repeat(c, [0]),
+        lambda c: sc.streams.RepeatNull(c, [-1]),
+        lambda c: sc.streams.Slice(c, sc.partitioner.strided(0)),
+    ):
+        with pytest.raises(sp.ScannerException, match=">= 1"):
+            make(sc.io.Input([tab]))
+
+
+def test_perf_params_validation():
+    with pytest.raises(sp.ScannerException, match=">= 1"):
+        sp.PerfParams.manual(0, 0)
+    with pytest.raises(sp.ScannerException, match="io_packet_size"):
+        sp.PerfParams.manual(8, 4)
+
+
+def test_output_table_is_input_rejected(sc):
+    """Writing a job's output over one of its own input tables would
+    destroy the input before it is read (Overwrite deletes first) — must
+    be rejected before any storage is touched, with the input intact."""
+    tab = int_table(sc, "ep9", 10)
+    col = sc.io.Input([tab])
+    out = sp.NamedStream(sc, "ep9")  # same table!
+    with pytest.raises(sp.ScannerException, match="also an input"):
+        sc.run(sc.io.Output(sc.ops.TestIncrement(ignore=col), [out]),
+               sp.PerfParams.manual(4, 8),
+               cache_mode=sp.CacheMode.Overwrite)
+    assert len(list(sp.NamedStream(sc, "ep9").load())) == 10  # intact
+
+
 def test_zero_row_job(sc):
     """An empty sampling result is a valid job: zero output rows, no
     error (the engine must handle tasks with nothing to produce)."""
