@@ -168,6 +168,15 @@ def register_python_op(name=None, device_type=DeviceType.CPU, batch=0,
     def deco(fn_or_cls):
         from . import _core
         opname = name or fn_or_cls.__name__
+        # Refuse to shadow a first-party C++ op (a python op silently
+        # replacing Histogram etc. breaks every pipeline using it);
+        # RE-registering a python op of the same name stays allowed —
+        # notebook-style iteration depends on it.
+        if (opname in _core.registered_ops()
+                and opname not in _PY_OP_PICKLES):
+            raise ScannerException(
+                f"op '{opname}' is already a registered C++ op; "
+                "pick a different name")
         is_cls = inspect.isclass(fn_or_cls)
         target = fn_or_cls.execute if is_cls else fn_or_cls
         sig = inspect.signature(target)

@@ -105,7 +105,16 @@ class NamedVideoStream(NamedStream):
 
     def _ingest(self, frames, codec, io_packet_size):
         from . import _core
-        frames = np.ascontiguousarray(frames, dtype=np.uint8)
+        frames = np.asarray(frames)
+        if frames.dtype != np.uint8:
+            # an implicit astype here silently mangled float frames into
+            # u8 garbage; video streams are byte-typed — make the caller
+            # quantize explicitly
+            raise ScannerException(
+                f"video frames must be uint8, got {frames.dtype} "
+                "(convert explicitly, e.g. (x*255).clip(0,255)"
+                ".astype(np.uint8))")
+        frames = np.ascontiguousarray(frames)
         if frames.ndim != 4:
             raise ScannerException("frames must be [N,H,W,C] u8")
         _core.write_video_table(self._sc._db, self.name, self._column or

@@ -107,6 +107,39 @@ def test_output_table_is_input_rejected(sc):
     assert len(list(sp.NamedStream(sc, "ep9").load())) == 10  # intact
 
 
+def test_float_frames_rejected(sc):
+    """Implicit astype used to mangle float frames into u8 garbage
+    silently; now an explicit error."""
+    import numpy as np
+    with pytest.raises(sp.ScannerException, match="uint8"):
+        sp.NamedVideoStream(sc, "epf",
+                            frames=np.zeros((2, 8, 8, 3), np.float32),
+                            codec="raw")
+
+
+def test_python_op_cannot_shadow_builtin():
+    with pytest.raises(sp.ScannerException, match="already a registered"):
+        @sp.register_python_op(name="Histogram")
+        def Histogram(col: bytes) -> bytes:  # pragma: no cover
+            return col
+
+
+def test_python_op_exception_message_preserved(sc):
+    """A python kernel's exception surfaces in the job failure with its
+    type, message, and traceback line."""
+    @sp.register_python_op(name="BoomEp")
+    def BoomEp(col: bytes) -> bytes:
+        raise ValueError("domain-specific detail 12345")
+
+    tab = int_table(sc, "ep10", 4)
+    col = sc.io.Input([tab])
+    out = sp.NamedStream(sc, "ep10_out")
+    with pytest.raises(Exception, match="domain-specific detail 12345"):
+        sc.run(sc.io.Output(sc.ops.BoomEp(col=col), [out]),
+               sp.PerfParams.manual(4, 8),
+               cache_mode=sp.CacheMode.Overwrite)
+
+
 def test_zero_row_job(sc):
     """An empty sampling result is a valid job: zero output rows, no
     error (the engine must handle tasks with nothing to produce)."""
