@@ -11,6 +11,7 @@ model. Fixed derandomized profile so CI runs are reproducible.
 import os
 import sys
 import tempfile
+from typing import Sequence
 
 from hypothesis import HealthCheck, given, settings, strategies as st
 
@@ -121,6 +122,81 @@ def test_sampler_matches_model(spec):
     got = [None if b is None else int.from_bytes(b, "little")
            for b in out.load()]
     assert got == model_rows(kind, args, N), (kind, args)
+
+
+@sp.register_python_op(name="PropDouble")
+def PropDouble(col: bytes) -> bytes:
+    return (2 * int.from_bytes(col, "little")).to_bytes(8, "little")
+
+
+@sp.register_python_op(name="PropStencilSum", stencil=[-1, 0, 1])
+def PropStencilSum(cols: Sequence[bytes]) -> bytes:
+    s = sum(int.from_bytes(c, "little") for c in cols)
+    return s.to_bytes(8, "little")
+
+
+@st.composite
+def pipeline_spec(draw):
+    """1-3 stages over a running stream length; samplers change the
+    length, ops keep it."""
+    stages = []
+    length = N
+    for _ in range(draw(st.integers(1, 3))):
+        kind = draw(st.sampled_from(
+            ["stride", "range", "gather", "double", "stencilsum"]))
+        if kind == "stride":
+            stages.append(("stride", draw(st.integers(1, 5))))
+            length = len(range(0, length, stages[-1][1]))
+        elif kind == "range":
+            a = draw(st.integers(0, length - 1))
+            b = draw(st.integers(a + 1, length))
+            stages.append(("range", (a, b)))
+            length = b - a
+        elif kind == "gather":
+            rows = draw(st.lists(st.integers(0, length - 1),
+                                 min_size=1, max_size=15))
+            stages.append(("gather", rows))
+            length = len(rows)
+        else:
+            stages.append((kind, None))
+    return stages
+
+
+@settings(max_examples=25, deadline=None, derandomize=True,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(spec=pipeline_spec())
+def test_composed_pipeline_matches_model(spec):
+    """Chains of samplers and (stateless / stencil) ops: the sampled row
+    spaces compose, stencil windows apply in the RESAMPLED space with
+    REPEAT_EDGE clamping at its bounds (reference semantics:
+    derive_stencil_requirements walks remaps then stencils,
+    dag_analysis.cpp:1328+)."""
+    sc, tab = client()
+    col = sc.io.Input([tab])
+    values = list(range(N))  # model: value per row at this stage
+    for kind, args in spec:
+        if kind == "stride":
+            col = sc.streams.Stride(col, [args])
+            values = values[::args]
+        elif kind == "range":
+            col = sc.streams.Range(col, [args])
+            values = values[args[0]:args[1]]
+        elif kind == "gather":
+            col = sc.streams.Gather(col, [args])
+            values = [values[r] for r in args]
+        elif kind == "double":
+            col = sc.ops.PropDouble(col=col)
+            values = [2 * v for v in values]
+        elif kind == "stencilsum":
+            col = sc.ops.PropStencilSum(cols=col)
+            m = len(values)
+            values = [values[max(i - 1, 0)] + values[i] +
+                      values[min(i + 1, m - 1)] for i in range(m)]
+    out = sp.NamedStream(sc, "prop_chain_out")
+    sc.run(sc.io.Output(col, [out]), sp.PerfParams.manual(4, 8),
+           cache_mode=sp.CacheMode.Overwrite)
+    got = [int.from_bytes(b, "little") for b in out.load()]
+    assert got == values, spec
 
 
 @settings(max_examples=15, deadline=None, derandomize=True,
