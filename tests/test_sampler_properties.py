@@ -199,6 +199,58 @@ def test_composed_pipeline_matches_model(spec):
     assert got == values, spec
 
 
+def _make_counter(name, warmup):
+    @sp.register_python_op(name=name, bounded_state=True, warmup=warmup)
+    class _Ctr(sp.Kernel):
+        def __init__(self, config, **kw):
+            self.c = 0
+
+        def reset(self):
+            self.c = 0
+
+        def execute(self, col: bytes) -> bytes:
+            self.c += 1
+            return (int.from_bytes(col, "little") +
+                    self.c).to_bytes(8, "little")
+
+
+for _w in (0, 1, 2, 3):
+    _make_counter(f"PropCtrW{_w}", _w)
+
+
+@settings(max_examples=25, deadline=None, derandomize=True,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(warmup=st.integers(0, 3), io=st.integers(1, 8),
+       work=st.integers(1, 8), n=st.integers(1, N))
+def test_bounded_state_warmup_matches_model(warmup, io, work, n):
+    """Bounded-state warmup across task boundaries for every (warmup,
+    io_packet, work_packet, stream length): at each task start the kernel
+    resets and re-feeds min(warmup, start) rows of history, so the
+    counter's contribution to row r is r - max(0, task_start - warmup) + 1
+    (reference: derive_stencil_requirements warmup handling,
+    dag_analysis.cpp:1328+; verified to match the engine for the pinned
+    example in this file's history)."""
+    work = min(work, io)
+    sc, _ = client()
+    tab_name = f"prop_bs_{n}"
+    try:
+        tab = sc.new_table(
+            tab_name, ["col"],
+            [[int(i).to_bytes(8, "little")] for i in range(n)])
+    except Exception:
+        from scanner_amd.storage import NamedStream as _NS
+        tab = _NS(sc, tab_name)  # already created by an earlier example
+    col = sc.io.Input([tab])
+    ctr = getattr(sc.ops, f"PropCtrW{warmup}")(col=col)
+    out = sp.NamedStream(sc, "prop_bs_out")
+    sc.run(sc.io.Output(ctr, [out]), sp.PerfParams.manual(work, io),
+           cache_mode=sp.CacheMode.Overwrite)
+    got = [int.from_bytes(b, "little") for b in out.load()]
+    expect = [r + (r - max(0, (r // io) * io - warmup) + 1)
+              for r in range(n)]
+    assert got == expect, (warmup, io, work, n)
+
+
 @settings(max_examples=15, deadline=None, derandomize=True,
           suppress_health_check=[HealthCheck.too_slow])
 @given(group_size=st.integers(1, N + 4))
