@@ -38,14 +38,11 @@ class Client:
         self._bucket = bucket or ""
         if self._storage_type == "posix":
             os.makedirs(self._db_path, exist_ok=True)
-            self._db = _core.Database(self._db_path)
-        else:
-            # object-store backend (S3 semantics emulated over a local
-            # bucket dir; config.py [storage] type="s3" bucket="...")
-            self._db = _core.Database(self._db_path, self._storage_type,
-                                      self._bucket)
+        self._db = self._open_db()
         if recover:
             self._db.recover()
+        # (metadata reloads after cluster runs go through _open_db so the
+        # object-store backend survives them — config.py [storage])
         self.ops = OpGenerator(self)
         self.streams = StreamsGenerator(self)
         self.partitioner = PartitionerGenerator()
@@ -102,6 +99,14 @@ class Client:
 
     def sequence(self, name):
         return NamedStream(self, name)
+
+    def _open_db(self):
+        if self._storage_type == "posix":
+            return _core.Database(self._db_path)
+        # object-store backend (S3 semantics emulated over a local bucket
+        # dir; config.py [storage] type="s3" bucket="...")
+        return _core.Database(self._db_path, self._storage_type,
+                              self._bucket)
 
     def summarize(self):
         lines = []
@@ -290,7 +295,7 @@ class Client:
                                            show_progress=show_progress)
             # reload metadata: output tables were created/committed by the
             # master process (shared storage)
-            self._db = _core.Database(self._db_path)
+            self._db = self._open_db()
             if result is not None:
                 self._last_profilers = result._profilers
             return result

@@ -53,9 +53,10 @@ class _BulkJob:
 class MasterServer:
     def __init__(self, db_path, addr="127.0.0.1:0",
                  no_workers_timeout=30.0, task_timeout=DEFAULT_TASK_TIMEOUT,
-                 checkpoint_frequency=1):
+                 checkpoint_frequency=1, storage_type="posix", bucket=""):
         self._db_path = db_path
-        self._db = _core.Database(db_path)
+        self._storage = (storage_type or "posix", bucket or "")
+        self._db = self._open_db()
         self._db.recover()
         self._workers = {}
         self._next_worker_id = 0
@@ -91,6 +92,14 @@ class MasterServer:
         self._monitor = threading.Thread(target=self._monitor_loop,
                                          daemon=True)
         self._monitor.start()
+
+    def _open_db(self):
+        # storage config threads to every Database open so a cluster on an
+        # object store (kube + S3 semantics) works the same as posix
+        st, bucket = self._storage
+        if st == "posix":
+            return _core.Database(self._db_path)
+        return _core.Database(self._db_path, st, bucket)
 
     # ---- rpc handlers ----
 
@@ -150,7 +159,7 @@ class MasterServer:
         # fresh Database handle: the client process may have ingested
         # tables since this one was opened (shared storage); table
         # creation itself is serialized by the db file lock
-        db = _core.Database(self._db_path)
+        db = self._open_db()
         job.ex = _core.LocalExecutor(
             db, job.graph, msgpack.packb(job.jobs), perf_m, [])
         job.ex.prepare(True)
@@ -452,10 +461,14 @@ def main():
     ap.add_argument("--addr", default="0.0.0.0:5001")
     ap.add_argument("--no-workers-timeout", type=float, default=30.0)
     ap.add_argument("--task-timeout", type=float, default=600.0)
+    ap.add_argument("--storage-type", default="posix",
+                    help="posix | s3 (object store; see config.py)")
+    ap.add_argument("--bucket", default="")
     args = ap.parse_args()
     m = MasterServer(args.db_path, args.addr,
                      no_workers_timeout=args.no_workers_timeout,
-                     task_timeout=args.task_timeout)
+                     task_timeout=args.task_timeout,
+                     storage_type=args.storage_type, bucket=args.bucket)
     print(f"master listening on {m.addr} (db: {args.db_path})", flush=True)
     m._server.wait()
 

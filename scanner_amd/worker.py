@@ -18,9 +18,11 @@ WATCHDOG_TIMEOUT = 60.0
 
 class WorkerServer:
     def __init__(self, master_addr, db_path, addr="127.0.0.1:0",
-                 pipeline_instances=1, gpu_ids=None, watchdog=True):
+                 pipeline_instances=1, gpu_ids=None, watchdog=True,
+                 storage_type="posix", bucket=""):
         self._db_path = db_path
-        self._db = _core.Database(db_path)
+        self._storage = (storage_type or "posix", bucket or "")
+        self._db = self._open_db()
         self._master = RpcClient(master_addr)
         self._instances = pipeline_instances
         self._gpu_ids = gpu_ids or []
@@ -68,6 +70,12 @@ class WorkerServer:
         self._last_poke = time.time()
         return {"ok": True}
 
+    def _open_db(self):
+        st, bucket = self._storage
+        if st == "posix":
+            return _core.Database(self._db_path)
+        return _core.Database(self._db_path, st, bucket)
+
     def _poke(self, req):
         self._last_poke = time.time()
         return {"ok": True}
@@ -112,7 +120,7 @@ class WorkerServer:
             perf = dict(params["perf"])
             perf["pipeline_instances"] = self._instances
             # reload metadata (ingests + master-created output tables)
-            self._db = _core.Database(self._db_path)
+            self._db = self._open_db()
             ex = _core.LocalExecutor(
                 self._db, params["graph"], msgpack.packb(params["jobs"]),
                 perf, self._gpu_ids)
@@ -227,6 +235,9 @@ def main():
                     help="comma-separated; default = all visible GPUs "
                          "(reference: default_machine_params)")
     ap.add_argument("--no-watchdog", action="store_true")
+    ap.add_argument("--storage-type", default="posix",
+                    help="posix | s3 (object store; see config.py)")
+    ap.add_argument("--bucket", default="")
     args = ap.parse_args()
     if args.gpu_ids is None:
         gpu_ids = list(range(_core.gpu_device_count()))
@@ -234,7 +245,8 @@ def main():
         gpu_ids = [int(x) for x in args.gpu_ids.split(",") if x != ""]
     start_worker(args.master, args.db_path, args.addr, block=True,
                  pipeline_instances=args.instances, gpu_ids=gpu_ids,
-                 watchdog=not args.no_watchdog)
+                 watchdog=not args.no_watchdog,
+                 storage_type=args.storage_type, bucket=args.bucket)
 
 
 if __name__ == "__main__":

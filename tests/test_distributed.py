@@ -368,3 +368,50 @@ def test_cluster_bootstrap_localhost(tmp_path):
             cluster.stop()
     finally:
         os.environ["PYTHONPATH"] = old_pp
+
+
+def test_distributed_on_object_store(tmp_path, monkeypatch):
+    """Full master + worker-subprocess job over the S3-semantics object
+    store: every table/metadata access on master, worker, and client goes
+    through the flat-keyspace backend (the kube + cloud-storage deployment
+    shape; profile files remain posix debug artifacts). The worker runs as
+    a REAL OS process with --storage-type s3."""
+    monkeypatch.chdir(tmp_path)  # posix profile artifacts land here
+    db = "db"
+    bucket = str(tmp_path / "bucket")
+    master = MasterServer(db, storage_type="s3", bucket=bucket)
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    wp = subprocess.Popen(
+        [sys.executable, "-m", "scanner_amd.worker", "--master",
+         master.addr, "--db-path", db, "--no-watchdog",
+         "--storage-type", "s3", "--bucket", bucket],
+        env=env, start_new_session=True)
+    try:
+        sc = sp.Client(db_path=db, master=master.addr,
+                       storage_type="s3", bucket=bucket)
+        vid = make_video(n=10, seed=3)
+        video = sp.NamedVideoStream(sc, "os_d", frames=vid, codec="svc")
+        frame = sc.io.Input([video])
+        hist = sc.ops.Histogram(frame=frame)
+        out = sp.NamedStream(sc, "os_d_hist")
+        sc.run(sc.io.Output(hist, [out]), sp.PerfParams.manual(2, 4),
+               cache_mode=sp.CacheMode.Overwrite)
+        rows = list(out.load())
+        assert len(rows) == 10
+        for r, blob in enumerate(rows):
+            got = np.frombuffer(blob, np.uint32).reshape(3, 256)
+            np.testing.assert_array_equal(got, ref_histogram(vid[r]))
+        # table data + metadata live in the flat bucket; the posix db path
+        # holds at most jobs/ profile debug artifacts
+        assert os.listdir(bucket)
+        posix_db = os.path.join(os.getcwd(), db)
+        if os.path.exists(posix_db):
+            assert set(os.listdir(posix_db)) <= {"jobs"}, \
+                os.listdir(posix_db)
+        sc.shutdown()
+    finally:
+        if wp.poll() is None:
+            os.killpg(wp.pid, signal.SIGKILL)
+        wp.wait()
+        master.shutdown()
