@@ -75,17 +75,21 @@ class Cluster:
 def bootstrap_cluster(db_path, master_host, worker_hosts, master_port=5001,
                       instances_per_worker=1, python=None, ssh_cmd=None,
                       log_dir="/tmp", connect_timeout=30.0,
-                      master_advertise=None):
+                      master_advertise=None, storage_type="posix",
+                      bucket=""):
     """Start master + workers over ssh; returns a Cluster handle whose
     master_addr plugs straight into Client(master=...).
 
     db_path must be a path valid on every host (shared filesystem), the
-    same contract as the reference's shared storage."""
+    same contract as the reference's shared storage. storage_type="s3"
+    (with bucket) runs the whole cluster on the object-store backend
+    instead — then db_path is a key prefix, not a filesystem path."""
     python = python or sys.executable
     master_addr = f"{master_advertise or master_host}:{master_port}"
     master = _launch(
         ssh_cmd, master_host, "scanner_amd.master",
-        ["--db-path", db_path, "--addr", f"0.0.0.0:{master_port}"],
+        ["--db-path", db_path, "--addr", f"0.0.0.0:{master_port}",
+         "--storage-type", storage_type, "--bucket", bucket],
         f"{log_dir}/scanner_master.log", python)
 
     # wait for the master to answer before launching workers
@@ -107,7 +111,8 @@ def bootstrap_cluster(db_path, master_host, worker_hosts, master_port=5001,
             workers.append(_launch(
                 ssh_cmd, host, "scanner_amd.worker",
                 ["--master", master_addr, "--db-path", db_path,
-                 "--instances", str(instances_per_worker)],
+                 "--instances", str(instances_per_worker),
+                 "--storage-type", storage_type, "--bucket", bucket],
                 f"{log_dir}/scanner_worker_{i}.log", python))
         # wait until every worker registered
         t0 = time.time()

@@ -32,7 +32,7 @@ class ClusterConfig:
                  master=None, worker=None, image=DEFAULT_IMAGE,
                  shared_claim="scanner-shared", db_path="/shared/db",
                  master_port=5001, namespace="default",
-                 autoscale_max=0):
+                 autoscale_max=0, storage_type="posix", bucket=""):
         self.id = id
         self.num_workers = num_workers
         self.master = master or MachineConfig(gpus_per_node=0)
@@ -44,6 +44,10 @@ class ClusterConfig:
         self.namespace = namespace
         # >0: also emit an HPA capping worker replicas
         self.autoscale_max = autoscale_max
+        # "s3": cluster runs on the object-store backend (bucket required;
+        # db_path becomes a key prefix); "posix": the RWX shared volume
+        self.storage_type = storage_type
+        self.bucket = bucket
 
 
 def _resources(mc):
@@ -70,7 +74,9 @@ def master_manifests(cfg):
                         "image": cfg.image,
                         "command": ["python", "-m", "scanner_amd.master",
                                     "--db-path", cfg.db_path, "--addr",
-                                    f"0.0.0.0:{cfg.master_port}"],
+                                    f"0.0.0.0:{cfg.master_port}",
+                                    "--storage-type", cfg.storage_type,
+                                    "--bucket", cfg.bucket],
                         "ports": [{"containerPort": cfg.master_port}],
                         "resources": _resources(cfg.master),
                         "volumeMounts": [{"name": "shared",
@@ -115,7 +121,9 @@ def worker_manifests(cfg):
                                     "--db-path", cfg.db_path,
                                     "--instances",
                                     str(max(1,
-                                            cfg.worker.gpus_per_node))],
+                                            cfg.worker.gpus_per_node)),
+                                    "--storage-type", cfg.storage_type,
+                                    "--bucket", cfg.bucket],
                         "resources": _resources(cfg.worker),
                         "volumeMounts": [{"name": "shared",
                                           "mountPath": "/shared"}],
