@@ -415,3 +415,32 @@ def test_distributed_on_object_store(tmp_path, monkeypatch):
             os.killpg(wp.pid, signal.SIGKILL)
         wp.wait()
         master.shutdown()
+
+
+def test_worker_multiple_pipeline_instances(tmp_path):
+    """One worker running 3 pipeline instances (the 8-GPU node shape:
+    --instances = gpus_per_node, each instance a full engine replica) —
+    tasks fan out across instances and results stay exact."""
+    db = _mk_db(tmp_path)
+    master = MasterServer(db)
+    wp = spawn_worker_proc(master.addr, db, instances=3)
+    try:
+        sc = sp.Client(db_path=db, master=master.addr)
+        vid = make_video(n=24, seed=9)
+        video = sp.NamedVideoStream(sc, "mi", frames=vid, codec="svc")
+        frame = sc.io.Input([video])
+        hist = sc.ops.Histogram(frame=frame)
+        out = sp.NamedStream(sc, "mi_hist")
+        sc.run(sc.io.Output(hist, [out]), sp.PerfParams.manual(2, 4),
+               cache_mode=sp.CacheMode.Overwrite)
+        rows = list(out.load())
+        assert len(rows) == 24
+        for r, blob in enumerate(rows):
+            got = np.frombuffer(blob, np.uint32).reshape(3, 256)
+            np.testing.assert_array_equal(got, ref_histogram(vid[r]))
+        sc.shutdown()
+    finally:
+        if wp.poll() is None:
+            os.killpg(wp.pid, signal.SIGKILL)
+        wp.wait()
+        master.shutdown()
