@@ -22,11 +22,6 @@ def _frame_variant(col, base):
     return base + ("Frame" if col.type == ColumnType.Video else "")
 
 
-def _per_stream(args, what):
-    """Normalize per-stream arg spec: a single value broadcasts."""
-    return args
-
-
 class StreamsGenerator:
     """sc.streams.* (parity: StreamsGenerator streams.py)."""
 
