@@ -116,7 +116,11 @@ class BlurKernelCPU : public BatchedKernel {
  public:
   explicit BlurKernelCPU(const KernelConfig& cfg) : BatchedKernel(cfg) {
     auto a = mp::decode(cfg.args);
-    radius_ = (i32)a.get_int("kernel_size", 3) / 2;
+    i32 ks = (i32)a.get_int("kernel_size", 3);
+    // ks < 1 made radius negative -> empty tap loop -> division by zero
+    // (UB; observed as a wedged pipeline instance)
+    SCA_CHECK(ks >= 1, "Blur kernel_size must be >= 1");
+    radius_ = ks / 2;
   }
   void execute_batch(const BatchedElements& in, BatchedElements& out) override {
     for (const Element& f : in[0]) {
@@ -348,6 +352,9 @@ class CropKernelCPU : public BatchedKernel {
     w_ = (i32)a.get_int("width", 0);
     h_ = (i32)a.get_int("height", 0);
     SCA_CHECK(w_ > 0 && h_ > 0, "Crop needs width/height args");
+    // negative offsets passed the (x + w <= frame_w) check and read
+    // before the source buffer
+    SCA_CHECK(x_ >= 0 && y_ >= 0, "Crop x/y must be >= 0");
   }
   void execute_batch(const BatchedElements& in, BatchedElements& out) override {
     for (const Element& f : in[0]) {

@@ -202,6 +202,7 @@ class CropKernelGPU : public BatchedKernel {
     w_ = (int)a.get_int("width", 0);
     h_ = (int)a.get_int("height", 0);
     SCA_CHECK(w_ > 0 && h_ > 0, "Crop needs width/height args");
+    SCA_CHECK(x_ >= 0 && y_ >= 0, "Crop x/y must be >= 0");
   }
   void execute_batch(const BatchedElements& in, BatchedElements& out) override {
     hipStream_t s = cur_stream();
@@ -233,7 +234,9 @@ class BlurKernelGPU : public BatchedKernel {
  public:
   explicit BlurKernelGPU(const KernelConfig& cfg) : BatchedKernel(cfg) {
     auto a = mp::decode(cfg.args);
-    radius_ = (int)a.get_int("kernel_size", 3) / 2;
+    int ks = (int)a.get_int("kernel_size", 3);
+    SCA_CHECK(ks >= 1, "Blur kernel_size must be >= 1");
+    radius_ = ks / 2;
   }
   void execute_batch(const BatchedElements& in, BatchedElements& out) override {
     hipStream_t s = cur_stream();

@@ -140,6 +140,36 @@ def test_python_op_exception_message_preserved(sc):
                cache_mode=sp.CacheMode.Overwrite)
 
 
+def frame_table(sc, name):
+    import numpy as np
+    frames = np.zeros((4, 32, 40, 3), np.uint8)
+    return sp.NamedVideoStream(sc, name, frames=frames, codec="raw")
+
+
+def test_blur_bad_kernel_size_rejected(sc):
+    """kernel_size < 1 made the tap radius negative -> division by zero
+    (UB, observed as a WEDGED pipeline instance pre-fix)."""
+    v = frame_table(sc, "ep11")
+    col = sc.io.Input([v])
+    out = sp.NamedStream(sc, "ep11_out")
+    with pytest.raises(Exception, match="kernel_size must be >= 1"):
+        sc.run(sc.io.Output(sc.ops.Blur(frame=col, kernel_size=-3), [out]),
+               sp.PerfParams.manual(2, 4),
+               cache_mode=sp.CacheMode.Overwrite)
+
+
+def test_crop_negative_offset_rejected(sc):
+    """Negative x/y passed the (x + w <= frame_w) bound and read before
+    the source buffer (silent OOB read pre-fix)."""
+    v = frame_table(sc, "ep12")
+    col = sc.io.Input([v])
+    out = sp.NamedStream(sc, "ep12_out")
+    with pytest.raises(Exception, match="x/y must be >= 0"):
+        sc.run(sc.io.Output(
+            sc.ops.Crop(frame=col, x=-2, y=0, width=10, height=10), [out]),
+            sp.PerfParams.manual(2, 4), cache_mode=sp.CacheMode.Overwrite)
+
+
 def test_zero_row_job(sc):
     """An empty sampling result is a valid job: zero output rows, no
     error (the engine must handle tasks with nothing to produce)."""
