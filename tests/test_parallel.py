@@ -33,6 +33,17 @@ if rank == 0:
 else:
     assert got is None
 
+# gather_column edge shapes: empty blobs, an empty rank, and a payload
+# far larger than max_bytes (multi-chunk wire messages)
+edge = ([b"", b"x" * 1000, b"", b"tail"] if rank == 0 else [])
+got = parallel.gather_column(edge, device, max_bytes=64)
+if rank == 0:
+    assert got == [b"", b"x" * 1000, b"", b"tail"], got
+edge2 = [bytes([rank]) * (300 + rank)]
+got = parallel.gather_column(edge2, device, max_bytes=128)
+if rank == 0:
+    assert got == [b"\x00" * 300, b"\x01" * 301], [len(g) for g in got]
+
 # broadcast_blob
 blob = b"weights-payload" if rank == 0 else None
 out = parallel.broadcast_blob(blob, device)
