@@ -297,3 +297,24 @@ def test_subprocess_kernel_parallelism(tmp_path):
     # ~24 x 100 ms = 2.4 s of compute. In-process serializes on the GIL
     # (>= ~2.4 s); 4 subprocess instances overlap it (+ ~1 s spawn cost).
     assert results["BurnP"] < results["BurnT"] * 0.75, results
+
+
+def test_multi_output_partial_consumption(sc):
+    """Only one of a multi-output op's columns is consumed: the engine's
+    liveness pass must discard the unused column's elements (executor
+    'unused output column' branch — reference: dead-column elimination,
+    dag_analysis.cpp:1145-1326) and the consumed column stays exact."""
+    from conftest import make_video
+    import numpy as np
+    frames = make_video(n=8, h=24, w=32)
+    video = sp.NamedVideoStream(sc, "mopc", frames=frames, codec="raw")
+    frame = sc.io.Input([video])
+    st = sc.ops.SplitStats(frame=frame)
+    out = sp.NamedStream(sc, "mopc_out")
+    # consume ONLY out1 (std); out0 (mean) is produced and must be dropped
+    sc.run(sc.io.Output(st["out1"], [out]),
+           sp.PerfParams.manual(2, 4), cache_mode=sp.CacheMode.Overwrite)
+    stds = [float.fromhex(b.decode()) for b in out.load()]
+    assert len(stds) == 8
+    for i in range(8):
+        assert abs(stds[i] - float(np.std(frames[i]))) < 1e-5
