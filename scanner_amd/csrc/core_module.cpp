@@ -32,11 +32,6 @@ mp::Value mp_from_pybytes(const py::bytes& b) {
   return mp::decode(reinterpret_cast<const u8*>(s.data()), s.size());
 }
 
-std::vector<u8> vec_from_pybytes(const py::bytes& b) {
-  std::string s = b;
-  return std::vector<u8>(s.begin(), s.end());
-}
-
 JobGraph graph_from_bytes(const py::bytes& b) {
   return JobGraph::from_msgpack(mp_from_pybytes(b));
 }

@@ -164,8 +164,6 @@ class DetectorKernelGPU : public BatchedKernel {
     u8* colbuf = bufs_[4];
     u8* loc = bufs_[5];
     u8* cls = bufs_[6];
-    size_t featpix = (size_t)n * kFeat * kFeat;
-
     std::vector<const u8*> ptrs(n);
     for (int i = 0; i < n; ++i) ptrs[i] = in[0][i].buffer;
     memcpy_buffer(d_ptrs, dev, (const u8*)ptrs.data(), CPU_DEVICE,
