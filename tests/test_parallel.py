@@ -9,6 +9,15 @@ import pytest
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
+def _free_port():
+    """A fresh OS-assigned port per test: the previous fixed ports hit
+    TIME_WAIT when the suite re-runs back-to-back (observed flake)."""
+    import socket
+    with socket.socket() as sk:
+        sk.bind(("127.0.0.1", 0))
+        return str(sk.getsockname()[1])
+
+
 WORKER = r"""
 import os, sys
 sys.path.insert(0, os.environ["SCA_REPO"])
@@ -60,6 +69,7 @@ def test_parallel_gloo_world2(tmp_path):
     script = tmp_path / "worker.py"
     script.write_text(WORKER)
     procs = []
+    port = _free_port()
     for rank in range(2):
         env = dict(os.environ)
         env.update({
@@ -68,7 +78,7 @@ def test_parallel_gloo_world2(tmp_path):
             "LOCAL_RANK": str(rank),
             "WORLD_SIZE": "2",
             "MASTER_ADDR": "127.0.0.1",
-            "MASTER_PORT": "29611",
+            "MASTER_PORT": port,
         })
         procs.append(subprocess.Popen(
             [sys.executable, str(script)], env=env,
@@ -89,6 +99,7 @@ def test_bench_cpu_world2(tmp_path):
     import json
     procs = []
     outs = []
+    port = _free_port()
     for rank in range(2):
         env = dict(os.environ)
         env.update({
@@ -97,7 +108,7 @@ def test_bench_cpu_world2(tmp_path):
             "LOCAL_RANK": str(rank),
             "WORLD_SIZE": "2",
             "MASTER_ADDR": "127.0.0.1",
-            "MASTER_PORT": "29617",
+            "MASTER_PORT": port,
         })
         procs.append(subprocess.Popen(
             [sys.executable, os.path.join(REPO, "bench.py"), "--pipeline",
@@ -184,6 +195,7 @@ def test_parallel_gloo_world8_engine_gather(tmp_path):
     script = tmp_path / "worker8.py"
     script.write_text(WORKER8)
     procs = []
+    port = _free_port()
     for rank in range(8):
         env = dict(os.environ)
         env.update({
@@ -192,7 +204,7 @@ def test_parallel_gloo_world8_engine_gather(tmp_path):
             "LOCAL_RANK": str(rank),
             "WORLD_SIZE": "8",
             "MASTER_ADDR": "127.0.0.1",
-            "MASTER_PORT": "29621",
+            "MASTER_PORT": port,
         })
         procs.append(subprocess.Popen(
             [sys.executable, str(script)], env=env,
