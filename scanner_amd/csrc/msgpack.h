@@ -227,12 +227,18 @@ class Decoder {
     return Value(std::move(b));
   }
   Value array(u64 n) {
+    // every element takes >= 1 byte, so a count beyond the remaining
+    // buffer is corrupt — reject BEFORE reserving (a trusted 4-byte
+    // count let one bad message attempt a ~100 GB allocation; caught by
+    // the ASan fuzz, tests/cpp/asan_parsers.cpp)
+    check(n);
     Array a;
-    a.reserve(n);
+    a.reserve((size_t)n);
     for (u64 i = 0; i < n; ++i) a.push_back(decode());
     return Value(std::move(a));
   }
   Value map(u64 n) {
+    check(n);  // >= 1 byte per entry minimum
     Map m;
     for (u64 i = 0; i < n; ++i) {
       Value k = decode();
@@ -241,7 +247,9 @@ class Decoder {
     return Value(std::move(m));
   }
   void check(u64 n) {
-    if (p_ + n > end_) throw ScannerError("msgpack: truncated");
+    // compare against the remaining length — `p_ + n` could overflow the
+    // pointer (UB) for hostile 64-bit counts
+    if (n > (u64)(end_ - p_)) throw ScannerError("msgpack: truncated");
   }
   const u8* p_;
   const u8* end_;

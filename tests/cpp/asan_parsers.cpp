@@ -17,6 +17,7 @@
 #include <string>
 #include <vector>
 
+#include "../../scanner_amd/csrc/msgpack.h"
 #include "../../scanner_amd/csrc/video/h264.h"
 #include "../../scanner_amd/csrc/video/mp4.h"
 #include "../../scanner_amd/csrc/video/svc.h"
@@ -122,6 +123,27 @@ int main(int argc, char** argv) {
         return 1;
       }
     }
+  }
+  // msgpack codec: job specs/op args reach the C++ engine as msgpack
+  // bytes over the RPC — decode of mutated/truncated payloads must
+  // reject, not read out of bounds. Corpus: a representative job-shaped
+  // value round-tripped through our own encoder.
+  {
+    sca::mp::Map m;
+    m["name"] = std::string("Histogram");
+    m["stride"] = (sca::i64)7;
+    m["flag"] = true;
+    m["rate"] = 29.97;
+    sca::mp::Array arr;
+    for (int i = 0; i < 20; ++i) arr.push_back(sca::mp::Value((sca::i64)i));
+    m["rows"] = std::move(arr);
+    m["blob"] = std::vector<u8>{1, 2, 3, 4, 5, 6, 7, 8};
+    sca::mp::Map inner;
+    inner["kind"] = std::string("Gather");
+    m["sampling"] = sca::mp::Value(std::move(inner));
+    std::vector<u8> enc = sca::mp::encode(sca::mp::Value(std::move(m)));
+    fuzz("msgpack", enc, iters,
+         [](const u8* p, size_t n) { (void)sca::mp::decode(p, n); });
   }
   std::printf("asan parser fuzz: OK\n");
   return 0;
