@@ -79,16 +79,25 @@ class BinReader {
   template <typename T>
   std::vector<T> vec_pod() {
     u64 n = u64v();
-    check(n * sizeof(T));
-    std::vector<T> v(n);
-    std::memcpy(v.data(), p_, n * sizeof(T));
-    p_ += n * sizeof(T);
+    // divide, don't multiply: n * sizeof(T) can wrap for a corrupt
+    // 64-bit count, passing the check and then allocating n elements
+    if (n > remaining() / sizeof(T))
+      throw ScannerError("binary deserialize out of bounds");
+    std::vector<T> v((size_t)n);
+    if (n) std::memcpy(v.data(), p_, (size_t)n * sizeof(T));
+    p_ += (size_t)n * sizeof(T);
     return v;
   }
   std::vector<std::string> vec_str() {
     u64 n = u64v();
+    // every string needs at least its 8-byte length; a count beyond the
+    // remaining bytes is corrupt — reject before reserving (same
+    // trusted-count DoS the msgpack decoder had; ASan fuzz corpus in
+    // tests/cpp/asan_parsers.cpp)
+    if (n > remaining() / 8)
+      throw ScannerError("binary deserialize out of bounds");
     std::vector<std::string> v;
-    v.reserve(n);
+    v.reserve((size_t)n);
     for (u64 i = 0; i < n; ++i) v.push_back(str());
     return v;
   }
@@ -103,8 +112,12 @@ class BinReader {
     p_ += sizeof(T);
     return v;
   }
+  u64 remaining() const { return (u64)(end_ - p_); }
   void check(u64 n) {
-    if (p_ + n > end_) throw ScannerError("binary deserialize out of bounds");
+    // compare against the remaining length — `p_ + n` could overflow the
+    // pointer (UB) for hostile 64-bit counts
+    if (n > remaining())
+      throw ScannerError("binary deserialize out of bounds");
   }
   const u8* p_;
   const u8* end_;

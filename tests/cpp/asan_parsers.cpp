@@ -17,6 +17,7 @@
 #include <string>
 #include <vector>
 
+#include "../../scanner_amd/csrc/metadata.h"
 #include "../../scanner_amd/csrc/msgpack.h"
 #include "../../scanner_amd/csrc/video/h264.h"
 #include "../../scanner_amd/csrc/video/mp4.h"
@@ -144,6 +145,40 @@ int main(int argc, char** argv) {
     std::vector<u8> enc = sca::mp::encode(sca::mp::Value(std::move(m)));
     fuzz("msgpack", enc, iters,
          [](const u8* p, size_t n) { (void)sca::mp::decode(p, n); });
+  }
+  // metadata records (BinWriter/BinReader, csrc/serialize.h): table and
+  // video descriptors are read back from storage bytes — the same
+  // corruption model as SVC packets. BinReader's count handling had the
+  // same trusted-prefix/overflow patterns as the msgpack decoder.
+  {
+    sca::VideoMetadata vm;
+    vm.width = 1920;
+    vm.height = 1080;
+    vm.channels = 3;
+    vm.codec = "svc";
+    vm.num_frames = 64;
+    for (int i = 0; i < 64; i += 16) vm.keyframe_indices.push_back(i);
+    for (int i = 0; i < 64; ++i) {
+      vm.sample_offsets.push_back((sca::u64)i * 1000);
+      vm.sample_sizes.push_back(1000);
+    }
+    std::vector<u8> enc = vm.serialize();
+    fuzz("video_meta", enc, iters, [](const u8* p, size_t n) {
+      (void)sca::VideoMetadata::deserialize(std::vector<u8>(p, p + n));
+    });
+
+    sca::TableMetadata tm;
+    tm.id = 7;
+    tm.name = "clip_table";
+    for (int i = 1; i <= 5; ++i) tm.end_rows.push_back(i * 128);
+    for (int c = 0; c < 3; ++c)
+      tm.columns.push_back(sca::ColumnMeta{
+          c, "col" + std::to_string(c),
+          c ? sca::ColumnType::Bytes : sca::ColumnType::Video});
+    std::vector<u8> enc2 = tm.serialize();
+    fuzz("table_meta", enc2, iters, [](const u8* p, size_t n) {
+      (void)sca::TableMetadata::deserialize(std::vector<u8>(p, p + n));
+    });
   }
   std::printf("asan parser fuzz: OK\n");
   return 0;

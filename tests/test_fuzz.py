@@ -46,9 +46,12 @@ def test_asan_parser_fuzz(tmp_path):
         paths[name] = str(p)
 
     src = os.path.join(REPO, "tests", "cpp", "asan_parsers.cpp")
-    vid = os.path.join(REPO, "scanner_amd", "csrc", "video")
+    csrc = os.path.join(REPO, "scanner_amd", "csrc")
+    vid = os.path.join(csrc, "video")
     tus = [os.path.join(vid, f) for f in ("mp4.cpp", "h264.cpp",
                                           "svc_cpu.cpp")]
+    tus += [os.path.join(csrc, "metadata.cpp"),
+            os.path.join(csrc, "storage.cpp")]
     binp = str(tmp_path / "asan_fuzz")
     r = subprocess.run(
         ["g++", "-O1", "-g", "-std=c++17",
