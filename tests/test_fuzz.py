@@ -45,6 +45,9 @@ def test_asan_parser_fuzz(tmp_path):
         p.write_bytes(bytes(blob))
         paths[name] = str(p)
 
+    import hashlib
+    import tempfile
+
     src = os.path.join(REPO, "tests", "cpp", "asan_parsers.cpp")
     csrc = os.path.join(REPO, "scanner_amd", "csrc")
     vid = os.path.join(csrc, "video")
@@ -52,14 +55,25 @@ def test_asan_parser_fuzz(tmp_path):
                                           "svc_cpu.cpp")]
     tus += [os.path.join(csrc, "metadata.cpp"),
             os.path.join(csrc, "storage.cpp")]
-    binp = str(tmp_path / "asan_fuzz")
-    r = subprocess.run(
-        ["g++", "-O1", "-g", "-std=c++17",
-         "-fsanitize=address,undefined", "-fno-sanitize-recover=all",
-         src, *tus, "-o", binp],
-        capture_output=True, text=True, timeout=300)
-    if r.returncode != 0:
-        pytest.fail(f"asan build failed:\n{r.stderr[-2000:]}")
+    # the ~20 s sanitizer build dominates suite time — cache the binary
+    # keyed by a hash of every input (headers included via the csrc tree)
+    h = hashlib.sha256()
+    for f in [src] + tus:
+        h.update(open(f, "rb").read())
+    for root, _, files in os.walk(csrc):
+        for f in sorted(files):
+            if f.endswith(".h"):
+                h.update(open(os.path.join(root, f), "rb").read())
+    binp = os.path.join(tempfile.gettempdir(),
+                        f"sca_asan_fuzz_{h.hexdigest()[:16]}")
+    if not os.path.exists(binp):
+        r = subprocess.run(
+            ["g++", "-O1", "-g", "-std=c++17",
+             "-fsanitize=address,undefined", "-fno-sanitize-recover=all",
+             src, *tus, "-o", binp],
+            capture_output=True, text=True, timeout=300)
+        if r.returncode != 0:
+            pytest.fail(f"asan build failed:\n{r.stderr[-2000:]}")
 
     r = subprocess.run(
         [binp, paths["c.mp4"], paths["c.264"], paths["c.sps"], "1500"],
