@@ -17,12 +17,26 @@ DEFAULT_IMAGE = "YOUR_REGISTRY/scanner-amd:latest"
 class MachineConfig:
     """Per-pod resources (parity: kube.py MachineType/MachineConfig)."""
 
+    # USD/hour rate card for cost estimation (parity: the reference's
+    # hardcoded GCP price tables, kube.py:124-177). Override per
+    # deployment — cloud MI3xx list prices move frequently.
+    RATES = {"cpu": 0.032, "mem_gb": 0.0043, "gpu": 2.50}
+    PREEMPTIBLE_DISCOUNT = 0.30  # spot/preemptible multiplier
+
     def __init__(self, cpus=8, memory_gb=32, gpus_per_node=8,
-                 gpu_resource="amd.com/gpu"):
+                 gpu_resource="amd.com/gpu", preemptible=False):
         self.cpus = cpus
         self.memory_gb = memory_gb
         self.gpus_per_node = gpus_per_node
         self.gpu_resource = gpu_resource
+        self.preemptible = preemptible
+
+    def price(self):
+        """Estimated USD/hour for one pod of this shape."""
+        p = (self.RATES["cpu"] * self.cpus +
+             self.RATES["mem_gb"] * self.memory_gb +
+             self.RATES["gpu"] * self.gpus_per_node)
+        return p * (self.PREEMPTIBLE_DISCOUNT if self.preemptible else 1.0)
 
 
 class ClusterConfig:
@@ -48,6 +62,12 @@ class ClusterConfig:
         # db_path becomes a key prefix); "posix": the RWX shared volume
         self.storage_type = storage_type
         self.bucket = bucket
+
+    def price(self, no_master=False):
+        """Estimated whole-cluster USD/hour (parity: reference
+        ClusterConfig.price kube.py:209)."""
+        p = 0.0 if no_master else self.master.price()
+        return p + self.worker.price() * self.num_workers
 
 
 def _resources(mc):

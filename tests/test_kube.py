@@ -52,3 +52,19 @@ def test_deploy_scale_delete_flow():
     deletes = [a for a, _ in fake.calls[2:]]
     assert len(deletes) == 4
     assert c.master_address() == "s-master:5001"
+
+
+def test_cluster_price_estimate():
+    """Cost estimation (parity: reference ClusterConfig.price)."""
+    from scanner_amd.kube import ClusterConfig, MachineConfig
+    cfg = ClusterConfig(
+        num_workers=4,
+        master=MachineConfig(cpus=4, memory_gb=16, gpus_per_node=0),
+        worker=MachineConfig(cpus=16, memory_gb=128, gpus_per_node=8))
+    m = 0.032 * 4 + 0.0043 * 16
+    w = 0.032 * 16 + 0.0043 * 128 + 2.50 * 8
+    assert abs(cfg.price() - (m + 4 * w)) < 1e-9
+    assert abs(cfg.price(no_master=True) - 4 * w) < 1e-9
+    spot = MachineConfig(cpus=16, memory_gb=128, gpus_per_node=8,
+                         preemptible=True)
+    assert abs(spot.price() - w * 0.30) < 1e-9
