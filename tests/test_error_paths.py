@@ -180,3 +180,35 @@ def test_zero_row_job(sc):
     sc.run(sc.io.Output(r, [out]), sp.PerfParams.manual(4, 8),
            cache_mode=sp.CacheMode.Overwrite)
     assert list(out.load()) == []
+
+
+def test_colorconvert_unknown_format(sc):
+    v = frame_table(sc, "ep13")
+    col = sc.io.Input([v])
+    out = sp.NamedStream(sc, "ep13_out")
+    with pytest.raises(Exception, match="gray"):
+        sc.run(sc.io.Output(
+            sc.ops.ColorConvert(frame=col, format="bogus"), [out]),
+            sp.PerfParams.manual(2, 4), cache_mode=sp.CacheMode.Overwrite)
+
+
+def test_resize_bad_dims(sc):
+    v = frame_table(sc, "ep14")
+    col = sc.io.Input([v])
+    out = sp.NamedStream(sc, "ep14_out")
+    for w, h in ((-5, 10), (0, 0)):
+        with pytest.raises(Exception, match="width/height"):
+            sc.run(sc.io.Output(
+                sc.ops.Resize(frame=col, width=w, height=h), [out]),
+                sp.PerfParams.manual(2, 4),
+                cache_mode=sp.CacheMode.Overwrite)
+
+
+def test_histogram_on_bytes_rejected(sc):
+    tab = int_table(sc, "ep15", 4)
+    col = sc.io.Input([tab])
+    out = sp.NamedStream(sc, "ep15_out")
+    with pytest.raises(Exception, match="frame input"):
+        sc.run(sc.io.Output(sc.ops.Histogram(frame=col), [out]),
+               sp.PerfParams.manual(2, 4),
+               cache_mode=sp.CacheMode.Overwrite)
