@@ -444,3 +444,62 @@ def test_worker_multiple_pipeline_instances(tmp_path):
             os.killpg(wp.pid, signal.SIGKILL)
         wp.wait()
         master.shutdown()
+
+
+@pytest.mark.parametrize("seed", [3, 11])
+def test_scheduler_chaos_seeded(tmp_path, seed):
+    """Seeded randomized churn: workers killed and joined at random times
+    while a Sleep-op job runs; the job must complete with exact results
+    whatever the interleaving (generalizes the fixed-timing churn test)."""
+    import random
+    import threading
+    rng = random.Random(seed)
+    db = _mk_db(tmp_path)
+    master = MasterServer(db, task_timeout=20)
+    procs = [spawn_worker_proc(master.addr, db)
+             for _ in range(rng.randint(2, 4))]
+    try:
+        sc = sp.Client(db_path=db, master=master.addr)
+        n = 30
+        tab = sc.new_table("ch", ["col"],
+                           [[int(i).to_bytes(8, "little")]
+                            for i in range(n)])
+        col = sc.io.Input([tab])
+        slow = sc.ops.Sleep(ignore=col, ms=rng.randint(15, 40))
+        out = sp.NamedStream(sc, "ch_out")
+        done = {}
+
+        def run():
+            try:
+                sc.run(sc.io.Output(slow, [out]), sp.PerfParams.manual(1, 2),
+                       cache_mode=sp.CacheMode.Overwrite)
+                done["ok"] = True
+            except Exception as e:  # pragma: no cover
+                done["err"] = e
+
+        th = threading.Thread(target=run)
+        th.start()
+        # random churn script: 2-3 events at random offsets
+        for _ in range(rng.randint(2, 3)):
+            time.sleep(rng.uniform(0.2, 0.8))
+            if rng.random() < 0.5 and len(procs) > 1:
+                victim = procs.pop(rng.randrange(len(procs)))
+                try:
+                    os.killpg(victim.pid, signal.SIGKILL)
+                except ProcessLookupError:
+                    pass
+            else:
+                procs.append(spawn_worker_proc(master.addr, db))
+        th.join(timeout=120)
+        assert not th.is_alive(), "job hung under churn"
+        assert done.get("ok"), f"job failed: {done.get('err')}"
+        vals = [int.from_bytes(b, "little") for b in out.load()]
+        assert vals == list(range(n))
+        sc.shutdown()
+    finally:
+        for p in procs:
+            try:
+                os.killpg(p.pid, signal.SIGKILL)
+            except ProcessLookupError:
+                pass
+        master.shutdown()
