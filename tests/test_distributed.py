@@ -281,7 +281,9 @@ def test_master_stress_8workers_2jobs_churn(tmp_path):
     #9 master hardening: analysis off the RPC lock, concurrent pings)."""
     import threading
     db = _mk_db(tmp_path)
-    master = MasterServer(db, task_timeout=30)
+    # generous timeouts: this test runs 10 OS processes and flaked under
+    # heavy machine load with tighter settings
+    master = MasterServer(db, task_timeout=60)
     procs = [spawn_worker_proc(master.addr, db) for _ in range(8)]
     try:
         sc = sp.Client(db_path=db, master=master.addr)
@@ -294,14 +296,17 @@ def test_master_stress_8workers_2jobs_churn(tmp_path):
 
         def run_job(tag):
             # separate client per thread (its own RPC connection)
-            scj = sp.Client(db_path=db, master=master.addr)
-            frame = scj.io.Input(streams)
-            slow = scj.ops.Sleep(ignore=frame, ms=20)
-            outs = [sp.NamedStream(scj, f"o{tag}_{i}") for i in range(2)]
-            scj.run(scj.io.Output(slow, outs),
-                    sp.PerfParams.manual(2, 4),
-                    cache_mode=sp.CacheMode.Overwrite)
-            results[tag] = [list(o.load()) for o in outs]
+            try:
+                scj = sp.Client(db_path=db, master=master.addr)
+                frame = scj.io.Input(streams)
+                slow = scj.ops.Sleep(ignore=frame, ms=20)
+                outs = [sp.NamedStream(scj, f"o{tag}_{i}") for i in range(2)]
+                scj.run(scj.io.Output(slow, outs),
+                        sp.PerfParams.manual(2, 4),
+                        cache_mode=sp.CacheMode.Overwrite)
+                results[tag] = [list(o.load()) for o in outs]
+            except Exception as e:  # surfaced by the assertion below
+                results[tag] = e
 
         threads = [threading.Thread(target=run_job, args=(t,))
                    for t in range(2)]
@@ -313,9 +318,11 @@ def test_master_stress_8workers_2jobs_churn(tmp_path):
             os.killpg(p.pid, signal.SIGKILL)
         procs.extend(spawn_worker_proc(master.addr, db) for _ in range(2))
         for t in threads:
-            t.join(timeout=180)
+            t.join(timeout=300)
             assert not t.is_alive(), "job thread hung"
         for tag in range(2):
+            assert not isinstance(results.get(tag), Exception), \
+                f"job {tag} failed: {results[tag]}"
             assert len(results[tag]) == 2
             for rows in results[tag]:
                 assert len(rows) == n
