@@ -906,3 +906,32 @@ def test_gather_unsorted_streaming(sc):
     for k, blob in enumerate(rows):
         got = np.frombuffer(blob, np.uint32).reshape(3, 256)
         np.testing.assert_array_equal(got, ref_histogram(frames[wanted[k]]))
+
+
+def test_concurrent_clients_table_creation(tmp_path):
+    """4 OS processes create disjoint tables on ONE posix db concurrently:
+    the cross-process db lock (ScopedDbLock, metadata.cpp) must serialize
+    metadata updates so no creation is lost."""
+    import subprocess
+    import sys as _sys
+    import os as _os
+    repo = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+    db = str(tmp_path / "db")
+    script = tmp_path / "mk.py"
+    script.write_text(
+        "import sys\n"
+        f"sys.path.insert(0, {repo!r})\n"
+        "import scanner_amd as sp\n"
+        "idx = int(sys.argv[1])\n"
+        f"sc = sp.Client(db_path={db!r}, recover=False)\n"
+        "for k in range(6):\n"
+        "    sc.new_table(f't{idx}_{k}', ['col'], [[b'12345678']])\n")
+    procs = [subprocess.Popen([_sys.executable, str(script), str(i)],
+                              stdout=subprocess.PIPE,
+                              stderr=subprocess.STDOUT)
+             for i in range(4)]
+    for p in procs:
+        out, _ = p.communicate(timeout=120)
+        assert p.returncode == 0, out.decode()
+    sc = sp.Client(db_path=db)
+    assert len(sc.table_names()) == 24
