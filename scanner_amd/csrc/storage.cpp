@@ -90,8 +90,15 @@ class PosixStorage : public StorageBackend {
 
   void write_all(const std::string& path, const u8* data,
                  size_t size) override {
-    // Write to temp + rename for atomicity (commit semantics depend on it).
-    std::string tmp = path + ".tmp";
+    // Write to temp + rename for atomicity (commit semantics depend on
+    // it). The temp name must be unique PER WRITER: two processes
+    // writing the same path (e.g. both creating a fresh db's metadata)
+    // collided on a shared ".tmp" — the first rename stole the second
+    // writer's file and its rename failed with ENOENT (found by
+    // test_concurrent_clients_table_creation).
+    static std::atomic<u64> seq{0};
+    std::string tmp = path + ".tmp." + std::to_string((u64)::getpid()) +
+                      "." + std::to_string(seq.fetch_add(1));
     int fd = ::open(tmp.c_str(), O_WRONLY | O_CREAT | O_TRUNC, 0644);
     if (fd < 0) throw ScannerError("open for write failed: " + tmp + ": " + strerror(errno));
     size_t off = 0;
@@ -106,7 +113,8 @@ class PosixStorage : public StorageBackend {
     if (fsync(fd) != 0 || ::close(fd) != 0)
       throw ScannerError("fsync/close failed: " + tmp);
     if (::rename(tmp.c_str(), path.c_str()) != 0)
-      throw ScannerError("rename failed: " + path);
+      throw ScannerError("rename failed: " + path + ": " +
+                         strerror(errno));
   }
 
   bool exists(const std::string& path) override {
