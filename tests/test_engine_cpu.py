@@ -935,3 +935,46 @@ def test_concurrent_clients_table_creation(tmp_path):
         assert p.returncode == 0, out.decode()
     sc = sp.Client(db_path=db)
     assert len(sc.table_names()) == 24
+
+
+def test_concurrent_jobs_shared_db(tmp_path):
+    """3 OS processes each run a full ingest+histogram job against ONE
+    shared posix db concurrently (separate output tables): metadata
+    commits serialize under the db lock and every job's results stay
+    exact."""
+    import subprocess
+    import sys as _sys
+    import os as _os
+    repo = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+    db = str(tmp_path / "db")
+    script = tmp_path / "job.py"
+    script.write_text(
+        "import sys\n"
+        f"sys.path.insert(0, {repo!r})\n"
+        "import numpy as np\n"
+        "import scanner_amd as sp\n"
+        "idx = int(sys.argv[1])\n"
+        f"sc = sp.Client(db_path={db!r}, recover=False)\n"
+        "rng = np.random.RandomState(idx)\n"
+        "frames = rng.randint(0, 255, (10, 32, 40, 3)).astype(np.uint8)\n"
+        "v = sp.NamedVideoStream(sc, f'v{idx}', frames=frames,\n"
+        "                        codec='svc')\n"
+        "col = sc.io.Input([v])\n"
+        "out = sp.NamedStream(sc, f'h{idx}')\n"
+        "sc.run(sc.io.Output(sc.ops.Histogram(frame=col), [out]),\n"
+        "       sp.PerfParams.manual(2, 4),\n"
+        "       cache_mode=sp.CacheMode.Overwrite)\n"
+        "rows = list(out.load())\n"
+        "assert len(rows) == 10\n"
+        "for i, b in enumerate(rows):\n"
+        "    got = np.frombuffer(b, np.uint32).reshape(3, 256)\n"
+        "    exp = np.stack([np.bincount(frames[i][:, :, c].ravel(),\n"
+        "                                minlength=256) for c in range(3)])\n"
+        "    assert (got == exp).all(), f'row {i} mismatch'\n")
+    procs = [subprocess.Popen([_sys.executable, str(script), str(i)],
+                              stdout=subprocess.PIPE,
+                              stderr=subprocess.STDOUT)
+             for i in range(3)]
+    for p in procs:
+        out, _ = p.communicate(timeout=200)
+        assert p.returncode == 0, out.decode()[-2000:]
