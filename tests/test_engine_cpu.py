@@ -978,3 +978,24 @@ def test_concurrent_jobs_shared_db(tmp_path):
     for p in procs:
         out, _ = p.communicate(timeout=200)
         assert p.returncode == 0, out.decode()[-2000:]
+
+
+def test_large_row_count_linear(sc):
+    """50k rows through plan/liveness/streaming in one job — guards the
+    per-row bookkeeping (task plans, read-count maps, liveness frees)
+    against accidental quadratic behavior; finishes in well under the
+    suite timeout or something regressed."""
+    import time
+    n = 50000
+    tab = sc.new_table("big_lin", ["col"],
+                       [[int(i).to_bytes(8, "little")] for i in range(n)])
+    col = sc.io.Input([tab])
+    out = sp.NamedStream(sc, "big_lin_out")
+    t0 = time.time()
+    sc.run(sc.io.Output(sc.ops.TestIncrement(ignore=col), [out]),
+           sp.PerfParams.manual(64, 512),
+           cache_mode=sp.CacheMode.Overwrite)
+    elapsed = time.time() - t0
+    vals = [int.from_bytes(b, "little") for b in out.load()]
+    assert vals == [i + (i % 512) + 1 for i in range(n)]
+    assert elapsed < 30, f"50k rows took {elapsed:.1f}s (expected < 1s warm)"
