@@ -38,20 +38,20 @@ def fresh(name):
     return f"{name}_{_serial[0]}"
 
 
-def gen_frames(content, n, h, w, seed):
+def gen_frames(content, n, h, w, seed, c=3):
     rng = np.random.RandomState(seed)
     if content == "const":
-        return np.full((n, h, w, 3), seed % 256, np.uint8)
+        return np.full((n, h, w, c), seed % 256, np.uint8)
     if content == "random":  # residual width 8 everywhere
-        return rng.randint(0, 256, size=(n, h, w, 3)).astype(np.uint8)
+        return rng.randint(0, 256, size=(n, h, w, c)).astype(np.uint8)
     if content == "gradient":  # small deltas -> narrow widths
-        base = np.arange(h * w * 3, dtype=np.uint32).reshape(h, w, 3)
+        base = np.arange(h * w * c, dtype=np.uint32).reshape(h, w, c)
         return np.stack([((base + 3 * f) % 256).astype(np.uint8)
                          for f in range(n)])
     if content == "step":  # mixed: half constant, half noisy
-        fr = np.full((n, h, w, 3), 50, np.uint8)
+        fr = np.full((n, h, w, c), 50, np.uint8)
         fr[:, h // 2:] = rng.randint(0, 256,
-                                     size=(n, h - h // 2, w, 3))
+                                     size=(n, h - h // 2, w, c))
         return fr
     raise AssertionError(content)
 
@@ -60,10 +60,10 @@ def gen_frames(content, n, h, w, seed):
           suppress_health_check=[HealthCheck.too_slow])
 @given(n=st.integers(1, 10), h=st.integers(4, 33), w=st.integers(4, 37),
        content=st.sampled_from(["const", "random", "gradient", "step"]),
-       seed=st.integers(0, 10))
-def test_svc_roundtrip_property(n, h, w, content, seed):
+       seed=st.integers(0, 10), c=st.sampled_from([1, 2, 3, 4]))
+def test_svc_roundtrip_property(n, h, w, content, seed, c):
     sc = client()
-    frames = gen_frames(content, n, h, w, seed)
+    frames = gen_frames(content, n, h, w, seed, c)
     video = sp.NamedVideoStream(sc, fresh("cprop"), frames=frames, codec="svc")
     got = np.stack(list(video.load()))
     np.testing.assert_array_equal(got, frames)
