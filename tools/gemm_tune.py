@@ -1,9 +1,15 @@
 #!/usr/bin/env python3
 """GEMM micro-benchmark at the DNN ops' real shapes (run on GPU via
 gpurun). Prints ms and TFLOP/s per shape."""
+import os
 import sys
-sys.path.insert(0, ".")
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 from scanner_amd import _core
+
+if not _core.have_gpu():
+    print("gemm_tune needs a GPU (run via gpurun)", file=sys.stderr)
+    sys.exit(2)
 
 # (label, M, N, K): ResNet-50 @ batch 16 and Pose @ batch 8 hot shapes
 SHAPES = [
