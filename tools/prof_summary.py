@@ -6,6 +6,9 @@ import sys
 
 
 def main():
+    if len(sys.argv) < 2:
+        print(__doc__.strip(), file=sys.stderr)
+        sys.exit(2)
     path = sys.argv[1]
     db = sqlite3.connect(path)
     cur = db.cursor()
