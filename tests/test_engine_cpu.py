@@ -215,7 +215,7 @@ def test_slice_unslice(sc):
     assert vals == [i + (i % 4) + 1 for i in range(n)]
 
 
-def test_overlapping_slices(sc):
+def test_overlapping_slices_single_task(sc):
     n = 10
     tab = sc.new_table("ov", ["col"],
                        [[int(i).to_bytes(8, "little")] for i in range(n)])
