@@ -253,6 +253,11 @@ def test_subprocess_kernel_parallelism(tmp_path):
     kernels serialize on the GIL; isolation='process' must overlap
     (VERDICT r01 weak #7 — the reference forked a child per kernel
     instance for exactly this)."""
+    import os as _os
+    if _os.environ.get("PYTEST_XDIST_WORKER"):
+        # wall-clock speedup assertion — meaningless when pytest-xdist
+        # workers compete for the same cores (stable in serial runs)
+        pytest.skip("timing assertion; serial runs only")
     import time
 
     import scanner_amd as sp
