@@ -42,6 +42,7 @@ class _BulkJob:
         self.done = set()
         self.failures = collections.Counter()
         self.blacklisted = set()    # stream indices
+        self.task_errors = {}       # stream -> last worker-reported error
         self.stream_tasks = collections.Counter()
         self.stream_done = collections.Counter()
         self.finalized = set()
@@ -234,6 +235,8 @@ class MasterServer:
                     job.stream_done[stream] += 1
                     self._maybe_finalize(job, stream)
             else:
+                if req.get("error"):
+                    job.task_errors[stream] = req["error"]
                 job.failures[key] += 1
                 if job.failures[key] >= TASK_FAILURES:
                     # poison stream: blacklist so one bad stream can't sink
@@ -260,6 +263,8 @@ class MasterServer:
                 "total_tasks": len(job.tasks),
                 "done_tasks": len(job.done),
                 "blacklisted_streams": sorted(job.blacklisted),
+                "task_errors": {str(k): v
+                                for k, v in job.task_errors.items()},
                 "error": job.error,
                 "n_workers": len(self._workers),
             }
@@ -426,9 +431,14 @@ class ClusterClient:
                 if st["error"]:
                     raise ScannerException(st["error"])
                 if st["blacklisted_streams"]:
+                    # include the last worker-reported error per stream so
+                    # the root cause is in the exception, not lost in a
+                    # worker subprocess log
+                    errs = st.get("task_errors", {})
                     raise ScannerException(
                         "streams failed (blacklisted after repeated task "
-                        f"failures): {st['blacklisted_streams']}")
+                        f"failures): {st['blacklisted_streams']}"
+                        + (f"; last errors: {errs}" if errs else ""))
                 return self._load_profiles(jid)
             time.sleep(0.1)
 
