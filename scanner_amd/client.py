@@ -20,7 +20,7 @@ from .streams import PartitionerGenerator, StreamsGenerator
 
 class Client:
     def __init__(self, db_path=None, master=None, workers=None,
-                 start_cluster=True, recover=True, config=None,
+                 start_cluster=True, recover=None, config=None,
                  config_path=None, storage_type=None, bucket=None):
         if config is None and (config_path is not None
                                or db_path is None or master is None):
@@ -39,6 +39,16 @@ class Client:
         if self._storage_type == "posix":
             os.makedirs(self._db_path, exist_ok=True)
         self._db = self._open_db()
+        if recover is None:
+            # Recovery (uncommitted-table GC) belongs to whoever OWNS the
+            # db: a standalone local client, or the MASTER at startup
+            # (reference: recover_and_init_database, master.cpp:1311). A
+            # client connecting to a cluster must NOT recover — a second
+            # client constructed while a job is running would garbage-
+            # collect the active job's uncommitted output tables (this
+            # exact race flaked the churn stress test: workers then fail
+            # with "no table 'o0_0'").
+            recover = master is None
         if recover:
             self._db.recover()
         # (metadata reloads after cluster runs go through _open_db so the
